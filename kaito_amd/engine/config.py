@@ -1,0 +1,73 @@
+"""Engine + model configuration.
+
+ModelConfig covers the llama-architecture family (Llama-3 8B/70B, Phi-4-mini
+class, Mistral, Qwen dense) — the architectures behind the reference's
+headline benchmarks (Llama-3-8B TP=1 / 70B TP=8,
+Phi-4-mini serving CSVs).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+
+@dataclass
+class ModelConfig:
+    name: str = "llama-3-8b"
+    hidden_size: int = 4096
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    intermediate_size: int = 14336
+    vocab_size: int = 128256
+    head_dim: int = 128
+    rope_theta: float = 500000.0
+    max_position: int = 8192
+    rms_eps: float = 1e-5
+    tie_word_embeddings: bool = False
+    dtype: torch.dtype = torch.bfloat16
+    # partial-rotary models (phi family): fraction of head_dim rotated
+    partial_rotary_factor: float = 1.0
+    # attention bias (qwen1/2 style)
+    attention_bias: bool = False
+
+    @property
+    def rotary_dim(self) -> int:
+        r = int(self.head_dim * self.partial_rotary_factor)
+        return r - (r % 2)
+
+    def kv_bytes_per_token(self, tp_size: int = 1) -> int:
+        """Per-token KV cache bytes across all layers (per TP rank)."""
+        kvh = max(self.num_kv_heads // tp_size, 1)
+        return 2 * self.num_layers * kvh * self.head_dim * 2  # k+v, bf16
+
+    def param_bytes(self, tp_size: int = 1) -> int:
+        h, i, v = self.hidden_size, self.intermediate_size, self.vocab_size
+        qkv = h * (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
+        o = self.num_heads * self.head_dim * h
+        mlp = 3 * h * i
+        per_layer = (qkv + o) // tp_size + mlp // tp_size + 2 * h
+        embed = v * h * (1 if self.tie_word_embeddings else 2)
+        return 2 * (self.num_layers * per_layer + embed + h)
+
+
+@dataclass
+class EngineConfig:
+    model: ModelConfig = field(default_factory=ModelConfig)
+    block_size: int = 16
+    max_num_seqs: int = 256
+    max_num_batched_tokens: int = 8192      # prefill token budget per step
+    max_model_len: Optional[int] = None     # None → "auto": fit KV budget
+    gpu_memory_utilization: float = 0.90
+    num_gpu_blocks: Optional[int] = None    # None → probe free VRAM
+    tensor_parallel_size: int = 1
+    enforce_eager: bool = False             # disable hipGraph decode capture
+    device: str = "cuda"
+    seed: int = 0
+    # decode graph buckets (batch sizes to capture)
+    graph_batch_sizes: tuple = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256)
+
+    def max_blocks_per_seq(self, max_len: int) -> int:
+        return (max_len + self.block_size - 1) // self.block_size

@@ -1,0 +1,191 @@
+"""Llama-architecture causal LM built on kaito_amd's gfx950 HIP ops.
+
+Covers Llama-3 8B/70B (BASELINE configs #2/#3), Mistral, Qwen-dense and the
+Phi-4-mini class (partial rotary). GEMMs go through hipBLASLt
+(torch.nn.functional.linear); everything between GEMMs is our fused HIP
+kernels (rmsnorm, rope, silu_mul, paged/prefill attention).
+
+Reference parity: this is the engine-side replacement for the vLLM model
+executor that KAITO launches via presets/workspace/inference/vllm/
+inference_api.py.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..engine.config import ModelConfig
+from ..parallel.state import get_state
+from ..parallel.layers import (ColumnParallelLinear, RowParallelLinear,
+                               VocabParallelEmbedding)
+
+
+@dataclass
+class AttnMetadata:
+    """Per-batch attention metadata. Exactly one of (prefill, decode)."""
+    is_prefill: bool
+    slot_mapping: torch.Tensor                  # [T] int64
+    # prefill:
+    cu_seqlens: Optional[torch.Tensor] = None   # [B+1] int32
+    max_seqlen: int = 0
+    # decode:
+    block_tables: Optional[torch.Tensor] = None  # [T, max_blocks] int32
+    seq_lens: Optional[torch.Tensor] = None      # [T] int32
+
+
+def build_cos_sin_cache(cfg: ModelConfig, device, max_pos: Optional[int] = None
+                        ) -> torch.Tensor:
+    """[max_pos, rot_dim] f32 = [cos | sin] table."""
+    rot = cfg.rotary_dim
+    max_pos = max_pos or cfg.max_position
+    inv = 1.0 / (cfg.rope_theta ** (
+        torch.arange(0, rot, 2, dtype=torch.float64, device=device) / rot))
+    t = torch.arange(max_pos, dtype=torch.float64, device=device)
+    freqs = torch.outer(t, inv)
+    return torch.cat([freqs.cos(), freqs.sin()], dim=-1).float().contiguous()
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        tp = get_state().tp_size
+        self.cfg = cfg
+        self.head_dim = cfg.head_dim
+        self.num_heads = cfg.num_heads // tp
+        self.num_kv_heads = max(cfg.num_kv_heads // tp, 1)
+        self.scale = 1.0 / math.sqrt(self.head_dim)
+        h = cfg.hidden_size
+        self.qkv_proj = ColumnParallelLinear(
+            h, (cfg.num_heads + 2 * cfg.num_kv_heads) * cfg.head_dim,
+            bias=cfg.attention_bias, dtype=cfg.dtype)
+        self.o_proj = RowParallelLinear(
+            cfg.num_heads * cfg.head_dim, h, bias=False, dtype=cfg.dtype)
+        self.q_size = self.num_heads * self.head_dim
+        self.kv_size = self.num_kv_heads * self.head_dim
+
+    def forward(self, x: torch.Tensor, positions: torch.Tensor,
+                kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]],
+                meta: AttnMetadata, cos_sin: torch.Tensor) -> torch.Tensor:
+        qkv = self.qkv_proj(x)
+        q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
+        q = q.contiguous()
+        k = k.contiguous()
+        q, k = ops.rotary_embedding(positions, q, k, self.head_dim, cos_sin)
+        T = x.size(0)
+        qh = q.view(T, self.num_heads, self.head_dim)
+        kh = k.view(T, self.num_kv_heads, self.head_dim)
+        vh = v.view(T, self.num_kv_heads, self.head_dim).contiguous()
+        if kv_cache is not None:
+            ops.reshape_and_cache(kh, vh, kv_cache[0], kv_cache[1],
+                                  meta.slot_mapping)
+        if meta.is_prefill:
+            out = ops.prefill_attention(qh, kh, vh, meta.cu_seqlens, self.scale,
+                                        meta.max_seqlen)
+        else:
+            out = ops.paged_attention(qh, kv_cache[0], kv_cache[1],
+                                      meta.block_tables, meta.seq_lens,
+                                      self.scale)
+        return self.o_proj(out.view(T, -1))
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.gate_up_proj = ColumnParallelLinear(
+            cfg.hidden_size, 2 * cfg.intermediate_size, dtype=cfg.dtype)
+        self.down_proj = RowParallelLinear(
+            cfg.intermediate_size, cfg.hidden_size, dtype=cfg.dtype)
+        tp = get_state().tp_size
+        self.inter_per_rank = cfg.intermediate_size // tp
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        gu = self.gate_up_proj(x)
+        return self.down_proj(ops.silu_and_mul(gu))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.self_attn = LlamaAttention(cfg)
+        self.mlp = LlamaMLP(cfg)
+        self.input_layernorm = nn.Parameter(
+            torch.empty(cfg.hidden_size, dtype=cfg.dtype), requires_grad=False)
+        self.post_attention_layernorm = nn.Parameter(
+            torch.empty(cfg.hidden_size, dtype=cfg.dtype), requires_grad=False)
+
+    def forward(self, hidden, residual, positions, kv_cache, meta, cos_sin):
+        if residual is None:
+            residual = hidden
+            hidden = ops.rms_norm(hidden, self.input_layernorm, self.cfg.rms_eps)
+        else:
+            hidden, residual = ops.fused_add_rms_norm(
+                hidden, residual, self.input_layernorm, self.cfg.rms_eps)
+        hidden = self.self_attn(hidden, positions, kv_cache, meta, cos_sin)
+        hidden, residual = ops.fused_add_rms_norm(
+            hidden, residual, self.post_attention_layernorm, self.cfg.rms_eps)
+        hidden = self.mlp(hidden)
+        return hidden, residual
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed_tokens = VocabParallelEmbedding(
+            cfg.vocab_size, cfg.hidden_size, dtype=cfg.dtype)
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(cfg) for _ in range(cfg.num_layers)])
+        self.norm = nn.Parameter(
+            torch.empty(cfg.hidden_size, dtype=cfg.dtype), requires_grad=False)
+        if cfg.tie_word_embeddings:
+            self.lm_head = None
+        else:
+            self.lm_head = ColumnParallelLinear(
+                cfg.hidden_size, cfg.vocab_size, dtype=cfg.dtype)
+        self.register_buffer("cos_sin_cache", torch.empty(0), persistent=False)
+
+    def init_rope(self, device, max_pos: Optional[int] = None):
+        self.cos_sin_cache = build_cos_sin_cache(self.cfg, device, max_pos)
+
+    def forward(self, input_ids: torch.Tensor, positions: torch.Tensor,
+                kv_caches: Optional[List[Tuple[torch.Tensor, torch.Tensor]]],
+                meta: AttnMetadata) -> torch.Tensor:
+        hidden = self.embed_tokens(input_ids)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            kv = kv_caches[i] if kv_caches is not None else None
+            hidden, residual = layer(hidden, residual, positions, kv, meta,
+                                     self.cos_sin_cache)
+        hidden, _ = ops.fused_add_rms_norm(hidden, residual, self.norm,
+                                           self.cfg.rms_eps)
+        return hidden
+
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        from ..parallel.state import tp_all_gather
+        if self.lm_head is not None:
+            logits = self.lm_head(hidden)
+        else:
+            logits = torch.nn.functional.linear(hidden, self.embed_tokens.weight)
+        return tp_all_gather(logits, dim=-1)
+
+    @torch.no_grad()
+    def random_init(self, seed: int = 0):
+        """Random-init weights of the right architecture (no network; bench
+        contract: synthetic data / random weights)."""
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        for name, p in self.named_parameters():
+            if "layernorm" in name or name == "norm":
+                p.fill_(1.0)
+            else:
+                std = 0.02 if "embed" in name or "lm_head" in name else \
+                    0.02 / math.sqrt(2 * self.cfg.num_layers)
+                cpu = torch.empty(p.shape, dtype=torch.float32)
+                cpu.normal_(0, std, generator=gen)
+                p.copy_(cpu.to(p.dtype))
+        return self

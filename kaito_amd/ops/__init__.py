@@ -1,0 +1,145 @@
+"""kaito_amd.ops — gfx950 HIP kernels with a CPU fp32 reference fallback.
+
+Dispatch policy (engine contract):
+  * CUDA/ROCm tensors → the in-tree HIP extension. If the extension is
+    missing on a GPU machine this raises loudly — there is NO silent
+    eager fallback on GPU.
+  * CPU tensors → torch_ref reference implementations (tests / no-GPU dev).
+"""
+from __future__ import annotations
+
+import torch
+
+from . import torch_ref
+from ._build import SO_PATH
+
+_LOADED = False
+
+
+def extension_available() -> bool:
+    return SO_PATH.exists()
+
+
+def load_extension() -> None:
+    """Load (and lazily build) the HIP extension library."""
+    global _LOADED
+    if _LOADED:
+        return
+    if not SO_PATH.exists():
+        from ._build import build
+        build()
+    torch.ops.load_library(str(SO_PATH))
+    _LOADED = True
+
+
+def _require_ext() -> None:
+    if not _LOADED:
+        if not torch.cuda.is_available():
+            raise RuntimeError(
+                "kaito_amd.ops: got a CUDA tensor but torch.cuda is unavailable")
+        load_extension()
+
+
+# ---------------------------------------------------------------- public ops
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty_like(x)
+        torch.ops.kaito.rms_norm(out, x, weight, eps)
+        return out
+    return torch_ref.rms_norm(x, weight, eps)
+
+
+def fused_add_rms_norm(x: torch.Tensor, residual: torch.Tensor,
+                       weight: torch.Tensor, eps: float):
+    """Returns (normed, residual). On GPU, `residual` is updated IN PLACE
+    (residual += x) and returned."""
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty_like(x)
+        torch.ops.kaito.fused_add_rms_norm(out, x, residual, weight, eps)
+        return out, residual
+    return torch_ref.fused_add_rms_norm(x, residual, weight, eps)
+
+
+def rotary_embedding(positions: torch.Tensor, q: torch.Tensor, k: torch.Tensor,
+                     head_dim: int, cos_sin_cache: torch.Tensor):
+    """In-place on GPU; returns (q, k)."""
+    if q.is_cuda:
+        _require_ext()
+        torch.ops.kaito.rotary_embedding(positions, q, k, head_dim, cos_sin_cache)
+        return q, k
+    return torch_ref.rotary_embedding(positions, q, k, head_dim, cos_sin_cache)
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext()
+        d = x.size(-1) // 2
+        out = torch.empty(*x.shape[:-1], d, dtype=x.dtype, device=x.device)
+        torch.ops.kaito.silu_and_mul(out, x)
+        return out
+    return torch_ref.silu_and_mul(x)
+
+
+def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, slot_mapping: torch.Tensor) -> None:
+    if k.is_cuda:
+        _require_ext()
+        torch.ops.kaito.reshape_and_cache(
+            k.reshape(k.size(0), -1), v.reshape(v.size(0), -1),
+            k_cache, v_cache, slot_mapping)
+        return
+    torch_ref.reshape_and_cache(
+        k.reshape(k.size(0), k_cache.size(1), k_cache.size(3)),
+        v.reshape(v.size(0), v_cache.size(1), v_cache.size(3)),
+        k_cache, v_cache, slot_mapping)
+
+
+def paged_attention(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
+                    block_tables: torch.Tensor, seq_lens: torch.Tensor,
+                    scale: float) -> torch.Tensor:
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty_like(q)
+        torch.ops.kaito.paged_attention(out, q, k_cache, v_cache,
+                                        block_tables, seq_lens, scale)
+        return out
+    return torch_ref.paged_attention(q, k_cache, v_cache, block_tables,
+                                     seq_lens, scale)
+
+
+def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                      cu_seqlens: torch.Tensor, scale: float,
+                      max_seqlen: int | None = None) -> torch.Tensor:
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty_like(q)
+        tile_seq, tile_qbase = _build_tiles(cu_seqlens)
+        torch.ops.kaito.prefill_attention(out, q, k, v, tile_seq, tile_qbase,
+                                          cu_seqlens, scale)
+        return out
+    return torch_ref.prefill_attention(q, k, v, cu_seqlens, scale)
+
+
+def mfma_tile_gemm(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    _require_ext()
+    return torch.ops.kaito.mfma_tile_gemm(a, b)
+
+
+QTILE = 64  # must match prefill_attention.hip
+
+
+def _build_tiles(cu_seqlens: torch.Tensor):
+    """Map varlen batch → list of 64-row Q tiles: (seq_idx, q_base)."""
+    cs = cu_seqlens.tolist()
+    seqs, bases = [], []
+    for i in range(len(cs) - 1):
+        L = cs[i + 1] - cs[i]
+        for q0 in range(0, L, QTILE):
+            seqs.append(i)
+            bases.append(q0)
+    dev = cu_seqlens.device
+    return (torch.tensor(seqs, dtype=torch.int32, device=dev),
+            torch.tensor(bases, dtype=torch.int32, device=dev))

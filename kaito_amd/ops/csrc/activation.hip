@@ -1,0 +1,49 @@
+// Fused SiLU-and-mul (SwiGLU gate) for gfx950: out = silu(x[:, :I]) * x[:, I:].
+// Memory-bound; short8-vectorized grid-stride loop.
+#include "common.h"
+#include <torch/library.h>
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace kaito {
+
+__global__ void silu_and_mul_kernel(
+    short* __restrict__ out,        // [T, I]
+    const short* __restrict__ x,    // [T, 2I]  (gate | up)
+    const int64_t T, const int I) {
+  const int nvec = I / 8;
+  const int64_t total = T * (int64_t)nvec;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = idx / nvec;
+    const int v = (int)(idx % nvec);
+    const short8_t g = *reinterpret_cast<const short8_t*>(x + row * 2 * I + v * 8);
+    const short8_t u = *reinterpret_cast<const short8_t*>(x + row * 2 * I + I + v * 8);
+    short8_t o;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float gf = bf16_to_f32(g[j]);
+      float uf = bf16_to_f32(u[j]);
+      float s = gf / (1.f + __expf(-gf));
+      o[j] = f32_to_bf16(s * uf);
+    }
+    *reinterpret_cast<short8_t*>(out + row * I + v * 8) = o;
+  }
+}
+
+void silu_and_mul(at::Tensor out, at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16);
+  TORCH_CHECK(x.is_contiguous() && out.is_contiguous());
+  const int I2 = x.size(-1);
+  TORCH_CHECK(I2 % 16 == 0);
+  const int I = I2 / 2;
+  const int64_t T = x.numel() / I2;
+  auto stream = at::hip::getCurrentHIPStream();
+  const int block = 256;
+  const int64_t work = T * (I / 8);
+  const int grid = (int)std::min<int64_t>((work + block - 1) / block, 2048);
+  hipLaunchKernelGGL(silu_and_mul_kernel, dim3(grid), dim3(block), 0, stream,
+      (short*)out.data_ptr(), (const short*)x.data_ptr(), T, I);
+}
+
+}  // namespace kaito

@@ -1,0 +1,44 @@
+// torch op registration for kaito_amd's gfx950 kernels.
+// Loaded via torch.ops.load_library; no pybind dependency.
+#include <torch/library.h>
+#include <ATen/ATen.h>
+
+namespace kaito {
+void rms_norm(at::Tensor out, at::Tensor input, at::Tensor weight, double eps);
+void fused_add_rms_norm(at::Tensor out, at::Tensor input, at::Tensor residual,
+                        at::Tensor weight, double eps);
+void rotary_embedding(at::Tensor positions, at::Tensor q, at::Tensor k,
+                      int64_t head_dim, at::Tensor cos_sin_cache);
+void silu_and_mul(at::Tensor out, at::Tensor x);
+void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
+                       at::Tensor v_cache, at::Tensor slot_mapping);
+void paged_attention(at::Tensor out, at::Tensor query, at::Tensor k_cache,
+                     at::Tensor v_cache, at::Tensor block_tables,
+                     at::Tensor seq_lens, double scale);
+void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
+                       at::Tensor tile_seq, at::Tensor tile_qbase,
+                       at::Tensor cu_seqlens, double scale);
+at::Tensor mfma_tile_gemm(at::Tensor a, at::Tensor b);
+}  // namespace kaito
+
+TORCH_LIBRARY(kaito, m) {
+  m.def("rms_norm(Tensor(a!) out, Tensor input, Tensor weight, float eps) -> ()");
+  m.def("fused_add_rms_norm(Tensor(a!) out, Tensor input, Tensor(b!) residual, Tensor weight, float eps) -> ()");
+  m.def("rotary_embedding(Tensor positions, Tensor(a!) q, Tensor(b!) k, int head_dim, Tensor cos_sin_cache) -> ()");
+  m.def("silu_and_mul(Tensor(a!) out, Tensor x) -> ()");
+  m.def("reshape_and_cache(Tensor k, Tensor v, Tensor(a!) k_cache, Tensor(b!) v_cache, Tensor slot_mapping) -> ()");
+  m.def("paged_attention(Tensor(a!) out, Tensor query, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, float scale) -> ()");
+  m.def("prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens, float scale) -> ()");
+  m.def("mfma_tile_gemm(Tensor a, Tensor b) -> Tensor");
+}
+
+TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
+  m.impl("rms_norm", &kaito::rms_norm);
+  m.impl("fused_add_rms_norm", &kaito::fused_add_rms_norm);
+  m.impl("rotary_embedding", &kaito::rotary_embedding);
+  m.impl("silu_and_mul", &kaito::silu_and_mul);
+  m.impl("reshape_and_cache", &kaito::reshape_and_cache);
+  m.impl("paged_attention", &kaito::paged_attention);
+  m.impl("prefill_attention", &kaito::prefill_attention);
+  m.impl("mfma_tile_gemm", &kaito::mfma_tile_gemm);
+}

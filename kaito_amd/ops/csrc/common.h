@@ -1,0 +1,90 @@
+// Common device helpers for kaito_amd CDNA4 (gfx950 / MI355X) kernels.
+//
+// Design notes (MI355X-first):
+//  - wavefront = 64 lanes; all cross-lane reductions use 64-wide shuffles.
+//  - bf16 global loads are ALWAYS vectorized (short4/short8 reinterpret):
+//    hipcc does not auto-vectorize scalar bf16 loads (2-2.5x penalty).
+//  - LDS = 32 banks x 4B; row-major tiles with 128B-multiple strides are
+//    32-way conflicts on ds_read_b128 -> XOR-swizzle byte offsets.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <stdint.h>
+
+#define KAITO_WAVE 64
+#define KAITO_DEV __device__ __forceinline__
+
+// ---- vector types for wide loads --------------------------------------
+typedef short  short4_t  __attribute__((ext_vector_type(4)));   // 8B  = 4 bf16
+typedef short  short8_t  __attribute__((ext_vector_type(8)));   // 16B = 8 bf16
+typedef float  float4_t  __attribute__((ext_vector_type(4)));
+typedef float  float2_t  __attribute__((ext_vector_type(2)));
+typedef short  bf16x8    __attribute__((ext_vector_type(8)));   // MFMA A/B frag (4 VGPRs)
+typedef float  f32x4     __attribute__((ext_vector_type(4)));   // MFMA C/D frag 16x16
+typedef __hip_bfloat16 bf16_t;
+
+KAITO_DEV float bf16_to_f32(short u) {
+  union { float f; uint32_t i; } v;
+  v.i = (uint32_t)(uint16_t)u << 16;
+  return v.f;
+}
+
+KAITO_DEV short f32_to_bf16(float f) {
+  union { float f; uint32_t i; } v;
+  v.f = f;
+  // round-to-nearest-even
+  uint32_t rounding = 0x7FFF + ((v.i >> 16) & 1);
+  return (short)((v.i + rounding) >> 16);
+}
+
+// ---- wave-level reductions (64 lanes) ---------------------------------
+KAITO_DEV float wave_reduce_sum(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, 64);
+  return x;
+}
+
+KAITO_DEV float wave_reduce_max(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, 64));
+  return x;
+}
+
+// Reduce within a group of `W` consecutive lanes (W power of 2, <= 64).
+template <int W>
+KAITO_DEV float group_reduce_sum(float x) {
+#pragma unroll
+  for (int off = W / 2; off > 0; off >>= 1) x += __shfl_xor(x, off, 64);
+  return x;
+}
+
+template <int W>
+KAITO_DEV float group_reduce_max(float x) {
+#pragma unroll
+  for (int off = W / 2; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, 64));
+  return x;
+}
+
+// Block-level reduce over all waves via LDS scratch (caller provides
+// __shared__ float scratch[nwaves]); valid for blockDim.x % 64 == 0.
+KAITO_DEV float block_reduce_sum(float x, float* scratch) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
+  x = wave_reduce_sum(x);
+  if (lane == 0) scratch[wave] = x;
+  __syncthreads();
+  x = (lane < nwaves) ? scratch[lane] : 0.f;
+  x = wave_reduce_sum(x);  // cheap: only first nwaves lanes carry data
+  return x;
+}
+
+#define HIP_CHECK_KAITO(expr)                                              \
+  do {                                                                     \
+    hipError_t _e = (expr);                                                \
+    if (_e != hipSuccess) {                                                \
+      TORCH_CHECK(false, "HIP error: ", hipGetErrorString(_e));            \
+    }                                                                      \
+  } while (0)

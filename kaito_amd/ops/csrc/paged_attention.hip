@@ -1,0 +1,220 @@
+// Paged-attention DECODE kernel for gfx950 (MI355X).
+//
+// Replaces the vLLM decode attention the reference delegates to
+// (SURVEY.md §2.3 row "Paged-attention prefill/decode"). MI355X-first
+// design, NOT a CUDA port:
+//  - memory-bound regime: one (seq, kv_head) workgroup streams the whole
+//    K/V history once and shares it across the GQA group's Q heads.
+//  - wave = 64 lanes: lane (tg,dc) = (lane>>3, lane&7) → 8 tokens/wave-step,
+//    16 dims/lane → 32 B/lane vectorized bf16 loads (2KB contiguous per
+//    wave-step from one [block, kv_head] slab).
+//  - flash-decoding online softmax per wave, LDS cross-wave combine,
+//    deferred rescale when the running max doesn't grow (T13).
+//
+// Cache layout: [num_blocks, KV_HEADS, BLOCK_SIZE, HEAD_DIM] bf16.
+#include "common.h"
+#include <torch/library.h>
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace kaito {
+
+template <int D, int G, int BS>
+__global__ __launch_bounds__(256, 2) void paged_attention_kernel(
+    short* __restrict__ out,             // [T, QH, D] bf16
+    const short* __restrict__ q,         // [T, QH, D] bf16
+    const short* __restrict__ k_cache,   // [B, KH, BS, D]
+    const short* __restrict__ v_cache,
+    const int* __restrict__ block_tables,// [T, max_blocks]
+    const int* __restrict__ seq_lens,    // [T]
+    const float scale, const int KH, const int max_blocks) {
+  constexpr int DL = D / 8;       // dims per lane (16 for D=128)
+  constexpr int NW = 4;           // waves per workgroup
+  const int seq = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int tg = lane >> 3;       // token slot in chunk (0..7)
+  const int dc = lane & 7;        // dim chunk (0..7)
+  const int seq_len = seq_lens[seq];
+  const int QH = KH * G;
+
+  const int* bt = block_tables + (int64_t)seq * max_blocks;
+
+  // ---- preload Q for this kv-head's group (f32) ----
+  float qreg[G][DL];
+#pragma unroll
+  for (int g = 0; g < G; g++) {
+    const short8_t* qp = reinterpret_cast<const short8_t*>(
+        q + ((int64_t)seq * QH + kvh * G + g) * D + dc * DL);
+#pragma unroll
+    for (int vv = 0; vv < DL / 8; vv++) {
+      short8_t x = qp[vv];
+#pragma unroll
+      for (int j = 0; j < 8; j++) qreg[g][vv * 8 + j] = bf16_to_f32(x[j]);
+    }
+  }
+
+  float m[G], l[G], acc[G][DL];
+#pragma unroll
+  for (int g = 0; g < G; g++) {
+    m[g] = -1e30f; l[g] = 0.f;
+#pragma unroll
+    for (int j = 0; j < DL; j++) acc[g][j] = 0.f;
+  }
+
+  const int nchunks = (seq_len + 7) / 8;  // 8 tokens per wave-step
+  for (int c = wave; c < nchunks; c += NW) {
+    const int tok = c * 8 + tg;
+    const bool valid = tok < seq_len;
+    const int tok_c = valid ? tok : (seq_len - 1);
+    const int blk = bt[tok_c / BS];
+    const int64_t base = (((int64_t)blk * KH + kvh) * BS + (tok_c % BS)) * D + dc * DL;
+
+    // ---- K tile: 16 dims/lane → f32 ----
+    float kreg[DL];
+    {
+      const short8_t* kp = reinterpret_cast<const short8_t*>(k_cache + base);
+#pragma unroll
+      for (int vv = 0; vv < DL / 8; vv++) {
+        short8_t x = kp[vv];
+#pragma unroll
+        for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(x[j]);
+      }
+    }
+
+    float s[G];
+#pragma unroll
+    for (int g = 0; g < G; g++) {
+      float p = 0.f;
+#pragma unroll
+      for (int j = 0; j < DL; j++) p += qreg[g][j] * kreg[j];
+      p = group_reduce_sum<8>(p) * scale;   // dot over 8 dc-lanes
+      s[g] = valid ? p : -1e30f;
+    }
+
+    // ---- V tile (reuse kreg registers) ----
+    {
+      const short8_t* vp = reinterpret_cast<const short8_t*>(v_cache + base);
+#pragma unroll
+      for (int vv = 0; vv < DL / 8; vv++) {
+        short8_t x = vp[vv];
+#pragma unroll
+        for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(x[j]);
+      }
+    }
+
+#pragma unroll
+    for (int g = 0; g < G; g++) {
+      // wave-max over the 8 token-groups (s uniform across dc lanes)
+      float cm = s[g];
+#pragma unroll
+      for (int off = 8; off < 64; off <<= 1) cm = fmaxf(cm, __shfl_xor(cm, off, 64));
+      if (cm > m[g]) {            // deferred rescale: skip when max static
+        const float corr = __expf(m[g] - cm);
+        l[g] *= corr;
+#pragma unroll
+        for (int j = 0; j < DL; j++) acc[g][j] *= corr;
+        m[g] = cm;
+      }
+      float p = __expf(s[g] - m[g]);        // masked lanes: exp(-inf)→0
+      float psum = p;
+#pragma unroll
+      for (int off = 8; off < 64; off <<= 1) psum += __shfl_xor(psum, off, 64);
+      l[g] += psum;
+#pragma unroll
+      for (int j = 0; j < DL; j++) acc[g][j] += p * kreg[j];
+    }
+  }
+
+  // ---- reduce acc across the 8 token-group lanes (same dc) ----
+#pragma unroll
+  for (int g = 0; g < G; g++)
+#pragma unroll
+    for (int j = 0; j < DL; j++)
+#pragma unroll
+      for (int off = 8; off < 64; off <<= 1)
+        acc[g][j] += __shfl_xor(acc[g][j], off, 64);
+
+  // ---- cross-wave combine via LDS ----
+  __shared__ float s_acc[NW][G][D];
+  __shared__ float s_ml[NW][G][2];
+  if (tg == 0) {
+#pragma unroll
+    for (int g = 0; g < G; g++) {
+#pragma unroll
+      for (int j = 0; j < DL; j++) s_acc[wave][g][dc * DL + j] = acc[g][j];
+      if (dc == 0) { s_ml[wave][g][0] = m[g]; s_ml[wave][g][1] = l[g]; }
+    }
+  }
+  __syncthreads();
+
+  // 256 threads: thread handles (g, d) pairs strided.
+  for (int idx = threadIdx.x; idx < G * D; idx += 256) {
+    const int g = idx / D;
+    const int d = idx % D;
+    float gm = -1e30f;
+#pragma unroll
+    for (int w = 0; w < NW; w++) gm = fmaxf(gm, s_ml[w][g][0]);
+    float num = 0.f, den = 0.f;
+#pragma unroll
+    for (int w = 0; w < NW; w++) {
+      const float e = __expf(s_ml[w][g][0] - gm);
+      num += e * s_acc[w][g][d];
+      den += e * s_ml[w][g][1];
+    }
+    out[((int64_t)seq * QH + kvh * G + g) * D + d] =
+        f32_to_bf16(num / fmaxf(den, 1e-20f));
+  }
+}
+
+#define PA_LAUNCH(D_, G_)                                                      \
+  hipLaunchKernelGGL((paged_attention_kernel<D_, G_, 16>), dim3(T, KH),        \
+      dim3(256), 0, stream, (short*)out.data_ptr(),                           \
+      (const short*)query.data_ptr(), (const short*)k_cache.data_ptr(),       \
+      (const short*)v_cache.data_ptr(), block_tables.data_ptr<int>(),         \
+      seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks)
+
+void paged_attention(at::Tensor out, at::Tensor query, at::Tensor k_cache,
+                     at::Tensor v_cache, at::Tensor block_tables,
+                     at::Tensor seq_lens, double scale) {
+  TORCH_CHECK(query.is_cuda() && query.dtype() == at::kBFloat16);
+  TORCH_CHECK(query.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(block_tables.dtype() == at::kInt && seq_lens.dtype() == at::kInt);
+  const int T = query.size(0);
+  const int QH = query.size(1);
+  const int D = query.size(2);
+  const int KH = k_cache.size(1);
+  const int BS = k_cache.size(2);
+  TORCH_CHECK(BS == 16, "block_size must be 16");
+  TORCH_CHECK(QH % KH == 0);
+  const int G = QH / KH;
+  const int max_blocks = block_tables.size(1);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (T == 0) return;
+  if (D == 128) {
+    switch (G) {
+      case 1: PA_LAUNCH(128, 1); break;
+      case 2: PA_LAUNCH(128, 2); break;
+      case 3: PA_LAUNCH(128, 3); break;
+      case 4: PA_LAUNCH(128, 4); break;
+      case 5: PA_LAUNCH(128, 5); break;
+      case 6: PA_LAUNCH(128, 6); break;
+      case 7: PA_LAUNCH(128, 7); break;
+      case 8: PA_LAUNCH(128, 8); break;
+      default: TORCH_CHECK(false, "unsupported GQA group ", G);
+    }
+  } else if (D == 64) {
+    switch (G) {
+      case 1: PA_LAUNCH(64, 1); break;
+      case 2: PA_LAUNCH(64, 2); break;
+      case 4: PA_LAUNCH(64, 4); break;
+      case 8: PA_LAUNCH(64, 8); break;
+      default: TORCH_CHECK(false, "unsupported GQA group ", G);
+    }
+  } else {
+    TORCH_CHECK(false, "unsupported head_dim ", D);
+  }
+}
+
+}  // namespace kaito

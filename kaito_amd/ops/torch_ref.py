@@ -1,0 +1,114 @@
+"""Pure-PyTorch fp32 reference implementations of every HIP op.
+
+Used (a) on CPU so the engine/scheduler stack is testable without a GPU and
+(b) as the numerics oracle for the GPU kernels (tests compare HIP bf16
+kernels against these fp32 references).
+"""
+from __future__ import annotations
+
+import torch
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    xf = x.float()
+    var = xf.pow(2).mean(-1, keepdim=True)
+    out = xf * torch.rsqrt(var + eps) * weight.float()
+    return out.to(x.dtype)
+
+
+def fused_add_rms_norm(x: torch.Tensor, residual: torch.Tensor,
+                       weight: torch.Tensor, eps: float):
+    """Returns (normed, new_residual). Residual update happens in bf16 to
+    match the kernel's residual-stream precision."""
+    new_res = (x.float() + residual.float()).to(x.dtype)
+    return rms_norm(new_res, weight, eps), new_res
+
+
+def rotary_embedding(positions: torch.Tensor, q: torch.Tensor, k: torch.Tensor,
+                     head_dim: int, cos_sin_cache: torch.Tensor):
+    """Neox/llama rotate-half RoPE. q: [T, QH*D] or [T, QH, D]; in-place-like
+    (returns rotated copies). cos_sin_cache: [max_pos, R] = [cos | sin]."""
+    rot = cos_sin_cache.size(1)
+    half = rot // 2
+    cs = cos_sin_cache[positions]          # [T, R]
+    cos = cs[:, :half].float()             # [T, half]
+    sin = cs[:, half:].float()
+
+    def _apply(t: torch.Tensor) -> torch.Tensor:
+        shp = t.shape
+        x = t.reshape(shp[0], -1, head_dim).float()
+        x1 = x[..., :half]
+        x2 = x[..., half:rot]
+        c = cos.unsqueeze(1)
+        s = sin.unsqueeze(1)
+        o1 = x1 * c - x2 * s
+        o2 = x2 * c + x1 * s
+        out = torch.cat([o1, o2, x[..., rot:]], dim=-1)
+        return out.reshape(shp).to(t.dtype)
+
+    return _apply(q), _apply(k)
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    d = x.size(-1) // 2
+    gate, up = x[..., :d].float(), x[..., d:].float()
+    return (torch.nn.functional.silu(gate) * up).to(x.dtype)
+
+
+def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, slot_mapping: torch.Tensor):
+    """k/v: [T, KH, D]; caches: [B, KH, BS, D]."""
+    bs = k_cache.size(2)
+    mask = slot_mapping >= 0
+    slots = slot_mapping[mask]
+    blocks = torch.div(slots, bs, rounding_mode="floor")
+    offs = slots % bs
+    k_cache[blocks, :, offs] = k[mask].to(k_cache.dtype)
+    v_cache[blocks, :, offs] = v[mask].to(v_cache.dtype)
+
+
+def paged_attention(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
+                    block_tables: torch.Tensor, seq_lens: torch.Tensor,
+                    scale: float) -> torch.Tensor:
+    """q: [T, QH, D] (one token/seq). Returns [T, QH, D]."""
+    T, QH, D = q.shape
+    KH = k_cache.size(1)
+    BS = k_cache.size(2)
+    G = QH // KH
+    out = torch.empty_like(q)
+    for i in range(T):
+        L = int(seq_lens[i])
+        nb = (L + BS - 1) // BS
+        blocks = block_tables[i, :nb].long()
+        keys = k_cache[blocks].permute(1, 0, 2, 3).reshape(KH, nb * BS, D)[:, :L]
+        vals = v_cache[blocks].permute(1, 0, 2, 3).reshape(KH, nb * BS, D)[:, :L]
+        qh = q[i].float()                                # [QH, D]
+        kx = keys.float().repeat_interleave(G, dim=0)    # [QH, L, D]
+        vx = vals.float().repeat_interleave(G, dim=0)
+        s = torch.einsum("hd,hld->hl", qh, kx) * scale
+        p = torch.softmax(s, dim=-1)
+        out[i] = torch.einsum("hl,hld->hd", p, vx).to(q.dtype)
+    return out
+
+
+def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                      cu_seqlens: torch.Tensor, scale: float) -> torch.Tensor:
+    """Varlen causal attention. q: [T, QH, D], k/v: [T, KH, D]."""
+    T, QH, D = q.shape
+    KH = k.size(1)
+    G = QH // KH
+    out = torch.empty_like(q)
+    cs = cu_seqlens.tolist()
+    for b in range(len(cs) - 1):
+        s0, s1 = cs[b], cs[b + 1]
+        L = s1 - s0
+        qs = q[s0:s1].float().transpose(0, 1)                      # [QH, L, D]
+        ks = k[s0:s1].float().transpose(0, 1).repeat_interleave(G, 0)
+        vs = v[s0:s1].float().transpose(0, 1).repeat_interleave(G, 0)
+        att = torch.einsum("hqd,hkd->hqk", qs, ks) * scale
+        mask = torch.triu(torch.ones(L, L, dtype=torch.bool, device=q.device), 1)
+        att = att.masked_fill(mask, float("-inf"))
+        p = torch.softmax(att, dim=-1)
+        o = torch.einsum("hqk,hkd->hqd", p, vs)
+        out[s0:s1] = o.transpose(0, 1).to(q.dtype)
+    return out

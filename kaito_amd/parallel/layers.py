@@ -1,0 +1,83 @@
+"""Tensor-parallel linear layers (RCCL over xGMI).
+
+Column-parallel: weight split along output dim, no comm on forward.
+Row-parallel: weight split along input dim, all-reduce on forward.
+At tp=1 these are plain GEMMs (hipBLASLt via torch.nn.functional.linear).
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .state import get_state, tp_all_reduce
+
+
+class ColumnParallelLinear(nn.Module):
+    def __init__(self, in_features: int, out_features: int, bias: bool = False,
+                 dtype: torch.dtype = torch.bfloat16):
+        super().__init__()
+        tp = get_state().tp_size
+        assert out_features % tp == 0, (out_features, tp)
+        self.in_features = in_features
+        self.out_features_per_rank = out_features // tp
+        self.weight = nn.Parameter(
+            torch.empty(self.out_features_per_rank, in_features, dtype=dtype),
+            requires_grad=False)
+        self.bias = nn.Parameter(
+            torch.empty(self.out_features_per_rank, dtype=dtype),
+            requires_grad=False) if bias else None
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F.linear(x, self.weight, self.bias)
+
+
+class RowParallelLinear(nn.Module):
+    def __init__(self, in_features: int, out_features: int, bias: bool = False,
+                 dtype: torch.dtype = torch.bfloat16):
+        super().__init__()
+        tp = get_state().tp_size
+        assert in_features % tp == 0, (in_features, tp)
+        self.in_features_per_rank = in_features // tp
+        self.out_features = out_features
+        self.weight = nn.Parameter(
+            torch.empty(out_features, self.in_features_per_rank, dtype=dtype),
+            requires_grad=False)
+        # bias added once (after reduce) on rank 0's shard only
+        self.bias = nn.Parameter(
+            torch.empty(out_features, dtype=dtype),
+            requires_grad=False) if bias else None
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        out = F.linear(x, self.weight)
+        out = tp_all_reduce(out)
+        if self.bias is not None:
+            out = out + self.bias
+        return out
+
+
+class VocabParallelEmbedding(nn.Module):
+    """Embedding split along vocab; out-of-shard ids contribute 0, summed by
+    all-reduce."""
+
+    def __init__(self, num_embeddings: int, embedding_dim: int,
+                 dtype: torch.dtype = torch.bfloat16):
+        super().__init__()
+        st = get_state()
+        tp = st.tp_size
+        assert num_embeddings % tp == 0
+        self.per_rank = num_embeddings // tp
+        self.start = st.tp_rank * self.per_rank
+        self.weight = nn.Parameter(
+            torch.empty(self.per_rank, embedding_dim, dtype=dtype),
+            requires_grad=False)
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        if get_state().tp_size == 1:
+            return F.embedding(ids, self.weight)
+        local = ids - self.start
+        mask = (local < 0) | (local >= self.per_rank)
+        local = local.clamp(0, self.per_rank - 1)
+        out = F.embedding(local, self.weight)
+        out = out.masked_fill(mask.unsqueeze(-1), 0)
+        return tp_all_reduce(out)

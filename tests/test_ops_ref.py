@@ -1,0 +1,95 @@
+"""CPU tests of the reference op implementations (the numerics oracle) and
+their basic invariants. GPU kernels are compared against these in
+tests/test_gpu_kernels.py.
+"""
+import math
+
+import pytest
+import torch
+
+from kaito_amd.ops import torch_ref as R
+
+
+def test_rms_norm_matches_manual():
+    x = torch.randn(4, 64)
+    w = torch.randn(64)
+    out = R.rms_norm(x, w, 1e-5)
+    row = x[0]
+    expect = row / math.sqrt(float((row ** 2).mean()) + 1e-5) * w
+    assert torch.allclose(out[0], expect, atol=1e-5)
+
+
+def test_fused_add_rms_norm_updates_residual():
+    x = torch.randn(4, 64, dtype=torch.bfloat16)
+    res = torch.randn(4, 64, dtype=torch.bfloat16)
+    w = torch.ones(64, dtype=torch.bfloat16)
+    out, new_res = R.fused_add_rms_norm(x, res, w, 1e-5)
+    assert torch.allclose(new_res.float(), (x.float() + res.float()), atol=0.02)
+    assert out.shape == x.shape
+
+
+def test_rope_rotates_pairs():
+    torch.manual_seed(0)
+    D = 8
+    cache = torch.cat([torch.zeros(4, D // 2), torch.ones(4, D // 2)], dim=-1)
+    # cos=0, sin=1 → o1 = -x2, o2 = x1
+    q = torch.randn(2, 1 * D)
+    k = torch.randn(2, 1 * D)
+    pos = torch.tensor([0, 1])
+    q2, k2 = R.rotary_embedding(pos, q.clone(), k.clone(), D, cache)
+    qq = q.view(2, 1, D)
+    assert torch.allclose(q2.view(2, 1, D)[..., :4], -qq[..., 4:], atol=1e-5)
+    assert torch.allclose(q2.view(2, 1, D)[..., 4:], qq[..., :4], atol=1e-5)
+
+
+def test_silu_and_mul():
+    x = torch.randn(3, 32)
+    out = R.silu_and_mul(x)
+    g, u = x[..., :16], x[..., 16:]
+    assert torch.allclose(out, torch.nn.functional.silu(g) * u, atol=1e-5)
+
+
+def test_reshape_and_cache_roundtrip():
+    T, KH, D, BS, NB = 5, 2, 16, 4, 8
+    k = torch.randn(T, KH, D)
+    v = torch.randn(T, KH, D)
+    kc = torch.zeros(NB, KH, BS, D)
+    vc = torch.zeros(NB, KH, BS, D)
+    slots = torch.tensor([0, 1, 5, 17, -1])
+    R.reshape_and_cache(k, v, kc, vc, slots)
+    assert torch.allclose(kc[0, :, 0], k[0])
+    assert torch.allclose(kc[0, :, 1], k[1])
+    assert torch.allclose(kc[1, :, 1], k[2])
+    assert torch.allclose(vc[4, :, 1], v[3])
+    assert kc[4, :, 2].abs().sum() == 0  # slot -1 skipped
+
+
+def test_paged_attention_equals_dense():
+    torch.manual_seed(0)
+    T, QH, KH, D, BS, NB = 2, 4, 2, 16, 4, 16
+    L = [7, 10]
+    kc = torch.randn(NB, KH, BS, D)
+    vc = torch.randn(NB, KH, BS, D)
+    bt = torch.tensor([[0, 1, 2, 0], [3, 4, 5, 0]], dtype=torch.int32)
+    q = torch.randn(T, QH, D)
+    out = R.paged_attention(q, kc, vc, bt, torch.tensor(L, dtype=torch.int32), 0.25)
+    # manual for seq 0, head 0 (kv head 0)
+    keys = torch.cat([kc[0, 0], kc[1, 0]], 0)[:7]
+    vals = torch.cat([vc[0, 0], vc[1, 0]], 0)[:7]
+    s = (q[0, 0] @ keys.T) * 0.25
+    expect = torch.softmax(s, -1) @ vals
+    assert torch.allclose(out[0, 0], expect, atol=1e-4)
+
+
+def test_prefill_attention_causal():
+    torch.manual_seed(0)
+    QH, KH, D = 4, 2, 16
+    cu = torch.tensor([0, 5, 12], dtype=torch.int32)
+    T = 12
+    q = torch.randn(T, QH, D)
+    k = torch.randn(T, KH, D)
+    v = torch.randn(T, KH, D)
+    out = R.prefill_attention(q, k, v, cu, 0.25)
+    # row 0 of each seq attends only to itself → out = v (broadcast over group)
+    assert torch.allclose(out[0, 0], v[0, 0], atol=1e-4)
+    assert torch.allclose(out[5, 3], v[5, 1], atol=1e-4)  # head 3 → kv head 1
