@@ -1,0 +1,169 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark (driver contract).
+
+Measures the reference's headline metric (BASELINE.json): output tokens/sec
+(+ p50 TTFT) for Llama-3-8B TP=1, saturated continuous batching, synthetic
+requests of input 200 / output 200 tokens (the reference's Phi-4 CSV
+methodology, website/docs/gpu-benchmarks.md), random-init weights.
+
+Scaling: weak — each rank runs an independent TP=1 engine replica (the
+reference's DP tier, pkg/model/interface.go:547-555); `value` aggregates
+output tokens/sec over all N GPUs.
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W]
+Multi-GPU (driver):  python -m torch.distributed.run --nnodes=1
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=300)
+    p.add_argument("--warmup", type=int, default=40)
+    p.add_argument("--model", default="llama-3-8b")
+    p.add_argument("--tp", type=int, default=1)
+    p.add_argument("--max-num-seqs", type=int, default=256)
+    p.add_argument("--in-tokens", type=int, default=200)
+    p.add_argument("--out-tokens", type=int, default=200)
+    p.add_argument("--eager", action="store_true")
+    p.add_argument("--profile", action="store_true",
+                   help="skip barriers/json (for rocprofv3 single-rank runs)")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+
+    from kaito_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from kaito_amd.models import get_model_config
+    from kaito_amd.parallel import state as ps
+
+    ps.init_parallel(tp_size=args.tp)
+
+    mc = get_model_config(args.model)
+    max_len = args.in_tokens + args.out_tokens + 16
+    cfg = EngineConfig(
+        model=mc,
+        device="cuda" if torch.cuda.is_available() else "cpu",
+        max_num_seqs=args.max_num_seqs,
+        max_model_len=max_len,
+        tensor_parallel_size=args.tp,
+        enforce_eager=args.eager,
+        seed=1234 + rank,
+    )
+    eng = LLMEngine(cfg)
+    if not args.eager:
+        eng.capture_graphs()
+
+    import numpy as np
+    rng = np.random.default_rng(42 + rank)
+
+    def new_prompt():
+        return rng.integers(10, mc.vocab_size - 10, args.in_tokens).tolist()
+
+    sp = SamplingParams(max_tokens=args.out_tokens, ignore_eos=True)
+
+    # saturate: queue 1.5x max_num_seqs requests; refill on finish.
+    inflight_target = int(args.max_num_seqs * 1.5)
+    for _ in range(inflight_target):
+        eng.add_request(new_prompt(), sp)
+    inflight = inflight_target
+
+    def run_steps(n):
+        nonlocal inflight
+        fin = []
+        for _ in range(n):
+            done = eng.step()
+            fin.extend(done)
+            for _ in done:
+                eng.add_request(new_prompt(), sp)
+        return fin
+
+    # ---- warmup ----
+    run_steps(args.warmup)
+
+    # ---- timed region ----
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    ps.barrier()
+    tok0 = eng.num_generation_tokens
+    t_start_mono = time.monotonic()
+    t0 = time.perf_counter()
+    finished = run_steps(args.steps)
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    ps.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    tokens = eng.num_generation_tokens - tok0
+    # TTFTs of requests that got their first token inside the timed region
+    ttfts = [
+        (s.first_token_time - s.arrival_time) * 1000.0
+        for s in eng.seqs.values()
+        if s.first_token_time is not None and s.first_token_time >= t_start_mono
+        and s.arrival_time >= t_start_mono
+    ]
+    ttft_p50 = statistics.median(ttfts) if ttfts else None
+
+    # aggregate across ranks: MAX(elapsed), SUM(tokens)
+    if world > 1:
+        import torch.distributed as dist
+        dev = "cuda" if torch.cuda.is_available() else "cpu"
+        te = torch.tensor([elapsed], device=dev)
+        tt = torch.tensor([float(tokens)], device=dev)
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        dist.all_reduce(tt, op=dist.ReduceOp.SUM)
+        elapsed = float(te.item())
+        tokens = int(tt.item())
+
+    if rank == 0:
+        value = tokens / elapsed
+        result = {
+            "metric": "output_tokens_per_sec",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "ttft_p50_ms": round(ttft_p50, 1) if ttft_p50 is not None else None,
+            "config": {
+                "model": args.model,
+                "parallelism": f"dp{world}(tp={args.tp})",
+                "max_num_seqs": args.max_num_seqs,
+                "in_tokens": args.in_tokens,
+                "out_tokens": args.out_tokens,
+                "seq_len": args.in_tokens + args.out_tokens,
+                "kv_blocks": eng.runner.num_gpu_blocks,
+                "decode_graphs": not args.eager,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if world > 1:
+        ps.destroy()
+
+
+if __name__ == "__main__":
+    main()

@@ -74,8 +74,11 @@ class LLMEngine:
         batch = self.scheduler.schedule()
         if batch is None:
             return []
-        logits = self.runner.execute(batch)
+        sampled = None if batch.is_prefill else getattr(self, "_last_decode_tokens", None)
+        logits = self.runner.execute(batch, sampled)
         tokens = self.sampler.sample(logits, batch.seqs)
+        if not batch.is_prefill:
+            self._last_decode_tokens = tokens
         tokens_cpu = tokens.tolist()
         finished: List[Sequence] = []
         for seq, tok in zip(batch.seqs, tokens_cpu):

@@ -164,33 +164,59 @@ class ModelRunner:
         return None
 
     @torch.no_grad()
-    def execute_decode(self, seqs: List[Sequence]) -> torch.Tensor:
-        """One token per sequence; returns hidden [B, H]."""
+    def execute_decode(self, seqs: List[Sequence],
+                       sampled: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """One token per sequence; returns hidden [B, H].
+
+        Fast path: when the running set is unchanged since the previous
+        decode step and `sampled` (last step's token ids, on device) is
+        given, all input updates happen device-side — no host round-trip
+        except dirty block-table rows (block-boundary crossings).
+        """
         if not hasattr(self, "_buf"):
             self._init_decode_buffers()
+            self._last_ids: List[int] = []
         bs = len(seqs)
-        ids = [s.last_token_id for s in seqs]
-        pos = [s.num_tokens - 1 for s in seqs]
-        slots = [self._slot(s, p) for s, p in zip(seqs, pos)]
-        lens = [s.num_tokens for s in seqs]
-
         b = self._buf
-        dev = self.device
-        b["input_ids"][:bs].copy_(
-            torch.tensor(ids, dtype=torch.long), non_blocking=True)
-        b["positions"][:bs].copy_(
-            torch.tensor(pos, dtype=torch.long), non_blocking=True)
-        b["slot_mapping"][:bs].copy_(
-            torch.tensor(slots, dtype=torch.long), non_blocking=True)
-        b["seq_lens"][:bs].copy_(
-            torch.tensor(lens, dtype=torch.int32), non_blocking=True)
-        bt = b["block_tables"]
-        # block tables: copy per-seq rows (padded)
-        flat = torch.zeros(bs, self._max_blocks, dtype=torch.int32)
-        for i, s in enumerate(seqs):
-            n = len(s.block_table)
-            flat[i, :n] = torch.tensor(s.block_table, dtype=torch.int32)
-        bt[:bs].copy_(flat, non_blocking=True)
+        ids_now = [s.seq_id for s in seqs]
+        reuse = (sampled is not None and ids_now == self._last_ids)
+        if reuse:
+            b["input_ids"][:bs].copy_(sampled[:bs])
+            b["positions"][:bs] += 1
+            b["seq_lens"][:bs] += 1
+            for i, s in enumerate(seqs):
+                if getattr(s, "_bt_dirty", False):
+                    n = len(s.block_table)
+                    b["block_tables"][i, :n].copy_(torch.tensor(
+                        s.block_table, dtype=torch.int32), non_blocking=True)
+                    s._bt_dirty = False
+            # slot = bt[i, pos // BS] * BS + pos % BS  (device-side gather)
+            pos = b["positions"][:bs]
+            blk = torch.gather(b["block_tables"][:bs].long(), 1,
+                               (pos // self.cfg.block_size).unsqueeze(1)).squeeze(1)
+            b["slot_mapping"][:bs].copy_(
+                blk * self.cfg.block_size + pos % self.cfg.block_size)
+        else:
+            ids = [s.last_token_id for s in seqs]
+            pos = [s.num_tokens - 1 for s in seqs]
+            slots = [self._slot(s, p) for s, p in zip(seqs, pos)]
+            lens = [s.num_tokens for s in seqs]
+            b["input_ids"][:bs].copy_(
+                torch.tensor(ids, dtype=torch.long), non_blocking=True)
+            b["positions"][:bs].copy_(
+                torch.tensor(pos, dtype=torch.long), non_blocking=True)
+            b["slot_mapping"][:bs].copy_(
+                torch.tensor(slots, dtype=torch.long), non_blocking=True)
+            b["seq_lens"][:bs].copy_(
+                torch.tensor(lens, dtype=torch.int32), non_blocking=True)
+            import numpy as np
+            flat = np.zeros((bs, self._max_blocks), dtype=np.int32)
+            for i, s in enumerate(seqs):
+                flat[i, :len(s.block_table)] = s.block_table
+                s._bt_dirty = False
+            b["block_tables"][:bs].copy_(
+                torch.from_numpy(flat), non_blocking=True)
+        self._last_ids = ids_now
 
         bucket = self._graph_bucket(bs) if self._graphs else None
         if bucket is not None:
@@ -205,10 +231,11 @@ class ModelRunner:
 
     # ------------------------------------------------------------- step
     @torch.no_grad()
-    def execute(self, batch: ScheduledBatch) -> torch.Tensor:
+    def execute(self, batch: ScheduledBatch,
+                sampled: Optional[torch.Tensor] = None) -> torch.Tensor:
         """Run the batch; returns logits [B, vocab] for the last tokens."""
         if batch.is_prefill:
             hidden = self.execute_prefill(batch.seqs)
         else:
-            hidden = self.execute_decode(batch.seqs)
+            hidden = self.execute_decode(batch.seqs, sampled)
         return self.model.compute_logits(hidden)

@@ -98,6 +98,7 @@ class Scheduler:
             if need > 0:
                 if self.pool.can_allocate(need):
                     seq.block_table.extend(self.pool.allocate(need))
+                    seq._bt_dirty = True
                 else:
                     victim = self.running.pop()  # newest
                     self.pool.free(victim.block_table)

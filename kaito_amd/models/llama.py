@@ -177,15 +177,17 @@ class LlamaForCausalLM(nn.Module):
     @torch.no_grad()
     def random_init(self, seed: int = 0):
         """Random-init weights of the right architecture (no network; bench
-        contract: synthetic data / random weights)."""
-        gen = torch.Generator(device="cpu").manual_seed(seed)
+        contract: synthetic data / random weights). Generates on the params'
+        device (device-side for 8B+ models; CPU init would take minutes)."""
+        dev = next(self.parameters()).device
+        gen = torch.Generator(device=dev).manual_seed(seed)
         for name, p in self.named_parameters():
             if "layernorm" in name or name == "norm":
                 p.fill_(1.0)
             else:
                 std = 0.02 if "embed" in name or "lm_head" in name else \
                     0.02 / math.sqrt(2 * self.cfg.num_layers)
-                cpu = torch.empty(p.shape, dtype=torch.float32)
-                cpu.normal_(0, std, generator=gen)
-                p.copy_(cpu.to(p.dtype))
+                tmp = torch.empty(p.shape, dtype=torch.float32, device=dev)
+                tmp.normal_(0, std, generator=gen)
+                p.copy_(tmp.to(p.dtype))
         return self

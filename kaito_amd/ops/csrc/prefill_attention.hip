@@ -211,12 +211,19 @@ void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
   TORCH_CHECK(QH % KH == 0);
   auto stream = at::hip::getCurrentHIPStream();
   if (ntiles == 0) return;
-  TORCH_CHECK(D == 128, "prefill: only head_dim=128 supported");
-  hipLaunchKernelGGL((prefill_attn_kernel<128>), dim3(ntiles, QH), dim3(256), 0,
-      stream, (short*)out.data_ptr(), (const short*)q.data_ptr(),
-      (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-      tile_seq.data_ptr<int>(), tile_qbase.data_ptr<int>(),
-      cu_seqlens.data_ptr<int>(), (float)scale, QH, KH);
+#define PF_LAUNCH(D_)                                                         \
+  hipLaunchKernelGGL((prefill_attn_kernel<D_>), dim3(ntiles, QH), dim3(256),  \
+      0, stream, (short*)out.data_ptr(), (const short*)q.data_ptr(),         \
+      (const short*)k.data_ptr(), (const short*)v.data_ptr(),                \
+      tile_seq.data_ptr<int>(), tile_qbase.data_ptr<int>(),                  \
+      cu_seqlens.data_ptr<int>(), (float)scale, QH, KH)
+  switch (D) {
+    case 128: PF_LAUNCH(128); break;
+    case 96:  PF_LAUNCH(96);  break;
+    case 64:  PF_LAUNCH(64);  break;
+    default: TORCH_CHECK(false, "prefill: unsupported head_dim ", D);
+  }
+#undef PF_LAUNCH
 }
 
 }  // namespace kaito
