@@ -36,6 +36,10 @@ def parse_args():
     p.add_argument("--in-tokens", type=int, default=200)
     p.add_argument("--out-tokens", type=int, default=200)
     p.add_argument("--eager", action="store_true")
+    p.add_argument("--tune-gemms", default=None, nargs="?", const="",
+                   metavar="OUT_CSV",
+                   help="run TunableOp GEMM tuning for all decode shapes, "
+                        "write results (default: in-tree csv), then exit")
     p.add_argument("--profile", action="store_true",
                    help="skip barriers/json (for rocprofv3 single-rank runs)")
     return p.parse_args()
@@ -67,6 +71,9 @@ def main():
         seed=1234 + rank,
     )
     eng = LLMEngine(cfg)
+    if args.tune_gemms is not None:
+        eng.runner.tune_gemms(args.tune_gemms or None)
+        return
     if not args.eager:
         eng.capture_graphs()
 

@@ -27,6 +27,7 @@ class LLMEngine:
     def __init__(self, cfg: EngineConfig, weights_path: Optional[str] = None):
         self.cfg = cfg
         self.runner = ModelRunner(cfg).load_model(weights_path, cfg.seed)
+        self.runner.setup_tunable()
         self.runner.profile_and_allocate_kv()
         self.pool = BlockPool(self.runner.num_gpu_blocks, cfg.block_size)
         self.scheduler = Scheduler(cfg, self.pool)

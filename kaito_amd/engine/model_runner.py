@@ -11,9 +11,16 @@ MI355X-first specifics:
 from __future__ import annotations
 
 import logging
+import os
+from pathlib import Path
 from typing import Dict, List, Optional, Tuple
 
 import torch
+
+# hipBLASLt algo selections tuned offline on MI355X (torch TunableOp);
+# shipped in-tree so every engine start gets the tuned GEMM kernels without
+# paying the tuning cost. Regenerate with: python bench.py --tune-gemms.
+TUNABLE_FILE = Path(__file__).resolve().parent.parent / "ops" / "tunableop_gfx950.csv"
 
 from .config import EngineConfig
 from .scheduler import ScheduledBatch
@@ -131,6 +138,44 @@ class ModelRunner:
             seq_lens=b["seq_lens"][:bs])
         return self.model(b["input_ids"][:bs], b["positions"][:bs],
                           self.kv_caches, meta)
+
+    def setup_tunable(self):
+        """Enable TunableOp lookups from the in-tree MI355X tuning cache
+        (hipBLASLt's default heuristic picks catastrophically bad kernels
+        for some skinny decode shapes — measured 557us vs 44us for
+        near-identical work, profiles/r01_decode_profile.md)."""
+        if not self.is_gpu:
+            return
+        import torch.cuda.tunable as tunable
+        tunable.enable(True)
+        tunable.tuning_enable(False)
+        if TUNABLE_FILE.exists():
+            tunable.read_file(str(TUNABLE_FILE))
+
+    def tune_gemms(self, out_file: Optional[str] = None):
+        """Tune hipBLASLt algo selection for every decode-bucket GEMM shape
+        (+ one full prefill shape) and persist the results."""
+        if not self.is_gpu:
+            return
+        import torch.cuda.tunable as tunable
+        tunable.enable(True)
+        if TUNABLE_FILE.exists():
+            tunable.read_file(str(TUNABLE_FILE))
+        tunable.tuning_enable(True)
+        if not hasattr(self, "_buf"):
+            self._init_decode_buffers()
+        self._buf["seq_lens"].fill_(1)
+        for bs in sorted(self.cfg.graph_batch_sizes, reverse=True):
+            if bs > self.cfg.max_num_seqs:
+                continue
+            hidden = self._decode_forward(bs)
+            self.model.compute_logits(hidden)
+            torch.cuda.synchronize()
+        self._buf["seq_lens"].zero_()
+        tunable.tuning_enable(False)
+        dest = out_file or str(TUNABLE_FILE)
+        tunable.write_file(dest)
+        logger.info("TunableOp results written to %s", dest)
 
     def capture_decode_graphs(self):
         """Capture hipGraphs for each decode bucket (largest first so the

@@ -72,14 +72,14 @@ class LlamaAttention(nn.Module):
                 kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]],
                 meta: AttnMetadata, cos_sin: torch.Tensor) -> torch.Tensor:
         qkv = self.qkv_proj(x)
+        # strided views into the fused qkv buffer — the HIP kernels take row
+        # strides, so no .contiguous() copies on the hot path.
         q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
-        q = q.contiguous()
-        k = k.contiguous()
         q, k = ops.rotary_embedding(positions, q, k, self.head_dim, cos_sin)
         T = x.size(0)
-        qh = q.view(T, self.num_heads, self.head_dim)
-        kh = k.view(T, self.num_kv_heads, self.head_dim)
-        vh = v.view(T, self.num_kv_heads, self.head_dim).contiguous()
+        qh = q.unflatten(-1, (self.num_heads, self.head_dim))
+        kh = k.unflatten(-1, (self.num_kv_heads, self.head_dim))
+        vh = v.unflatten(-1, (self.num_kv_heads, self.head_dim))
         if kv_cache is not None:
             ops.reshape_and_cache(kh, vh, kv_cache[0], kv_cache[1],
                                   meta.slot_mapping)

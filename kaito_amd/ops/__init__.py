@@ -85,16 +85,12 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
 
 def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, slot_mapping: torch.Tensor) -> None:
+    """k/v: [T, KH, D] (rows may be strided views into a fused qkv buffer)."""
     if k.is_cuda:
         _require_ext()
-        torch.ops.kaito.reshape_and_cache(
-            k.reshape(k.size(0), -1), v.reshape(v.size(0), -1),
-            k_cache, v_cache, slot_mapping)
+        torch.ops.kaito.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
         return
-    torch_ref.reshape_and_cache(
-        k.reshape(k.size(0), k_cache.size(1), k_cache.size(3)),
-        v.reshape(v.size(0), v_cache.size(1), v_cache.size(3)),
-        k_cache, v_cache, slot_mapping)
+    torch_ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
 
 
 def paged_attention(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,

@@ -16,7 +16,8 @@ __global__ void reshape_and_cache_kernel(
     short* __restrict__ k_cache,       // [B, KH, BS, D]
     short* __restrict__ v_cache,
     const int64_t* __restrict__ slots, // [T] = block*BS + off ; <0 = skip
-    const int T, const int KH, const int D, const int BS) {
+    const int T, const int KH, const int D, const int BS,
+    const int64_t kv_stride) {
   const int token = blockIdx.x;
   if (token >= T) return;
   const int64_t slot = slots[token];
@@ -24,8 +25,8 @@ __global__ void reshape_and_cache_kernel(
   const int64_t block = slot / BS;
   const int off = (int)(slot % BS);
   const int nvec = KH * D / 8;
-  const short8_t* kv = reinterpret_cast<const short8_t*>(k + (int64_t)token * KH * D);
-  const short8_t* vv = reinterpret_cast<const short8_t*>(v + (int64_t)token * KH * D);
+  const short8_t* kv = reinterpret_cast<const short8_t*>(k + (int64_t)token * kv_stride);
+  const short8_t* vv = reinterpret_cast<const short8_t*>(v + (int64_t)token * kv_stride);
   for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
     const int h = (i * 8) / D;
     const int d = (i * 8) % D;
@@ -38,7 +39,10 @@ __global__ void reshape_and_cache_kernel(
 void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor slot_mapping) {
   TORCH_CHECK(k.is_cuda() && k.dtype() == at::kBFloat16);
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(k.stride(-1) == 1 && v.stride(-1) == 1);
+  TORCH_CHECK(k.dim() == 3 && k.stride(1) == k.size(2),
+              "k must be [T, KH, D] with contiguous heads");
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share row stride");
   TORCH_CHECK(slot_mapping.dtype() == at::kLong);
   const int T = slot_mapping.size(0);
   const int KH = k_cache.size(1);
@@ -51,7 +55,7 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
   hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(T), dim3(block), 0, stream,
       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
       (short*)k_cache.data_ptr(), (short*)v_cache.data_ptr(),
-      slot_mapping.data_ptr<int64_t>(), T, KH, D, BS);
+      slot_mapping.data_ptr<int64_t>(), T, KH, D, BS, k.stride(0));
 }
 
 }  // namespace kaito

@@ -27,7 +27,8 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
     const short* __restrict__ v_cache,
     const int* __restrict__ block_tables,// [T, max_blocks]
     const int* __restrict__ seq_lens,    // [T]
-    const float scale, const int KH, const int max_blocks) {
+    const float scale, const int KH, const int max_blocks,
+    const int64_t q_stride) {
   constexpr int DL = D / 8;       // dims per lane (16 for D=128)
   constexpr int NW = 4;           // waves per workgroup
   const int seq = blockIdx.x;
@@ -46,7 +47,7 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
 #pragma unroll
   for (int g = 0; g < G; g++) {
     const short8_t* qp = reinterpret_cast<const short8_t*>(
-        q + ((int64_t)seq * QH + kvh * G + g) * D + dc * DL);
+        q + (int64_t)seq * q_stride + (kvh * G + g) * D + dc * DL);
 #pragma unroll
     for (int vv = 0; vv < DL / 8; vv++) {
       short8_t x = qp[vv];
@@ -173,13 +174,16 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
       dim3(256), 0, stream, (short*)out.data_ptr(),                           \
       (const short*)query.data_ptr(), (const short*)k_cache.data_ptr(),       \
       (const short*)v_cache.data_ptr(), block_tables.data_ptr<int>(),         \
-      seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks)
+      seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks,                 \
+      query.stride(0))
 
 void paged_attention(at::Tensor out, at::Tensor query, at::Tensor k_cache,
                      at::Tensor v_cache, at::Tensor block_tables,
                      at::Tensor seq_lens, double scale) {
   TORCH_CHECK(query.is_cuda() && query.dtype() == at::kBFloat16);
-  TORCH_CHECK(query.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(out.is_contiguous());
+  TORCH_CHECK(query.stride(-1) == 1 && query.stride(1) == query.size(2),
+              "query must be [T, QH, D] with contiguous heads");
   TORCH_CHECK(block_tables.dtype() == at::kInt && seq_lens.dtype() == at::kInt);
   const int T = query.size(0);
   const int QH = query.size(1);

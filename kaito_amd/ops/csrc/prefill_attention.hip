@@ -36,7 +36,8 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
     const int* __restrict__ tile_seq,   // [ntiles] seq index
     const int* __restrict__ tile_qbase, // [ntiles] q row base within seq
     const int* __restrict__ cu_seqlens, // [batch+1]
-    const float scale, const int QH, const int KH) {
+    const float scale, const int QH, const int KH,
+    const int64_t q_stride, const int64_t kv_stride) {
   constexpr int KK = D / 32;          // MFMA k-steps over head_dim (4 for 128)
   constexpr int DT = D / 16;          // output d-tiles (8 for 128)
   const int tile = blockIdx.x;
@@ -67,7 +68,7 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
 #pragma unroll
   for (int kk = 0; kk < KK; kk++) {
     const short8_t* qp = reinterpret_cast<const short8_t*>(
-        q + ((int64_t)qtok * QH + qh) * D + kk * 32 + hi * 8);
+        q + (int64_t)qtok * q_stride + qh * D + kk * 32 + hi * 8);
     qfrag[kk] = *reinterpret_cast<const bf16x8*>(qp);
   }
 
@@ -91,14 +92,14 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
         const int kvp = kv0 + row;
         const int tok = tok0 + min(kvp, slen - 1);
         short8_t kd = *reinterpret_cast<const short8_t*>(
-            k + ((int64_t)tok * KH + kvh) * D + col);
+            k + (int64_t)tok * kv_stride + kvh * D + col);
         // swizzle byte offset within row: 16B-granular XOR of row bits
         const int bo = col * 2;
         const int swz = bo ^ ((row & 7) << 4);
         *reinterpret_cast<short8_t*>(&lds_k[row * D + swz / 2]) = kd;
         // V: transpose into Vt[d][kv]
         short8_t vd = *reinterpret_cast<const short8_t*>(
-            v + ((int64_t)tok * KH + kvh) * D + col);
+            v + (int64_t)tok * kv_stride + kvh * D + col);
 #pragma unroll
         for (int j = 0; j < 8; j++) lds_vt[(col + j) * VT_PAD + row] = vd[j];
       }
@@ -202,7 +203,10 @@ void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
                        at::Tensor tile_seq, at::Tensor tile_qbase,
                        at::Tensor cu_seqlens, double scale) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == at::kBFloat16);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(out.is_contiguous());
+  TORCH_CHECK(q.stride(-1) == 1 && q.stride(1) == q.size(2));
+  TORCH_CHECK(k.stride(-1) == 1 && k.stride(1) == k.size(2));
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share row stride");
   TORCH_CHECK(tile_seq.dtype() == at::kInt && cu_seqlens.dtype() == at::kInt);
   const int QH = q.size(1);
   const int D = q.size(2);
@@ -216,7 +220,8 @@ void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
       0, stream, (short*)out.data_ptr(), (const short*)q.data_ptr(),         \
       (const short*)k.data_ptr(), (const short*)v.data_ptr(),                \
       tile_seq.data_ptr<int>(), tile_qbase.data_ptr<int>(),                  \
-      cu_seqlens.data_ptr<int>(), (float)scale, QH, KH)
+      cu_seqlens.data_ptr<int>(), (float)scale, QH, KH,                       \
+      q.stride(0), k.stride(0))
   switch (D) {
     case 128: PF_LAUNCH(128); break;
     case 96:  PF_LAUNCH(96);  break;

@@ -18,7 +18,8 @@ __global__ void rope_kernel(
     short* __restrict__ q, short* __restrict__ k,
     const int64_t* __restrict__ positions,
     const float* __restrict__ cos_sin,  // [max_pos, R]
-    const int T, const int QH, const int KH, const int D, const int R) {
+    const int T, const int QH, const int KH, const int D, const int R,
+    const int64_t q_stride, const int64_t k_stride) {
   const int token = blockIdx.x;
   if (token >= T) return;
   const int64_t pos = positions[token];
@@ -32,8 +33,8 @@ __global__ void rope_kernel(
     const int h = pair / half;
     const int i = pair % half;
     short* base = (h < QH)
-        ? q + (int64_t)token * QH * D + (int64_t)h * D
-        : k + (int64_t)token * KH * D + (int64_t)(h - QH) * D;
+        ? q + (int64_t)token * q_stride + (int64_t)h * D
+        : k + (int64_t)token * k_stride + (int64_t)(h - QH) * D;
     // load x[i..i+1] and x[i+half..i+half+1]
     short2 xa = *reinterpret_cast<short2*>(base + i);
     short2 xb = *reinterpret_cast<short2*>(base + i + half);
@@ -55,19 +56,20 @@ void rotary_embedding(at::Tensor positions, at::Tensor q, at::Tensor k,
                       int64_t head_dim, at::Tensor cos_sin_cache) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == at::kBFloat16);
   TORCH_CHECK(positions.dtype() == at::kLong);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous());
+  TORCH_CHECK(q.stride(-1) == 1 && k.stride(-1) == 1,
+              "rope: innermost dim must be contiguous");
   const int T = positions.size(0);
   const int R = cos_sin_cache.size(1);
   const int D = (int)head_dim;
-  const int QH = q.numel() / ((int64_t)T * D);
-  const int KH = k.numel() / ((int64_t)T * D);
+  const int QH = (int)(q.size(-1) / D) * (q.dim() == 3 ? (int)q.size(1) : 1);
+  const int KH = (int)(k.size(-1) / D) * (k.dim() == 3 ? (int)k.size(1) : 1);
   TORCH_CHECK(R <= D && R % 4 == 0, "rot_dim must be <= head_dim, mult of 4");
   auto stream = at::hip::getCurrentHIPStream();
   const int block = 256;
   hipLaunchKernelGGL(rope_kernel, dim3(T), dim3(block), 0, stream,
       (short*)q.data_ptr(), (short*)k.data_ptr(),
       positions.data_ptr<int64_t>(), cos_sin_cache.data_ptr<float>(),
-      T, QH, KH, D, R);
+      T, QH, KH, D, R, q.stride(0), k.stride(0));
 }
 
 }  // namespace kaito

@@ -1,0 +1,196 @@
+"""OpenAI-compatible HTTP front-end on :5000.
+
+Endpoint surface matches what the reference's ecosystem expects from a
+Workspace pod: /health, /metrics, /v1/models,
+/v1/completions, /v1/chat/completions (with SSE streaming). Metric names
+are vLLM-compatible so benchmark_entrypoint / EPP / KEDA integrations work
+unchanged.
+"""
+from __future__ import annotations
+
+import json
+import time
+import uuid
+from typing import AsyncIterator, List, Optional
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, Response, StreamingResponse
+from pydantic import BaseModel, Field
+
+from ..engine.sequence import SamplingParams
+from . import metrics
+from .async_engine import AsyncLLMEngine
+from .rate_limit import RateLimitMiddleware
+
+
+class CompletionRequest(BaseModel):
+    model: str = ""
+    prompt: object = ""           # str | list[str] | list[int]
+    max_tokens: int = 16
+    temperature: float = 1.0
+    top_p: float = 1.0
+    top_k: int = 0
+    n: int = 1
+    stream: bool = False
+    stop: Optional[object] = None
+    ignore_eos: bool = False
+    seed: Optional[int] = None
+
+
+class ChatMessage(BaseModel):
+    role: str
+    content: str
+
+
+class ChatCompletionRequest(BaseModel):
+    model: str = ""
+    messages: List[ChatMessage] = Field(default_factory=list)
+    max_tokens: Optional[int] = None
+    max_completion_tokens: Optional[int] = None
+    temperature: float = 1.0
+    top_p: float = 1.0
+    stream: bool = False
+    ignore_eos: bool = False
+
+
+def _sampling(max_tokens, temperature, top_p, top_k=0, ignore_eos=False,
+              eos_id=None):
+    return SamplingParams(
+        max_tokens=max_tokens or 16,
+        temperature=temperature if temperature is not None else 1.0,
+        top_p=top_p, top_k=top_k, ignore_eos=ignore_eos)
+
+
+def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
+              max_queue: Optional[int] = None) -> FastAPI:
+    app = FastAPI(title="kaito-amd inference engine")
+    eng_cfg = async_engine.engine.cfg
+    app.add_middleware(
+        RateLimitMiddleware,
+        get_queue_depth=lambda: async_engine.num_waiting,
+        max_queue=max_queue or eng_cfg.max_num_seqs)
+
+    metrics.set_cache_config(eng_cfg.block_size,
+                             async_engine.engine.runner.num_gpu_blocks)
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok"}
+
+    @app.get("/metrics")
+    async def prom_metrics():
+        data, ctype = metrics.render()
+        return Response(content=data, media_type=ctype)
+
+    @app.get("/v1/models")
+    async def models():
+        return {"object": "list", "data": [{
+            "id": model_name, "object": "model",
+            "owned_by": "kaito-amd", "created": int(time.time())}]}
+
+    # ------------------------------------------------------------ completions
+    @app.post("/v1/completions")
+    async def completions(req: CompletionRequest, raw: Request):
+        if isinstance(req.prompt, list) and req.prompt and \
+                isinstance(req.prompt[0], int):
+            prompt_ids = [int(x) for x in req.prompt]
+            prompt_text = None
+        else:
+            text = req.prompt if isinstance(req.prompt, str) else str(req.prompt)
+            prompt_ids = tokenizer.encode(text)
+            prompt_text = text
+        sp = _sampling(req.max_tokens, req.temperature, req.top_p, req.top_k,
+                       req.ignore_eos)
+        rid = f"cmpl-{uuid.uuid4().hex[:24]}"
+        if req.stream:
+            return StreamingResponse(
+                _stream_completion(rid, prompt_ids, sp),
+                media_type="text/event-stream")
+        toks: List[int] = []
+        finish = "length"
+        async for item in async_engine.generate(prompt_ids, sp):
+            if item.finished:
+                finish = item.finish_reason or "stop"
+            else:
+                toks.append(item.token_id)
+        return {
+            "id": rid, "object": "text_completion",
+            "created": int(time.time()), "model": model_name,
+            "choices": [{"index": 0, "text": tokenizer.decode(toks),
+                         "finish_reason": finish, "logprobs": None}],
+            "usage": {"prompt_tokens": len(prompt_ids),
+                      "completion_tokens": len(toks),
+                      "total_tokens": len(prompt_ids) + len(toks)}}
+
+    async def _stream_completion(rid, prompt_ids, sp) -> AsyncIterator[str]:
+        async for item in async_engine.generate(prompt_ids, sp):
+            if item.finished:
+                payload = {"id": rid, "object": "text_completion",
+                           "model": model_name,
+                           "choices": [{"index": 0, "text": "",
+                                        "finish_reason": item.finish_reason}]}
+                yield f"data: {json.dumps(payload)}\n\n"
+                yield "data: [DONE]\n\n"
+            else:
+                payload = {"id": rid, "object": "text_completion",
+                           "model": model_name,
+                           "choices": [{"index": 0,
+                                        "text": tokenizer.decode([item.token_id]),
+                                        "finish_reason": None}]}
+                yield f"data: {json.dumps(payload)}\n\n"
+
+    # ------------------------------------------------------------ chat
+    @app.post("/v1/chat/completions")
+    async def chat(req: ChatCompletionRequest):
+        text = tokenizer.apply_chat_template(
+            [m.model_dump() for m in req.messages],
+            add_generation_prompt=True, tokenize=False)
+        prompt_ids = tokenizer.encode(text)
+        max_toks = req.max_completion_tokens or req.max_tokens or 128
+        sp = _sampling(max_toks, req.temperature, req.top_p,
+                       ignore_eos=req.ignore_eos)
+        rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+        if req.stream:
+            return StreamingResponse(_stream_chat(rid, prompt_ids, sp),
+                                     media_type="text/event-stream")
+        toks: List[int] = []
+        finish = "length"
+        async for item in async_engine.generate(prompt_ids, sp):
+            if item.finished:
+                finish = item.finish_reason or "stop"
+            else:
+                toks.append(item.token_id)
+        return {
+            "id": rid, "object": "chat.completion",
+            "created": int(time.time()), "model": model_name,
+            "choices": [{"index": 0,
+                         "message": {"role": "assistant",
+                                     "content": tokenizer.decode(toks)},
+                         "finish_reason": finish}],
+            "usage": {"prompt_tokens": len(prompt_ids),
+                      "completion_tokens": len(toks),
+                      "total_tokens": len(prompt_ids) + len(toks)}}
+
+    async def _stream_chat(rid, prompt_ids, sp) -> AsyncIterator[str]:
+        first = {"id": rid, "object": "chat.completion.chunk",
+                 "model": model_name,
+                 "choices": [{"index": 0, "delta": {"role": "assistant"},
+                              "finish_reason": None}]}
+        yield f"data: {json.dumps(first)}\n\n"
+        async for item in async_engine.generate(prompt_ids, sp):
+            if item.finished:
+                payload = {"id": rid, "object": "chat.completion.chunk",
+                           "model": model_name,
+                           "choices": [{"index": 0, "delta": {},
+                                        "finish_reason": item.finish_reason}]}
+                yield f"data: {json.dumps(payload)}\n\n"
+                yield "data: [DONE]\n\n"
+            else:
+                payload = {"id": rid, "object": "chat.completion.chunk",
+                           "model": model_name,
+                           "choices": [{"index": 0,
+                                        "delta": {"content": tokenizer.decode([item.token_id])},
+                                        "finish_reason": None}]}
+                yield f"data: {json.dumps(payload)}\n\n"
+
+    return app

@@ -1,0 +1,129 @@
+"""AsyncLLMEngine: the background engine loop bridging asyncio request
+handlers to the synchronous continuous-batching engine.
+
+The engine loop owns the GPU and runs in one dedicated thread (one process
+per GPU — DP replicas are separate processes behind the k8s Service). Token
+streams flow back through per-request asyncio queues.
+"""
+from __future__ import annotations
+
+import asyncio
+import queue
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import AsyncIterator, Dict, List, Optional
+
+from ..engine.engine import LLMEngine
+from ..engine.sequence import SamplingParams, Sequence
+from . import metrics
+
+
+@dataclass
+class StreamItem:
+    token_id: int
+    finished: bool = False
+    finish_reason: Optional[str] = None
+
+
+@dataclass
+class _Pending:
+    prompt: List[int]
+    sampling: SamplingParams
+    out_q: asyncio.Queue
+    loop: asyncio.AbstractEventLoop
+    seq_id: Optional[int] = None
+
+
+class AsyncLLMEngine:
+    def __init__(self, engine: LLMEngine):
+        self.engine = engine
+        self._submit: "queue.Queue[_Pending]" = queue.Queue()
+        self._streams: Dict[int, _Pending] = {}
+        self._emitted: Dict[int, int] = {}
+        self._wake = threading.Event()
+        self._stop = False
+        self._thread: Optional[threading.Thread] = None
+
+    # ---------------------------------------------------------------- loop
+    def start(self):
+        self._thread = threading.Thread(target=self._loop, daemon=True,
+                                        name="kaito-engine-loop")
+        self._thread.start()
+        return self
+
+    def shutdown(self):
+        self._stop = True
+        self._wake.set()
+        if self._thread:
+            self._thread.join(timeout=10)
+
+    def _drain_submissions(self):
+        while True:
+            try:
+                p = self._submit.get_nowait()
+            except queue.Empty:
+                return
+            sid = self.engine.add_request(p.prompt, p.sampling)
+            p.seq_id = sid
+            self._streams[sid] = p
+            self._emitted[sid] = 0
+
+    def _loop(self):
+        eng = self.engine
+        while not self._stop:
+            self._drain_submissions()
+            if not eng.has_unfinished():
+                metrics.REQUESTS_RUNNING.set(0)
+                self._wake.wait(timeout=0.05)
+                self._wake.clear()
+                continue
+            finished = eng.step()
+            metrics.REQUESTS_RUNNING.set(eng.scheduler.num_running)
+            metrics.REQUESTS_WAITING.set(eng.scheduler.num_waiting)
+            # push fresh tokens to streams
+            for sid, p in list(self._streams.items()):
+                seq = eng.seqs.get(sid)
+                if seq is None:
+                    continue
+                n = len(seq.output_token_ids)
+                e = self._emitted[sid]
+                if n > e:
+                    metrics.GENERATION_TOKENS.inc(n - e)
+                    for tok in seq.output_token_ids[e:n]:
+                        self._push(p, StreamItem(tok))
+                    self._emitted[sid] = n
+            for seq in finished:
+                p = self._streams.pop(seq.seq_id, None)
+                self._emitted.pop(seq.seq_id, None)
+                if p is not None:
+                    metrics.PROMPT_TOKENS.inc(seq.num_prompt_tokens)
+                    metrics.E2E_LATENCY.inc(
+                        (seq.finish_time or time.monotonic()) - seq.arrival_time)
+                    self._push(p, StreamItem(-1, True, seq.finish_reason))
+
+    @staticmethod
+    def _push(p: _Pending, item: StreamItem):
+        p.loop.call_soon_threadsafe(p.out_q.put_nowait, item)
+
+    # ---------------------------------------------------------------- API
+    @property
+    def num_waiting(self) -> int:
+        return self.engine.scheduler.num_waiting + self._submit.qsize()
+
+    @property
+    def num_running(self) -> int:
+        return self.engine.scheduler.num_running
+
+    async def generate(self, prompt_ids: List[int], sampling: SamplingParams
+                       ) -> AsyncIterator[StreamItem]:
+        loop = asyncio.get_running_loop()
+        out_q: asyncio.Queue = asyncio.Queue()
+        p = _Pending(prompt_ids, sampling, out_q, loop)
+        self._submit.put(p)
+        self._wake.set()
+        while True:
+            item = await out_q.get()
+            yield item
+            if item.finished:
+                return
