@@ -1,0 +1,215 @@
+"""Kubernetes manifest generation (plain dicts → YAML).
+
+Python re-implementation of the reference's manifest layer
+(pkg/workspace/inference/preset_inferences.go:179-292 GeneratePresetInference,
+pkg/workspace/manifests/manifests.go:43-200 services/statefulset,
+pkg/workspace/tuning/preset_tuning.go:145 CreatePresetTuning), with the pod
+spec retargeted at MI355X nodes: amd.com/gpu resources, ROCm env, our
+kaito_amd.server.entrypoint command.
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional
+
+from ..engine.config import ModelConfig
+from .api_types import (LABEL_WORKSPACE_NAME, LABEL_WORKSPACE_NAMESPACE,
+                        Workspace)
+from .planner import (ParallelPlan, build_inference_command,
+                      build_multinode_command, configure_parallelism)
+from .sku import GPUConfig
+
+INFERENCE_PORT = 5000
+GPU_RESOURCE = "amd.com/gpu"
+SHM_VOLUME = {"name": "dshm", "emptyDir": {"medium": "Memory"}}
+CONFIG_MOUNT = "/mnt/config"
+ADAPTER_MOUNT = "/mnt/adapter"
+WEIGHTS_MOUNT = "/workspace/weights"
+NVME_STORAGE_CLASS = "kaito-local-nvme-disk"
+
+
+def workspace_selector(ws: Workspace) -> Dict[str, str]:
+    return {LABEL_WORKSPACE_NAME: ws.name,
+            LABEL_WORKSPACE_NAMESPACE: ws.namespace}
+
+
+def generate_service(ws: Workspace, headless: bool = False) -> Dict[str, Any]:
+    """manifests.go:43-133: ClusterIP service + headless variant for the
+    multi-node rendezvous DNS."""
+    name = ws.name + ("-headless" if headless else "")
+    spec: Dict[str, Any] = {
+        "selector": workspace_selector(ws),
+        "ports": [{"name": "http", "port": 80,
+                   "targetPort": INFERENCE_PORT, "protocol": "TCP"}],
+    }
+    if headless:
+        spec["clusterIP"] = "None"
+        spec["publishNotReadyAddresses"] = True
+    else:
+        spec["type"] = "ClusterIP"
+    return {"apiVersion": "v1", "kind": "Service",
+            "metadata": {"name": name, "namespace": ws.namespace,
+                         "labels": workspace_selector(ws)},
+            "spec": spec}
+
+
+def _probes(readiness_timeout_s: int = 600) -> Dict[str, Any]:
+    """preset_inferences.go:403-422: startup probe sized by model readiness
+    timeout, liveness/readiness on /health."""
+    return {
+        "startupProbe": {
+            "httpGet": {"path": "/health", "port": INFERENCE_PORT},
+            "failureThreshold": max(readiness_timeout_s // 10, 6),
+            "periodSeconds": 10},
+        "livenessProbe": {
+            "httpGet": {"path": "/health", "port": INFERENCE_PORT},
+            "periodSeconds": 10, "failureThreshold": 6},
+        "readinessProbe": {
+            "httpGet": {"path": "/health", "port": INFERENCE_PORT},
+            "periodSeconds": 10},
+    }
+
+
+def generate_inference_pod_spec(ws: Workspace, model: ModelConfig,
+                                gpu: GPUConfig, plan: ParallelPlan,
+                                image: str,
+                                readiness_timeout_s: int = 600
+                                ) -> Dict[str, Any]:
+    """preset_inferences.go:522-674 GenerateInferencePodSpec for MI355X."""
+    gpus = gpu.gpu_count
+    if plan.num_nodes > 1:
+        command = ["/bin/sh", "-c", build_multinode_command(
+            model, gpu, plan, f"{ws.name}-headless.{ws.namespace}.svc")]
+    else:
+        command = build_inference_command(
+            model, gpu, plan,
+            config_file=(f"{CONFIG_MOUNT}/inference_config.yaml"
+                         if ws.inference and ws.inference.config else None))
+    container: Dict[str, Any] = {
+        "name": ws.name,
+        "image": image,
+        "command": command,
+        "resources": {
+            "requests": {GPU_RESOURCE: str(gpus)},
+            "limits": {GPU_RESOURCE: str(gpus)},
+        },
+        "ports": [{"containerPort": INFERENCE_PORT, "name": "http"}],
+        "env": [
+            {"name": "POD_INDEX", "valueFrom": {"fieldRef": {
+                "fieldPath": "metadata.labels['apps.kubernetes.io/pod-index']"}}},
+            {"name": "ROCM_PATH", "value": "/opt/rocm"},
+            {"name": "HSA_ENABLE_IPC_MODE_LEGACY", "value": "0"},
+            {"name": "KAITO_PROCESSOR", "value": "gpu"},
+        ],
+        "volumeMounts": [{"name": "dshm", "mountPath": "/dev/shm"}],
+        **_probes(readiness_timeout_s),
+    }
+    volumes: List[Dict[str, Any]] = [SHM_VOLUME]
+    if ws.inference and ws.inference.config:
+        volumes.append({"name": "config-volume",
+                        "configMap": {"name": ws.inference.config}})
+        container["volumeMounts"].append(
+            {"name": "config-volume", "mountPath": CONFIG_MOUNT})
+    init_containers = []
+    for ad in (ws.inference.adapters if ws.inference else []):
+        nm = ad.source.get("name", "adapter")
+        init_containers.append({
+            "name": f"adapter-{nm}",
+            "image": ad.source.get("image", ""),
+            "command": ["/bin/sh", "-c",
+                        f"cp -r /data/* {ADAPTER_MOUNT}/{nm}/"],
+            "volumeMounts": [{"name": "adapter-volume",
+                              "mountPath": ADAPTER_MOUNT}],
+        })
+    if init_containers:
+        volumes.append({"name": "adapter-volume", "emptyDir": {}})
+        container["volumeMounts"].append(
+            {"name": "adapter-volume", "mountPath": ADAPTER_MOUNT})
+    spec = {
+        "containers": [container],
+        "initContainers": init_containers,
+        "volumes": volumes,
+        "tolerations": [
+            {"key": "sku", "operator": "Equal", "value": "gpu",
+             "effect": "NoSchedule"},
+            {"key": GPU_RESOURCE, "operator": "Exists",
+             "effect": "NoSchedule"},
+        ],
+    }
+    if ws.resource.instanceType:
+        spec["nodeSelector"] = {
+            "node.kubernetes.io/instance-type": ws.resource.instanceType}
+    return spec
+
+
+def generate_statefulset(ws: Workspace, model: ModelConfig, gpu: GPUConfig,
+                         image: str,
+                         plan: Optional[ParallelPlan] = None
+                         ) -> Dict[str, Any]:
+    """manifests.go:135-200 GenerateStatefulSetManifest."""
+    plan = plan or configure_parallelism(model, gpu)
+    sel = workspace_selector(ws)
+    pod_spec = generate_inference_pod_spec(ws, model, gpu, plan, image)
+    return {
+        "apiVersion": "apps/v1",
+        "kind": "StatefulSet",
+        "metadata": {"name": ws.name, "namespace": ws.namespace,
+                     "labels": sel},
+        "spec": {
+            "replicas": plan.num_nodes,
+            "selector": {"matchLabels": sel},
+            "serviceName": ws.name + "-headless",
+            "podManagementPolicy": "Parallel",
+            "template": {
+                "metadata": {"labels": sel},
+                "spec": pod_spec,
+            },
+        },
+    }
+
+
+def generate_tuning_job(ws: Workspace, model: ModelConfig, gpu: GPUConfig,
+                        image: str) -> Dict[str, Any]:
+    """preset_tuning.go:145 CreatePresetTuning → batch/v1 Job with
+    data-source init container and result output volume."""
+    assert ws.tuning is not None
+    sel = workspace_selector(ws)
+    data_init: Dict[str, Any] = {
+        "name": "data-downloader", "image": "busybox",
+        "command": ["sh", "-c", " && ".join(
+            f"wget -O /mnt/data/{i}.dat {u}"
+            for i, u in enumerate(ws.tuning.input.urls or ["none"]))],
+        "volumeMounts": [{"name": "data-volume", "mountPath": "/mnt/data"}],
+    }
+    container = {
+        "name": ws.name,
+        "image": image,
+        "command": ["python3", "-m", "kaito_amd.tuning.fine_tuning",
+                    "--model", model.name,
+                    "--method", ws.tuning.method,
+                    "--num-processes", str(gpu.gpu_count)],
+        "resources": {"requests": {GPU_RESOURCE: str(gpu.gpu_count)},
+                      "limits": {GPU_RESOURCE: str(gpu.gpu_count)}},
+        "volumeMounts": [
+            {"name": "data-volume", "mountPath": "/mnt/data"},
+            {"name": "results-volume", "mountPath": "/mnt/results"},
+            {"name": "dshm", "mountPath": "/dev/shm"},
+        ],
+    }
+    return {
+        "apiVersion": "batch/v1",
+        "kind": "Job",
+        "metadata": {"name": ws.name, "namespace": ws.namespace, "labels": sel},
+        "spec": {
+            "backoffLimit": 2,
+            "template": {"metadata": {"labels": sel}, "spec": {
+                "restartPolicy": "Never",
+                "initContainers": [data_init],
+                "containers": [container],
+                "volumes": [
+                    {"name": "data-volume", "emptyDir": {}},
+                    {"name": "results-volume", "emptyDir": {}},
+                    SHM_VOLUME,
+                ],
+            }},
+        },
+    }

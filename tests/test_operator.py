@@ -1,0 +1,222 @@
+"""Operator planning-layer tests: SKU table, estimator memory solve,
+3-tier parallelism planner, golden command/manifest assertions — the test
+style of the reference's estimator_test.go (table-driven) and
+preset_inferences_test.go (golden-command).
+"""
+import pytest
+
+from kaito_amd.models import get_model_config
+from kaito_amd.operator import api_types as at
+from kaito_amd.operator import manifests as mf
+from kaito_amd.operator.estimator import NodeEstimateRequest, estimate_node_count
+from kaito_amd.operator.planner import (build_inference_command,
+                                        build_multinode_command,
+                                        configure_parallelism)
+from kaito_amd.operator.sku import (get_sku_handler,
+                                    gpu_config_from_node_labels)
+
+
+def _gpu(sku="Standard_ND96isr_MI355X_v1"):
+    return get_sku_handler("azure").get_gpu_config(sku)
+
+
+# -------------------------------------------------------------------- SKU
+def test_sku_table_mi355x():
+    g = _gpu()
+    assert g.gpu_count == 8 and g.gpu_mem_gib == 288
+    assert g.gfx_arch == "gfx950" and g.xgmi_links == 7
+    assert g.supports_bfloat16()
+    assert g.total_gpu_mem_gib == 2304
+
+
+def test_sku_unknown_cloud():
+    with pytest.raises(ValueError):
+        get_sku_handler("gcp")
+
+
+def test_byo_node_labels():
+    g = gpu_config_from_node_labels({
+        "amd.com/gpu.count": "8",
+        "amd.com/gpu.vram": "288G",
+        "amd.com/gpu.product": "AMD-Instinct-MI355X",
+        "amd.com/gpu.family": "gfx950"})
+    assert g.gpu_count == 8 and g.gpu_mem_gib == 288 and g.xgmi_links == 7
+    assert gpu_config_from_node_labels({}) is None
+
+
+# -------------------------------------------------------------- estimator
+@pytest.mark.parametrize("model,expect_gpus,expect_nodes", [
+    ("llama-3-8b", 1, 1),        # 16 GiB weights ≪ 288 GiB
+    ("llama-3-70b", 1, 1),       # 141 GiB weights < 1× MI355X budget!
+    ("phi-4-mini-instruct", 1, 1),
+    ("qwen2.5-72b", 1, 1),
+])
+def test_estimator_single_node(model, expect_gpus, expect_nodes):
+    res = estimate_node_count(NodeEstimateRequest(
+        model=get_model_config(model), gpu=_gpu()))
+    assert res.min_gpus == expect_gpus
+    assert res.nodes_per_replica == expect_nodes
+    assert res.avail_mem_per_gpu_gib > 0
+
+
+def test_estimator_replicas_multiply_nodes():
+    res = estimate_node_count(NodeEstimateRequest(
+        model=get_model_config("llama-3-8b"), gpu=_gpu(), replicas=3))
+    assert res.target_node_count == 3 * res.nodes_per_replica
+
+
+def test_estimator_long_context_forces_more_gpus():
+    # 131072-token context × 64 seqs of KV forces sharding on 1-GPU SKU
+    small = _gpu("Standard_NC12s_MI355X_v1")
+    res = estimate_node_count(NodeEstimateRequest(
+        model=get_model_config("llama-3.1-8b"), gpu=small,
+        max_model_len=131072))
+    # kv budget 131072×64×128KiB ≈ 1 TiB → cannot fit on one 288 GiB GPU
+    assert res.min_gpus > 1 or res.kv_budget_gib > 288
+
+
+# ---------------------------------------------------------------- planner
+def test_tier1_dp_for_small_model():
+    plan = configure_parallelism(get_model_config("llama-3-8b"), _gpu())
+    assert plan.data_parallel == 8 and plan.tensor_parallel == 1
+    assert not plan.kv_offload
+
+
+def test_tier2_tp_for_70b():
+    plan = configure_parallelism(get_model_config("llama-3-70b"), _gpu())
+    # 141 GiB weights > 50% of one 288 GiB GPU budget → TP over xGMI
+    assert plan.tensor_parallel == 8 and plan.data_parallel == 1
+
+
+def test_tier3_pp_multinode():
+    plan = configure_parallelism(get_model_config("llama-3-70b"), _gpu(),
+                                 num_nodes=2)
+    assert plan.pipeline_parallel == 2 and plan.tensor_parallel == 8
+    assert plan.world_size == 16
+
+
+# ------------------------------------------------------- golden commands
+def test_inference_command_golden_8b():
+    cmd = build_inference_command(get_model_config("llama-3-8b"), _gpu())
+    assert cmd[:3] == ["python3", "-m", "kaito_amd.server.entrypoint"]
+    assert "--model" in cmd and "llama-3-8b" in cmd
+    assert cmd[cmd.index("--tensor-parallel-size") + 1] == "1"
+    assert cmd[cmd.index("--data-parallel-size") + 1] == "8"
+    assert cmd[cmd.index("--max-model-len") + 1] == "auto"
+
+
+def test_inference_command_golden_70b_tp8():
+    mc = get_model_config("llama-3-70b")
+    cmd = build_inference_command(mc, _gpu())
+    assert cmd[cmd.index("--tensor-parallel-size") + 1] == "8"
+    assert "--data-parallel-size" not in cmd
+
+
+def test_multinode_command_uses_pod_index_rendezvous():
+    mc = get_model_config("llama-3-70b")
+    plan = configure_parallelism(mc, _gpu(), num_nodes=2)
+    cmd = build_multinode_command(mc, _gpu(), plan, "ws-headless.default.svc")
+    assert "--nnodes=2" in cmd and "--nproc-per-node=8" in cmd
+    assert "--node-rank=${POD_INDEX}" in cmd
+    assert "--master-addr=ws-headless.default.svc" in cmd
+
+
+# ----------------------------------------------------------- API types
+def _ws(**kw):
+    base = dict(
+        name="ws1",
+        resource=at.ResourceSpec(instanceType="Standard_ND96isr_MI355X_v1"),
+        inference=at.InferenceSpec(preset=at.PresetSpec(name="llama-3-8b")))
+    base.update(kw)
+    return at.Workspace(**base)
+
+
+def test_workspace_validation_ok():
+    _ws().validate(sku_handler=get_sku_handler("azure"),
+                   known_presets={"llama-3-8b"})
+
+
+def test_workspace_validation_rejects_both_modes():
+    ws = _ws(tuning=at.TuningSpec(input=at.DataSource(urls=["u"]),
+                                  output=at.DataDestination(image="i")))
+    with pytest.raises(at.ValidationError):
+        ws.validate()
+
+
+def test_workspace_validation_rejects_bad_sku():
+    ws = _ws(resource=at.ResourceSpec(instanceType="Standard_NC24ads_A100_v4"))
+    with pytest.raises(at.ValidationError):
+        ws.validate(sku_handler=get_sku_handler("azure"))
+
+
+def test_workspace_validation_bypass_annotation():
+    ws = _ws(resource=at.ResourceSpec(instanceType="weird-sku"),
+             annotations={at.ANNOTATION_BYPASS_RESOURCE_CHECKS: "true"})
+    ws.validate(sku_handler=get_sku_handler("azure"))
+
+
+def test_workspace_validation_duplicate_adapters():
+    ws = _ws()
+    ws.inference.adapters = [at.AdapterSpec(source={"name": "a"}),
+                             at.AdapterSpec(source={"name": "a"})]
+    with pytest.raises(at.ValidationError):
+        ws.validate()
+
+
+def test_inferenceset_validation():
+    iset = at.InferenceSet("is1", spec=at.InferenceSetSpec(
+        replicas=2, workspaceTemplate=_ws()))
+    iset.validate()
+    iset.spec.upgradeStrategy = "YOLO"
+    with pytest.raises(at.ValidationError):
+        iset.validate()
+
+
+# -------------------------------------------------------- golden manifests
+def test_statefulset_golden_8b():
+    ws = _ws()
+    mc = get_model_config("llama-3-8b")
+    ss = mf.generate_statefulset(ws, mc, _gpu(), image="kaito/engine:v1")
+    assert ss["kind"] == "StatefulSet"
+    assert ss["spec"]["replicas"] == 1
+    pod = ss["spec"]["template"]["spec"]
+    c = pod["containers"][0]
+    assert c["resources"]["limits"]["amd.com/gpu"] == "8"
+    assert any(e["name"] == "HSA_ENABLE_IPC_MODE_LEGACY" for e in c["env"])
+    assert c["startupProbe"]["httpGet"]["path"] == "/health"
+    assert pod["nodeSelector"]["node.kubernetes.io/instance-type"] == \
+        "Standard_ND96isr_MI355X_v1"
+    assert {"name": "dshm", "mountPath": "/dev/shm"} in c["volumeMounts"]
+
+
+def test_statefulset_multinode_70b():
+    ws = _ws(inference=at.InferenceSpec(preset=at.PresetSpec(name="llama-3-70b")))
+    mc = get_model_config("llama-3-70b")
+    plan = configure_parallelism(mc, _gpu(), num_nodes=2)
+    ss = mf.generate_statefulset(ws, mc, _gpu(), "img", plan)
+    assert ss["spec"]["replicas"] == 2
+    cmd = ss["spec"]["template"]["spec"]["containers"][0]["command"]
+    assert cmd[0] == "/bin/sh" and "torch.distributed.run" in cmd[2]
+
+
+def test_service_and_headless():
+    ws = _ws()
+    svc = mf.generate_service(ws)
+    assert svc["spec"]["ports"][0]["targetPort"] == 5000
+    hl = mf.generate_service(ws, headless=True)
+    assert hl["spec"]["clusterIP"] == "None"
+    assert hl["metadata"]["name"] == "ws1-headless"
+
+
+def test_tuning_job_golden():
+    ws = _ws(inference=None, tuning=at.TuningSpec(
+        preset=at.PresetSpec(name="llama-3-8b"), method="qlora",
+        input=at.DataSource(urls=["http://x/data.json"]),
+        output=at.DataDestination(image="reg/out:v1")))
+    job = mf.generate_tuning_job(ws, get_model_config("llama-3-8b"), _gpu(),
+                                 "img")
+    assert job["kind"] == "Job"
+    spec = job["spec"]["template"]["spec"]
+    assert spec["restartPolicy"] == "Never"
+    assert "--method" in spec["containers"][0]["command"]
+    assert spec["initContainers"][0]["name"] == "data-downloader"
