@@ -112,6 +112,7 @@ def main():
     t_start_mono = time.monotonic()
     t0 = time.perf_counter()
     finished = run_steps(args.steps)
+    eng.flush()  # drain the pipelined step so token counts are exact
     if torch.cuda.is_available():
         torch.cuda.synchronize()
     ps.barrier()

@@ -9,6 +9,7 @@ from __future__ import annotations
 import itertools
 import logging
 import time
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 import torch
@@ -23,6 +24,16 @@ from .sequence import SamplingParams, Sequence, SeqStatus
 logger = logging.getLogger(__name__)
 
 
+@dataclass
+class _PendingStep:
+    seqs: List[Sequence]
+    epochs: List[int]
+    tokens: torch.Tensor                 # [bs] device (feeds the next step)
+    host: Optional[torch.Tensor]         # pinned copy (GPU) or None (CPU)
+    event: Optional[object]              # cuda Event or None
+    index: Dict[int, int]                # seq_id -> batch index
+
+
 class LLMEngine:
     def __init__(self, cfg: EngineConfig, weights_path: Optional[str] = None):
         self.cfg = cfg
@@ -35,6 +46,7 @@ class LLMEngine:
         self.eos_token_id: Optional[int] = None
         self._next_id = itertools.count()
         self.seqs: Dict[int, Sequence] = {}
+        self._pending: Optional[_PendingStep] = None
         # counters for /metrics (names consumed by the benchmark probe)
         self.num_generation_tokens = 0
         self.num_prompt_tokens = 0
@@ -66,27 +78,100 @@ class LLMEngine:
         seq.finish_reason = "abort"
 
     def has_unfinished(self) -> bool:
-        return self.scheduler.has_work()
+        return self.scheduler.has_work() or self._pending is not None
 
     # ------------------------------------------------------------- stepping
+    #
+    # Pipelined decode: step N is LAUNCHED before step N-1's sampled tokens
+    # reach the host. The GPU stays busy through the host-side bookkeeping
+    # that used to sit between steps (~ms per step at bs=256). Token VALUES
+    # flow device-side between steps (runner fast path / pending_map);
+    # host bookkeeping (append, EOS/length finish) happens one step late at
+    # _resolve_pending(). A sequence that emits EOS costs one speculative
+    # extra decode step (its KV write lands in its own still-allocated spare
+    # block — scheduler pre-allocates +1 token, so it's harmless).
     @torch.no_grad()
     def step(self) -> List[Sequence]:
         """One engine iteration. Returns sequences that FINISHED this step."""
         batch = self.scheduler.schedule()
         if batch is None:
-            return []
-        sampled = None if batch.is_prefill else getattr(self, "_last_decode_tokens", None)
-        logits = self.runner.execute(batch, sampled)
-        tokens = self.sampler.sample(logits, batch.seqs)
-        if not batch.is_prefill:
-            self._last_decode_tokens = tokens
-        tokens_cpu = tokens.tolist()
+            return self._resolve_pending()
         finished: List[Sequence] = []
-        for seq, tok in zip(batch.seqs, tokens_cpu):
+        if batch.is_prefill:
+            # prefill steps are synchronous: drain the pipeline first so the
+            # prefill batch sees fully-committed state.
+            finished += self._resolve_pending()
+            logits = self.runner.execute(batch)
+            tokens = self.sampler.sample(logits, batch.seqs)
+            for seq in batch.seqs:
+                seq.sched_len = seq.num_prompt_tokens + 1
+                self.num_prompt_tokens += seq.num_prompt_tokens
+            finished += self._commit(batch.seqs, tokens.tolist(),
+                                     [s.epoch for s in batch.seqs])
+            return finished
+
+        pend = self._pending
+        sampled = pend.tokens if pend is not None else None
+        pending_map = (pend.tokens, pend.index) if pend is not None else None
+        logits = self.runner.execute(batch, sampled, pending_map)
+        tokens = self.sampler.sample(logits, batch.seqs)
+        for seq in batch.seqs:
+            seq.sched_len = seq.sched_tokens + 1
+        host_copy = None
+        if tokens.is_cuda:
+            host_copy = self._pinned(len(tokens))
+            host_copy.copy_(tokens, non_blocking=True)
+            event = torch.cuda.Event()
+            event.record()
+        else:
+            event = None
+        new_pend = _PendingStep(
+            seqs=list(batch.seqs),
+            epochs=[s.epoch for s in batch.seqs],
+            tokens=tokens,
+            host=host_copy,
+            event=event,
+            index={s.seq_id: i for i, s in enumerate(batch.seqs)})
+        # resolve the PREVIOUS step while the GPU runs this one
+        finished += self._resolve_pending()
+        self._pending = new_pend
+        return finished
+
+    def _pinned(self, n: int) -> torch.Tensor:
+        buf = getattr(self, "_pin_bufs", None)
+        if buf is None:
+            self._pin_bufs = {}
+            buf = self._pin_bufs
+        t = buf.get(n)
+        if t is None:
+            t = torch.empty(n, dtype=torch.long, pin_memory=True)
+            buf[n] = t
+        return t
+
+    def _resolve_pending(self) -> List[Sequence]:
+        p = self._pending
+        if p is None:
+            return []
+        self._pending = None
+        if p.event is not None:
+            p.event.synchronize()
+            vals = p.host.tolist()
+        else:
+            vals = p.tokens.tolist()
+        return self._commit(p.seqs, vals, p.epochs)
+
+    def flush(self) -> List[Sequence]:
+        """Drain the pipelined step (bench/end-of-stream)."""
+        return self._resolve_pending()
+
+    def _commit(self, seqs: List[Sequence], vals: List[int],
+                epochs: List[int]) -> List[Sequence]:
+        finished: List[Sequence] = []
+        for seq, tok, ep in zip(seqs, vals, epochs):
+            if seq.epoch != ep or seq.status == SeqStatus.FINISHED:
+                continue  # preempted or already finished: drop stale token
             seq.append_token(int(tok))
             self.num_generation_tokens += 1
-            if batch.is_prefill:
-                self.num_prompt_tokens += seq.num_prompt_tokens
             if seq.check_finished(self.eos_token_id):
                 self.scheduler.finish(seq)
                 finished.append(seq)

@@ -117,7 +117,7 @@ class ModelRunner:
     # ------------------------------------------------------------- decode
     def _init_decode_buffers(self):
         cfg = self.cfg
-        max_bs = max(cfg.graph_batch_sizes)
+        max_bs = max(max(cfg.graph_batch_sizes), cfg.max_num_seqs)
         mb = cfg.max_blocks_per_seq(self.max_model_len)
         dev = self.device
         self._buf = {
@@ -211,13 +211,20 @@ class ModelRunner:
 
     @torch.no_grad()
     def execute_decode(self, seqs: List[Sequence],
-                       sampled: Optional[torch.Tensor] = None) -> torch.Tensor:
+                       sampled: Optional[torch.Tensor] = None,
+                       pending_map: Optional[Tuple[torch.Tensor, dict]] = None
+                       ) -> torch.Tensor:
         """One token per sequence; returns hidden [B, H].
 
         Fast path: when the running set is unchanged since the previous
         decode step and `sampled` (last step's token ids, on device) is
         given, all input updates happen device-side — no host round-trip
         except dirty block-table rows (block-boundary crossings).
+
+        Rebuild path: lengths come from seq.sched_tokens (pipelined steps may
+        be unresolved); input token VALUES for sequences with an in-flight
+        step are sourced device-side from `pending_map` = (tokens_tensor,
+        {seq_id: index}).
         """
         if not hasattr(self, "_buf"):
             self._init_decode_buffers()
@@ -243,12 +250,25 @@ class ModelRunner:
             b["slot_mapping"][:bs].copy_(
                 blk * self.cfg.block_size + pos % self.cfg.block_size)
         else:
-            ids = [s.last_token_id for s in seqs]
-            pos = [s.num_tokens - 1 for s in seqs]
+            ptoks, pmap = pending_map if pending_map is not None else (None, {})
+            ids, dst, src = [], [], []
+            for i, s in enumerate(seqs):
+                j = pmap.get(s.seq_id)
+                if j is not None and s.sched_tokens > s.num_tokens:
+                    ids.append(0)
+                    dst.append(i)
+                    src.append(j)
+                else:
+                    ids.append(s.last_token_id)
+            pos = [s.sched_tokens - 1 for s in seqs]
             slots = [self._slot(s, p) for s, p in zip(seqs, pos)]
-            lens = [s.num_tokens for s in seqs]
+            lens = [s.sched_tokens for s in seqs]
             b["input_ids"][:bs].copy_(
                 torch.tensor(ids, dtype=torch.long), non_blocking=True)
+            if dst:
+                didx = torch.tensor(dst, dtype=torch.long, device=self.device)
+                sidx = torch.tensor(src, dtype=torch.long, device=self.device)
+                b["input_ids"][:bs].index_copy_(0, didx, ptoks[sidx])
             b["positions"][:bs].copy_(
                 torch.tensor(pos, dtype=torch.long), non_blocking=True)
             b["slot_mapping"][:bs].copy_(
@@ -278,10 +298,11 @@ class ModelRunner:
     # ------------------------------------------------------------- step
     @torch.no_grad()
     def execute(self, batch: ScheduledBatch,
-                sampled: Optional[torch.Tensor] = None) -> torch.Tensor:
+                sampled: Optional[torch.Tensor] = None,
+                pending_map=None) -> torch.Tensor:
         """Run the batch; returns logits [B, vocab] for the last tokens."""
         if batch.is_prefill:
             hidden = self.execute_prefill(batch.seqs)
         else:
-            hidden = self.execute_decode(batch.seqs, sampled)
+            hidden = self.execute_decode(batch.seqs, sampled, pending_map)
         return self.model.compute_logits(hidden)

@@ -90,27 +90,34 @@ class Scheduler:
     def _schedule_decode(self) -> Optional[ScheduledBatch]:
         if not self.running:
             return None
+        # skip seqs whose output is fully scheduled (pipelined step still in
+        # flight — the engine resolves & finishes them)
+        batch = [s for s in self.running
+                 if s.sched_output_tokens < s.sampling.max_tokens]
         # grow block tables; preempt newest if pool dry
         i = 0
-        while i < len(self.running):
-            seq = self.running[i]
-            need = self.pool.blocks_needed(seq.num_tokens + 1) - len(seq.block_table)
+        while i < len(batch):
+            seq = batch[i]
+            need = self.pool.blocks_needed(seq.sched_tokens + 1) - len(seq.block_table)
             if need > 0:
                 if self.pool.can_allocate(need):
                     seq.block_table.extend(self.pool.allocate(need))
                     seq._bt_dirty = True
                 else:
-                    victim = self.running.pop()  # newest
+                    victim = batch.pop()  # newest scheduled
+                    self.running.remove(victim)
                     self.pool.free(victim.block_table)
                     victim.block_table = []
                     victim.output_token_ids = []
+                    victim.sched_len = 0
+                    victim.epoch += 1
                     victim.status = SeqStatus.WAITING
                     self.waiting.appendleft(victim)
-                    continue
+                    continue  # retry same i (batch[i] is seq or its successor)
             i += 1
-        if not self.running:
+        if not batch:
             return None
-        return ScheduledBatch(is_prefill=False, seqs=list(self.running))
+        return ScheduledBatch(is_prefill=False, seqs=batch)
 
     def finish(self, seq: Sequence) -> None:
         self.pool.free(seq.block_table)

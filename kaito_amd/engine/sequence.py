@@ -37,6 +37,12 @@ class Sequence:
     finish_time: Optional[float] = None
     finish_reason: Optional[str] = None
 
+    # tokens scheduled on-device but not yet resolved to host (pipelined
+    # decode); sched_tokens is the authoritative length for KV/blocks.
+    sched_len: int = 0
+    # bumped on preemption so stale in-flight results are dropped at resolve
+    epoch: int = 0
+
     @property
     def num_prompt_tokens(self) -> int:
         return len(self.prompt_token_ids)
@@ -44,6 +50,14 @@ class Sequence:
     @property
     def num_tokens(self) -> int:
         return len(self.prompt_token_ids) + len(self.output_token_ids)
+
+    @property
+    def sched_tokens(self) -> int:
+        return max(self.num_tokens, self.sched_len)
+
+    @property
+    def sched_output_tokens(self) -> int:
+        return self.sched_tokens - len(self.prompt_token_ids)
 
     @property
     def last_token_id(self) -> int:

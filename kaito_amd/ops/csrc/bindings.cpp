@@ -19,6 +19,8 @@ void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
                        at::Tensor tile_seq, at::Tensor tile_qbase,
                        at::Tensor cu_seqlens, double scale);
 at::Tensor mfma_tile_gemm(at::Tensor a, at::Tensor b);
+void topk(at::Tensor out_vals, at::Tensor out_idx, at::Tensor scores,
+          int64_t k);
 }  // namespace kaito
 
 TORCH_LIBRARY(kaito, m) {
@@ -30,6 +32,7 @@ TORCH_LIBRARY(kaito, m) {
   m.def("paged_attention(Tensor(a!) out, Tensor query, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, float scale) -> ()");
   m.def("prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens, float scale) -> ()");
   m.def("mfma_tile_gemm(Tensor a, Tensor b) -> Tensor");
+  m.def("topk(Tensor(a!) out_vals, Tensor(b!) out_idx, Tensor scores, int k) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
@@ -41,4 +44,5 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("paged_attention", &kaito::paged_attention);
   m.impl("prefill_attention", &kaito::prefill_attention);
   m.impl("mfma_tile_gemm", &kaito::mfma_tile_gemm);
+  m.impl("topk", &kaito::topk);
 }
