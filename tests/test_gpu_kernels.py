@@ -197,3 +197,35 @@ def test_engine_gpu_graphs_match_eager():
     gg = graph.generate(prompts, sp)
     for a, b in zip(ge, gg):
         assert a.output_token_ids == b.output_token_ids
+
+
+# ------------------------------------------------------------------ topk
+@pytest.mark.parametrize("Q,N,k", [(1, 1000, 5), (7, 4096, 10), (3, 50, 32),
+                                   (2, 100000, 16)])
+def test_topk_matches_torch(Q, N, k):
+    torch.manual_seed(0)
+    import kaito_amd.ops as O
+    O.load_extension()
+    scores = torch.randn(Q, N, device=DEV).float().contiguous()
+    vals = torch.empty(Q, k, dtype=torch.float32, device=DEV)
+    idx = torch.empty(Q, k, dtype=torch.int32, device=DEV)
+    torch.ops.kaito.topk(vals, idx, scores, k)
+    tv, ti = torch.topk(scores, k, dim=-1)
+    assert torch.allclose(vals, tv, atol=1e-6), (vals, tv)
+    assert torch.equal(idx.long(), ti)
+
+
+def test_flat_index_gpu_matches_cpu():
+    import numpy as np
+    from kaito_amd.ragengine.vector_store import FlatIndex
+    rng = np.random.default_rng(0)
+    vecs = rng.standard_normal((500, 64)).astype(np.float32)
+    q = rng.standard_normal(64).astype(np.float32)
+    cpu = FlatIndex(64, use_gpu=False)
+    gpu = FlatIndex(64, use_gpu=True)
+    for i, v in enumerate(vecs):
+        cpu.add(f"d{i}", v)
+        gpu.add(f"d{i}", v)
+    hc = cpu.search(q, 8)
+    hg = gpu.search(q, 8)
+    assert [h[0] for h in hc] == [h[0] for h in hg]

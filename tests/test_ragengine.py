@@ -1,0 +1,144 @@
+"""RAGEngine tests: vector store, BM25, hybrid fusion, persistence, and
+the FastAPI endpoint surface (reference: presets/ragengine/tests)."""
+import numpy as np
+import pytest
+from fastapi.testclient import TestClient
+
+from kaito_amd.ragengine.bm25 import BM25Index
+from kaito_amd.ragengine.config import RagConfig
+from kaito_amd.ragengine.embeddings import HashEmbedding
+from kaito_amd.ragengine.service import build_rag_app
+from kaito_amd.ragengine.vector_store import (FlatIndex, VectorStoreIndex,
+                                              VectorStoreManager, doc_id_for)
+
+DOCS = [
+    "the quick brown fox jumps over the lazy dog",
+    "kubernetes operators reconcile desired state",
+    "mi355x has 288 gigabytes of hbm3e memory",
+    "paged attention stores kv cache in fixed blocks",
+    "a fox is a small wild canine animal",
+]
+
+
+@pytest.fixture
+def index():
+    idx = VectorStoreIndex("t", HashEmbedding(), use_gpu=False)
+    idx.index_documents(DOCS)
+    return idx
+
+
+def test_doc_id_is_text_hash():
+    assert doc_id_for("abc") == doc_id_for("abc")
+    assert doc_id_for("abc") != doc_id_for("abd")
+    assert len(doc_id_for("abc")) == 64
+
+
+def test_flat_index_add_remove_search():
+    f = FlatIndex(4, use_gpu=False)
+    f.add("a", np.array([1, 0, 0, 0], dtype=np.float32))
+    f.add("b", np.array([0, 1, 0, 0], dtype=np.float32))
+    f.add("c", np.array([0.9, 0.1, 0, 0], dtype=np.float32))
+    hits = f.search(np.array([1, 0, 0, 0], dtype=np.float32), 2)
+    assert hits[0][0] == "a" and hits[1][0] == "c"
+    f.remove("a")
+    hits = f.search(np.array([1, 0, 0, 0], dtype=np.float32), 2)
+    assert hits[0][0] == "c"
+    assert len(f) == 2
+
+
+def test_bm25_ranks_term_matches():
+    b = BM25Index()
+    for i, d in enumerate(DOCS):
+        b.add(str(i), d)
+    hits = b.search("fox", top_k=3)
+    ids = [h[0] for h in hits]
+    assert set(ids[:2]) == {"0", "4"}
+    b.remove("0")
+    hits = b.search("fox", 3)
+    assert hits[0][0] == "4"
+
+
+def test_hybrid_retrieval_prefers_exact_terms(index):
+    out = index.retrieve("fox jumps", top_k=3)
+    assert out
+    assert "fox" in out[0].text
+
+
+def test_update_and_delete(index):
+    did = doc_id_for(DOCS[0])
+    new_id = index.update_document(did, "completely new text about cats")
+    assert new_id != did
+    assert index.delete_document(new_id)
+    assert not index.delete_document("nonexistent")
+
+
+def test_persist_load_roundtrip(tmp_path, index):
+    index.persist(str(tmp_path / "t"))
+    idx2 = VectorStoreIndex("t", HashEmbedding(), use_gpu=False)
+    idx2.load(str(tmp_path / "t"))
+    assert len(idx2.docs) == len(DOCS)
+    out = idx2.retrieve("kubernetes operators", top_k=1)
+    assert "kubernetes" in out[0].text
+
+
+# ------------------------------------------------------------ service
+@pytest.fixture
+def client(tmp_path):
+    cfg = RagConfig()
+    cfg.persist_dir = str(tmp_path)
+    emb = HashEmbedding()
+    app = build_rag_app(cfg, emb, VectorStoreManager(emb, use_gpu=False))
+    return TestClient(app)
+
+
+def test_service_index_and_retrieve(client):
+    r = client.post("/index", json={
+        "index_name": "kb",
+        "documents": [{"text": d} for d in DOCS]})
+    assert r.status_code == 200
+    assert len(r.json()) == len(DOCS)
+    assert client.get("/indexes").json() == ["kb"]
+    r = client.post("/retrieve", json={"index_name": "kb",
+                                       "query": "kv cache blocks", "top_k": 2})
+    assert r.status_code == 200
+    res = r.json()["results"]
+    assert res and "kv cache" in res[0]["text"]
+
+
+def test_service_documents_crud(client):
+    client.post("/index", json={"index_name": "kb",
+                                "documents": [{"text": "hello world"}]})
+    docs = client.get("/indexes/kb/documents").json()
+    assert docs["count"] == 1
+    did = docs["documents"][0]["doc_id"]
+    r = client.post(f"/indexes/kb/documents/{did}",
+                    json={"text": "goodbye world"})
+    assert r.status_code == 200
+    r = client.delete(f"/indexes/kb/documents/{r.json()['doc_id']}")
+    assert r.status_code == 200
+    assert client.get("/indexes/kb/documents").json()["count"] == 0
+
+
+def test_service_persist_load(client):
+    client.post("/index", json={"index_name": "kb",
+                                "documents": [{"text": "persist me"}]})
+    assert client.post("/persist/kb").status_code == 200
+    assert client.delete("/indexes/kb").status_code == 200
+    assert client.post("/load/kb").status_code == 200
+    r = client.post("/retrieve", json={"index_name": "kb",
+                                       "query": "persist", "top_k": 1})
+    assert r.json()["results"][0]["text"] == "persist me"
+
+
+def test_service_404s(client):
+    assert client.post("/retrieve", json={
+        "index_name": "nope", "query": "x"}).status_code == 404
+    assert client.get("/indexes/nope/documents").status_code == 404
+    assert client.delete("/indexes/nope").status_code == 404
+
+
+def test_service_health_metrics(client):
+    assert client.get("/health").json() == {"status": "ok"}
+    client.post("/index", json={"index_name": "kb",
+                                "documents": [{"text": "m"}]})
+    assert "kaito_rag_request_latency_seconds" in client.get("/metrics").text
