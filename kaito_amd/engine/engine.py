@@ -138,14 +138,18 @@ class LLMEngine:
         return finished
 
     def _pinned(self, n: int) -> torch.Tensor:
-        buf = getattr(self, "_pin_bufs", None)
-        if buf is None:
-            self._pin_bufs = {}
-            buf = self._pin_bufs
-        t = buf.get(n)
+        """Pinned host staging buffers, DOUBLE-buffered: step N's async D2H
+        copy must not overwrite step N-1's still-unresolved values."""
+        bufs = getattr(self, "_pin_bufs", None)
+        if bufs is None:
+            bufs = self._pin_bufs = {}
+            self._pin_flip = 0
+        self._pin_flip ^= 1
+        key = (n, self._pin_flip)
+        t = bufs.get(key)
         if t is None:
             t = torch.empty(n, dtype=torch.long, pin_memory=True)
-            buf[n] = t
+            bufs[key] = t
         return t
 
     def _resolve_pending(self) -> List[Sequence]:

@@ -23,12 +23,16 @@ class Sampler:
 
     @torch.no_grad()
     def sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> torch.Tensor:
-        """logits: [n, vocab] (float). Returns [n] long on logits.device."""
-        temps = torch.tensor([s.sampling.temperature for s in seqs],
-                             device=logits.device, dtype=torch.float32)
-        greedy = temps <= 0.0
-        if bool(greedy.all()):
+        """logits: [n, vocab] (float). Returns [n] long on logits.device.
+
+        All branch decisions come from HOST-side sampling params — no
+        device reads, so the call enqueues async (pipelined decode relies
+        on sample() never synchronizing)."""
+        temps_l = [s.sampling.temperature for s in seqs]
+        if all(t <= 0.0 for t in temps_l):
             return logits.argmax(dim=-1)
+        temps = torch.tensor(temps_l, device=logits.device, dtype=torch.float32)
+        greedy = temps <= 0.0
 
         logits = logits.float()
         scaled = logits / temps.clamp(min=1e-5).unsqueeze(1)
@@ -42,8 +46,9 @@ class Sampler:
                     scaled[i][scaled[i] < kth] = float("-inf")
 
         # top-p (nucleus)
-        ps = torch.tensor([s.sampling.top_p for s in seqs], device=logits.device)
-        if bool((ps < 1.0).any()):
+        ps_l = [s.sampling.top_p for s in seqs]
+        ps = torch.tensor(ps_l, device=logits.device)
+        if any(p < 1.0 for p in ps_l):
             sorted_logits, idx = torch.sort(scaled, descending=True, dim=-1)
             probs = torch.softmax(sorted_logits, dim=-1)
             cum = probs.cumsum(dim=-1)
