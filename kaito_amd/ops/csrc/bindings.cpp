@@ -21,6 +21,9 @@ void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
 at::Tensor mfma_tile_gemm(at::Tensor a, at::Tensor b);
 void topk(at::Tensor out_vals, at::Tensor out_idx, at::Tensor scores,
           int64_t k);
+void lora_shrink(at::Tensor tmp, at::Tensor x, at::Tensor A, at::Tensor idx,
+                 double scale);
+void lora_expand(at::Tensor y, at::Tensor tmp, at::Tensor B, at::Tensor idx);
 }  // namespace kaito
 
 TORCH_LIBRARY(kaito, m) {
@@ -33,6 +36,8 @@ TORCH_LIBRARY(kaito, m) {
   m.def("prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens, float scale) -> ()");
   m.def("mfma_tile_gemm(Tensor a, Tensor b) -> Tensor");
   m.def("topk(Tensor(a!) out_vals, Tensor(b!) out_idx, Tensor scores, int k) -> ()");
+  m.def("lora_shrink(Tensor(a!) tmp, Tensor x, Tensor A, Tensor idx, float scale) -> ()");
+  m.def("lora_expand(Tensor(a!) y, Tensor tmp, Tensor B, Tensor idx) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
@@ -45,4 +50,6 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("prefill_attention", &kaito::prefill_attention);
   m.impl("mfma_tile_gemm", &kaito::mfma_tile_gemm);
   m.impl("topk", &kaito::topk);
+  m.impl("lora_shrink", &kaito::lora_shrink);
+  m.impl("lora_expand", &kaito::lora_expand);
 }
