@@ -64,6 +64,9 @@ class EngineConfig:
     num_gpu_blocks: Optional[int] = None    # None → probe free VRAM
     tensor_parallel_size: int = 1
     enforce_eager: bool = False             # disable hipGraph decode capture
+    enable_lora: bool = False
+    max_loras: int = 8
+    max_lora_rank: int = 64
     device: str = "cuda"
     seed: int = 0
     # decode graph buckets (batch sizes to capture)

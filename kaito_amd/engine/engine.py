@@ -58,10 +58,18 @@ class LLMEngine:
     # ------------------------------------------------------------- requests
     def add_request(self, prompt_token_ids: List[int],
                     sampling: Optional[SamplingParams] = None,
-                    seq_id: Optional[int] = None) -> int:
+                    seq_id: Optional[int] = None,
+                    lora_name: Optional[str] = None) -> int:
         sid = seq_id if seq_id is not None else next(self._next_id)
         seq = Sequence(sid, list(prompt_token_ids),
                        sampling or SamplingParams())
+        if lora_name is not None:
+            mgr = self.runner.lora_manager
+            if mgr is None:
+                raise ValueError("LoRA not enabled (EngineConfig.enable_lora)")
+            seq.lora_id = mgr.slot(lora_name)
+            if seq.lora_id < 0:
+                raise KeyError(f"unknown LoRA adapter {lora_name!r}")
         self.seqs[sid] = seq
         self.scheduler.add(seq)
         return sid

@@ -25,6 +25,7 @@ TUNABLE_FILE = Path(__file__).resolve().parent.parent / "ops" / "tunableop_gfx95
 from .config import EngineConfig
 from .scheduler import ScheduledBatch
 from .sequence import Sequence
+from . import lora as lora_mod
 from ..models.llama import AttnMetadata, LlamaForCausalLM
 
 logger = logging.getLogger(__name__)
@@ -42,6 +43,11 @@ class ModelRunner:
         self.max_model_len = cfg.max_model_len or cfg.model.max_position
         self._graphs: Dict[int, tuple] = {}
         self._graph_pool = None
+        self.lora_manager = None
+        if cfg.enable_lora:
+            self.lora_manager = lora_mod.LoRAManager(
+                self.model, cfg.max_loras, cfg.max_lora_rank,
+                device=cfg.device, dtype=cfg.model.dtype)
 
     # ------------------------------------------------------------- setup
     def load_model(self, weights_path: Optional[str] = None, seed: int = 0):
@@ -110,7 +116,15 @@ class ModelRunner:
             slot_mapping=torch.tensor(slots, dtype=torch.long, device=dev),
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
             max_seqlen=max(s.num_prompt_tokens for s in seqs))
+        if self.lora_manager is not None:
+            tok_ids = []
+            for seq in seqs:
+                tok_ids.extend([seq.lora_id] * seq.num_prompt_tokens)
+            lora_mod.set_active(self.lora_manager, torch.tensor(
+                tok_ids, dtype=torch.int32, device=dev))
         hidden = self.model(input_ids, positions, self.kv_caches, meta)
+        if self.lora_manager is not None:
+            lora_mod.set_active(None, None)
         last_idx = torch.tensor([c - 1 for c in cu[1:]], device=dev)
         return hidden[last_idx]
 
@@ -126,11 +140,14 @@ class ModelRunner:
             "slot_mapping": torch.full((max_bs,), -1, dtype=torch.long, device=dev),
             "block_tables": torch.zeros(max_bs, mb, dtype=torch.int32, device=dev),
             "seq_lens": torch.zeros(max_bs, dtype=torch.int32, device=dev),
+            "lora_ids": torch.full((max_bs,), -1, dtype=torch.int32, device=dev),
         }
         self._max_blocks = mb
 
     def _decode_forward(self, bs: int) -> torch.Tensor:
         b = self._buf
+        if self.lora_manager is not None:
+            lora_mod.set_active(self.lora_manager, b["lora_ids"][:bs])
         meta = AttnMetadata(
             is_prefill=False,
             slot_mapping=b["slot_mapping"][:bs],
@@ -282,6 +299,10 @@ class ModelRunner:
                 s._bt_dirty = False
             b["block_tables"][:bs].copy_(
                 torch.from_numpy(flat), non_blocking=True)
+            if self.lora_manager is not None:
+                b["lora_ids"][:bs].copy_(torch.tensor(
+                    [s.lora_id for s in seqs], dtype=torch.int32),
+                    non_blocking=True)
         self._last_ids = ids_now
 
         bucket = self._graph_bucket(bs) if self._graphs else None

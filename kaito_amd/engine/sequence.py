@@ -42,6 +42,7 @@ class Sequence:
     sched_len: int = 0
     # bumped on preemption so stale in-flight results are dropped at resolve
     epoch: int = 0
+    lora_id: int = -1            # adapter slot (-1 = base model)
 
     @property
     def num_prompt_tokens(self) -> int:

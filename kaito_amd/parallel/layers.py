@@ -11,6 +11,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .state import get_state, tp_all_reduce
+from ..engine import lora as lora_mod
 
 
 class ColumnParallelLinear(nn.Module):
@@ -29,7 +30,8 @@ class ColumnParallelLinear(nn.Module):
             requires_grad=False) if bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(x, self.weight, self.bias)
+        out = F.linear(x, self.weight, self.bias)
+        return lora_mod.maybe_apply(self, x, out)
 
 
 class RowParallelLinear(nn.Module):
@@ -50,6 +52,7 @@ class RowParallelLinear(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         out = F.linear(x, self.weight)
+        out = lora_mod.maybe_apply(self, x, out)
         out = tp_all_reduce(out)
         if self.bias is not None:
             out = out + self.bias

@@ -229,3 +229,42 @@ def test_flat_index_gpu_matches_cpu():
     hc = cpu.search(q, 8)
     hg = gpu.search(q, 8)
     assert [h[0] for h in hc] == [h[0] for h in hg]
+
+
+# ------------------------------------------------------------------ LoRA
+def test_lora_kernels_match_cpu_manager():
+    from kaito_amd.engine.lora import LoRAAdapter, LoRAManager
+    from kaito_amd.models import get_model_config
+    from kaito_amd.models.llama import LlamaForCausalLM
+    from kaito_amd.parallel.state import init_parallel
+    init_parallel(1)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(get_model_config("tiny-llama-test"))
+    cpu_mgr = LoRAManager(model, max_adapters=2, max_rank=8, device="cpu")
+    gpu_mgr = LoRAManager(model, max_adapters=2, max_rank=8, device="cuda")
+    key = "layers.0.mlp.down_proj"
+    kin, out = cpu_mgr.module_shapes[key]
+    A = torch.randn(8, kin) * 0.1
+    B = torch.randn(out, 8) * 0.1
+    for mgr in (cpu_mgr, gpu_mgr):
+        mgr.register(LoRAAdapter("x", 8, 16.0, {key: (A, B)}))
+    T = 9
+    x = torch.randn(T, kin).to(torch.bfloat16)
+    y = torch.randn(T, out).to(torch.bfloat16)
+    ids = torch.tensor([0, -1, 0, 0, -1, 0, 0, 0, -1], dtype=torch.int32)
+    out_cpu = cpu_mgr.apply(key, x, y.clone(), ids)
+    out_gpu = gpu_mgr.apply(key, x.cuda(), y.clone().cuda(), ids.cuda())
+    _close(out_gpu, out_cpu.to(DEV), atol=0.05, rtol=0.05)
+
+
+def test_engine_gpu_lora_matches_cpu_tokens():
+    from kaito_amd.engine import SamplingParams
+    eng = _gpu_engine(enforce_eager=True, enable_lora=True, max_lora_rank=16)
+    eng.runner.lora_manager.register_random("a", rank=8, seed=3, scale=0.3)
+    prompt = list(range(30, 50))
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+    sid = eng.add_request(prompt, sp, lora_name="a")
+    base = eng.add_request(prompt, sp)
+    while eng.has_unfinished():
+        eng.step()
+    assert eng.seqs[sid].output_token_ids != eng.seqs[base].output_token_ids
