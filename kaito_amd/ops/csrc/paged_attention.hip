@@ -65,24 +65,46 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
   }
 
   const int nchunks = (seq_len + 7) / 8;  // 8 tokens per wave-step
-  for (int c = wave; c < nchunks; c += NW) {
-    const int tok = c * 8 + tg;
-    const bool valid = tok < seq_len;
-    const int tok_c = valid ? tok : (seq_len - 1);
-    const int blk = bt[tok_c / BS];
-    const int64_t base = (((int64_t)blk * KH + kvh) * BS + (tok_c % BS)) * D + dc * DL;
 
-    // ---- K tile: 16 dims/lane → f32 ----
-    float kreg[DL];
-    {
-      const short8_t* kp = reinterpret_cast<const short8_t*>(k_cache + base);
+  // software-pipelined K/V streaming: issue chunk c+NW's loads (raw bf16
+  // registers) before computing chunk c, so HBM latency hides under the
+  // dot/softmax VALU work (guide T14 async-split).
+  auto chunk_base = [&](int c) -> int64_t {
+    const int tok = c * 8 + tg;
+    const int tok_c = (tok < seq_len) ? tok : (seq_len - 1);
+    const int blk = bt[tok_c / BS];
+    return (((int64_t)blk * KH + kvh) * BS + (tok_c % BS)) * D + dc * DL;
+  };
+  constexpr int NV = DL / 8;              // short8 vectors per lane (2 for D=128)
+  short8_t kraw[NV], vraw[NV];
+  if (wave < nchunks) {
+    const int64_t b0 = chunk_base(wave);
 #pragma unroll
-      for (int vv = 0; vv < DL / 8; vv++) {
-        short8_t x = kp[vv];
+    for (int vv = 0; vv < NV; vv++) {
+      kraw[vv] = reinterpret_cast<const short8_t*>(k_cache + b0)[vv];
+      vraw[vv] = reinterpret_cast<const short8_t*>(v_cache + b0)[vv];
+    }
+  }
+
+  for (int c = wave; c < nchunks; c += NW) {
+    const bool valid = (c * 8 + tg) < seq_len;
+    // issue next chunk's loads now; first use is next iteration
+    short8_t knext[NV], vnext[NV];
+    const int cn = c + NW;
+    if (cn < nchunks) {
+      const int64_t bn = chunk_base(cn);
 #pragma unroll
-        for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(x[j]);
+      for (int vv = 0; vv < NV; vv++) {
+        knext[vv] = reinterpret_cast<const short8_t*>(k_cache + bn)[vv];
+        vnext[vv] = reinterpret_cast<const short8_t*>(v_cache + bn)[vv];
       }
     }
+
+    float kreg[DL];
+#pragma unroll
+    for (int vv = 0; vv < NV; vv++)
+#pragma unroll
+      for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(kraw[vv][j]);
 
     float s[G];
 #pragma unroll
@@ -95,15 +117,12 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
     }
 
     // ---- V tile (reuse kreg registers) ----
-    {
-      const short8_t* vp = reinterpret_cast<const short8_t*>(v_cache + base);
 #pragma unroll
-      for (int vv = 0; vv < DL / 8; vv++) {
-        short8_t x = vp[vv];
+    for (int vv = 0; vv < NV; vv++)
 #pragma unroll
-        for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(x[j]);
-      }
-    }
+      for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(vraw[vv][j]);
+#pragma unroll
+    for (int vv = 0; vv < NV; vv++) { kraw[vv] = knext[vv]; vraw[vv] = vnext[vv]; }
 
 #pragma unroll
     for (int g = 0; g < G; g++) {
