@@ -102,8 +102,11 @@ class LoRAManager:
         for key, (A, B) in adapter.weights.items():
             if key not in self.module_shapes:
                 continue
-            st = self._ensure_stack(key)
+            k, o = self.module_shapes[key]
             r = A.shape[0]
+            if A.shape[1] != k or B.shape[0] != o or r > self.max_rank:
+                continue  # adapter trained for a different base shape
+            st = self._ensure_stack(key)
             # fold the adapter's alpha/r scaling into A once
             st["A"][slot, :r].copy_((A * adapter.scaling).to(self.dtype))
             st["A"][slot, r:].zero_()

@@ -79,7 +79,7 @@ void lora_shrink(at::Tensor tmp, at::Tensor x, at::Tensor A, at::Tensor idx,
   const int T = x.size(0);
   const int K = x.size(-1);
   const int R = A.size(1);
-  TORCH_CHECK(A.size(2) == K && K % 512 == 0, "K must be multiple of 512");
+  TORCH_CHECK(A.size(2) == K && K % 8 == 0, "K must be multiple of 8");
   TORCH_CHECK(R <= 64);
   if (T == 0) return;
   auto stream = at::hip::getCurrentHIPStream();

@@ -3,7 +3,11 @@
 hot kernels at bench-like shapes. Run under gpurun."""
 import argparse
 import math
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
