@@ -1,0 +1,151 @@
+"""InferenceSet reconciler — Python re-implementation of the reference's
+pkg/inferenceset/inferenceset_controller.go (765 L): N replicas of a
+Workspace template, upgrade-aware scale-down (old-revision-first, keep
+Ready >= desired), label propagation, TPM aggregation, HPA/KEDA selector.
+"""
+from __future__ import annotations
+
+import copy
+import hashlib
+import json
+from dataclasses import dataclass
+from typing import Dict, List, Optional
+
+from ..api_types import (COND_INFERENCESET_READY, Condition, InferenceSet,
+                         LABEL_INFERENCESET_CREATED_BY, LABEL_UPGRADE_TO_VERSION,
+                         LABEL_WORKSPACE_NAME, Workspace)
+from ..kubeclient import KubeClient, NotFound
+
+
+@dataclass
+class ISReconcileResult:
+    requeue_after_s: float = 0.0
+    created: int = 0
+    deleted: int = 0
+
+
+def _ws_obj(iset: InferenceSet, index: int, revision: str) -> Dict:
+    """Materialize the workspace template as a stored object dict."""
+    tpl = iset.spec.workspaceTemplate
+    return {
+        "apiVersion": "kaito.sh/v1beta1",
+        "kind": "Workspace",
+        "metadata": {
+            "name": f"{iset.name}-{index}",
+            "namespace": iset.namespace,
+            "labels": {
+                LABEL_INFERENCESET_CREATED_BY: iset.name,
+                "inferenceset.kaito.io/revision": revision,
+            },
+        },
+        "spec": {
+            "resource": {
+                "instanceType": tpl.resource.instanceType,
+                "count": tpl.resource.count,
+            },
+            "inference": {
+                "preset": tpl.inference.preset.name
+                if tpl.inference and tpl.inference.preset else None,
+            },
+        },
+        "status": {},
+    }
+
+
+class InferenceSetReconciler:
+    def __init__(self, client: KubeClient):
+        self.client = client
+
+    def _revision(self, iset: InferenceSet) -> str:
+        tpl = iset.spec.workspaceTemplate
+        payload = json.dumps({
+            "instanceType": tpl.resource.instanceType,
+            "preset": tpl.inference.preset.name
+            if tpl.inference and tpl.inference.preset else None,
+        }, sort_keys=True)
+        return hashlib.sha256(payload.encode()).hexdigest()[:10]
+
+    def _list_children(self, iset: InferenceSet) -> List[Dict]:
+        return self.client.list("Workspace", iset.namespace, {
+            LABEL_INFERENCESET_CREATED_BY: iset.name})
+
+    @staticmethod
+    def _is_ready(ws_obj: Dict) -> bool:
+        return ws_obj.get("status", {}).get("state") == "Running"
+
+    def select_workspaces_to_delete(self, children: List[Dict], excess: int,
+                                    revision: str) -> List[Dict]:
+        """Reference parity: selectWorkspacesToDelete
+        (inferenceset_controller.go:225-300) — delete old-revision first,
+        then not-ready, then newest index; keep Ready >= desired."""
+        def sort_key(obj):
+            old_rev = obj["metadata"]["labels"].get(
+                "inferenceset.kaito.io/revision") != revision
+            ready = self._is_ready(obj)
+            idx = int(obj["metadata"]["name"].rsplit("-", 1)[1])
+            # delete priority: old revision first, then not-ready, then
+            # highest index
+            return (not old_rev, ready, -idx)
+
+        return sorted(children, key=sort_key)[:excess]
+
+    def reconcile(self, iset: InferenceSet) -> ISReconcileResult:
+        iset.validate()
+        revision = self._revision(iset)
+        children = self._list_children(iset)
+        desired = iset.spec.replicas
+        res = ISReconcileResult()
+
+        if len(children) > desired:
+            for obj in self.select_workspaces_to_delete(
+                    children, len(children) - desired, revision):
+                self.client.delete("Workspace", iset.namespace,
+                                   obj["metadata"]["name"])
+                res.deleted += 1
+        elif len(children) < desired:
+            used = {int(o["metadata"]["name"].rsplit("-", 1)[1])
+                    for o in children}
+            idx = 0
+            for _ in range(desired - len(children)):
+                while idx in used:
+                    idx += 1
+                used.add(idx)
+                self.client.create(_ws_obj(iset, idx, revision))
+                res.created += 1
+
+        # upgrade: children on an old revision get the upgrade label
+        # (consumed by shouldUpgradeBaseImage, workspace_controller.go:695)
+        for obj in self._list_children(iset):
+            labels = obj["metadata"].setdefault("labels", {})
+            if labels.get("inferenceset.kaito.io/revision") != revision and \
+                    LABEL_UPGRADE_TO_VERSION not in labels:
+                labels[LABEL_UPGRADE_TO_VERSION] = revision
+                self.client.update(obj)
+
+        # status
+        children = self._list_children(iset)
+        ready = sum(1 for o in children if self._is_ready(o))
+        iset.status.replicas = len(children)
+        iset.status.readyReplicas = ready
+        iset.status.selector = f"{LABEL_INFERENCESET_CREATED_BY}={iset.name}"
+        # aggregated TPM (:177-193,429-514)
+        tpm = 0.0
+        for o in children:
+            for m in o.get("status", {}).get("performance", {}) \
+                    .get("metrics", []):
+                if m.get("name") == "peakTokensPerMinute" and m.get("value"):
+                    tpm += float(m["value"])
+        iset.status.aggregatedPeakTokensPerMinute = tpm
+        ok = ready >= desired
+        found = False
+        for c in iset.status.conditions:
+            if c.type == COND_INFERENCESET_READY:
+                c.status = "True" if ok else "False"
+                found = True
+        if not found:
+            iset.status.conditions.append(Condition(
+                COND_INFERENCESET_READY, "True" if ok else "False",
+                "Ready" if ok else "ScalingOrWaiting"))
+        if not ok:
+            res.requeue_after_s = 5.0
+        return res

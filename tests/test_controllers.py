@@ -1,0 +1,257 @@
+"""Controller reconcile tests against FakeKubeClient — the reference's
+MockClient unit-test strategy: no cluster, full reconcile
+paths, golden assertions on created objects and conditions."""
+import json
+
+import pytest
+
+from kaito_amd.operator import api_types as at
+from kaito_amd.operator.controllers.inferenceset import InferenceSetReconciler
+from kaito_amd.operator.controllers.workspace import (WorkspaceReconciler,
+                                                      classify_pod_failure)
+from kaito_amd.operator.kubeclient import FakeKubeClient, NotFound
+from kaito_amd.operator.nodeprovision import (BYOProvisioner,
+                                              KarpenterProvisioner,
+                                              make_provisioner)
+from kaito_amd.operator.sku import get_sku_handler
+
+SKU = "Standard_ND96isr_MI355X_v1"
+
+
+def _node(name, ready=True, labels=None):
+    return {"apiVersion": "v1", "kind": "Node",
+            "metadata": {"name": name, "namespace": "", "labels": labels or {}},
+            "status": {"conditions": [
+                {"type": "Ready", "status": "True" if ready else "False"}]}}
+
+
+def _ws(name="ws1", preset="llama-3-8b", count=1):
+    return at.Workspace(
+        name=name,
+        resource=at.ResourceSpec(instanceType=SKU, count=count),
+        inference=at.InferenceSpec(preset=at.PresetSpec(name=preset)))
+
+
+def _store_ws(client, ws):
+    client.create({"apiVersion": "kaito.sh/v1beta1", "kind": "Workspace",
+                   "metadata": {"name": ws.name, "namespace": ws.namespace},
+                   "spec": {}, "status": {}})
+
+
+@pytest.fixture
+def client():
+    return FakeKubeClient()
+
+
+def _reconciler(client, **kw):
+    prov = BYOProvisioner(client)
+    return WorkspaceReconciler(client, get_sku_handler("azure"), prov, **kw)
+
+
+# ------------------------------------------------------------ workspace
+def test_workspace_waits_for_nodes(client):
+    ws = _ws()
+    _store_ws(client, ws)
+    r = _reconciler(client)
+    res = r.reconcile(ws)
+    assert res.requeue and res.requeue_after_s == 2.0
+    assert ws.status.state == "Pending"
+    conds = {c.type: c.status for c in ws.status.conditions}
+    assert conds["NodesReady"] == "False"
+
+
+def test_workspace_full_reconcile_to_running(client):
+    ws = _ws()
+    _store_ws(client, ws)
+    client.create(_node("n1", labels={
+        "node.kubernetes.io/instance-type": SKU}))
+    r = _reconciler(client)
+    res = r.reconcile(ws)  # creates statefulset; pods not ready yet
+    assert not res.done
+    ss = client.get("StatefulSet", "default", "ws1")
+    assert ss["spec"]["replicas"] == 1
+    assert client.get("Service", "default", "ws1")
+    assert client.get("Service", "default", "ws1-headless")
+    # simulate statefulset becoming ready
+    ss["status"] = {"readyReplicas": 1}
+    client.update(ss)
+    res = r.reconcile(ws)
+    assert res.done
+    assert ws.status.state == "Running"
+    assert ws.status.workerNodes == ["n1"]
+    conds = {c.type: c.status for c in ws.status.conditions}
+    assert conds["InferenceReady"] == "True"
+    # status pushed to the stored object
+    stored = client.get("Workspace", "default", "ws1")
+    assert stored["status"]["state"] == "Running"
+
+
+def test_workspace_estimator_sets_target_node_count(client):
+    ws = _ws(preset="llama-3-70b")
+    _store_ws(client, ws)
+    r = _reconciler(client)
+    r.reconcile(ws)
+    assert ws.status.targetNodeCount == 1  # 70B fits one MI355X node
+
+
+def test_workspace_pod_failure_classification(client):
+    ws = _ws()
+    _store_ws(client, ws)
+    client.create(_node("n1", labels={
+        "node.kubernetes.io/instance-type": SKU}))
+    client.create({
+        "apiVersion": "v1", "kind": "Pod",
+        "metadata": {"name": "ws1-0", "namespace": "default",
+                     "labels": {at.LABEL_WORKSPACE_NAME: "ws1"}},
+        "status": {"phase": "Running", "containerStatuses": [{
+            "state": {"waiting": {"reason": "CrashLoopBackOff"}},
+            "lastState": {"terminated": {"reason": "OOMKilled"}}}]},
+    })
+    r = _reconciler(client)
+    res = r.reconcile(ws)
+    assert ws.status.state == "Failed"
+    conds = {c.type: c.reason for c in ws.status.conditions}
+    assert conds["InferenceReady"] == "OOMKilled"
+
+
+def test_classify_variants():
+    assert classify_pod_failure({"status": {"phase": "Pending", "conditions": [
+        {"type": "PodScheduled", "status": "False"}]}}) == "Unschedulable"
+    assert classify_pod_failure({"status": {"containerStatuses": [
+        {"state": {"waiting": {"reason": "ImagePullBackOff"}}}]}}) == \
+        "ImagePullFailure"
+    assert classify_pod_failure({"status": {"phase": "Running"}}) is None
+
+
+def test_workspace_benchmark_ingestion(client):
+    ws = _ws()
+    _store_ws(client, ws)
+    client.create(_node("n1", labels={
+        "node.kubernetes.io/instance-type": SKU}))
+    logs = ("startup...\n"
+            'KAITO_BENCHMARK_CONFIG: {"engine": "kaito-amd", "engineVersion": "0.1"}\n'
+            'KAITO_BENCHMARK_RESULT: {"peakTokensPerMinute": 1523400}\n')
+    r = _reconciler(client, get_pod_logs=lambda ns, name: logs)
+    r.reconcile(ws)
+    ss = client.get("StatefulSet", "default", "ws1")
+    ss["status"] = {"readyReplicas": 1}
+    client.update(ss)
+    r.reconcile(ws)
+    metrics = ws.status.performance["metrics"]
+    assert metrics[0]["name"] == "peakTokensPerMinute"
+    assert metrics[0]["value"] == 1523400
+    assert metrics[0]["unit"] == "tokens/min"
+
+
+def test_workspace_tuning_job(client):
+    ws = at.Workspace(
+        name="tune1",
+        resource=at.ResourceSpec(instanceType=SKU),
+        tuning=at.TuningSpec(preset=at.PresetSpec(name="llama-3-8b"),
+                             method="qlora",
+                             input=at.DataSource(urls=["http://d/x.json"]),
+                             output=at.DataDestination(image="out:v1")))
+    _store_ws(client, ws)
+    client.create(_node("n1", labels={
+        "node.kubernetes.io/instance-type": SKU}))
+    r = _reconciler(client)
+    res = r.reconcile(ws)
+    assert not res.done
+    job = client.get("Job", "default", "tune1")
+    job["status"] = {"succeeded": 1}
+    client.update(job)
+    res = r.reconcile(ws)
+    assert res.done and ws.status.state == "Succeeded"
+
+
+# ------------------------------------------------------------ provisioners
+def test_karpenter_provisioner_creates_claims(client):
+    ws = _ws()
+    prov = make_provisioner("karpenter", client)
+    created = prov.provision_nodes(ws, 2)
+    assert len(created) == 2
+    claims = client.list("NodeClaim", "default")
+    assert len(claims) == 2
+    assert claims[0]["spec"]["requirements"][0]["values"] == [SKU]
+    # idempotent
+    assert prov.provision_nodes(ws, 2) == []
+    # readiness: bind a node
+    c = claims[0]
+    c["status"] = {"nodeName": "nk1"}
+    client.update(c)
+    client.create(_node("nk1"))
+    assert prov.ensure_nodes_ready(ws, 2) == ["nk1"]
+    prov.delete_nodes(ws)
+    assert client.list("NodeClaim", "default") == []
+
+
+def test_byo_prefers_preferred_nodes(client):
+    ws = _ws()
+    ws.resource.preferredNodes = ["p2"]
+    client.create(_node("p1", labels={
+        "node.kubernetes.io/instance-type": SKU}))
+    client.create(_node("p2", labels={
+        "node.kubernetes.io/instance-type": SKU}))
+    prov = BYOProvisioner(client)
+    assert prov.ensure_nodes_ready(ws, 1) == ["p2"]
+
+
+# ------------------------------------------------------------ inferenceset
+def _iset(replicas=2):
+    return at.InferenceSet("is1", spec=at.InferenceSetSpec(
+        replicas=replicas, workspaceTemplate=_ws("tpl")))
+
+
+def test_inferenceset_scale_up_down(client):
+    r = InferenceSetReconciler(client)
+    iset = _iset(3)
+    res = r.reconcile(iset)
+    assert res.created == 3
+    names = [o["metadata"]["name"]
+             for o in client.list("Workspace", "default")]
+    assert names == ["is1-0", "is1-1", "is1-2"]
+    iset.spec.replicas = 1
+    res = r.reconcile(iset)
+    assert res.deleted == 2
+    assert len(client.list("Workspace", "default")) == 1
+
+
+def test_inferenceset_deletes_old_revision_first(client):
+    r = InferenceSetReconciler(client)
+    iset = _iset(2)
+    r.reconcile(iset)
+    # mark is1-0 as old revision + Running, is1-1 current + Running
+    for i, obj in enumerate(client.list("Workspace", "default")):
+        obj["status"] = {"state": "Running"}
+        if obj["metadata"]["name"] == "is1-0":
+            obj["metadata"]["labels"]["inferenceset.kaito.io/revision"] = "old"
+        client.update(obj)
+    iset.spec.replicas = 1
+    r.reconcile(iset)
+    remaining = client.list("Workspace", "default")
+    assert [o["metadata"]["name"] for o in remaining] == ["is1-1"]
+
+
+def test_inferenceset_aggregates_tpm_and_selector(client):
+    r = InferenceSetReconciler(client)
+    iset = _iset(2)
+    r.reconcile(iset)
+    for obj in client.list("Workspace", "default"):
+        obj["status"] = {"state": "Running", "performance": {
+            "metrics": [{"name": "peakTokensPerMinute", "value": 100.0}]}}
+        client.update(obj)
+    r.reconcile(iset)
+    assert iset.status.readyReplicas == 2
+    assert iset.status.aggregatedPeakTokensPerMinute == 200.0
+    assert iset.status.selector == "inferenceset.kaito.sh/created-by=is1"
+
+
+def test_inferenceset_marks_upgrade_label(client):
+    r = InferenceSetReconciler(client)
+    iset = _iset(1)
+    r.reconcile(iset)
+    # change the template → new revision
+    iset.spec.workspaceTemplate.inference.preset.name = "llama-3-70b"
+    r.reconcile(iset)
+    obj = client.list("Workspace", "default")[0]
+    assert at.LABEL_UPGRADE_TO_VERSION in obj["metadata"]["labels"]
