@@ -50,6 +50,8 @@ class LLMEngine:
         # counters for /metrics (names consumed by the benchmark probe)
         self.num_generation_tokens = 0
         self.num_prompt_tokens = 0
+        # optional KV event bus (EPP KVCache-aware routing surface)
+        self.kv_publisher = None
 
     def capture_graphs(self):
         self.runner.capture_decode_graphs()
@@ -114,6 +116,9 @@ class LLMEngine:
             for seq in batch.seqs:
                 seq.sched_len = seq.num_prompt_tokens + 1
                 self.num_prompt_tokens += seq.num_prompt_tokens
+            if self.kv_publisher is not None:
+                blocks = [b for s2 in batch.seqs for b in s2.block_table]
+                self.kv_publisher.block_stored(blocks)
             finished += self._commit(batch.seqs, tokens.tolist(),
                                      [s.epoch for s in batch.seqs])
             return finished
@@ -185,6 +190,8 @@ class LLMEngine:
             seq.append_token(int(tok))
             self.num_generation_tokens += 1
             if seq.check_finished(self.eos_token_id):
+                if self.kv_publisher is not None:
+                    self.kv_publisher.block_removed(list(seq.block_table))
                 self.scheduler.finish(seq)
                 finished.append(seq)
         return finished

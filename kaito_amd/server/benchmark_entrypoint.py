@@ -1,0 +1,150 @@
+"""Self-benchmark startup probe — parity with the reference's
+presets/workspace/inference/vllm/benchmark_entrypoint.py (644 L):
+
+  1. wait for /health;
+  2. read /metrics → vllm:cache_config_info num_gpu_blocks → compute the
+     KV-pool saturation concurrency (reference :149-265);
+  3. drive the engine at that concurrency (input 2048 / output 256,
+     default 60 s — reference :48-50 uses guidellm; ours is a built-in
+     async load generator);
+  4. compute peakTokensPerMinute from vllm:generation_tokens deltas;
+  5. emit KAITO_BENCHMARK_CONFIG / KAITO_BENCHMARK_RESULT JSON lines to
+     /proc/1/fd/1 (pod log) so the Workspace controller ingests them
+     (pkg/workspace/controllers/benchmark.go:47-73).
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import re
+import sys
+import time
+
+import httpx
+
+CONFIG_TAG = "KAITO_BENCHMARK_CONFIG"
+RESULT_TAG = "KAITO_BENCHMARK_RESULT"
+
+
+def emit(tag: str, payload: dict) -> None:
+    line = f"{tag}: {json.dumps(payload)}"
+    try:
+        with open("/proc/1/fd/1", "w") as f:   # pod's main stdout
+            f.write(line + "\n")
+    except OSError:
+        pass
+    print(line, flush=True)
+
+
+def parse_metric(text: str, name: str) -> float:
+    for line in text.splitlines():
+        if line.startswith(name):
+            try:
+                return float(line.rsplit(" ", 1)[1])
+            except ValueError:
+                continue
+    return 0.0
+
+
+def parse_cache_config(text: str):
+    m = re.search(r'vllm:cache_config_info\{([^}]*)\}', text)
+    if not m:
+        return None
+    labels = dict(kv.split("=", 1) for kv in m.group(1).split(","))
+    return {k: v.strip('"') for k, v in labels.items()}
+
+
+async def wait_healthy(base: str, timeout_s: float) -> bool:
+    deadline = time.monotonic() + timeout_s
+    async with httpx.AsyncClient() as client:
+        while time.monotonic() < deadline:
+            try:
+                r = await client.get(f"{base}/health", timeout=5)
+                if r.status_code == 200:
+                    return True
+            except httpx.HTTPError:
+                pass
+            await asyncio.sleep(2)
+    return False
+
+
+async def run_benchmark(base: str, duration_s: int, in_tokens: int,
+                        out_tokens: int, max_concurrency: int) -> dict:
+    async with httpx.AsyncClient(timeout=300) as client:
+        metrics = (await client.get(f"{base}/metrics")).text
+        cache = parse_cache_config(metrics) or {}
+        block_size = int(cache.get("block_size", 16))
+        num_blocks = int(cache.get("num_gpu_blocks", 1024))
+        pool_tokens = block_size * num_blocks
+        concurrency = max(1, min(pool_tokens // (in_tokens + out_tokens),
+                                 max_concurrency))
+        tok0 = parse_metric(metrics, "vllm:generation_tokens")
+        t0 = time.monotonic()
+        stop = t0 + duration_s
+        stats = {"completed": 0, "errors": 0}
+
+        async def worker(wid: int):
+            rng_base = wid * 1009
+            i = 0
+            while time.monotonic() < stop:
+                i += 1
+                prompt = [(rng_base + j * 31 + i) % 30000 + 10
+                          for j in range(in_tokens)]
+                try:
+                    r = await client.post(f"{base}/v1/completions", json={
+                        "prompt": prompt, "max_tokens": out_tokens,
+                        "temperature": 0.0, "ignore_eos": True})
+                    if r.status_code == 200:
+                        stats["completed"] += 1
+                    elif r.status_code == 429:
+                        await asyncio.sleep(0.5)
+                    else:
+                        stats["errors"] += 1
+                except httpx.HTTPError:
+                    stats["errors"] += 1
+        await asyncio.gather(*(worker(w) for w in range(concurrency)))
+        elapsed = time.monotonic() - t0
+        metrics2 = (await client.get(f"{base}/metrics")).text
+        tok1 = parse_metric(metrics2, "vllm:generation_tokens")
+        tpm = (tok1 - tok0) / elapsed * 60.0
+        return {
+            "concurrency": concurrency,
+            "elapsedSeconds": round(elapsed, 1),
+            "completedRequests": stats["completed"],
+            "errors": stats["errors"],
+            "peakTokensPerMinute": round(tpm, 1),
+        }
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser()
+    p.add_argument("--base-url", default="http://127.0.0.1:5000")
+    p.add_argument("--duration", type=int, default=60)
+    p.add_argument("--input-tokens", type=int, default=2048)
+    p.add_argument("--output-tokens", type=int, default=256)
+    p.add_argument("--max-concurrency", type=int, default=512)
+    p.add_argument("--health-timeout", type=float, default=3600)
+    args = p.parse_args(argv)
+
+    if not asyncio.run(wait_healthy(args.base_url, args.health_timeout)):
+        print("engine never became healthy", file=sys.stderr)
+        return 1
+    emit(CONFIG_TAG, {
+        "engine": "kaito-amd",
+        "engineVersion": "0.1.0",
+        "quantization": "none",
+        "inputTokens": args.input_tokens,
+        "outputTokens": args.output_tokens,
+        "description": "stress/high-concurrency",
+    })
+    result = asyncio.run(run_benchmark(
+        args.base_url, args.duration, args.input_tokens, args.output_tokens,
+        args.max_concurrency))
+    emit(RESULT_TAG, result)
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -111,3 +111,46 @@ def test_rate_limit_429():
         assert c.get("/health").status_code == 200
         m = c.get("/metrics").text
         assert "kaito_ratelimit_rejected" in m
+
+
+def test_kv_event_bus_roundtrip():
+    from kaito_amd.engine.kv_events import (KVEventPublisher,
+                                            KVEventSubscriber, BLOCK_STORED)
+    pub = KVEventPublisher(host="127.0.0.1", port=0)
+    sub = KVEventSubscriber(port=pub.port)
+    time.sleep(0.2)  # let subscription register
+    pub.block_stored([1, 2, 3])
+    pub.block_removed([2])
+    pub.all_cleared()
+    deadline = time.monotonic() + 3
+    while len(sub.events) < 3 and time.monotonic() < deadline:
+        time.sleep(0.05)
+    assert len(sub.events) == 3
+    assert sub.events[0]["event"] == BLOCK_STORED
+    assert sub.events[0]["block_hashes"] == [1, 2, 3]
+    assert sub.events[2]["event"] == "AllBlocksCleared"
+    sub.close()
+    pub.close()
+
+
+def test_engine_publishes_kv_events():
+    from kaito_amd.engine.kv_events import KVEventPublisher, KVEventSubscriber
+    from kaito_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from kaito_amd.models import get_model_config
+    init_parallel(1)
+    cfg = EngineConfig(model=get_model_config("tiny-llama-test"), device="cpu",
+                       max_num_seqs=4, num_gpu_blocks=64, enforce_eager=True,
+                       max_model_len=64)
+    eng = LLMEngine(cfg)
+    pub = KVEventPublisher(host="127.0.0.1", port=0)
+    eng.kv_publisher = pub
+    sub = KVEventSubscriber(port=pub.port)
+    time.sleep(0.2)
+    eng.generate([[1, 2, 3, 4]], SamplingParams(max_tokens=3, ignore_eos=True))
+    deadline = time.monotonic() + 3
+    while len(sub.events) < 2 and time.monotonic() < deadline:
+        time.sleep(0.05)
+    kinds = [e["event"] for e in sub.events]
+    assert "BlockStored" in kinds and "BlockRemoved" in kinds
+    sub.close()
+    pub.close()
