@@ -67,6 +67,8 @@ class EngineConfig:
     enable_lora: bool = False
     max_loras: int = 8
     max_lora_rank: int = 64
+    kv_offload: bool = False                # LMCache-style host KV cache
+    kv_offload_bytes: Optional[int] = None  # None → 0.5 * available RAM
     device: str = "cuda"
     seed: int = 0
     # decode graph buckets (batch sizes to capture)

@@ -52,6 +52,12 @@ class LLMEngine:
         self.num_prompt_tokens = 0
         # optional KV event bus (EPP KVCache-aware routing surface)
         self.kv_publisher = None
+        self.kv_offload = None
+        if cfg.kv_offload:
+            from .kv_offload import KVOffloadManager
+            self.kv_offload = KVOffloadManager(
+                self.runner.kv_caches, cfg.block_size,
+                cfg.kv_offload_bytes, cfg.device)
 
     def capture_graphs(self):
         self.runner.capture_decode_graphs()
@@ -111,6 +117,19 @@ class LLMEngine:
             # prefill steps are synchronous: drain the pipeline first so the
             # prefill batch sees fully-committed state.
             finished += self._resolve_pending()
+            if self.kv_offload is not None:
+                fresh = []
+                for seq in batch.seqs:
+                    if self.kv_offload.restore(seq.prompt_token_ids,
+                                               seq.block_table):
+                        # full-prompt KV hit: skip prefill; the next decode
+                        # step feeds the last prompt token over restored KV
+                        seq.sched_len = seq.num_prompt_tokens
+                    else:
+                        fresh.append(seq)
+                batch.seqs = fresh
+                if not batch.seqs:
+                    return finished
             logits = self.runner.execute(batch)
             tokens = self.sampler.sample(logits, batch.seqs)
             for seq in batch.seqs:
@@ -190,6 +209,10 @@ class LLMEngine:
             seq.append_token(int(tok))
             self.num_generation_tokens += 1
             if seq.check_finished(self.eos_token_id):
+                if self.kv_offload is not None:
+                    self.kv_offload.offload(
+                        seq.prompt_token_ids + seq.output_token_ids,
+                        seq.block_table)
                 if self.kv_publisher is not None:
                     self.kv_publisher.block_removed(list(seq.block_table))
                 self.scheduler.finish(seq)

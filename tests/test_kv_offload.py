@@ -1,0 +1,64 @@
+"""KV host-offload tests: full-prompt restore skips prefill and reproduces
+identical greedy outputs; LRU eviction under budget."""
+import pytest
+import torch
+
+from kaito_amd.engine import EngineConfig, LLMEngine, SamplingParams
+from kaito_amd.models import get_model_config
+from kaito_amd.parallel.state import init_parallel
+
+
+@pytest.fixture(autouse=True)
+def _p():
+    init_parallel(1)
+
+
+def _engine(**kw):
+    base = dict(model=get_model_config("tiny-llama-test"), device="cpu",
+                max_num_seqs=8, num_gpu_blocks=64, enforce_eager=True,
+                max_model_len=128, kv_offload=True,
+                kv_offload_bytes=64 << 20)
+    base.update(kw)
+    return LLMEngine(EngineConfig(**base))
+
+
+def test_restore_skips_prefill_same_tokens():
+    eng = _engine()
+    prompt = list(range(20, 52))
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+    first = eng.generate([prompt], sp)[0].output_token_ids
+    assert eng.kv_offload.hits == 0
+    # resubmit the identical prompt: restore path must hit and skip prefill
+    calls = {"prefill": 0}
+    orig = eng.runner.execute_prefill
+
+    def counting(seqs):
+        calls["prefill"] += 1
+        return orig(seqs)
+
+    eng.runner.execute_prefill = counting
+    second = eng.generate([prompt], sp)[0].output_token_ids
+    assert eng.kv_offload.hits == 1
+    assert calls["prefill"] == 0, "prefill should be skipped on full hit"
+    assert second == first
+
+
+def test_miss_on_different_prompt():
+    eng = _engine()
+    sp = SamplingParams(max_tokens=4, ignore_eos=True)
+    eng.generate([[1, 2, 3, 4]], sp)
+    eng.generate([[1, 2, 3, 5]], sp)
+    assert eng.kv_offload.hits == 0
+    assert eng.kv_offload.misses >= 1
+
+
+def test_lru_eviction_under_budget():
+    eng = _engine(kv_offload_bytes=1 << 15)  # tiny: ~1 seq worth
+    sp = SamplingParams(max_tokens=2, ignore_eos=True)
+    p1 = list(range(10, 30))
+    p2 = list(range(40, 60))
+    eng.generate([p1], sp)
+    eng.generate([p2], sp)  # evicts p1's entry
+    eng.generate([p1], sp)
+    assert eng.kv_offload.hits == 0  # p1 was evicted
+    assert eng.kv_offload.used_bytes <= 1 << 15
