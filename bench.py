@@ -28,8 +28,8 @@ import torch
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=300)
-    p.add_argument("--warmup", type=int, default=40)
+    p.add_argument("--steps", type=int, default=400)
+    p.add_argument("--warmup", type=int, default=250)
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--tp", type=int, default=1)
     p.add_argument("--max-num-seqs", type=int, default=512)

@@ -76,36 +76,32 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
     return (((int64_t)blk * KH + kvh) * BS + (tok_c % BS)) * D + dc * DL;
   };
   constexpr int NV = DL / 8;              // short8 vectors per lane (2 for D=128)
-  short8_t kraw[NV], vraw[NV];
-  if (wave < nchunks) {
-    const int64_t b0 = chunk_base(wave);
+
+  // 2-wide chunk unroll: two INDEPENDENT softmax/accumulate chains per
+  // iteration (ILP across the shfl/exp latency chains) + prefetch of the
+  // following pair (HBM latency hidden under both chains).
+  short8_t kraw[2][NV], vraw[2][NV];
+  auto load_chunk = [&](int c, short8_t (&kd)[NV], short8_t (&vd)[NV]) {
+    const int64_t b = chunk_base(c);
 #pragma unroll
     for (int vv = 0; vv < NV; vv++) {
-      kraw[vv] = reinterpret_cast<const short8_t*>(k_cache + b0)[vv];
-      vraw[vv] = reinterpret_cast<const short8_t*>(v_cache + b0)[vv];
+      kd[vv] = reinterpret_cast<const short8_t*>(k_cache + b)[vv];
+      vd[vv] = reinterpret_cast<const short8_t*>(v_cache + b)[vv];
     }
-  }
+  };
+  if (wave < nchunks) load_chunk(wave, kraw[0], vraw[0]);
+  if (wave + NW < nchunks) load_chunk(wave + NW, kraw[1], vraw[1]);
 
-  for (int c = wave; c < nchunks; c += NW) {
+  // one chunk's full update: dot → online-softmax → accumulate
+  auto process = [&](int c, short8_t (&kd)[NV], short8_t (&vd)[NV]) {
+    const bool valid_c = c < nchunks;
+    if (!valid_c) return;
     const bool valid = (c * 8 + tg) < seq_len;
-    // issue next chunk's loads now; first use is next iteration
-    short8_t knext[NV], vnext[NV];
-    const int cn = c + NW;
-    if (cn < nchunks) {
-      const int64_t bn = chunk_base(cn);
-#pragma unroll
-      for (int vv = 0; vv < NV; vv++) {
-        knext[vv] = reinterpret_cast<const short8_t*>(k_cache + bn)[vv];
-        vnext[vv] = reinterpret_cast<const short8_t*>(v_cache + bn)[vv];
-      }
-    }
-
     float kreg[DL];
 #pragma unroll
     for (int vv = 0; vv < NV; vv++)
 #pragma unroll
-      for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(kraw[vv][j]);
-
+      for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(kd[vv][j]);
     float s[G];
 #pragma unroll
     for (int g = 0; g < G; g++) {
@@ -115,18 +111,12 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
       p = group_reduce_sum<8>(p) * scale;   // dot over 8 dc-lanes
       s[g] = valid ? p : -1e30f;
     }
-
-    // ---- V tile (reuse kreg registers) ----
 #pragma unroll
     for (int vv = 0; vv < NV; vv++)
 #pragma unroll
-      for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(vraw[vv][j]);
-#pragma unroll
-    for (int vv = 0; vv < NV; vv++) { kraw[vv] = knext[vv]; vraw[vv] = vnext[vv]; }
-
+      for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(vd[vv][j]);
 #pragma unroll
     for (int g = 0; g < G; g++) {
-      // wave-max over the 8 token-groups (s uniform across dc lanes)
       float cm = s[g];
 #pragma unroll
       for (int off = 8; off < 64; off <<= 1) cm = fmaxf(cm, __shfl_xor(cm, off, 64));
@@ -144,6 +134,19 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
       l[g] += psum;
 #pragma unroll
       for (int j = 0; j < DL; j++) acc[g][j] += p * kreg[j];
+    }
+  };
+
+  for (int c = wave; c < nchunks; c += 2 * NW) {
+    short8_t kn[2][NV], vn[2][NV];
+    if (c + 2 * NW < nchunks) load_chunk(c + 2 * NW, kn[0], vn[0]);
+    if (c + 3 * NW < nchunks) load_chunk(c + 3 * NW, kn[1], vn[1]);
+    process(c, kraw[0], vraw[0]);
+    process(c + NW, kraw[1], vraw[1]);
+#pragma unroll
+    for (int vv = 0; vv < NV; vv++) {
+      kraw[0][vv] = kn[0][vv]; vraw[0][vv] = vn[0][vv];
+      kraw[1][vv] = kn[1][vv]; vraw[1][vv] = vn[1][vv];
     }
   }
 
