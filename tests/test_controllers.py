@@ -255,3 +255,123 @@ def test_inferenceset_marks_upgrade_label(client):
     r.reconcile(iset)
     obj = client.list("Workspace", "default")[0]
     assert at.LABEL_UPGRADE_TO_VERSION in obj["metadata"]["labels"]
+
+
+# ------------------------------------------------------------ modelmirror
+def test_modelmirror_managed_flow(client):
+    from kaito_amd.operator.controllers.modelmirror import (
+        ModelMirror, ModelMirrorReconciler, PHASE_DOWNLOADING, PHASE_READY)
+    logs = {"text": "downloading... 42%\n"}
+    r = ModelMirrorReconciler(client,
+                              get_pod_logs=lambda ns, n: logs["text"])
+    mm = ModelMirror(name="llama8b", modelName="llama-3-8b")
+    # first reconcile: creates PVC + job, phase pending→downloading
+    assert r.reconcile(mm) == "Pending"
+    assert client.get("PersistentVolumeClaim", "kaito-system",
+                      "modelmirror-llama8b")
+    job = client.get("Job", "kaito-system", "modelmirror-llama8b-download")
+    job["status"] = {"active": 1}
+    client.update(job)
+    assert r.reconcile(mm) == PHASE_DOWNLOADING
+    assert mm.status["progress"] == 42
+    job["status"] = {"succeeded": 1}
+    client.update(job)
+    assert r.reconcile(mm) == PHASE_READY
+
+
+def test_modelmirror_static_and_failure(client):
+    from kaito_amd.operator.controllers.modelmirror import (
+        ModelMirror, ModelMirrorReconciler, classify_download_failure)
+    r = ModelMirrorReconciler(client)
+    mm = ModelMirror(name="s", mode="Static", staticVolumePath="/weights/x")
+    assert r.reconcile(mm) == "Ready"
+    assert classify_download_failure("HTTP 401 Unauthorized") == "AuthFailure"
+    assert classify_download_failure("model not found 404") == "ModelNotFound"
+    assert classify_download_failure("no space left on device") == "OutOfDisk"
+
+
+# ------------------------------------------------------------ multirole
+def test_multirole_creates_role_sets_and_pool(client):
+    from kaito_amd.operator.controllers.multirole import (
+        MultiRoleInference, MultiRoleInferenceReconciler, RoleSpec)
+    r = MultiRoleInferenceReconciler(client)
+    mri = MultiRoleInference(name="pd", preset="llama-3-8b",
+                             prefill=RoleSpec(replicas=2, instanceType=SKU),
+                             decode=RoleSpec(replicas=3, instanceType=SKU))
+    st = r.reconcile(mri)
+    pre = client.get("InferenceSet", "default", "pd-prefill")
+    dec = client.get("InferenceSet", "default", "pd-decode")
+    assert pre["spec"]["replicas"] == 2 and dec["spec"]["replicas"] == 3
+    assert pre["metadata"]["labels"][at.LABEL_INFERENCE_ROLE] == "prefill"
+    pool = client.get("InferencePool", "default", "pd-pool")
+    plugins = [p["name"] for p in pool["spec"]["eppConfig"]["plugins"]]
+    assert "prefill-filter" in plugins and "kv-cache-utilization-scorer" in plugins
+    assert not st["ready"]
+    # roles become ready
+    for obj in (pre, dec):
+        obj["status"] = {"readyReplicas": obj["spec"]["replicas"]}
+        client.update(obj)
+    st = r.reconcile(mri)
+    assert st["ready"]
+    # scale decode
+    mri.decode.replicas = 5
+    r.reconcile(mri)
+    assert client.get("InferenceSet", "default",
+                      "pd-decode")["spec"]["replicas"] == 5
+
+
+# ------------------------------------------------------------ drift/upgrade
+def test_drift_serializes_remediation(client):
+    from kaito_amd.operator.controllers.lifecycle import DriftReconciler
+    for i in range(2):
+        client.create({
+            "apiVersion": "kaito.sh/v1beta1", "kind": "Workspace",
+            "metadata": {"name": f"is1-{i}", "namespace": "default",
+                         "labels": {at.LABEL_INFERENCESET_CREATED_BY: "is1"},
+                         "annotations": {"kaito.sh/node-drifted": "true"}},
+            "spec": {}, "status": {}})
+    d = DriftReconciler(client)
+    active = d.reconcile("is1")
+    assert active == "is1-0"
+    np = client.get("NodePool", "default", "is1-0-nodepool")
+    assert np["spec"]["disruption"]["budgets"] == [{"nodes": "1"}]
+    # second tick: still the same one (serialized)
+    assert d.reconcile("is1") == "is1-0"
+    # remediation completes: annotation cleared
+    ws = client.get("Workspace", "default", "is1-0")
+    ws["metadata"]["annotations"] = {}
+    client.update(ws)
+    assert d.reconcile("is1") == "is1-1"
+    np = client.get("NodePool", "default", "is1-0-nodepool")
+    assert np["spec"]["disruption"]["budgets"] == [{"nodes": "0"}]
+
+
+def test_autoupgrade_surge_and_inplace(client):
+    from kaito_amd.operator.controllers.lifecycle import AutoUpgradeRunner
+    for i in range(2):
+        client.create({
+            "apiVersion": "kaito.sh/v1beta1", "kind": "Workspace",
+            "metadata": {"name": f"is2-{i}", "namespace": "default",
+                         "labels": {
+                             at.LABEL_INFERENCESET_CREATED_BY: "is2",
+                             "inferenceset.kaito.io/revision": "old"}},
+            "spec": {}, "status": {}})
+    surge = AutoUpgradeRunner(client, "v2", strategy="Surge")
+    assert sorted(surge.poll("is2")) == ["is2-0", "is2-1"]
+    ws = client.get("Workspace", "default", "is2-0")
+    assert ws["metadata"]["labels"][at.LABEL_UPGRADE_TO_VERSION] == "v2"
+    inplace = AutoUpgradeRunner(client, "v3", strategy="InPlace")
+    inplace.poll("is2")
+    ws = client.get("Workspace", "default", "is2-0")
+    assert ws["metadata"]["labels"]["inferenceset.kaito.io/revision"] == "v3"
+
+
+def test_maintenance_window():
+    import time as _t
+    from kaito_amd.operator.controllers.lifecycle import in_maintenance_window
+    now = _t.struct_time((2026, 1, 1, 3, 0, 0, 2, 1, 0))  # 03:00, Wed
+    assert in_maintenance_window("0 3 * * *", now)
+    assert not in_maintenance_window("0 4 * * *", now)
+    assert in_maintenance_window("", now)
+    assert in_maintenance_window("0 3 * * 2", now)
+    assert not in_maintenance_window("0 3 * * 5", now)
