@@ -43,14 +43,17 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
   const int* bt = block_tables + (int64_t)seq * max_blocks;
 
   // ---- preload Q for this kv-head's group (f32) ----
+  // lane-dim mapping: vector vv holds dims vv*64 + dc*8 + [0..7] so each
+  // 16B lane-load is CONTIGUOUS across the 8 dc-lanes (dense 128B/token
+  // transactions; the naive dc*DL mapping strides 32B between lanes and
+  // halves effective HBM bandwidth).
   float qreg[G][DL];
 #pragma unroll
   for (int g = 0; g < G; g++) {
-    const short8_t* qp = reinterpret_cast<const short8_t*>(
-        q + (int64_t)seq * q_stride + (kvh * G + g) * D + dc * DL);
 #pragma unroll
     for (int vv = 0; vv < DL / 8; vv++) {
-      short8_t x = qp[vv];
+      short8_t x = *reinterpret_cast<const short8_t*>(
+          q + (int64_t)seq * q_stride + (kvh * G + g) * D + vv * 64 + dc * 8);
 #pragma unroll
       for (int j = 0; j < 8; j++) qreg[g][vv * 8 + j] = bf16_to_f32(x[j]);
     }
@@ -73,7 +76,7 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
     const int tok = c * 8 + tg;
     const int tok_c = (tok < seq_len) ? tok : (seq_len - 1);
     const int blk = bt[tok_c / BS];
-    return (((int64_t)blk * KH + kvh) * BS + (tok_c % BS)) * D + dc * DL;
+    return (((int64_t)blk * KH + kvh) * BS + (tok_c % BS)) * D + dc * 8;
   };
   constexpr int NV = DL / 8;              // short8 vectors per lane (2 for D=128)
 
@@ -85,8 +88,8 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
     const int64_t b = chunk_base(c);
 #pragma unroll
     for (int vv = 0; vv < NV; vv++) {
-      kd[vv] = reinterpret_cast<const short8_t*>(k_cache + b)[vv];
-      vd[vv] = reinterpret_cast<const short8_t*>(v_cache + b)[vv];
+      kd[vv] = *reinterpret_cast<const short8_t*>(k_cache + b + vv * 64);
+      vd[vv] = *reinterpret_cast<const short8_t*>(v_cache + b + vv * 64);
     }
   };
   if (wave < nchunks) load_chunk(wave, kraw[0], vraw[0]);
@@ -166,7 +169,10 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
 #pragma unroll
     for (int g = 0; g < G; g++) {
 #pragma unroll
-      for (int j = 0; j < DL; j++) s_acc[wave][g][dc * DL + j] = acc[g][j];
+      for (int vv = 0; vv < NV; vv++)
+#pragma unroll
+        for (int j = 0; j < 8; j++)
+          s_acc[wave][g][vv * 64 + dc * 8 + j] = acc[g][vv * 8 + j];
       if (dc == 0) { s_ml[wave][g][0] = m[g]; s_ml[wave][g][1] = l[g]; }
     }
   }
