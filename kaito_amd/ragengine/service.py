@@ -21,6 +21,7 @@ from pydantic import BaseModel, Field
 
 from .config import RagConfig
 from .embeddings import make_embedding
+from .guardrails import BufferWindowScanner, PolicyLoader, Scanner
 from .vector_store import VectorStoreManager
 
 RAG_REGISTRY = CollectorRegistry()
@@ -52,6 +53,13 @@ class ChatRequest(BaseModel):
     stream: bool = False
 
 
+def _sse_text_chunk(text: str) -> str:
+    payload = {"object": "chat.completion.chunk",
+               "choices": [{"index": 0, "delta": {"content": text},
+                            "finish_reason": None}]}
+    return f"data: {json.dumps(payload)}\n\n"
+
+
 def build_rag_app(cfg: Optional[RagConfig] = None, embedding=None,
                   manager: Optional[VectorStoreManager] = None) -> FastAPI:
     cfg = cfg or RagConfig()
@@ -60,6 +68,11 @@ def build_rag_app(cfg: Optional[RagConfig] = None, embedding=None,
     app = FastAPI(title="kaito-amd ragengine")
     app.state.manager = manager
     app.state.cfg = cfg
+    scanner = None
+    if cfg.guardrails_enabled:
+        scanner = Scanner(PolicyLoader(cfg.guardrails_policy_path,
+                                       cfg.guardrails_hot_reload))
+    app.state.scanner = scanner
 
     @app.get("/health")
     async def health():
@@ -196,17 +209,56 @@ def build_rag_app(cfg: Optional[RagConfig] = None, embedding=None,
             headers["Authorization"] = f"Bearer {cfg.llm_access_secret}"
         if req.stream:
             async def relay():
+                bw = BufferWindowScanner(scanner) if scanner else None
                 async with httpx.AsyncClient(timeout=300) as client:
                     async with client.stream("POST", cfg.llm_inference_url,
                                              json=payload,
                                              headers=headers) as r:
                         async for line in r.aiter_lines():
-                            if line:
+                            if not line:
+                                continue
+                            if bw is None:
                                 yield line + "\n\n"
+                                continue
+                            # guardrails: re-chunk SSE through the buffer
+                            # window (reference streaming/guardrails.py)
+                            if not line.startswith("data:") or \
+                                    line.strip() == "data: [DONE]":
+                                tail = bw.flush()
+                                if tail:
+                                    yield _sse_text_chunk(tail)
+                                yield line + "\n\n"
+                                continue
+                            try:
+                                obj = json.loads(line[5:].strip())
+                                delta = obj["choices"][0].get(
+                                    "delta", {}).get("content", "")
+                            except (json.JSONDecodeError, KeyError,
+                                    IndexError):
+                                yield line + "\n\n"
+                                continue
+                            cleared = bw.feed(delta)
+                            if bw.blocked:
+                                yield _sse_text_chunk(cleared)
+                                yield "data: [DONE]\n\n"
+                                return
+                            if cleared:
+                                yield _sse_text_chunk(cleared)
             return StreamingResponse(relay(), media_type="text/event-stream")
         async with httpx.AsyncClient(timeout=300) as client:
             r = await client.post(cfg.llm_inference_url, json=payload,
                                   headers=headers)
+            if scanner is not None and r.status_code == 200:
+                try:
+                    body = r.json()
+                    msg = body["choices"][0]["message"]
+                    res = scanner.scan(msg.get("content", ""))
+                    msg["content"] = res.text
+                    return Response(json.dumps(body),
+                                    status_code=200,
+                                    media_type="application/json")
+                except (KeyError, IndexError, ValueError):
+                    pass
             return Response(r.content, status_code=r.status_code,
                             media_type="application/json")
 

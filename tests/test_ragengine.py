@@ -142,3 +142,52 @@ def test_service_health_metrics(client):
     client.post("/index", json={"index_name": "kb",
                                 "documents": [{"text": "m"}]})
     assert "kaito_rag_request_latency_seconds" in client.get("/metrics").text
+
+
+# ------------------------------------------------------------ guardrails
+def test_guardrails_scan_block_and_redact(tmp_path):
+    from kaito_amd.ragengine.guardrails import PolicyLoader, Scanner
+    pol = tmp_path / "policy.yaml"
+    pol.write_text("""
+blocked_keywords: ["forbidden"]
+redactions:
+  - pattern: "\\\\b\\\\d{3}-\\\\d{2}-\\\\d{4}\\\\b"
+    replacement: "[SSN]"
+""")
+    sc = Scanner(PolicyLoader(str(pol)))
+    res = sc.scan("my ssn is 123-45-6789 ok")
+    assert res.ok and "[SSN]" in res.text and "123-45" not in res.text
+    res = sc.scan("this mentions Forbidden things")
+    assert not res.ok and "blocked" in res.text
+
+
+def test_guardrails_hot_reload(tmp_path):
+    import os, time
+    from kaito_amd.ragengine.guardrails import PolicyLoader, Scanner
+    pol = tmp_path / "p.yaml"
+    pol.write_text("blocked_keywords: []\n")
+    sc = Scanner(PolicyLoader(str(pol), hot_reload=True))
+    assert sc.scan("hello bad").ok
+    pol.write_text("blocked_keywords: ['bad']\n")
+    os.utime(pol, (time.time() + 5, time.time() + 5))
+    assert not sc.scan("hello bad").ok
+
+
+def test_buffer_window_catches_split_matches(tmp_path):
+    from kaito_amd.ragengine.guardrails import (BufferWindowScanner,
+                                                PolicyLoader, Scanner)
+    pol = tmp_path / "p.yaml"
+    pol.write_text("blocked_keywords: ['topsecret']\n")
+    bw = BufferWindowScanner(Scanner(PolicyLoader(str(pol))), window=16)
+    out = bw.feed("this is top")
+    out += bw.feed("secret info and much more text to push the window")
+    assert bw.blocked
+    assert "blocked" in out
+    # clean stream passes through
+    bw2 = BufferWindowScanner(Scanner(PolicyLoader(str(pol))), window=8)
+    text = "a perfectly clean sentence streaming through"
+    got = ""
+    for i in range(0, len(text), 7):
+        got += bw2.feed(text[i:i + 7])
+    got += bw2.flush()
+    assert got == text
