@@ -375,3 +375,40 @@ def test_maintenance_window():
     assert in_maintenance_window("", now)
     assert in_maintenance_window("0 3 * * 2", now)
     assert not in_maintenance_window("0 3 * * 5", now)
+
+
+# ------------------------------------------------------------ operator loop
+def test_feature_gates_parse():
+    from kaito_amd.operator.featuregates import (FeatureGateError,
+                                                 parse_feature_gates)
+    g = parse_feature_gates("ModelMirror=true, enableMIG=false")
+    assert g["ModelMirror"] and not g["enableMIG"]
+    assert g["enableInferenceSetController"]  # default preserved
+    with pytest.raises(FeatureGateError):
+        parse_feature_gates("bogusGate=true")
+    with pytest.raises(FeatureGateError):
+        parse_feature_gates("ModelMirror=maybe")
+
+
+def test_operator_loop_end_to_end(client):
+    """CR dict in store → loop tick reconciles it to Running."""
+    from kaito_amd.operator.main import OperatorLoop, workspace_from_obj
+    client.create({
+        "apiVersion": "kaito.sh/v1beta1", "kind": "Workspace",
+        "metadata": {"name": "ws-e2e", "namespace": "default"},
+        "spec": {"resource": {"instanceType": SKU},
+                 "inference": {"preset": {"name": "llama-3-8b"}}},
+        "status": {}})
+    client.create(_node("n1", labels={
+        "node.kubernetes.io/instance-type": SKU}))
+    loop = OperatorLoop(client, provisioner="byo")
+    assert loop.tick() == 1
+    ss = client.get("StatefulSet", "default", "ws-e2e")
+    ss["status"] = {"readyReplicas": 1}
+    client.update(ss)
+    loop.tick()
+    stored = client.get("Workspace", "default", "ws-e2e")
+    assert stored["status"]["state"] == "Running"
+    # typed roundtrip
+    ws = workspace_from_obj(stored)
+    assert ws.inference.preset.name == "llama-3-8b"
