@@ -24,6 +24,8 @@ void topk(at::Tensor out_vals, at::Tensor out_idx, at::Tensor scores,
 void lora_shrink(at::Tensor tmp, at::Tensor x, at::Tensor A, at::Tensor idx,
                  double scale);
 void lora_expand(at::Tensor y, at::Tensor tmp, at::Tensor B, at::Tensor idx);
+void paged_read_bw(at::Tensor out, at::Tensor k_cache, at::Tensor v_cache,
+                   at::Tensor block_tables, at::Tensor seq_lens);
 }  // namespace kaito
 
 TORCH_LIBRARY(kaito, m) {
@@ -38,6 +40,7 @@ TORCH_LIBRARY(kaito, m) {
   m.def("topk(Tensor(a!) out_vals, Tensor(b!) out_idx, Tensor scores, int k) -> ()");
   m.def("lora_shrink(Tensor(a!) tmp, Tensor x, Tensor A, Tensor idx, float scale) -> ()");
   m.def("lora_expand(Tensor(a!) y, Tensor tmp, Tensor B, Tensor idx) -> ()");
+  m.def("paged_read_bw(Tensor(a!) out, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
@@ -52,4 +55,5 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("topk", &kaito::topk);
   m.impl("lora_shrink", &kaito::lora_shrink);
   m.impl("lora_expand", &kaito::lora_expand);
+  m.impl("paged_read_bw", &kaito::paged_read_bw);
 }

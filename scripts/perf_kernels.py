@@ -50,6 +50,12 @@ def main():
     print(f"paged_attention bs={T} L={L}: {t*1e6:.1f} us  "
           f"{bytes_moved/t/1e12:.2f} TB/s")
 
+    # pure-read diagnostic: same grid/address walk, no softmax
+    outd = torch.empty(T * KH, dtype=torch.float32, device=dev)
+    t = bench(lambda: torch.ops.kaito.paged_read_bw(outd, kc, vc, bt, sl))
+    print(f"paged_read_bw  bs={T} L={L}: {t*1e6:.1f} us  "
+          f"{bytes_moved/t/1e12:.2f} TB/s")
+
     # rmsnorm
     x = torch.randn(T, 4096, device=dev).to(torch.bfloat16)
     r = torch.randn_like(x)
