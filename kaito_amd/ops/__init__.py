@@ -119,6 +119,24 @@ def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return torch_ref.prefill_attention(q, k, v, cu_seqlens, scale)
 
 
+def context_attention(q: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, cu_seqlens_q: torch.Tensor,
+                      kv_lens: torch.Tensor, block_tables: torch.Tensor,
+                      scale: float) -> torch.Tensor:
+    """Suffix-query causal attention over the paged cache (chunked
+    prefill / prefix-cache continuation)."""
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty_like(q)
+        tile_seq, tile_qbase = _build_tiles(cu_seqlens_q)
+        torch.ops.kaito.context_attention(out, q, k_cache, v_cache, tile_seq,
+                                          tile_qbase, cu_seqlens_q, kv_lens,
+                                          block_tables, scale)
+        return out
+    return torch_ref.context_attention(q, k_cache, v_cache, cu_seqlens_q,
+                                       kv_lens, block_tables, scale)
+
+
 def mfma_tile_gemm(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     _require_ext()
     return torch.ops.kaito.mfma_tile_gemm(a, b)

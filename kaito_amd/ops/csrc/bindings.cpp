@@ -18,6 +18,11 @@ void paged_attention(at::Tensor out, at::Tensor query, at::Tensor k_cache,
 void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
                        at::Tensor tile_seq, at::Tensor tile_qbase,
                        at::Tensor cu_seqlens, double scale);
+void context_attention(at::Tensor out, at::Tensor q, at::Tensor k_cache,
+                       at::Tensor v_cache, at::Tensor tile_seq,
+                       at::Tensor tile_qbase, at::Tensor cu_seqlens_q,
+                       at::Tensor kv_lens, at::Tensor block_tables,
+                       double scale);
 at::Tensor mfma_tile_gemm(at::Tensor a, at::Tensor b);
 void topk(at::Tensor out_vals, at::Tensor out_idx, at::Tensor scores,
           int64_t k);
@@ -36,6 +41,7 @@ TORCH_LIBRARY(kaito, m) {
   m.def("reshape_and_cache(Tensor k, Tensor v, Tensor(a!) k_cache, Tensor(b!) v_cache, Tensor slot_mapping) -> ()");
   m.def("paged_attention(Tensor(a!) out, Tensor query, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, float scale) -> ()");
   m.def("prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens, float scale) -> ()");
+  m.def("context_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens_q, Tensor kv_lens, Tensor block_tables, float scale) -> ()");
   m.def("mfma_tile_gemm(Tensor a, Tensor b) -> Tensor");
   m.def("topk(Tensor(a!) out_vals, Tensor(b!) out_idx, Tensor scores, int k) -> ()");
   m.def("lora_shrink(Tensor(a!) tmp, Tensor x, Tensor A, Tensor idx, float scale) -> ()");
@@ -51,6 +57,7 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("reshape_and_cache", &kaito::reshape_and_cache);
   m.impl("paged_attention", &kaito::paged_attention);
   m.impl("prefill_attention", &kaito::prefill_attention);
+  m.impl("context_attention", &kaito::context_attention);
   m.impl("mfma_tile_gemm", &kaito::mfma_tile_gemm);
   m.impl("topk", &kaito::topk);
   m.impl("lora_shrink", &kaito::lora_shrink);

@@ -112,3 +112,36 @@ def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         o = torch.einsum("hqk,hkd->hqd", p, vs)
         out[s0:s1] = o.transpose(0, 1).to(q.dtype)
     return out
+
+
+def context_attention(q: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, cu_seqlens_q: torch.Tensor,
+                      kv_lens: torch.Tensor, block_tables: torch.Tensor,
+                      scale: float) -> torch.Tensor:
+    """Suffix-query causal attention over the paged cache. q: [Tq, QH, D];
+    cache holds each sequence's FULL kv_len tokens (suffix included)."""
+    Tq, QH, D = q.shape
+    KH = k_cache.size(1)
+    BS = k_cache.size(2)
+    G = QH // KH
+    out = torch.empty_like(q)
+    cs = cu_seqlens_q.tolist()
+    for b in range(len(cs) - 1):
+        s0, s1 = cs[b], cs[b + 1]
+        q_len = s1 - s0
+        L = int(kv_lens[b])
+        nb = (L + BS - 1) // BS
+        blocks = block_tables[b, :nb].long()
+        keys = k_cache[blocks].permute(1, 0, 2, 3).reshape(KH, nb * BS, D)[:, :L]
+        vals = v_cache[blocks].permute(1, 0, 2, 3).reshape(KH, nb * BS, D)[:, :L]
+        qs = q[s0:s1].float().transpose(0, 1)                  # [QH, q_len, D]
+        kx = keys.float().repeat_interleave(G, 0)
+        vx = vals.float().repeat_interleave(G, 0)
+        att = torch.einsum("hqd,hkd->hqk", qs, kx) * scale
+        qpos = torch.arange(L - q_len, L, device=q.device).unsqueeze(1)
+        kpos = torch.arange(L, device=q.device).unsqueeze(0)
+        att = att.masked_fill((kpos > qpos).unsqueeze(0), float("-inf"))
+        p = torch.softmax(att, dim=-1)
+        o = torch.einsum("hqk,hkd->hqd", p, vx)
+        out[s0:s1] = o.transpose(0, 1).to(q.dtype)
+    return out

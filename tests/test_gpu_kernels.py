@@ -268,3 +268,38 @@ def test_engine_gpu_lora_matches_cpu_tokens():
     while eng.has_unfinished():
         eng.step()
     assert eng.seqs[sid].output_token_ids != eng.seqs[base].output_token_ids
+
+
+@pytest.mark.parametrize("cfgs", [
+    # (kv_len, q_len) pairs per sequence in the batch
+    [(40, 40)],              # full prefill through cache
+    [(100, 30)],             # chunked continuation
+    [(33, 1)],               # single-token (decode-like)
+    [(200, 64), (17, 17), (90, 25)],
+])
+def test_context_attention_gpu(cfgs):
+    torch.manual_seed(11)
+    QH, KH, D, BS = 8, 2, 128, 16
+    kv_lens = [c[0] for c in cfgs]
+    q_lens = [c[1] for c in cfgs]
+    mb = max((L + BS - 1) // BS for L in kv_lens)
+    NB = sum((L + BS - 1) // BS for L in kv_lens) + 1
+    kc = _bf16(NB, KH, BS, D)
+    vc = _bf16(NB, KH, BS, D)
+    bt = torch.zeros(len(cfgs), mb, dtype=torch.int32, device=DEV)
+    nxt = 1
+    for i, L in enumerate(kv_lens):
+        n = (L + BS - 1) // BS
+        bt[i, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    Tq = sum(q_lens)
+    q = _bf16(Tq, QH, D)
+    cu = torch.tensor([0] + list(torch.tensor(q_lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    kvl = torch.tensor(kv_lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.context_attention(q, kc, vc, cu, kvl, bt, scale)
+    ref = R.context_attention(q.cpu().float(), kc.cpu().float(),
+                              vc.cpu().float(), cu.cpu(), kvl.cpu(),
+                              bt.cpu(), scale)
+    _close(out, ref.to(DEV), atol=2e-2)

@@ -93,3 +93,28 @@ def test_prefill_attention_causal():
     # row 0 of each seq attends only to itself → out = v (broadcast over group)
     assert torch.allclose(out[0, 0], v[0, 0], atol=1e-4)
     assert torch.allclose(out[5, 3], v[5, 1], atol=1e-4)  # head 3 → kv head 1
+
+
+def test_context_attention_suffix_matches_full():
+    """Suffix-query context attention == the suffix rows of full prefill."""
+    torch.manual_seed(3)
+    QH, KH, D, BS = 4, 2, 16, 4
+    L, q_len = 22, 6
+    nb = (L + BS - 1) // BS
+    k = torch.randn(L, KH, D)
+    v = torch.randn(L, KH, D)
+    q_full = torch.randn(L, QH, D)
+    full = R.prefill_attention(q_full, k, v,
+                               torch.tensor([0, L], dtype=torch.int32), 0.25)
+    kc = torch.zeros(nb + 1, KH, BS, D)
+    vc = torch.zeros_like(kc)
+    bt = torch.arange(1, nb + 1, dtype=torch.int32).unsqueeze(0)
+    toks = torch.arange(L)
+    slots = (bt[0][toks // BS].long() * BS + toks % BS)
+    R.reshape_and_cache(k, v, kc, vc, slots)
+    ctx = R.context_attention(q_full[-q_len:],
+                              kc, vc,
+                              torch.tensor([0, q_len], dtype=torch.int32),
+                              torch.tensor([L], dtype=torch.int32),
+                              bt, 0.25)
+    assert torch.allclose(ctx, full[-q_len:], atol=1e-4)
