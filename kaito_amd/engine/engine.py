@@ -120,12 +120,14 @@ class LLMEngine:
             if self.kv_offload is not None:
                 fresh = []
                 for seq in batch.seqs:
-                    if self.kv_offload.restore(seq.prompt_token_ids,
-                                               seq.block_table):
+                    covered = self.kv_offload.restore_prefix(
+                        seq.prompt_token_ids, seq.block_table)
+                    if covered >= seq.num_prompt_tokens:
                         # full-prompt KV hit: skip prefill; the next decode
                         # step feeds the last prompt token over restored KV
                         seq.sched_len = seq.num_prompt_tokens
                     else:
+                        seq._restored_prefix = covered  # 0 = fresh
                         fresh.append(seq)
                 batch.seqs = fresh
                 if not batch.seqs:

@@ -119,6 +119,61 @@ class KVOffloadManager:
                 return key
         return None
 
+    MIN_PREFIX = 16  # don't bother restoring less than one block
+
+    def longest_prefix(self, prompt: List[int]) -> Tuple[Optional[str], int]:
+        """Entry with the longest exact common prefix with `prompt`;
+        returns (key, covered_tokens)."""
+        best_key, best_n = None, 0
+        p = tuple(prompt)
+        for key, (toks, _) in self._store.items():
+            n = 0
+            for a, b in zip(p, toks):
+                if a != b:
+                    break
+                n += 1
+            if n > best_n:
+                best_key, best_n = key, n
+        return best_key, best_n
+
+    def restore_prefix(self, prompt: List[int],
+                       block_table: List[int]) -> int:
+        """Restore the longest matching prefix into the sequence's blocks.
+        Returns the covered token count (0 = miss / below threshold).
+        Partially-restored boundary blocks are safe: the suffix prefill
+        rewrites its own slots in stream order before attention reads."""
+        key, covered = self.longest_prefix(prompt)
+        if key is None or covered < self.MIN_PREFIX:
+            self.misses += 1
+            return 0
+        self._store.move_to_end(key)
+        self._restore_tokens(key, covered, block_table)
+        self.hits += 1
+        return covered
+
+    def _restore_tokens(self, key: str, n: int,
+                        block_table: List[int]) -> None:
+        bs = self.block_size
+        nb = (n + bs - 1) // bs
+        dev = self.kv_caches[0][0].device
+        blocks = torch.tensor(block_table[:nb], dtype=torch.long, device=dev)
+        _toks, host = self._store[key]
+        ctx = torch.cuda.stream(self._stream) if self._stream else _nullctx()
+        with ctx:
+            for (kc, vc), (kh, vh) in zip(self.kv_caches, host):
+                kvh, d = kc.shape[1], kc.shape[3]
+                pad = nb * bs
+                kg = torch.zeros(kvh, pad, d, dtype=kc.dtype, device="cpu")
+                vg = torch.zeros_like(kg)
+                kg[:, :n] = kh[:, :n]
+                vg[:, :n] = vh[:, :n]
+                kdev = kg.to(dev, non_blocking=self.is_gpu)
+                vdev = vg.to(dev, non_blocking=self.is_gpu)
+                kc[blocks] = kdev.reshape(kvh, nb, bs, d).permute(1, 0, 2, 3)
+                vc[blocks] = vdev.reshape(kvh, nb, bs, d).permute(1, 0, 2, 3)
+        if self._stream:
+            self._stream.synchronize()
+
     def restore(self, prompt: List[int], block_table: List[int]) -> bool:
         """Copy cached KV for `prompt` into the sequence's allocated blocks.
         Returns True on a full-prompt hit."""

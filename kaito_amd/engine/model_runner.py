@@ -100,22 +100,43 @@ class ModelRunner:
     # ------------------------------------------------------------- prefill
     @torch.no_grad()
     def execute_prefill(self, seqs: List[Sequence]) -> torch.Tensor:
-        """Returns hidden states of each sequence's LAST token: [B, H]."""
+        """Returns hidden states of each sequence's LAST token: [B, H].
+
+        Sequences with seq._restored_prefix > 0 (prefix-cache hit) compute
+        only the SUFFIX tokens: attention runs in context mode over the
+        paged cache."""
         ids, pos, slots, cu = [], [], [], [0]
+        any_restored = any(getattr(s, "_restored_prefix", 0) > 0
+                           for s in seqs)
         for seq in seqs:
-            toks = seq.prompt_token_ids
+            start = getattr(seq, "_restored_prefix", 0)
+            toks = seq.prompt_token_ids[start:]
             ids.extend(toks)
-            pos.extend(range(len(toks)))
-            slots.extend(self._slot(seq, p) for p in range(len(toks)))
+            pos.extend(range(start, start + len(toks)))
+            slots.extend(self._slot(seq, p)
+                         for p in range(start, start + len(toks)))
             cu.append(cu[-1] + len(toks))
         dev = self.device
         input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
         positions = torch.tensor(pos, dtype=torch.long, device=dev)
+        kv_lens = None
+        block_tables = None
+        if any_restored:
+            import numpy as np
+            mb = max(len(s.block_table) for s in seqs)
+            flat = np.zeros((len(seqs), mb), dtype=np.int32)
+            for i, s in enumerate(seqs):
+                flat[i, :len(s.block_table)] = s.block_table
+            block_tables = torch.from_numpy(flat).to(dev)
+            kv_lens = torch.tensor([s.num_prompt_tokens for s in seqs],
+                                   dtype=torch.int32, device=dev)
         meta = AttnMetadata(
             is_prefill=True,
             slot_mapping=torch.tensor(slots, dtype=torch.long, device=dev),
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
-            max_seqlen=max(s.num_prompt_tokens for s in seqs))
+            max_seqlen=max(s.num_prompt_tokens for s in seqs),
+            block_tables=block_tables,
+            kv_lens=kv_lens)
         if self.lora_manager is not None:
             tok_ids = []
             for seq in seqs:

@@ -31,11 +31,14 @@ class AttnMetadata:
     is_prefill: bool
     slot_mapping: torch.Tensor                  # [T] int64
     # prefill:
-    cu_seqlens: Optional[torch.Tensor] = None   # [B+1] int32
+    cu_seqlens: Optional[torch.Tensor] = None   # [B+1] int32 (q suffix lens)
     max_seqlen: int = 0
     # decode:
     block_tables: Optional[torch.Tensor] = None  # [T, max_blocks] int32
     seq_lens: Optional[torch.Tensor] = None      # [T] int32
+    # context-prefill (chunked prefill / prefix-cache continuation):
+    # q covers only suffix tokens; K/V read from the paged cache.
+    kv_lens: Optional[torch.Tensor] = None       # [B] int32 total ctx len
 
 
 def build_cos_sin_cache(cfg: ModelConfig, device, max_pos: Optional[int] = None
@@ -84,8 +87,15 @@ class LlamaAttention(nn.Module):
             ops.reshape_and_cache(kh, vh, kv_cache[0], kv_cache[1],
                                   meta.slot_mapping)
         if meta.is_prefill:
-            out = ops.prefill_attention(qh, kh, vh, meta.cu_seqlens, self.scale,
-                                        meta.max_seqlen)
+            if meta.kv_lens is not None:
+                # suffix-query attention over the paged cache (the suffix
+                # K/V was just written by reshape_and_cache above)
+                out = ops.context_attention(qh, kv_cache[0], kv_cache[1],
+                                            meta.cu_seqlens, meta.kv_lens,
+                                            meta.block_tables, self.scale)
+            else:
+                out = ops.prefill_attention(qh, kh, vh, meta.cu_seqlens,
+                                            self.scale, meta.max_seqlen)
         else:
             out = ops.paged_attention(qh, kv_cache[0], kv_cache[1],
                                       meta.block_tables, meta.seq_lens,

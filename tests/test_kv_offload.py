@@ -62,3 +62,34 @@ def test_lru_eviction_under_budget():
     eng.generate([p1], sp)
     assert eng.kv_offload.hits == 0  # p1 was evicted
     assert eng.kv_offload.used_bytes <= 1 << 15
+
+
+def test_partial_prefix_restore_matches_fresh():
+    """A prompt sharing a 32-token prefix with a cached sequence restores
+    the prefix and context-prefills only the suffix — outputs must match a
+    fresh engine exactly."""
+    base = list(range(100, 140))            # 40-token prompt A
+    sp = SamplingParams(max_tokens=5, ignore_eos=True)
+    eng = _engine(seed=5)
+    eng.generate([base], sp)                # A + gen now cached
+
+    new_prompt = base[:32] + list(range(300, 316))   # A[:32] + C
+    calls = {"ctx": 0}
+    import kaito_amd.ops as O
+    orig_ctx = O.context_attention
+
+    def counting(*a, **k):
+        calls["ctx"] += 1
+        return orig_ctx(*a, **k)
+
+    O.context_attention = counting
+    try:
+        out = eng.generate([new_prompt], sp)[0].output_token_ids
+    finally:
+        O.context_attention = orig_ctx
+    assert eng.kv_offload.hits >= 1
+    assert calls["ctx"] > 0, "suffix prefill should use context attention"
+
+    fresh = _engine(seed=5)
+    expect = fresh.generate([new_prompt], sp)[0].output_token_ids
+    assert out == expect
