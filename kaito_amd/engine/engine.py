@@ -41,7 +41,17 @@ class LLMEngine:
         self.runner.setup_tunable()
         self.runner.profile_and_allocate_kv()
         self.pool = BlockPool(self.runner.num_gpu_blocks, cfg.block_size)
-        self.scheduler = Scheduler(cfg, self.pool)
+        self.kv_offload = None
+        if cfg.kv_offload:
+            from .kv_offload import KVOffloadManager
+            self.kv_offload = KVOffloadManager(
+                self.runner.kv_caches, cfg.block_size,
+                cfg.kv_offload_bytes, cfg.device)
+        restore_cb = None
+        if self.kv_offload is not None:
+            restore_cb = lambda seq: self.kv_offload.restore_prefix(  # noqa: E731
+                seq.prompt_token_ids, seq.block_table)
+        self.scheduler = Scheduler(cfg, self.pool, restore_cb)
         self.sampler = Sampler(cfg.device, cfg.seed)
         self.eos_token_id: Optional[int] = None
         self._next_id = itertools.count()
@@ -52,12 +62,6 @@ class LLMEngine:
         self.num_prompt_tokens = 0
         # optional KV event bus (EPP KVCache-aware routing surface)
         self.kv_publisher = None
-        self.kv_offload = None
-        if cfg.kv_offload:
-            from .kv_offload import KVOffloadManager
-            self.kv_offload = KVOffloadManager(
-                self.runner.kv_caches, cfg.block_size,
-                cfg.kv_offload_bytes, cfg.device)
 
     def capture_graphs(self):
         self.runner.capture_decode_graphs()
@@ -117,31 +121,21 @@ class LLMEngine:
             # prefill steps are synchronous: drain the pipeline first so the
             # prefill batch sees fully-committed state.
             finished += self._resolve_pending()
-            if self.kv_offload is not None:
-                fresh = []
-                for seq in batch.seqs:
-                    covered = self.kv_offload.restore_prefix(
-                        seq.prompt_token_ids, seq.block_table)
-                    if covered >= seq.num_prompt_tokens:
-                        # full-prompt KV hit: skip prefill; the next decode
-                        # step feeds the last prompt token over restored KV
-                        seq.sched_len = seq.num_prompt_tokens
-                    else:
-                        seq._restored_prefix = covered  # 0 = fresh
-                        fresh.append(seq)
-                batch.seqs = fresh
-                if not batch.seqs:
-                    return finished
-            logits = self.runner.execute(batch)
-            tokens = self.sampler.sample(logits, batch.seqs)
-            for seq in batch.seqs:
+            hidden = self.runner.execute_prefill(batch.chunks)
+            self.scheduler.finish_prefill_chunks(batch)
+            samp = batch.sampling_seqs
+            if not samp:
+                return finished          # all chunks partial: no sampling
+            logits = self.runner.model.compute_logits(hidden)
+            tokens = self.sampler.sample(logits, samp)
+            for seq in samp:
                 seq.sched_len = seq.num_prompt_tokens + 1
                 self.num_prompt_tokens += seq.num_prompt_tokens
             if self.kv_publisher is not None:
-                blocks = [b for s2 in batch.seqs for b in s2.block_table]
+                blocks = [b for s2 in samp for b in s2.block_table]
                 self.kv_publisher.block_stored(blocks)
-            finished += self._commit(batch.seqs, tokens.tolist(),
-                                     [s.epoch for s in batch.seqs])
+            finished += self._commit(samp, tokens.tolist(),
+                                     [s.epoch for s in samp])
             return finished
 
         pend = self._pending

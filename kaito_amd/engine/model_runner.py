@@ -99,55 +99,55 @@ class ModelRunner:
 
     # ------------------------------------------------------------- prefill
     @torch.no_grad()
-    def execute_prefill(self, seqs: List[Sequence]) -> torch.Tensor:
-        """Returns hidden states of each sequence's LAST token: [B, H].
-
-        Sequences with seq._restored_prefix > 0 (prefix-cache hit) compute
-        only the SUFFIX tokens: attention runs in context mode over the
-        paged cache."""
+    def execute_prefill(self, chunks) -> torch.Tensor:
+        """Run a (possibly chunked) prefill batch. Each chunk covers prompt
+        tokens [start, start+length) of its sequence; chunks with start>0
+        or incomplete prompts use context attention over the paged cache.
+        Returns hidden states of the LAST token of each COMPLETING chunk,
+        in chunk order (aligned with batch.sampling_seqs)."""
         ids, pos, slots, cu = [], [], [], [0]
-        any_restored = any(getattr(s, "_restored_prefix", 0) > 0
-                           for s in seqs)
-        for seq in seqs:
-            start = getattr(seq, "_restored_prefix", 0)
-            toks = seq.prompt_token_ids[start:]
+        fresh = all(c.start == 0 and c.completes for c in chunks)
+        for c in chunks:
+            toks = c.seq.prompt_token_ids[c.start:c.start + c.length]
             ids.extend(toks)
-            pos.extend(range(start, start + len(toks)))
-            slots.extend(self._slot(seq, p)
-                         for p in range(start, start + len(toks)))
-            cu.append(cu[-1] + len(toks))
+            pos.extend(range(c.start, c.start + c.length))
+            slots.extend(self._slot(c.seq, p)
+                         for p in range(c.start, c.start + c.length))
+            cu.append(cu[-1] + c.length)
         dev = self.device
         input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
         positions = torch.tensor(pos, dtype=torch.long, device=dev)
         kv_lens = None
         block_tables = None
-        if any_restored:
+        if not fresh:
             import numpy as np
-            mb = max(len(s.block_table) for s in seqs)
-            flat = np.zeros((len(seqs), mb), dtype=np.int32)
-            for i, s in enumerate(seqs):
-                flat[i, :len(s.block_table)] = s.block_table
+            mb = max(len(c.seq.block_table) for c in chunks)
+            flat = np.zeros((len(chunks), mb), dtype=np.int32)
+            for i, c in enumerate(chunks):
+                flat[i, :len(c.seq.block_table)] = c.seq.block_table
             block_tables = torch.from_numpy(flat).to(dev)
-            kv_lens = torch.tensor([s.num_prompt_tokens for s in seqs],
+            kv_lens = torch.tensor([c.start + c.length for c in chunks],
                                    dtype=torch.int32, device=dev)
         meta = AttnMetadata(
             is_prefill=True,
             slot_mapping=torch.tensor(slots, dtype=torch.long, device=dev),
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
-            max_seqlen=max(s.num_prompt_tokens for s in seqs),
+            max_seqlen=max(c.length for c in chunks),
             block_tables=block_tables,
             kv_lens=kv_lens)
         if self.lora_manager is not None:
             tok_ids = []
-            for seq in seqs:
-                tok_ids.extend([seq.lora_id] * seq.num_prompt_tokens)
+            for c in chunks:
+                tok_ids.extend([c.seq.lora_id] * c.length)
             lora_mod.set_active(self.lora_manager, torch.tensor(
                 tok_ids, dtype=torch.int32, device=dev))
         hidden = self.model(input_ids, positions, self.kv_caches, meta)
         if self.lora_manager is not None:
             lora_mod.set_active(None, None)
-        last_idx = torch.tensor([c - 1 for c in cu[1:]], device=dev)
-        return hidden[last_idx]
+        last = [cu[i + 1] - 1 for i, c in enumerate(chunks) if c.completes]
+        if not last:
+            return hidden[:0]
+        return hidden[torch.tensor(last, device=dev)]
 
     # ------------------------------------------------------------- decode
     def _init_decode_buffers(self):
@@ -344,7 +344,7 @@ class ModelRunner:
                 pending_map=None) -> torch.Tensor:
         """Run the batch; returns logits [B, vocab] for the last tokens."""
         if batch.is_prefill:
-            hidden = self.execute_prefill(batch.seqs)
+            hidden = self.execute_prefill(batch.chunks)
         else:
             hidden = self.execute_decode(batch.seqs, sampled, pending_map)
         return self.model.compute_logits(hidden)

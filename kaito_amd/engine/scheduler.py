@@ -23,21 +23,47 @@ from .sequence import Sequence, SeqStatus
 
 
 @dataclass
+class PrefillChunk:
+    seq: Sequence
+    start: int          # first prompt token index in this chunk
+    length: int
+
+    @property
+    def completes(self) -> bool:
+        return self.start + self.length >= self.seq.num_prompt_tokens
+
+
+@dataclass
 class ScheduledBatch:
     is_prefill: bool
     seqs: List[Sequence] = field(default_factory=list)
+    # prefill batches: per-seq chunk ranges (chunked prefill); seqs holds
+    # the chunk owners in order
+    chunks: List[PrefillChunk] = field(default_factory=list)
 
     @property
     def num_seqs(self) -> int:
         return len(self.seqs)
 
+    @property
+    def sampling_seqs(self) -> List[Sequence]:
+        """Sequences that get a token sampled this step (decode: all;
+        prefill: only chunks completing their prompt)."""
+        if not self.is_prefill:
+            return self.seqs
+        return [c.seq for c in self.chunks if c.completes]
+
 
 class Scheduler:
-    def __init__(self, cfg: EngineConfig, pool: BlockPool):
+    def __init__(self, cfg: EngineConfig, pool: BlockPool,
+                 restore_cb=None):
         self.cfg = cfg
         self.pool = pool
         self.waiting: deque[Sequence] = deque()
+        self.prefilling: List[Sequence] = []   # admitted, prompt KV partial
         self.running: List[Sequence] = []
+        # called at admission: restore_cb(seq) -> covered prefix tokens
+        self.restore_cb = restore_cb
 
     # ---- queue state (serves the rate limiter / metrics) ----
     @property
@@ -46,10 +72,10 @@ class Scheduler:
 
     @property
     def num_running(self) -> int:
-        return len(self.running)
+        return len(self.running) + len(self.prefilling)
 
     def has_work(self) -> bool:
-        return bool(self.waiting or self.running)
+        return bool(self.waiting or self.prefilling or self.running)
 
     def add(self, seq: Sequence) -> None:
         seq.status = SeqStatus.WAITING
@@ -63,29 +89,62 @@ class Scheduler:
         return self._schedule_decode()
 
     def _schedule_prefill(self) -> Optional[ScheduledBatch]:
-        if not self.waiting:
+        """Chunked prefill: each step processes up to max_num_batched_tokens
+        of prompt tokens; long prompts span multiple steps (context
+        attention handles suffix chunks). New sequences are admitted after
+        in-flight prefills continue."""
+        if not self.waiting and not self.prefilling:
             return None
         budget = self.cfg.max_num_batched_tokens
-        room = self.cfg.max_num_seqs - len(self.running)
-        picked: List[Sequence] = []
-        while self.waiting and room > 0:
-            seq = self.waiting[0]
-            n_tok = seq.num_prompt_tokens
-            if picked and n_tok > budget:
+        chunks: List[PrefillChunk] = []
+
+        # 1. continue partially-prefilled sequences
+        for seq in list(self.prefilling):
+            if budget <= 0:
                 break
+            remaining = seq.num_prompt_tokens - seq.prefilled_len
+            n = min(remaining, budget)
+            chunks.append(PrefillChunk(seq, seq.prefilled_len, n))
+            budget -= n
+
+        # 2. admit new sequences (blocks for the whole prompt upfront)
+        room = self.cfg.max_num_seqs - len(self.running) - len(self.prefilling)
+        while self.waiting and room > 0 and budget > 0:
+            seq = self.waiting[0]
             need = self.pool.blocks_needed(seq.num_tokens + 1)
             if not self.pool.can_allocate(need):
                 break
             self.waiting.popleft()
             seq.block_table = self.pool.allocate(need)
             seq.status = SeqStatus.RUNNING
-            picked.append(seq)
-            budget -= n_tok
+            if self.restore_cb is not None:
+                seq.prefilled_len = self.restore_cb(seq)
+            if seq.prefilled_len >= seq.num_prompt_tokens:
+                # full prefix-cache hit: straight to decode (the next decode
+                # step feeds the last prompt token over restored KV)
+                seq.sched_len = seq.num_prompt_tokens
+                self.running.append(seq)
+                room -= 1
+                continue
+            n = min(seq.num_prompt_tokens - seq.prefilled_len, budget)
+            chunks.append(PrefillChunk(seq, seq.prefilled_len, n))
+            self.prefilling.append(seq)
+            budget -= n
             room -= 1
-        if not picked:
+
+        if not chunks:
             return None
-        self.running.extend(picked)
-        return ScheduledBatch(is_prefill=True, seqs=picked)
+        # completion bookkeeping happens in the engine after execution
+        return ScheduledBatch(is_prefill=True,
+                              seqs=[c.seq for c in chunks], chunks=chunks)
+
+    def finish_prefill_chunks(self, batch: ScheduledBatch) -> None:
+        """Advance prefilled_len; move completed sequences to running."""
+        for c in batch.chunks:
+            c.seq.prefilled_len = c.start + c.length
+            if c.completes and c.seq in self.prefilling:
+                self.prefilling.remove(c.seq)
+                self.running.append(c.seq)
 
     def _schedule_decode(self) -> Optional[ScheduledBatch]:
         if not self.running:
@@ -110,6 +169,7 @@ class Scheduler:
                     victim.block_table = []
                     victim.output_token_ids = []
                     victim.sched_len = 0
+                    victim.prefilled_len = 0
                     victim.epoch += 1
                     victim.status = SeqStatus.WAITING
                     self.waiting.appendleft(victim)
@@ -124,3 +184,5 @@ class Scheduler:
         seq.block_table = []
         if seq in self.running:
             self.running.remove(seq)
+        if seq in self.prefilling:
+            self.prefilling.remove(seq)

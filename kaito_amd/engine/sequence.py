@@ -43,6 +43,9 @@ class Sequence:
     # bumped on preemption so stale in-flight results are dropped at resolve
     epoch: int = 0
     lora_id: int = -1            # adapter slot (-1 = base model)
+    # prompt tokens whose KV is in the cache (chunked prefill / prefix
+    # restore); prompt fully prefilled when == num_prompt_tokens
+    prefilled_len: int = 0
 
     @property
     def num_prompt_tokens(self) -> int:

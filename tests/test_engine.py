@@ -57,6 +57,8 @@ def test_scheduler_prefill_then_decode():
     s.add(Sequence(1, list(range(5))))
     b = s.schedule()
     assert b.is_prefill and b.num_seqs == 2
+    assert all(c.completes for c in b.chunks)
+    s.finish_prefill_chunks(b)
     assert s.num_waiting == 0 and s.num_running == 2
     b2 = s.schedule()
     assert not b2.is_prefill and b2.num_seqs == 2
@@ -70,6 +72,7 @@ def test_scheduler_preempts_on_pool_exhaustion():
     s.add(Sequence(1, list(range(16))))
     b = s.schedule()
     assert b.is_prefill and b.num_seqs == 2   # 2+2 blocks
+    s.finish_prefill_chunks(b)
     # grow both beyond pool: each at 32 tokens now needs 3rd block
     for seq in b.seqs:
         seq.output_token_ids = list(range(16))
@@ -79,14 +82,24 @@ def test_scheduler_preempts_on_pool_exhaustion():
     assert s.num_waiting == 1
 
 
-def test_scheduler_respects_token_budget():
+def test_scheduler_chunked_prefill_respects_token_budget():
     cfg = _cfg(max_num_batched_tokens=32)
     pool = BlockPool(64, cfg.block_size)
     s = Scheduler(cfg, pool)
     s.add(Sequence(0, list(range(30))))
     s.add(Sequence(1, list(range(30))))
     b = s.schedule()
-    assert b.num_seqs == 1   # second exceeds budget
+    # chunked: 30 tokens of seq0 + first 2 of seq1 fill the 32 budget
+    assert [c.length for c in b.chunks] == [30, 2]
+    assert b.chunks[0].completes and not b.chunks[1].completes
+    assert [s2.seq_id for s2 in b.sampling_seqs] == [0]
+    s.finish_prefill_chunks(b)
+    b2 = s.schedule()
+    assert b2.is_prefill  # prefill-priority: seq1's remaining 28 tokens
+    assert [c.length for c in b2.chunks] == [28]
+    assert b2.chunks[0].start == 2 and b2.chunks[0].completes
+    s.finish_prefill_chunks(b2)
+    assert s.num_running == 2 and not s.prefilling
 
 
 # --------------------------------------------------------------- e2e decode
@@ -145,6 +158,34 @@ def test_engine_continuous_batching_join():
     expect_b = _naive_generate(eng.runner.model, cfg, [35, 89, 79], 5)
     assert eng.seqs[a].output_token_ids == expect_a
     assert eng.seqs[b].output_token_ids == expect_b
+
+
+def test_engine_chunked_prefill_matches_oracle():
+    """A prompt longer than max_num_batched_tokens prefills over several
+    chunked steps (context attention) — greedy output must match the
+    full-recompute oracle."""
+    cfg = _cfg(max_num_batched_tokens=24)
+    eng = LLMEngine(cfg)
+    prompt = list(range(5, 85))   # 80 tokens → 4 chunks of 24/24/24/8
+    outs = eng.generate([prompt], SamplingParams(max_tokens=6, ignore_eos=True))
+    expect = _naive_generate(eng.runner.model, cfg, prompt, 6)
+    assert outs[0].output_token_ids == expect
+
+
+def test_engine_chunked_prefill_mixed_batch():
+    cfg = _cfg(max_num_batched_tokens=16)
+    eng = LLMEngine(cfg)
+    pa = list(range(10, 50))      # 40 tokens, chunked
+    pb = [3, 5, 7]
+    sp = SamplingParams(max_tokens=5, ignore_eos=True)
+    ia = eng.add_request(pa, sp)
+    ib = eng.add_request(pb, sp)
+    while eng.has_unfinished():
+        eng.step()
+    assert eng.seqs[ia].output_token_ids == _naive_generate(
+        eng.runner.model, cfg, pa, 5)
+    assert eng.seqs[ib].output_token_ids == _naive_generate(
+        eng.runner.model, cfg, pb, 5)
 
 
 def test_sampling_params_stop():
