@@ -199,3 +199,25 @@ def test_sampling_params_stop():
                           SamplingParams(max_tokens=64, stop_token_ids=(tok,)))
     assert outs2[0].output_token_ids == [tok]
     assert outs2[0].finish_reason == "stop"
+
+
+@pytest.mark.parametrize("preset_kw", [
+    dict(partial_rotary_factor=0.75, tie_word_embeddings=True),  # phi-4-mini
+    dict(num_kv_heads=4),                                        # denser GQA
+])
+def test_engine_model_variants_match_oracle(preset_kw):
+    """Partial-rotary + tied-embedding (Phi-4-mini class) and other
+    architecture variants run the same engine path correctly."""
+    from kaito_amd.engine.config import ModelConfig
+    kw = dict(name="variant-test", hidden_size=256, num_layers=2,
+              num_heads=4, num_kv_heads=2, intermediate_size=512,
+              vocab_size=512, head_dim=64, rope_theta=10000.0,
+              max_position=512)
+    kw.update(preset_kw)
+    mc = ModelConfig(**kw)
+    cfg = _cfg(model=mc)
+    eng = LLMEngine(cfg)
+    prompt = [7, 9, 11, 13, 15, 17]
+    outs = eng.generate([prompt], SamplingParams(max_tokens=6, ignore_eos=True))
+    expect = _naive_generate(eng.runner.model, cfg, prompt, 6)
+    assert outs[0].output_token_ids == expect
