@@ -53,6 +53,8 @@ class LLMEngine:
                 seq.prompt_token_ids, seq.block_table)
         self.scheduler = Scheduler(cfg, self.pool, restore_cb)
         self.sampler = Sampler(cfg.device, cfg.seed)
+        from ..parallel.state import get_state
+        self._pp = get_state().pp_size > 1
         self.eos_token_id: Optional[int] = None
         self._next_id = itertools.count()
         self.seqs: Dict[int, Sequence] = {}
@@ -126,8 +128,7 @@ class LLMEngine:
             samp = batch.sampling_seqs
             if not samp:
                 return finished          # all chunks partial: no sampling
-            logits = self.runner.model.compute_logits(hidden)
-            tokens = self.sampler.sample(logits, samp)
+            tokens = self._sample_maybe_pp(hidden, samp)
             for seq in samp:
                 seq.sched_len = seq.num_prompt_tokens + 1
                 self.num_prompt_tokens += seq.num_prompt_tokens
@@ -136,6 +137,21 @@ class LLMEngine:
                 self.kv_publisher.block_stored(blocks)
             finished += self._commit(samp, tokens.tolist(),
                                      [s.epoch for s in samp])
+            return finished
+
+        if self._pp:
+            # pipeline-parallel decode runs lockstep-synchronous (no
+            # speculative pipelining): every stage executes; tokens are
+            # sampled on the last stage and broadcast.
+            sampled = getattr(self, "_pp_prev_tokens", None)
+            logits = self.runner.execute(batch, sampled)
+            tokens = self._sample_maybe_pp(
+                logits, batch.seqs, precomputed_logits=True)
+            self._pp_prev_tokens = tokens
+            for seq in batch.seqs:
+                seq.sched_len = seq.sched_tokens + 1
+            finished += self._commit(batch.seqs, tokens.tolist(),
+                                     [s.epoch for s in batch.seqs])
             return finished
 
         pend = self._pending
@@ -164,6 +180,25 @@ class LLMEngine:
         finished += self._resolve_pending()
         self._pending = new_pend
         return finished
+
+    def _sample_maybe_pp(self, hidden_or_logits, seqs,
+                         precomputed_logits: bool = False) -> torch.Tensor:
+        """Sample on the last PP stage and broadcast token ids to every
+        stage (lockstep schedulers). Single-stage: plain sampling."""
+        from ..parallel import state as ps
+        st = ps.get_state()
+        if st.pp_size == 1:
+            logits = hidden_or_logits if precomputed_logits else \
+                self.runner.model.compute_logits(hidden_or_logits)
+            return self.sampler.sample(logits, seqs)
+        dev = self.runner.device
+        if st.is_last_stage:
+            logits = hidden_or_logits if precomputed_logits else \
+                self.runner.model.compute_logits(hidden_or_logits)
+            toks = self.sampler.sample(logits, seqs).to(dev)
+        else:
+            toks = torch.zeros(len(seqs), dtype=torch.long, device=dev)
+        return ps.pp_broadcast_from_last(toks)
 
     def _pinned(self, n: int) -> torch.Tensor:
         """Pinned host staging buffers, DOUBLE-buffered: step N's async D2H

@@ -27,6 +27,7 @@ from .scheduler import ScheduledBatch
 from .sequence import Sequence
 from . import lora as lora_mod
 from ..models.llama import AttnMetadata, LlamaForCausalLM
+from ..parallel import state as ps
 
 logger = logging.getLogger(__name__)
 
@@ -81,7 +82,7 @@ class ModelRunner:
         n_blocks = min(n_blocks, cap)
         self.num_gpu_blocks = n_blocks
         kvs = []
-        for _ in range(m.num_layers):
+        for _ in range(self.model.num_local_layers):
             k = torch.zeros(n_blocks, kvh, cfg.block_size, m.head_dim,
                             dtype=m.dtype, device=self.device)
             v = torch.zeros_like(k)
@@ -141,9 +142,11 @@ class ModelRunner:
                 tok_ids.extend([c.seq.lora_id] * c.length)
             lora_mod.set_active(self.lora_manager, torch.tensor(
                 tok_ids, dtype=torch.int32, device=dev))
-        hidden = self.model(input_ids, positions, self.kv_caches, meta)
+        hidden = self._model_forward(input_ids, positions, meta)
         if self.lora_manager is not None:
             lora_mod.set_active(None, None)
+        if hidden is None:
+            return None  # intermediate PP stage
         last = [cu[i + 1] - 1 for i, c in enumerate(chunks) if c.completes]
         if not last:
             return hidden[:0]
@@ -165,6 +168,26 @@ class ModelRunner:
         }
         self._max_blocks = mb
 
+    def _model_forward(self, input_ids, positions, meta):
+        """PP-aware forward: non-first stages receive the previous stage's
+        hidden; non-last stages send theirs on. Returns the final hidden on
+        the LAST stage, None elsewhere."""
+        st = ps.get_state()
+        if st.pp_size == 1:
+            return self.model(input_ids, positions, self.kv_caches, meta)
+        T = input_ids.shape[0]
+        H = self.cfg.model.hidden_size
+        if st.is_first_stage:
+            out = self.model(input_ids, positions, self.kv_caches, meta)
+        else:
+            hin = ps.pp_recv_prev((T, H), self.cfg.model.dtype, self.device)
+            out = self.model(input_ids, positions, self.kv_caches, meta,
+                             hidden_in=hin)
+        if not st.is_last_stage:
+            ps.pp_send_next(out)
+            return None
+        return out
+
     def _decode_forward(self, bs: int) -> torch.Tensor:
         b = self._buf
         if self.lora_manager is not None:
@@ -174,8 +197,8 @@ class ModelRunner:
             slot_mapping=b["slot_mapping"][:bs],
             block_tables=b["block_tables"][:bs],
             seq_lens=b["seq_lens"][:bs])
-        return self.model(b["input_ids"][:bs], b["positions"][:bs],
-                          self.kv_caches, meta)
+        return self._model_forward(b["input_ids"][:bs], b["positions"][:bs],
+                                   meta)
 
     def setup_tunable(self):
         """Enable TunableOp lookups from the in-tree MI355X tuning cache
@@ -347,4 +370,6 @@ class ModelRunner:
             hidden = self.execute_prefill(batch.chunks)
         else:
             hidden = self.execute_decode(batch.seqs, sampled, pending_map)
+        if hidden is None:
+            return None  # intermediate PP stage
         return self.model.compute_logits(hidden)

@@ -144,37 +144,67 @@ class LlamaDecoderLayer(nn.Module):
 
 
 class LlamaForCausalLM(nn.Module):
+    """Pipeline-aware: with pp_size>1 each stage builds only its layer
+    slice; stage 0 owns the embedding, the last stage owns norm + lm_head.
+    forward() takes token ids on stage 0 and hidden states elsewhere."""
+
     def __init__(self, cfg: ModelConfig):
         super().__init__()
         self.cfg = cfg
+        st = get_state()
+        self.pp_rank, self.pp_size = st.pp_rank, st.pp_size
+        per = (cfg.num_layers + self.pp_size - 1) // self.pp_size
+        self.layer_start = self.pp_rank * per
+        self.layer_end = min(cfg.num_layers, self.layer_start + per)
+        self.is_first = st.is_first_stage
+        self.is_last = st.is_last_stage
         self.embed_tokens = VocabParallelEmbedding(
-            cfg.vocab_size, cfg.hidden_size, dtype=cfg.dtype)
+            cfg.vocab_size, cfg.hidden_size, dtype=cfg.dtype) \
+            if (self.is_first or cfg.tie_word_embeddings) else None
         self.layers = nn.ModuleList(
-            [LlamaDecoderLayer(cfg) for _ in range(cfg.num_layers)])
+            [LlamaDecoderLayer(cfg)
+             for _ in range(self.layer_end - self.layer_start)])
         self.norm = nn.Parameter(
-            torch.empty(cfg.hidden_size, dtype=cfg.dtype), requires_grad=False)
-        if cfg.tie_word_embeddings:
-            self.lm_head = None
-        else:
+            torch.empty(cfg.hidden_size, dtype=cfg.dtype),
+            requires_grad=False) if self.is_last else None
+        self.lm_head = None
+        if self.is_last and not cfg.tie_word_embeddings:
             self.lm_head = ColumnParallelLinear(
                 cfg.hidden_size, cfg.vocab_size, dtype=cfg.dtype)
         self.register_buffer("cos_sin_cache", torch.empty(0), persistent=False)
+
+    @property
+    def num_local_layers(self) -> int:
+        return len(self.layers)
 
     def init_rope(self, device, max_pos: Optional[int] = None):
         self.cos_sin_cache = build_cos_sin_cache(self.cfg, device, max_pos)
 
     def forward(self, input_ids: torch.Tensor, positions: torch.Tensor,
                 kv_caches: Optional[List[Tuple[torch.Tensor, torch.Tensor]]],
-                meta: AttnMetadata) -> torch.Tensor:
-        hidden = self.embed_tokens(input_ids)
-        residual = None
+                meta: AttnMetadata,
+                hidden_in: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Stage 0: embeds input_ids; later stages take hidden_in. Returns
+        the stage output hidden (+ residual folded in): intermediate stages
+        return the value to SEND; the last stage returns final normed
+        hidden."""
+        if self.is_first:
+            hidden = self.embed_tokens(input_ids)
+            residual = None
+        else:
+            assert hidden_in is not None, "non-first PP stage needs hidden_in"
+            hidden = hidden_in
+            residual = None
         for i, layer in enumerate(self.layers):
             kv = kv_caches[i] if kv_caches is not None else None
             hidden, residual = layer(hidden, residual, positions, kv, meta,
                                      self.cos_sin_cache)
-        hidden, _ = ops.fused_add_rms_norm(hidden, residual, self.norm,
-                                           self.cfg.rms_eps)
-        return hidden
+        if self.is_last:
+            hidden, _ = ops.fused_add_rms_norm(hidden, residual, self.norm,
+                                               self.cfg.rms_eps)
+            return hidden
+        # fold the residual stream so one tensor crosses the stage boundary
+        return (hidden.float() + residual.float()).to(hidden.dtype)
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         from ..parallel.state import tp_all_gather

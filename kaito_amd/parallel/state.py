@@ -28,6 +28,26 @@ class ParallelState:
     tp_group: Optional[object] = None    # ProcessGroup
     dp_size: int = 1
     dp_rank: int = 0
+    # pipeline parallel (tier 3): ranks laid out TP-minor —
+    # rank = pp_rank * tp_size + tp_rank
+    pp_size: int = 1
+    pp_rank: int = 0
+
+    @property
+    def is_first_stage(self) -> bool:
+        return self.pp_rank == 0
+
+    @property
+    def is_last_stage(self) -> bool:
+        return self.pp_rank == self.pp_size - 1
+
+    @property
+    def prev_stage_rank(self) -> int:
+        return self.rank - self.tp_size
+
+    @property
+    def next_stage_rank(self) -> int:
+        return self.rank + self.tp_size
 
 
 _STATE = ParallelState()
@@ -41,13 +61,12 @@ def is_initialized() -> bool:
     return dist.is_available() and dist.is_initialized()
 
 
-def init_parallel(tp_size: int = 1, backend: Optional[str] = None,
+def init_parallel(tp_size: int = 1, pp_size: int = 1,
+                  backend: Optional[str] = None,
                   timeout_s: int = 600) -> ParallelState:
-    """Initialise torch.distributed (if WORLD_SIZE>1) and carve TP groups.
-
-    Ranks are laid out TP-major: ranks [i*tp, (i+1)*tp) form TP group i;
-    the remaining dimension is data parallel (one engine replica per group).
-    """
+    """Initialise torch.distributed (if WORLD_SIZE>1) and carve TP/PP
+    groups. Ranks are TP-minor: rank = (dp*pp_size + pp)*tp_size + tp; TP
+    groups are contiguous rank ranges; pipeline stages are tp_size apart."""
     global _STATE
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -61,7 +80,8 @@ def init_parallel(tp_size: int = 1, backend: Optional[str] = None,
                                 timeout=datetime.timedelta(seconds=timeout_s))
 
     tp_size = max(1, min(tp_size, world))
-    assert world % tp_size == 0, f"world {world} not divisible by tp {tp_size}"
+    pp_size = max(1, min(pp_size, world // tp_size))
+    assert world % (tp_size * pp_size) == 0,         f"world {world} not divisible by tp*pp {tp_size * pp_size}"
     tp_group = None
     tp_rank = 0
     if world > 1:
@@ -72,11 +92,32 @@ def init_parallel(tp_size: int = 1, backend: Optional[str] = None,
             if rank in ranks:
                 tp_group = grp
                 tp_rank = rank - g * tp_size
+    pp_rank = (rank // tp_size) % pp_size
     _STATE = ParallelState(
         world_size=world, rank=rank, local_rank=local_rank,
         tp_size=tp_size, tp_rank=tp_rank, tp_group=tp_group,
-        dp_size=world // tp_size, dp_rank=rank // tp_size)
+        dp_size=world // (tp_size * pp_size),
+        dp_rank=rank // (tp_size * pp_size),
+        pp_size=pp_size, pp_rank=pp_rank)
     return _STATE
+
+
+def pp_send_next(t: torch.Tensor) -> None:
+    dist.send(t.contiguous(), _STATE.next_stage_rank)
+
+
+def pp_recv_prev(shape, dtype, device) -> torch.Tensor:
+    t = torch.empty(*shape, dtype=dtype, device=device)
+    dist.recv(t, _STATE.prev_stage_rank)
+    return t
+
+
+def pp_broadcast_from_last(t: torch.Tensor) -> torch.Tensor:
+    """Broadcast a tensor from the LAST pipeline stage (tp rank 0 of that
+    stage) to all ranks — the sampled token ids each step."""
+    src = (_STATE.pp_size - 1) * _STATE.tp_size
+    dist.broadcast(t, src)
+    return t
 
 
 def tp_all_reduce(t: torch.Tensor) -> torch.Tensor:

@@ -143,3 +143,70 @@ def test_vocab_parallel_embedding_single():
     ids = torch.tensor([0, 5, 63])
     out = emb(ids)
     assert torch.allclose(out, emb.weight[ids])
+
+
+# -------------------------------------------------------- pipeline parallel
+def _body_pp_engine(rank, world):
+    """2-stage pipeline engine must reproduce the single-rank greedy output
+    exactly (weights sharded from one reference model)."""
+    import torch
+    from kaito_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from kaito_amd.models.llama import LlamaForCausalLM
+    from kaito_amd.parallel import state as ps
+
+    # re-init with pp=2 (worker already init'd dist with tp_size=world)
+    st = ps.get_state()
+    st.tp_size, st.tp_rank, st.tp_group = 1, 0, None
+    st.pp_size, st.pp_rank = world, rank
+
+    cfg_model = get_model_config("tiny-llama-test")
+    # reference monolithic weights (same on both ranks via same seed)
+    saved = st.pp_size, st.pp_rank
+    st.pp_size, st.pp_rank = 1, 0
+    ref = LlamaForCausalLM(cfg_model).random_init(11)
+    ref.init_rope("cpu")
+    st.pp_size, st.pp_rank = saved
+
+    cfg = EngineConfig(model=cfg_model, device="cpu", max_num_seqs=4,
+                       num_gpu_blocks=64, enforce_eager=True,
+                       max_model_len=96)
+    eng = LLMEngine(cfg)
+    # copy the stage's layer slice + shared tensors from the reference
+    model = eng.runner.model
+    rp = dict(ref.named_parameters())
+    with torch.no_grad():
+        for name, p in model.named_parameters():
+            if name.startswith("layers."):
+                idx = int(name.split(".")[1])
+                src = rp["layers.%d.%s" % (idx + model.layer_start,
+                                           name.split(".", 2)[2])]
+            else:
+                src = rp[name]
+            p.copy_(src)
+
+    prompt = [3, 14, 15, 92, 65, 35, 89]
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+    outs = eng.generate([prompt], sp)
+    got = outs[0].output_token_ids
+
+    # single-rank oracle using the reference model weights
+    if rank == 0:
+        from kaito_amd.models.llama import AttnMetadata
+        toks = list(prompt)
+        for _ in range(6):
+            T = len(toks)
+            meta = AttnMetadata(
+                is_prefill=True,
+                slot_mapping=torch.full((T,), -1, dtype=torch.long),
+                cu_seqlens=torch.tensor([0, T], dtype=torch.int32),
+                max_seqlen=T)
+            st.pp_size, st.pp_rank = 1, 0
+            h = ref(torch.tensor(toks), torch.arange(T), None, meta)
+            nxt = int(ref.compute_logits(h[-1:]).argmax(-1))
+            st.pp_size, st.pp_rank = saved
+            toks.append(nxt)
+        assert got == toks[len(prompt):], (got, toks[len(prompt):])
+
+
+def test_pp_engine_world2():
+    _spawn("_body_pp_engine", port=29617)
