@@ -53,8 +53,9 @@ def _ws_obj(iset: InferenceSet, index: int, revision: str) -> Dict:
 
 
 class InferenceSetReconciler:
-    def __init__(self, client: KubeClient):
+    def __init__(self, client: KubeClient, gateway_api: bool = False):
         self.client = client
+        self.gateway_api = gateway_api  # gatewayAPIInferenceExtension gate
 
     def _revision(self, iset: InferenceSet) -> str:
         tpl = iset.spec.workspaceTemplate
@@ -121,6 +122,14 @@ class InferenceSetReconciler:
                     LABEL_UPGRADE_TO_VERSION not in labels:
                 labels[LABEL_UPGRADE_TO_VERSION] = revision
                 self.client.update(obj)
+
+        if self.gateway_api:
+            from ..manifests import (generate_inference_pool_helm_release,
+                                     generate_inference_pool_oci_repository)
+            self.client.apply(generate_inference_pool_oci_repository(
+                iset.name, iset.namespace))
+            self.client.apply(generate_inference_pool_helm_release(
+                iset.name, iset.namespace))
 
         # status
         children = self._list_children(iset)

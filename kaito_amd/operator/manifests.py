@@ -213,3 +213,58 @@ def generate_tuning_job(ws: Workspace, model: ModelConfig, gpu: GPUConfig,
             }},
         },
     }
+
+
+def generate_inference_pool_oci_repository(iset_name: str, namespace: str,
+                                           chart_url: str =
+                                           "oci://ghcr.io/llm-d/charts"
+                                           ) -> Dict[str, Any]:
+    """Reference parity: GenerateInferencePoolOCIRepository
+    (pkg/workspace/manifests/manifests.go:393) — Flux source for the
+    llm-d router chart."""
+    return {
+        "apiVersion": "source.toolkit.fluxcd.io/v1",
+        "kind": "OCIRepository",
+        "metadata": {"name": f"{iset_name}-router", "namespace": namespace},
+        "spec": {
+            "interval": "10m",
+            "url": chart_url,
+            "ref": {"tag": "latest"},
+        },
+    }
+
+
+def generate_inference_pool_helm_release(iset_name: str, namespace: str,
+                                         epp_image: str =
+                                         "ghcr.io/kaito-amd/epp:latest"
+                                         ) -> Dict[str, Any]:
+    """Reference parity: GenerateInferencePoolHelmRelease (manifests.go:421)
+    — InferencePool + EPP with KVCache-aware routing; model servers matched
+    by the InferenceSet child label + pod-index=0 (EPP pin, :431)."""
+    from .api_types import LABEL_INFERENCESET_CREATED_BY
+    return {
+        "apiVersion": "helm.toolkit.fluxcd.io/v2",
+        "kind": "HelmRelease",
+        "metadata": {"name": f"{iset_name}-router", "namespace": namespace},
+        "spec": {
+            "interval": "10m",
+            "chartRef": {"kind": "OCIRepository",
+                         "name": f"{iset_name}-router"},
+            "values": {
+                "inferencePool": {
+                    "targetPort": INFERENCE_PORT,
+                    "modelServers": {"matchLabels": {
+                        LABEL_INFERENCESET_CREATED_BY: iset_name,
+                        "apps.kubernetes.io/pod-index": "0",
+                    }},
+                },
+                "epp": {
+                    "image": epp_image,
+                    "args": ["--kv-cache-events-port", "5557"],
+                    "plugins": ["load-aware-scorer",
+                                "kv-cache-utilization-scorer",
+                                "prefix-cache-scorer"],
+                },
+            },
+        },
+    }

@@ -412,3 +412,33 @@ def test_operator_loop_end_to_end(client):
     # typed roundtrip
     ws = workspace_from_obj(stored)
     assert ws.inference.preset.name == "llama-3-8b"
+
+
+def test_operator_metrics_poller():
+    from kaito_amd.operator.metrics import monitor_workspaces, render
+    monitor_workspaces([
+        {"status": {"state": "Running"},
+         "spec": {"inference": {"preset": {"name": "llama-3-8b"}}}},
+        {"status": {"state": "Running"}, "spec": {}},
+        {"status": {}, "spec": {}},
+    ])
+    body = render().decode()
+    assert 'kaito_workspace_count{phase="Running"} 2.0' in body
+    assert 'kaito_workspace_count{phase="Pending"} 1.0' in body
+    assert 'kaito_workspace_preset_count{preset="llama-3-8b"} 1.0' in body
+
+
+def test_inferenceset_gateway_manifests(client):
+    from kaito_amd.operator.controllers.inferenceset import \
+        InferenceSetReconciler
+    r = InferenceSetReconciler(client, gateway_api=True)
+    iset = _iset(1)
+    r.reconcile(iset)
+    oci = client.get("OCIRepository", "default", "is1-router")
+    assert oci["spec"]["url"].startswith("oci://")
+    hr = client.get("HelmRelease", "default", "is1-router")
+    values = hr["spec"]["values"]
+    ml = values["inferencePool"]["modelServers"]["matchLabels"]
+    assert ml["apps.kubernetes.io/pod-index"] == "0"
+    assert ml["inferenceset.kaito.sh/created-by"] == "is1"
+    assert "kv-cache-utilization-scorer" in values["epp"]["plugins"]
