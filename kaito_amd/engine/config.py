@@ -32,6 +32,10 @@ class ModelConfig:
     partial_rotary_factor: float = 1.0
     # attention bias (qwen1/2 style)
     attention_bias: bool = False
+    # mixture-of-experts (mixtral / qwen-moe / gpt-oss class); 0 = dense
+    num_experts: int = 0
+    num_experts_per_tok: int = 2
+    moe_intermediate_size: int = 0       # per-expert FFN width (0 → dense I)
 
     @property
     def rotary_dim(self) -> int:
@@ -47,7 +51,11 @@ class ModelConfig:
         h, i, v = self.hidden_size, self.intermediate_size, self.vocab_size
         qkv = h * (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
         o = self.num_heads * self.head_dim * h
-        mlp = 3 * h * i
+        if self.num_experts > 0:
+            ie = self.moe_intermediate_size or i
+            mlp = self.num_experts * 3 * h * ie + h * self.num_experts
+        else:
+            mlp = 3 * h * i
         per_layer = (qkv + o) // tp_size + mlp // tp_size + 2 * h
         embed = v * h * (1 if self.tie_word_embeddings else 2)
         return 2 * (self.num_layers * per_layer + embed + h)

@@ -123,7 +123,11 @@ class LlamaDecoderLayer(nn.Module):
         super().__init__()
         self.cfg = cfg
         self.self_attn = LlamaAttention(cfg)
-        self.mlp = LlamaMLP(cfg)
+        if cfg.num_experts > 0:
+            from .moe import MoEMLP
+            self.mlp = MoEMLP(cfg)
+        else:
+            self.mlp = LlamaMLP(cfg)
         self.input_layernorm = nn.Parameter(
             torch.empty(cfg.hidden_size, dtype=cfg.dtype), requires_grad=False)
         self.post_attention_layernorm = nn.Parameter(

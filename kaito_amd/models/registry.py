@@ -73,8 +73,21 @@ register(ModelConfig(
     num_kv_heads=8, intermediate_size=29568, vocab_size=152064, head_dim=128,
     rope_theta=1000000.0, max_position=32768, attention_bias=True))
 
+# ---- MoE family (mixtral-class; fused-MoE presets in the reference
+# catalog, e.g. supported_models.yaml mixtral/gpt-oss rows) ------------------
+register(ModelConfig(
+    name="mixtral-8x7b", hidden_size=4096, num_layers=32, num_heads=32,
+    num_kv_heads=8, intermediate_size=14336, vocab_size=32000, head_dim=128,
+    rope_theta=1000000.0, max_position=32768, num_experts=8,
+    num_experts_per_tok=2, moe_intermediate_size=14336))
+
 # ---- tiny configs for tests ------------------------------------------------
 register(ModelConfig(
     name="tiny-llama-test", hidden_size=256, num_layers=2, num_heads=4,
     num_kv_heads=2, intermediate_size=512, vocab_size=512, head_dim=64,
     rope_theta=10000.0, max_position=512))
+register(ModelConfig(
+    name="tiny-moe-test", hidden_size=256, num_layers=2, num_heads=4,
+    num_kv_heads=2, intermediate_size=512, vocab_size=512, head_dim=64,
+    rope_theta=10000.0, max_position=512, num_experts=4,
+    num_experts_per_tok=2, moe_intermediate_size=256))
