@@ -91,11 +91,16 @@ def main():
         eng.add_request(new_prompt(), sp)
     inflight = inflight_target
 
-    def run_steps(n):
+    step_ms = []
+
+    def run_steps(n, record=False):
         nonlocal inflight
         fin = []
         for _ in range(n):
+            t0s = time.perf_counter()
             done = eng.step()
+            if record:
+                step_ms.append((time.perf_counter() - t0s) * 1000)
             fin.extend(done)
             for _ in done:
                 eng.add_request(new_prompt(), sp)
@@ -111,7 +116,7 @@ def main():
     tok0 = eng.num_generation_tokens
     t_start_mono = time.monotonic()
     t0 = time.perf_counter()
-    finished = run_steps(args.steps)
+    finished = run_steps(args.steps, record=True)
     eng.flush()  # drain the pipelined step so token counts are exact
     if torch.cuda.is_available():
         torch.cuda.synchronize()
@@ -156,6 +161,7 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "ttft_p50_ms": round(ttft_p50, 1) if ttft_p50 is not None else None,
+            "itl_p50_ms": round(statistics.median(step_ms), 2) if step_ms else None,
             "config": {
                 "model": args.model,
                 "parallelism": f"dp{world}(tp={args.tp})",

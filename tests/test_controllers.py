@@ -442,3 +442,35 @@ def test_inferenceset_gateway_manifests(client):
     assert ml["apps.kubernetes.io/pod-index"] == "0"
     assert ml["inferenceset.kaito.sh/created-by"] == "is1"
     assert "kv-cache-utilization-scorer" in values["epp"]["plugins"]
+
+
+def test_admission_webhook_allows_and_denies():
+    from fastapi.testclient import TestClient
+    from kaito_amd.operator.webhooks import build_webhook_app
+    app = build_webhook_app(sku_handler=get_sku_handler("azure"),
+                            known_presets={"llama-3-8b"})
+    c = TestClient(app)
+
+    def review(obj, group="workspace.kaito.sh"):
+        return c.post(f"/validate/{group}", json={
+            "request": {"uid": "u1", "object": obj}}).json()["response"]
+
+    good = {"metadata": {"name": "w", "namespace": "default"},
+            "spec": {"resource": {"instanceType": SKU},
+                     "inference": {"preset": {"name": "llama-3-8b"}}}}
+    assert review(good)["allowed"]
+    bad = {"metadata": {"name": "w"},
+           "spec": {"resource": {"instanceType": SKU},
+                    "inference": {"preset": {"name": "nope"}}}}
+    r = review(bad)
+    assert not r["allowed"] and "nope" in r["status"]["message"]
+    both = {"metadata": {"name": "w"},
+            "spec": {"resource": {"instanceType": SKU},
+                     "inference": {"preset": {"name": "llama-3-8b"}},
+                     "tuning": {"method": "lora",
+                                "input": {"urls": ["u"]},
+                                "output": {"image": "i"}}}}
+    assert not review(both)["allowed"]
+    iset = {"metadata": {"name": "i"},
+            "spec": {"replicas": 2, "workspaceTemplate": good["spec"]}}
+    assert review(iset, "inferenceset.kaito.sh")["allowed"]
