@@ -30,7 +30,8 @@ void lora_shrink(at::Tensor tmp, at::Tensor x, at::Tensor A, at::Tensor idx,
                  double scale);
 void lora_expand(at::Tensor y, at::Tensor tmp, at::Tensor B, at::Tensor idx);
 void paged_read_bw(at::Tensor out, at::Tensor k_cache, at::Tensor v_cache,
-                   at::Tensor block_tables, at::Tensor seq_lens);
+                   at::Tensor block_tables, at::Tensor seq_lens,
+                   int64_t mode);
 }  // namespace kaito
 
 TORCH_LIBRARY(kaito, m) {
@@ -46,7 +47,7 @@ TORCH_LIBRARY(kaito, m) {
   m.def("topk(Tensor(a!) out_vals, Tensor(b!) out_idx, Tensor scores, int k) -> ()");
   m.def("lora_shrink(Tensor(a!) tmp, Tensor x, Tensor A, Tensor idx, float scale) -> ()");
   m.def("lora_expand(Tensor(a!) y, Tensor tmp, Tensor B, Tensor idx) -> ()");
-  m.def("paged_read_bw(Tensor(a!) out, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens) -> ()");
+  m.def("paged_read_bw(Tensor(a!) out, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, int mode) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(kaito, CUDA, m) {

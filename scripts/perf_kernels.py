@@ -50,10 +50,13 @@ def main():
     print(f"paged_attention bs={T} L={L}: {t*1e6:.1f} us  "
           f"{bytes_moved/t/1e12:.2f} TB/s")
 
-    # pure-read diagnostic: same grid/address walk, no softmax
+    # diagnostics: mode0 = pure read; mode1 = +dot+accumulate (no shfl)
     outd = torch.empty(T * KH, dtype=torch.float32, device=dev)
-    t = bench(lambda: torch.ops.kaito.paged_read_bw(outd, kc, vc, bt, sl))
+    t = bench(lambda: torch.ops.kaito.paged_read_bw(outd, kc, vc, bt, sl, 0))
     print(f"paged_read_bw  bs={T} L={L}: {t*1e6:.1f} us  "
+          f"{bytes_moved/t/1e12:.2f} TB/s")
+    t = bench(lambda: torch.ops.kaito.paged_read_bw(outd, kc, vc, bt, sl, 1))
+    print(f"paged_dot_bw   bs={T} L={L}: {t*1e6:.1f} us  "
           f"{bytes_moved/t/1e12:.2f} TB/s")
 
     # rmsnorm
