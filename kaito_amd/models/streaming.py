@@ -1,0 +1,93 @@
+"""Model weight streaming — the MI355X analog of the reference's
+model-streaming subsystem (pkg/workspace/inference/modelstreaming: az://
+path resolution, SAS-token auth via fetch_sas.py init container, RunAI
+streamer --load-format=runai_streamer; SURVEY.md §2 "Model streaming" row).
+
+Sources:
+  file://path or plain path  → used in place
+  http(s)://...              → chunked streaming download with progress
+                               callbacks (drives kaito_model_download_*)
+  az://account/container/blob → resolved to the blob HTTPS endpoint; a SAS
+                               token from AZURE_STORAGE_SAS_TOKEN (the
+                               fetch-sas init-container contract) is
+                               appended
+Downloads go to a local cache dir (the NVMe PVC mount in-cluster).
+"""
+from __future__ import annotations
+
+import os
+import shutil
+from pathlib import Path
+from typing import Callable, List, Optional
+from urllib.parse import urlparse
+
+ProgressCb = Callable[[float], None]   # 0.0 - 1.0
+
+
+def resolve_azure_url(url: str) -> str:
+    """az://account/container/path → https://account.blob.core.windows.net/
+    container/path[?sas] (reference: sasblob.go path resolution)."""
+    p = urlparse(url)
+    account = p.netloc
+    path = p.path.lstrip("/")
+    base = f"https://{account}.blob.core.windows.net/{path}"
+    sas = os.environ.get("AZURE_STORAGE_SAS_TOKEN", "")
+    if sas:
+        base += ("" if sas.startswith("?") else "?") + sas
+    return base
+
+
+def _download_http(url: str, dest: Path, progress: Optional[ProgressCb],
+                   chunk_bytes: int = 8 << 20) -> Path:
+    import httpx
+    dest.parent.mkdir(parents=True, exist_ok=True)
+    tmp = dest.with_suffix(dest.suffix + ".part")
+    with httpx.stream("GET", url, follow_redirects=True, timeout=600) as r:
+        r.raise_for_status()
+        total = int(r.headers.get("content-length", 0)) or None
+        got = 0
+        with open(tmp, "wb") as f:
+            for chunk in r.iter_bytes(chunk_bytes):
+                f.write(chunk)
+                got += len(chunk)
+                if progress and total:
+                    progress(min(got / total, 1.0))
+    tmp.rename(dest)
+    if progress:
+        progress(1.0)
+    return dest
+
+
+def fetch_weights(source: str, cache_dir: str = "/workspace/weights",
+                  files: Optional[List[str]] = None,
+                  progress: Optional[ProgressCb] = None) -> str:
+    """Materialize a weights directory from `source`; returns a local path
+    suitable for load_safetensors_weights()."""
+    p = urlparse(source)
+    if p.scheme in ("", "file"):
+        path = p.path if p.scheme == "file" else source
+        if not os.path.exists(path):
+            raise FileNotFoundError(path)
+        if progress:
+            progress(1.0)
+        return path
+    if p.scheme == "az":
+        source = resolve_azure_url(source)
+        p = urlparse(source)
+    if p.scheme in ("http", "https"):
+        base = source.rstrip("/")
+        names = files or ["model.safetensors", "config.json",
+                          "tokenizer.json", "tokenizer_config.json"]
+        out = Path(cache_dir)
+        n = len(names)
+        for i, name in enumerate(names):
+            def sub(frac, i=i):
+                if progress:
+                    progress((i + frac) / n)
+            try:
+                _download_http(f"{base}/{name}", out / name, sub)
+            except Exception:  # noqa: BLE001 — optional aux files
+                if name.endswith(".safetensors"):
+                    raise
+        return str(out)
+    raise ValueError(f"unsupported weight source scheme {p.scheme!r}")

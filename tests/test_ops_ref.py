@@ -118,3 +118,23 @@ def test_context_attention_suffix_matches_full():
                               torch.tensor([L], dtype=torch.int32),
                               bt, 0.25)
     assert torch.allclose(ctx, full[-q_len:], atol=1e-4)
+
+
+def test_streaming_weight_source(tmp_path, monkeypatch):
+    from kaito_amd.models.streaming import fetch_weights, resolve_azure_url
+    # file path passthrough + progress completion
+    d = tmp_path / "w"
+    d.mkdir()
+    (d / "model.safetensors").write_bytes(b"x")
+    seen = []
+    out = fetch_weights(str(d), progress=seen.append)
+    assert out == str(d) and seen[-1] == 1.0
+    out = fetch_weights(f"file://{d}")
+    assert out == str(d)
+    with pytest.raises(FileNotFoundError):
+        fetch_weights(str(tmp_path / "missing"))
+    # az:// resolution with SAS from env (fetch-sas contract)
+    monkeypatch.setenv("AZURE_STORAGE_SAS_TOKEN", "sig=abc")
+    url = resolve_azure_url("az://myacct/models/llama/model.safetensors")
+    assert url == ("https://myacct.blob.core.windows.net/models/llama/"
+                   "model.safetensors?sig=abc")
