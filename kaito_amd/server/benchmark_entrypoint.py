@@ -31,8 +31,13 @@ RESULT_TAG = "KAITO_BENCHMARK_RESULT"
 def emit(tag: str, payload: dict) -> None:
     line = f"{tag}: {json.dumps(payload)}"
     try:
-        with open("/proc/1/fd/1", "w") as f:   # pod's main stdout
-            f.write(line + "\n")
+        # pod's main stdout; O_NONBLOCK so an unread pipe (non-pod envs)
+        # can't block the probe
+        fd = os.open("/proc/1/fd/1", os.O_WRONLY | os.O_NONBLOCK)
+        try:
+            os.write(fd, (line + "\n").encode())
+        finally:
+            os.close(fd)
     except OSError:
         pass
     print(line, flush=True)

@@ -220,3 +220,39 @@ def test_tuning_job_golden():
     assert spec["restartPolicy"] == "Never"
     assert "--method" in spec["containers"][0]["command"]
     assert spec["initContainers"][0]["name"] == "data-downloader"
+
+
+# ------------------------------------------------------------ partitioning
+def test_partition_profiles_and_validation():
+    from kaito_amd.operator.partition import (MI355X_PROFILES,
+                                              partitioned_gpu_config,
+                                              validate_partition)
+    gpu = _gpu()
+    prof = validate_partition(at.PartitionSpec("cpx", 16), gpu)
+    assert prof.partitions_per_gpu == 8 and prof.mem_gib_per_partition == 36
+    with pytest.raises(at.ValidationError):
+        validate_partition(at.PartitionSpec("mig-1g", 1), gpu)
+    with pytest.raises(at.ValidationError):
+        validate_partition(at.PartitionSpec("cpx", 100), gpu)  # > 64
+    assert validate_partition(None, gpu) is None
+    pg = partitioned_gpu_config(gpu, prof)
+    assert pg.gpu_count == 64 and pg.gpu_mem_gib == 36
+    assert pg.xgmi_links == 0
+
+
+def test_partition_estimator_single_slice_rule():
+    """8B fits a 36 GiB CPX partition; 70B must not (MIG single-slice
+    analog)."""
+    from kaito_amd.operator.partition import (MI355X_PROFILES,
+                                              partitioned_gpu_config)
+    from kaito_amd.operator.estimator import (NodeEstimateRequest,
+                                              estimate_node_count)
+    pg = partitioned_gpu_config(_gpu(), MI355X_PROFILES["cpx"])
+    res = estimate_node_count(NodeEstimateRequest(
+        model=get_model_config("llama-3-8b"), gpu=pg, max_model_len=2048,
+        max_num_seqs=8))
+    assert res.min_gpus == 1   # one partition suffices for 8B
+    res70 = estimate_node_count(NodeEstimateRequest(
+        model=get_model_config("llama-3-70b"), gpu=pg, max_model_len=2048,
+        max_num_seqs=8))
+    assert res70.min_gpus > 1  # 70B cannot fit one partition
