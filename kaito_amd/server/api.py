@@ -13,7 +13,7 @@ import time
 import uuid
 from typing import AsyncIterator, List, Optional
 
-from fastapi import FastAPI, Request
+from fastapi import FastAPI, HTTPException, Request
 from fastapi.responses import JSONResponse, Response, StreamingResponse
 from pydantic import BaseModel, Field
 
@@ -65,6 +65,16 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
               max_queue: Optional[int] = None) -> FastAPI:
     app = FastAPI(title="kaito-amd inference engine")
     eng_cfg = async_engine.engine.cfg
+    vocab_size = eng_cfg.model.vocab_size
+    max_len = async_engine.engine.runner.max_model_len
+
+    def _check_prompt(ids):
+        if any(i < 0 or i >= vocab_size for i in ids):
+            raise HTTPException(422, f"prompt token id out of vocab "
+                                     f"(vocab_size={vocab_size})")
+        if len(ids) >= max_len:
+            raise HTTPException(422, f"prompt longer than max_model_len "
+                                     f"({max_len})")
     app.add_middleware(
         RateLimitMiddleware,
         get_queue_depth=lambda: async_engine.num_waiting,
@@ -86,7 +96,9 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
     async def models():
         return {"object": "list", "data": [{
             "id": model_name, "object": "model",
-            "owned_by": "kaito-amd", "created": int(time.time())}]}
+            "owned_by": "kaito-amd", "created": int(time.time()),
+            "vocab_size": vocab_size,
+            "max_model_len": max_len}]}
 
     # ------------------------------------------------------------ completions
     @app.post("/v1/completions")
@@ -94,11 +106,10 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
         if isinstance(req.prompt, list) and req.prompt and \
                 isinstance(req.prompt[0], int):
             prompt_ids = [int(x) for x in req.prompt]
-            prompt_text = None
         else:
             text = req.prompt if isinstance(req.prompt, str) else str(req.prompt)
             prompt_ids = tokenizer.encode(text)
-            prompt_text = text
+        _check_prompt(prompt_ids)
         sp = _sampling(req.max_tokens, req.temperature, req.top_p, req.top_k,
                        req.ignore_eos)
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
@@ -146,6 +157,7 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
             [m.model_dump() for m in req.messages],
             add_generation_prompt=True, tokenize=False)
         prompt_ids = tokenizer.encode(text)
+        _check_prompt(prompt_ids)
         max_toks = req.max_completion_tokens or req.max_tokens or 128
         sp = _sampling(max_toks, req.temperature, req.top_p,
                        ignore_eos=req.ignore_eos)

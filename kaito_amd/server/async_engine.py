@@ -69,6 +69,15 @@ class AsyncLLMEngine:
             self._streams[sid] = p
             self._emitted[sid] = 0
 
+    def _fail_all(self, reason: str):
+        """Engine-loop exception: fail every in-flight stream (the loop
+        must survive bad requests — a dead loop hangs all clients)."""
+        for sid, p in list(self._streams.items()):
+            self._push(p, StreamItem(-1, True, reason))
+            self.engine.abort(sid)
+            self._streams.pop(sid, None)
+            self._emitted.pop(sid, None)
+
     def _loop(self):
         eng = self.engine
         while not self._stop:
@@ -78,7 +87,14 @@ class AsyncLLMEngine:
                 self._wake.wait(timeout=0.05)
                 self._wake.clear()
                 continue
-            finished = eng.step()
+            try:
+                finished = eng.step()
+            except Exception:  # noqa: BLE001
+                import logging
+                logging.getLogger("kaito_amd.server").exception(
+                    "engine step failed; failing in-flight requests")
+                self._fail_all("error")
+                continue
             metrics.REQUESTS_RUNNING.set(eng.scheduler.num_running)
             metrics.REQUESTS_WAITING.set(eng.scheduler.num_waiting)
             # push fresh tokens to streams

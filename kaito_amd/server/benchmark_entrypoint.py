@@ -78,6 +78,14 @@ async def wait_healthy(base: str, timeout_s: float) -> bool:
 async def run_benchmark(base: str, duration_s: int, in_tokens: int,
                         out_tokens: int, max_concurrency: int) -> dict:
     async with httpx.AsyncClient(timeout=300) as client:
+        # vocab bound for synthetic prompts (served model metadata)
+        vocab = 30000
+        try:
+            md = (await client.get(f"{base}/v1/models")).json()
+            vocab = int(md["data"][0].get("vocab_size", vocab))
+        except (httpx.HTTPError, KeyError, ValueError, IndexError):
+            pass
+        hi = max(vocab - 10, 12)
         metrics = (await client.get(f"{base}/metrics")).text
         cache = parse_cache_config(metrics) or {}
         block_size = int(cache.get("block_size", 16))
@@ -95,7 +103,7 @@ async def run_benchmark(base: str, duration_s: int, in_tokens: int,
             i = 0
             while time.monotonic() < stop:
                 i += 1
-                prompt = [(rng_base + j * 31 + i) % 30000 + 10
+                prompt = [(rng_base + j * 31 + i) % (hi - 10) + 10
                           for j in range(in_tokens)]
                 try:
                     r = await client.post(f"{base}/v1/completions", json={
