@@ -154,3 +154,26 @@ def test_engine_publishes_kv_events():
     assert "BlockStored" in kinds and "BlockRemoved" in kinds
     sub.close()
     pub.close()
+
+
+def test_entrypoint_config_file_merge(tmp_path, monkeypatch):
+    """--kaito-config-file YAML merge: file overrides defaults, explicit
+    CLI flags override the file (reference inference_api.py:128-154)."""
+    import sys
+    from kaito_amd.server.entrypoint import build_parser, merge_config_file
+    cfg = tmp_path / "inference_config.yaml"
+    cfg.write_text("""
+vllm:
+  max-num-seqs: 64
+  gpu-memory-utilization: 0.8
+  max-model-len: 2048
+""")
+    monkeypatch.setattr(sys, "argv",
+                        ["prog", "--model", "llama-3-8b",
+                         "--max-num-seqs", "128"])
+    args = build_parser().parse_args(
+        ["--model", "llama-3-8b", "--max-num-seqs", "128"])
+    merge_config_file(args, str(cfg))
+    assert args.max_num_seqs == 128        # CLI wins
+    assert args.gpu_memory_utilization == 0.8   # file wins over default
+    assert str(args.max_model_len) == "2048"
