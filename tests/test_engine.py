@@ -258,7 +258,9 @@ def test_engine_preemption_recovers_exact_outputs():
     sequences recompute and still produce oracle-exact outputs."""
     cfg = _cfg(max_num_seqs=4, num_gpu_blocks=8, max_model_len=64)
     eng = LLMEngine(cfg)
-    sp = SamplingParams(max_tokens=18, ignore_eos=True)
+    # 14 prompt + 22 output = 36 tokens → 3 blocks each; 3x3 > 8 forces
+    # preemption when all sequences cross the 32-token boundary together
+    sp = SamplingParams(max_tokens=22, ignore_eos=True)
     prompts = [list(range(10, 24)), list(range(30, 44)),
                list(range(50, 64))]
     ids = [eng.add_request(p, sp) for p in prompts]
@@ -270,5 +272,5 @@ def test_engine_preemption_recovers_exact_outputs():
     assert any(eng.seqs[sid].epoch > 0 for sid in ids), \
         "test did not actually exercise preemption"
     for p, sid in zip(prompts, ids):
-        expect = _naive_generate(eng.runner.model, cfg, p, 18)
+        expect = _naive_generate(eng.runner.model, cfg, p, 22)
         assert eng.seqs[sid].output_token_ids == expect
