@@ -225,6 +225,7 @@ class RAGEngineSpec:
     embedding: Dict[str, Any] = field(default_factory=dict)
     inferenceService: Dict[str, Any] = field(default_factory=dict)
     storage: Dict[str, Any] = field(default_factory=dict)
+    guardrails: Dict[str, Any] = field(default_factory=dict)
     indexServiceName: str = ""
     queryServiceName: str = ""
 

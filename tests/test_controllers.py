@@ -474,3 +474,50 @@ def test_admission_webhook_allows_and_denies():
     iset = {"metadata": {"name": "i"},
             "spec": {"replicas": 2, "workspaceTemplate": good["spec"]}}
     assert review(iset, "inferenceset.kaito.sh")["allowed"]
+
+
+# ------------------------------------------------------------ ragengine
+def test_ragengine_reconcile(client):
+    from kaito_amd.operator.controllers.ragengine import (
+        RAGEngineReconciler, rag_env_from_spec)
+    rag = at.RAGEngine("rag1", spec=at.RAGEngineSpec(
+        compute=at.ResourceSpec(instanceType=SKU),
+        embedding={"local": {"modelID": "BAAI/bge-small-en-v1.5"}},
+        inferenceService={"url": "http://ws1/v1/chat/completions",
+                          "contextWindow": 4096},
+        storage={"vectorDBType": "faiss"},
+        guardrails={"enabled": True, "policy": "blocked_keywords: ['x']\n",
+                    "hotReload": True}))
+    r = RAGEngineReconciler(client)
+    assert r.reconcile(rag) is False          # deployment not ready yet
+    dep = client.get("Deployment", "default", "rag1")
+    c = dep["spec"]["template"]["spec"]["containers"][0]
+    env = {e["name"]: e["value"] for e in c["env"]}
+    assert env["EMBEDDING_SOURCE_TYPE"] == "local"
+    assert env["LOCAL_EMBEDDING_MODEL_ID"] == "BAAI/bge-small-en-v1.5"
+    assert env["LLM_INFERENCE_URL"] == "http://ws1/v1/chat/completions"
+    assert env["LLM_CONTEXT_WINDOW"] == "4096"
+    assert env["OUTPUT_GUARDRAILS_ENABLED"] == "true"
+    assert c["resources"]["limits"]["amd.com/gpu"] == "1"  # local embedding
+    assert client.get("ConfigMap", "default", "rag1-guardrails")
+    assert client.get("Service", "default", "rag1")
+    # becomes ready
+    dep["status"] = {"readyReplicas": 1}
+    client.update(dep)
+    assert r.reconcile(rag) is True
+    assert rag.status["state"] == "Ready"
+    conds = {c["type"]: c["status"] for c in rag.status["conditions"]}
+    assert conds["ServiceReady"] == "True"
+
+
+def test_ragengine_remote_embedding_no_gpu(client):
+    from kaito_amd.operator.controllers.ragengine import RAGEngineReconciler
+    rag = at.RAGEngine("rag2", spec=at.RAGEngineSpec(
+        embedding={"remote": {"url": "http://emb/v1/embeddings"}},
+        inferenceService={"url": "http://ws1"}))
+    RAGEngineReconciler(client).reconcile(rag)
+    c = client.get("Deployment", "default",
+                   "rag2")["spec"]["template"]["spec"]["containers"][0]
+    assert "resources" not in c            # no GPU for remote embedding
+    env = {e["name"]: e["value"] for e in c["env"]}
+    assert env["EMBEDDING_SOURCE_TYPE"] == "remote"

@@ -256,3 +256,33 @@ def test_partition_estimator_single_slice_rule():
         model=get_model_config("llama-3-70b"), gpu=pg, max_model_len=2048,
         max_num_seqs=8))
     assert res70.min_gpus > 1  # 70B cannot fit one partition
+
+
+def test_preset_generator_metadata(tmp_path):
+    import json
+    from kaito_amd.utils.preset_generator import (disk_storage_gib,
+                                                  generate_preset_metadata,
+                                                  kv_bytes_per_token)
+    cfg = {"hidden_size": 4096, "num_hidden_layers": 32,
+           "num_attention_heads": 32, "num_key_value_heads": 8,
+           "intermediate_size": 14336, "vocab_size": 128256,
+           "max_position_embeddings": 8192, "torch_dtype": "bfloat16"}
+    p = tmp_path / "config.json"
+    p.write_text(json.dumps(cfg))
+    md = generate_preset_metadata(str(p), name="llama-3-8b")
+    # KV bytes/token: 2*32*8*128*2 = 131072 (matches the reference formula)
+    assert md.bytes_per_token == 131072
+    assert kv_bytes_per_token(32, 8, 128) == 131072
+    assert 14 < md.total_param_bytes / (1 << 30) < 17   # ~16 GiB bf16
+    assert md.disk_storage_gib == disk_storage_gib(md.total_param_bytes)
+    assert md.disk_storage_gib % 10 == 0
+    assert md.model_token_limit == 8192
+
+
+def test_tuning_metrics_server():
+    from fastapi.testclient import TestClient
+    from kaito_amd.tuning.metrics_server import build_metrics_app
+    c = TestClient(build_metrics_app())
+    assert c.get("/health").json()["status"] == "ok"
+    body = c.get("/metrics").text
+    assert "tuning_cpu_percent" in body or body.strip() == ""
