@@ -39,10 +39,27 @@ KAITO_DEV short f32_to_bf16(float f) {
   return (short)((v.i + rounding) >> 16);
 }
 
+// ---- DPP cross-lane adds (VALU-speed; __shfl_xor lowers to
+// ds_bpermute_b32 which costs a ~30-60 cyc LDS-crossbar trip each).
+// xor1 = quad_perm(1,0,3,2)=0xB1; xor2 = quad_perm(2,3,0,1)=0x4E;
+// xor4 = ROW_HALF_MIRROR=0x141; xor8 = ROW_ROR:8=0x128 (i^8 == i+8 mod 16
+// within a row).
+template <int CTRL>
+KAITO_DEV float dpp_xor_add(float x) {
+  union { float f; int i; } u, v;
+  u.f = x;
+  v.i = __builtin_amdgcn_update_dpp(0, u.i, CTRL, 0xf, 0xf, true);
+  return x + v.f;
+}
+
 // ---- wave-level reductions (64 lanes) ---------------------------------
 KAITO_DEV float wave_reduce_sum(float x) {
+  x = dpp_xor_add<0xB1>(x);    // xor1 (quad_perm)
+  x = dpp_xor_add<0x4E>(x);    // xor2 (quad_perm)
+  x = dpp_xor_add<0x141>(x);   // xor4-equivalent (row_half_mirror)
+  x = dpp_xor_add<0x128>(x);   // xor8-equivalent (row_ror:8)
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, 64);
+  for (int off = 16; off < 64; off <<= 1) x += __shfl_xor(x, off, 64);
   return x;
 }
 
@@ -53,10 +70,15 @@ KAITO_DEV float wave_reduce_max(float x) {
 }
 
 // Reduce within a group of `W` consecutive lanes (W power of 2, <= 64).
+// W<=16 uses pure-DPP butterflies (every lane ends with the group sum).
 template <int W>
 KAITO_DEV float group_reduce_sum(float x) {
+  if constexpr (W >= 2) x = dpp_xor_add<0xB1>(x);
+  if constexpr (W >= 4) x = dpp_xor_add<0x4E>(x);
+  if constexpr (W >= 8) x = dpp_xor_add<0x141>(x);
+  if constexpr (W >= 16) x = dpp_xor_add<0x128>(x);
 #pragma unroll
-  for (int off = W / 2; off > 0; off >>= 1) x += __shfl_xor(x, off, 64);
+  for (int off = 16; off < W; off <<= 1) x += __shfl_xor(x, off, 64);
   return x;
 }
 
