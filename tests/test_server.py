@@ -177,3 +177,21 @@ vllm:
     assert args.max_num_seqs == 128        # CLI wins
     assert args.gpu_memory_utilization == 0.8   # file wins over default
     assert str(args.max_model_len) == "2048"
+
+
+def test_transformers_fallback_runtime():
+    """Fallback runtime serves phi-3-mini-class models (head_dim=96 — outside
+    the native decode kernel's support matrix)."""
+    from kaito_amd.server.transformers_runtime import (FallbackGenerator,
+                                                       build_fallback_app)
+    gen = FallbackGenerator("tiny-llama-test", device="cpu")
+    app = build_fallback_app(gen, "tiny-llama-test")
+    with TestClient(app) as c:
+        assert c.get("/health").json()["status"] == "ok"
+        assert c.get("/v1/models").json()["data"][0]["runtime"] == \
+            "transformers"
+        r = c.post("/v1/completions", json={"prompt": [4, 5, 6],
+                                            "max_tokens": 3,
+                                            "temperature": 0})
+        assert r.status_code == 200
+        assert r.json()["usage"]["completion_tokens"] == 3
