@@ -274,3 +274,17 @@ def test_engine_preemption_recovers_exact_outputs():
     for p, sid in zip(prompts, ids):
         expect = _naive_generate(eng.runner.model, cfg, p, 22)
         assert eng.seqs[sid].output_token_ids == expect
+
+
+def test_engine_stochastic_sampling_paths():
+    """temperature/top-k/top-p exercise the non-greedy sampler through the
+    pipelined engine (finite, in-vocab tokens)."""
+    cfg = _cfg()
+    eng = LLMEngine(cfg)
+    sp = SamplingParams(max_tokens=6, ignore_eos=True, temperature=0.8,
+                        top_k=20, top_p=0.9)
+    outs = eng.generate([[5, 6, 7, 8]] * 3, sp)
+    for seq in outs:
+        assert len(seq.output_token_ids) == 6
+        assert all(0 <= t < cfg.model.vocab_size
+                   for t in seq.output_token_ids)
