@@ -17,9 +17,8 @@ from __future__ import annotations
 import pickle
 import socket
 import struct
-import threading
 from dataclasses import dataclass
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import torch
 

@@ -11,7 +11,6 @@ MI355X-first specifics:
 from __future__ import annotations
 
 import logging
-import os
 from pathlib import Path
 from typing import Dict, List, Optional, Tuple
 

@@ -6,8 +6,6 @@ no network, synthetic weights).
 """
 from __future__ import annotations
 
-import json
-import os
 from pathlib import Path
 from typing import Dict
 

@@ -16,7 +16,6 @@ Downloads go to a local cache dir (the NVMe PVC mount in-cluster).
 from __future__ import annotations
 
 import os
-import shutil
 from pathlib import Path
 from typing import Callable, List, Optional
 from urllib.parse import urlparse

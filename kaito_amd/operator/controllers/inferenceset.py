@@ -5,16 +5,15 @@ Ready >= desired), label propagation, TPM aggregation, HPA/KEDA selector.
 """
 from __future__ import annotations
 
-import copy
 import hashlib
 import json
 from dataclasses import dataclass
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from ..api_types import (COND_INFERENCESET_READY, Condition, InferenceSet,
-                         LABEL_INFERENCESET_CREATED_BY, LABEL_UPGRADE_TO_VERSION,
-                         LABEL_WORKSPACE_NAME, Workspace)
-from ..kubeclient import KubeClient, NotFound
+                         LABEL_INFERENCESET_CREATED_BY,
+                         LABEL_UPGRADE_TO_VERSION)
+from ..kubeclient import KubeClient
 
 
 @dataclass

@@ -13,7 +13,6 @@ maintenance window (api/v1beta1/inferenceset_types.go:48-110).
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 from ..api_types import (LABEL_INFERENCESET_CREATED_BY,

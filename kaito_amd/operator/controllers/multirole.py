@@ -11,10 +11,10 @@ sidecar on :5000; KV blocks move prefill→decode over the transfer channel
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Dict, List
+from typing import Dict
 
-from ..api_types import (InferenceSet, InferenceSetSpec, LABEL_INFERENCE_ROLE,
-                         LABEL_MRI_CREATED_BY, ValidationError, Workspace)
+from ..api_types import (LABEL_INFERENCE_ROLE, LABEL_MRI_CREATED_BY,
+                         ValidationError)
 from ..kubeclient import KubeClient, NotFound
 
 ROLE_PREFILL = "prefill"

@@ -11,7 +11,7 @@ from __future__ import annotations
 import time
 from typing import Dict, List, Optional
 
-from ..api_types import (COND_RAGENGINE_SERVICE_READY, Condition, RAGEngine)
+from ..api_types import COND_RAGENGINE_SERVICE_READY, RAGEngine
 from ..kubeclient import KubeClient, NotFound
 
 RAG_PORT = 5000

@@ -10,15 +10,14 @@ import hashlib
 import json
 import time
 from dataclasses import dataclass
-from typing import Callable, Dict, List, Optional
+from typing import Callable, Dict, Optional
 
 from ..api_types import (ANNOTATION_DISABLE_BENCHMARK,
                          ANNOTATION_WORKSPACE_HASH, COND_INFERENCE_READY,
                          COND_JOB_STARTED, COND_NODECLAIM_READY,
                          COND_NODES_READY, COND_RESOURCE_READY,
                          COND_WORKSPACE_SUCCEEDED, Condition,
-                         LABEL_WORKSPACE_NAME, LABEL_WORKSPACE_NAMESPACE,
-                         ValidationError, Workspace)
+                         LABEL_WORKSPACE_NAME, ValidationError, Workspace)
 from ..estimator import NodeEstimateRequest, estimate_node_count
 from ..kubeclient import KubeClient, NotFound
 from ..manifests import (generate_service, generate_statefulset,

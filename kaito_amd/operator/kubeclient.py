@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import copy
 import threading
-from typing import Any, Callable, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
 Obj = Dict[str, Any]
 

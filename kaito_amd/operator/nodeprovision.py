@@ -10,7 +10,7 @@ Implementations:
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from .api_types import LABEL_WORKSPACE_NAME, LABEL_WORKSPACE_NAMESPACE, Workspace
 from .kubeclient import KubeClient, NotFound

@@ -15,12 +15,11 @@ and buildVLLMInferenceCommand (:394-528), retargeted at an 8-GPU xGMI node:
 from __future__ import annotations
 
 import shlex
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from dataclasses import dataclass
+from typing import List, Optional
 
 from ..engine.config import ModelConfig
-from .estimator import (GIB, GPU_MEMORY_UTILIZATION, BASE_OVERHEAD_GIB,
-                        NodeEstimateRequest, estimate_node_count)
+from .estimator import (GIB, GPU_MEMORY_UTILIZATION, BASE_OVERHEAD_GIB)
 from .sku import GPUConfig
 
 

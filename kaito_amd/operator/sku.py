@@ -9,7 +9,7 @@ pkg/sku/helpers.go:75-119 for nvidia.com/*).
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 

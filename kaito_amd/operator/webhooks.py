@@ -10,7 +10,7 @@ from typing import Callable, Dict, Optional
 
 from fastapi import FastAPI, Request
 
-from .api_types import (InferenceSet, RAGEngine, ValidationError, Workspace)
+from .api_types import ValidationError
 from .main import workspace_from_obj
 from .sku import CloudSKUHandler
 

@@ -9,7 +9,6 @@ a deterministic feature-hash embedder stands in.
 from __future__ import annotations
 
 import hashlib
-import math
 import re
 from typing import List, Optional
 

@@ -21,7 +21,7 @@ import os
 import re
 import threading
 from dataclasses import dataclass, field
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import yaml
 

@@ -14,7 +14,7 @@ import uuid
 from typing import AsyncIterator, List, Optional
 
 from fastapi import FastAPI, HTTPException, Request
-from fastapi.responses import JSONResponse, Response, StreamingResponse
+from fastapi.responses import Response, StreamingResponse
 from pydantic import BaseModel, Field
 
 from ..engine.sequence import SamplingParams

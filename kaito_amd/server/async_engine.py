@@ -11,11 +11,11 @@ import asyncio
 import queue
 import threading
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import AsyncIterator, Dict, List, Optional
 
 from ..engine.engine import LLMEngine
-from ..engine.sequence import SamplingParams, Sequence
+from ..engine.sequence import SamplingParams
 from . import metrics
 
 

@@ -14,7 +14,7 @@ from __future__ import annotations
 import json
 import math
 import os
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 import torch.nn as nn
