@@ -32,7 +32,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=250)
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--tp", type=int, default=1)
-    p.add_argument("--max-num-seqs", type=int, default=512)
+    p.add_argument("--max-num-seqs", type=int, default=1024)
     p.add_argument("--in-tokens", type=int, default=200)
     p.add_argument("--out-tokens", type=int, default=200)
     p.add_argument("--eager", action="store_true")

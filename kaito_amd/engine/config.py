@@ -65,7 +65,7 @@ class ModelConfig:
 class EngineConfig:
     model: ModelConfig = field(default_factory=ModelConfig)
     block_size: int = 16
-    max_num_seqs: int = 512
+    max_num_seqs: int = 1024
     max_num_batched_tokens: int = 8192      # prefill token budget per step
     max_model_len: Optional[int] = None     # None → "auto": fit KV budget
     gpu_memory_utilization: float = 0.90
@@ -81,7 +81,8 @@ class EngineConfig:
     seed: int = 0
     # decode graph buckets (batch sizes to capture)
     graph_batch_sizes: tuple = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,
-                                192, 256, 320, 384, 448, 512)
+                                192, 256, 320, 384, 448, 512, 640, 768,
+                                896, 1024)
 
     def max_blocks_per_seq(self, max_len: int) -> int:
         return (max_len + self.block_size - 1) // self.block_size
