@@ -13,8 +13,9 @@ import logging
 import time
 from typing import Dict, Optional
 
-from .api_types import Workspace, ResourceSpec, InferenceSpec, TuningSpec, \
-    PresetSpec, DataSource, DataDestination
+from .api_types import (Workspace, ResourceSpec, InferenceSpec, TuningSpec,
+                        PresetSpec, DataSource, DataDestination,
+                        InferenceSet, InferenceSetSpec)
 from .controllers.inferenceset import InferenceSetReconciler
 from .controllers.workspace import WorkspaceReconciler
 from .featuregates import parse_feature_gates
@@ -60,6 +61,22 @@ def workspace_from_obj(obj: Dict) -> Workspace:
     return ws
 
 
+def inferenceset_from_obj(obj: Dict) -> InferenceSet:
+    spec = obj.get("spec", {})
+    tpl_obj = {"metadata": {"name": obj["metadata"]["name"] + "-tpl",
+                            "namespace": obj["metadata"].get("namespace",
+                                                             "default")},
+               "spec": spec.get("workspaceTemplate", {})}
+    return InferenceSet(
+        name=obj["metadata"]["name"],
+        namespace=obj["metadata"].get("namespace", "default"),
+        spec=InferenceSetSpec(
+            replicas=spec.get("replicas", 1),
+            workspaceTemplate=workspace_from_obj(tpl_obj),
+            upgradeStrategy=spec.get("upgradeStrategy", "Surge"),
+            maintenanceWindow=spec.get("maintenanceWindow", "")))
+
+
 class OperatorLoop:
     """Polling reconcile driver (controller-runtime informer analog)."""
 
@@ -77,6 +94,14 @@ class OperatorLoop:
     def tick(self) -> int:
         """One reconcile pass over all stored CRs. Returns CR count."""
         n = 0
+        if self.inferenceset is not None:
+            for obj in self.client.list("InferenceSet"):
+                try:
+                    self.inferenceset.reconcile(inferenceset_from_obj(obj))
+                except Exception:  # noqa: BLE001
+                    logger.exception("inferenceset %s reconcile failed",
+                                     obj["metadata"]["name"])
+                n += 1
         for obj in self.client.list("Workspace"):
             ws = workspace_from_obj(obj)
             try:
@@ -84,6 +109,8 @@ class OperatorLoop:
             except Exception:  # noqa: BLE001
                 logger.exception("workspace %s reconcile failed", ws.name)
             n += 1
+        from .metrics import monitor_workspaces
+        monitor_workspaces(self.client.list("Workspace"))
         return n
 
     def run(self, interval_s: float = 5.0, max_ticks: Optional[int] = None):

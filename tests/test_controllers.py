@@ -521,3 +521,23 @@ def test_ragengine_remote_embedding_no_gpu(client):
     assert "resources" not in c            # no GPU for remote embedding
     env = {e["name"]: e["value"] for e in c["env"]}
     assert env["EMBEDDING_SOURCE_TYPE"] == "remote"
+
+
+def test_operator_loop_reconciles_inferencesets(client):
+    from kaito_amd.operator.main import OperatorLoop
+    client.create({
+        "apiVersion": "kaito.sh/v1beta1", "kind": "InferenceSet",
+        "metadata": {"name": "is-loop", "namespace": "default"},
+        "spec": {"replicas": 2, "workspaceTemplate": {
+            "resource": {"instanceType": SKU},
+            "inference": {"preset": {"name": "llama-3-8b"}}}},
+        "status": {}})
+    client.create(_node("n1", labels={
+        "node.kubernetes.io/instance-type": SKU}))
+    loop = OperatorLoop(client, provisioner="byo")
+    loop.tick()   # creates child workspaces
+    children = client.list("Workspace", "default",
+                           {at.LABEL_INFERENCESET_CREATED_BY: "is-loop"})
+    assert len(children) == 2
+    loop.tick()   # reconciles the children to statefulsets
+    assert client.get("StatefulSet", "default", "is-loop-0")
