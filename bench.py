@@ -86,15 +86,12 @@ def main():
     sp = SamplingParams(max_tokens=args.out_tokens, ignore_eos=True)
 
     # saturate: queue 1.5x max_num_seqs requests; refill on finish.
-    inflight_target = int(args.max_num_seqs * 1.5)
-    for _ in range(inflight_target):
+    for _ in range(int(args.max_num_seqs * 1.5)):
         eng.add_request(new_prompt(), sp)
-    inflight = inflight_target
 
     step_ms = []
 
     def run_steps(n, record=False):
-        nonlocal inflight
         fin = []
         for _ in range(n):
             t0s = time.perf_counter()

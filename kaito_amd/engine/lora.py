@@ -8,7 +8,6 @@ HIP shrink/expand kernels. CPU fallback mirrors the math for tests.
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 

@@ -8,8 +8,9 @@
 //  - wave = 64 lanes: lane (tg,dc) = (lane>>3, lane&7) → 8 tokens/wave-step,
 //    16 dims/lane → 32 B/lane vectorized bf16 loads (2KB contiguous per
 //    wave-step from one [block, kv_head] slab).
-//  - flash-decoding online softmax per wave, LDS cross-wave combine,
-//    deferred rescale when the running max doesn't grow (T13).
+//  - flash-decoding online softmax kept PER-LANE inside the K/V stream
+//    (single cross-lane merge at the end; in-loop cross-lane work is just
+//    the 8-lane dot reduce, done as DPP adds), LDS cross-wave combine.
 //
 // Cache layout: [num_blocks, KV_HEADS, BLOCK_SIZE, HEAD_DIM] bf16.
 #include "common.h"
