@@ -61,6 +61,22 @@ def _sampling(max_tokens, temperature, top_p, top_k=0, ignore_eos=False,
         top_p=top_p, top_k=top_k, ignore_eos=ignore_eos)
 
 
+def _stop_list(stop) -> List[str]:
+    if stop is None:
+        return []
+    return [stop] if isinstance(stop, str) else [str(x) for x in stop]
+
+
+def _truncate_at_stop(text: str, stops: List[str]):
+    """Returns (text, hit): text cut before the first stop string."""
+    best = None
+    for st in stops:
+        i = text.find(st)
+        if i >= 0 and (best is None or i < best):
+            best = i
+    return (text[:best], True) if best is not None else (text, False)
+
+
 def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
               max_queue: Optional[int] = None) -> FastAPI:
     app = FastAPI(title="kaito-amd inference engine")
@@ -119,15 +135,25 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
                 media_type="text/event-stream")
         toks: List[int] = []
         finish = "length"
+        stops = _stop_list(req.stop)
+        text = ""
         async for item in async_engine.generate(prompt_ids, sp):
             if item.finished:
                 finish = item.finish_reason or "stop"
             else:
                 toks.append(item.token_id)
+                if stops:
+                    text = tokenizer.decode(toks)
+                    cut, hit = _truncate_at_stop(text, stops)
+                    if hit:
+                        text, finish = cut, "stop"
+                        break
+        if not stops:
+            text = tokenizer.decode(toks)
         return {
             "id": rid, "object": "text_completion",
             "created": int(time.time()), "model": model_name,
-            "choices": [{"index": 0, "text": tokenizer.decode(toks),
+            "choices": [{"index": 0, "text": text,
                          "finish_reason": finish, "logprobs": None}],
             "usage": {"prompt_tokens": len(prompt_ids),
                       "completion_tokens": len(toks),

@@ -131,6 +131,14 @@ class AsyncLLMEngine:
     def num_running(self) -> int:
         return self.engine.scheduler.num_running
 
+    def abort(self, seq_id: Optional[int]) -> None:
+        """Client-side cancel (stop-string hit / disconnect)."""
+        if seq_id is None:
+            return
+        self._streams.pop(seq_id, None)
+        self._emitted.pop(seq_id, None)
+        self.engine.abort(seq_id)
+
     async def generate(self, prompt_ids: List[int], sampling: SamplingParams
                        ) -> AsyncIterator[StreamItem]:
         loop = asyncio.get_running_loop()
@@ -138,8 +146,12 @@ class AsyncLLMEngine:
         p = _Pending(prompt_ids, sampling, out_q, loop)
         self._submit.put(p)
         self._wake.set()
-        while True:
-            item = await out_q.get()
-            yield item
-            if item.finished:
-                return
+        try:
+            while True:
+                item = await out_q.get()
+                yield item
+                if item.finished:
+                    return
+        finally:
+            # consumer stopped early (stop string / client disconnect)
+            self.abort(p.seq_id)

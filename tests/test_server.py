@@ -195,3 +195,22 @@ def test_transformers_fallback_runtime():
                                             "temperature": 0})
         assert r.status_code == 200
         assert r.json()["usage"]["completion_tokens"] == 3
+
+
+def test_stop_strings_truncate(client):
+    """'stop' strings cut the completion and abort the sequence."""
+    # byte tokenizer: generated ids decode to bytes; pick the first
+    # generated char as the stop string
+    r0 = client.post("/v1/completions", json={
+        "prompt": "ab", "max_tokens": 6, "temperature": 0,
+        "ignore_eos": True})
+    full = r0.json()["choices"][0]["text"]
+    assert len(full) > 1
+    stop = full[1]
+    r = client.post("/v1/completions", json={
+        "prompt": "ab", "max_tokens": 6, "temperature": 0,
+        "ignore_eos": True, "stop": [stop]})
+    data = r.json()
+    assert data["choices"][0]["finish_reason"] == "stop"
+    assert stop not in data["choices"][0]["text"]
+    assert data["choices"][0]["text"] == full.split(stop)[0]
