@@ -21,7 +21,7 @@
 namespace kaito {
 
 template <int D, int G, int BS>
-__global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
+__global__ __launch_bounds__(256, 2) void paged_attention_kernel(
     short* __restrict__ out,             // [T, QH, D] bf16
     const short* __restrict__ q,         // [T, QH, D] bf16
     const short* __restrict__ k_cache,   // [B, KH, BS, D]
@@ -43,18 +43,22 @@ __global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
 
   const int* bt = block_tables + (int64_t)seq * max_blocks;
 
-  // ---- preload Q for this kv-head's group (PACKED bf16: occupancy) ----
+  // ---- preload Q for this kv-head's group (f32) ----
   // lane-dim mapping: vector vv holds dims vv*64 + dc*8 + [0..7] so each
   // 16B lane-load is CONTIGUOUS across the 8 dc-lanes (dense 128B/token
-  // transactions). Q stays bf16-packed (f32 qreg cost 2x the registers and
-  // dropped occupancy to 2 waves/SIMD); it is unpacked fused into the dot.
-  short8_t qpk[G][D / 64];
+  // transactions; the naive dc*DL mapping strides 32B between lanes and
+  // halves effective HBM bandwidth).
+  float qreg[G][DL];
 #pragma unroll
-  for (int g = 0; g < G; g++)
+  for (int g = 0; g < G; g++) {
 #pragma unroll
-    for (int vv = 0; vv < D / 64; vv++)
-      qpk[g][vv] = *reinterpret_cast<const short8_t*>(
+    for (int vv = 0; vv < DL / 8; vv++) {
+      short8_t x = *reinterpret_cast<const short8_t*>(
           q + (int64_t)seq * q_stride + (kvh * G + g) * D + vv * 64 + dc * 8);
+#pragma unroll
+      for (int j = 0; j < 8; j++) qreg[g][vv * 8 + j] = bf16_to_f32(x[j]);
+    }
+  }
 
   float m[G], l[G], acc[G][DL];
 #pragma unroll
@@ -77,10 +81,10 @@ __global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
   };
   constexpr int NV = DL / 8;              // short8 vectors per lane (2 for D=128)
 
-  // software-pipelined: prefetch the next chunk's K/V (raw bf16) while the
-  // current chunk's dot/softmax runs; occupancy (3 waves/SIMD at G<=4)
-  // supplies the cross-wave latency hiding.
-  short8_t kraw[NV], vraw[NV];
+  // 2-wide chunk unroll: two INDEPENDENT softmax/accumulate chains per
+  // iteration (ILP across the shfl/exp latency chains) + prefetch of the
+  // following pair (HBM latency hidden under both chains).
+  short8_t kraw[2][NV], vraw[2][NV];
   auto load_chunk = [&](int c, short8_t (&kd)[NV], short8_t (&vd)[NV]) {
     const int64_t b = chunk_base(c);
 #pragma unroll
@@ -89,7 +93,8 @@ __global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
       vd[vv] = *reinterpret_cast<const short8_t*>(v_cache + b + vv * 64);
     }
   };
-  if (wave < nchunks) load_chunk(wave, kraw, vraw);
+  if (wave < nchunks) load_chunk(wave, kraw[0], vraw[0]);
+  if (wave + NW < nchunks) load_chunk(wave + NW, kraw[1], vraw[1]);
 
   // one chunk's update: dot → PER-LANE online softmax → accumulate.
   // The dot's 8-lane reduce is the ONLY cross-lane op in the loop; each
@@ -111,10 +116,7 @@ __global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
     for (int g = 0; g < G; g++) {
       float p = 0.f;
 #pragma unroll
-      for (int vv = 0; vv < NV; vv++)
-#pragma unroll
-        for (int j = 0; j < 8; j++)
-          p += bf16_to_f32(qpk[g][vv][j]) * kreg[vv * 8 + j];
+      for (int j = 0; j < DL; j++) p += qreg[g][j] * kreg[j];
       p = group_reduce_sum<8>(p) * scale;   // dot over 8 dc-lanes
       s[g] = valid ? p : -1e30f;
     }
@@ -142,12 +144,17 @@ __global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
     }
   };
 
-  for (int c = wave; c < nchunks; c += NW) {
-    short8_t kn[NV], vn[NV];
-    if (c + NW < nchunks) load_chunk(c + NW, kn, vn);
-    process(c, kraw, vraw);
+  for (int c = wave; c < nchunks; c += 2 * NW) {
+    short8_t kn[2][NV], vn[2][NV];
+    if (c + 2 * NW < nchunks) load_chunk(c + 2 * NW, kn[0], vn[0]);
+    if (c + 3 * NW < nchunks) load_chunk(c + 3 * NW, kn[1], vn[1]);
+    process(c, kraw[0], vraw[0]);
+    process(c + NW, kraw[1], vraw[1]);
 #pragma unroll
-    for (int vv = 0; vv < NV; vv++) { kraw[vv] = kn[vv]; vraw[vv] = vn[vv]; }
+    for (int vv = 0; vv < NV; vv++) {
+      kraw[0][vv] = kn[0][vv]; vraw[0][vv] = vn[0][vv];
+      kraw[1][vv] = kn[1][vv]; vraw[1][vv] = vn[1][vv];
+    }
   }
 
   // ---- merge lane-local softmax state (once, not per chunk) ----
