@@ -21,7 +21,7 @@
 namespace kaito {
 
 template <int D, int G, int BS>
-__global__ __launch_bounds__(256, 2) void paged_attention_kernel(
+__global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
     short* __restrict__ out,             // [T, QH, D] bf16
     const short* __restrict__ q,         // [T, QH, D] bf16
     const short* __restrict__ k_cache,   // [B, KH, BS, D]
@@ -81,10 +81,9 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
   };
   constexpr int NV = DL / 8;              // short8 vectors per lane (2 for D=128)
 
-  // 2-wide chunk unroll: two INDEPENDENT softmax/accumulate chains per
-  // iteration (ILP across the shfl/exp latency chains) + prefetch of the
-  // following pair (HBM latency hidden under both chains).
-  short8_t kraw[2][NV], vraw[2][NV];
+  // minimal-register variant: no explicit prefetch regs — loads are issued
+  // at the top of each iteration and occupancy (3 waves/SIMD) hides the
+  // latency across waves.
   auto load_chunk = [&](int c, short8_t (&kd)[NV], short8_t (&vd)[NV]) {
     const int64_t b = chunk_base(c);
 #pragma unroll
@@ -93,8 +92,6 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
       vd[vv] = *reinterpret_cast<const short8_t*>(v_cache + b + vv * 64);
     }
   };
-  if (wave < nchunks) load_chunk(wave, kraw[0], vraw[0]);
-  if (wave + NW < nchunks) load_chunk(wave + NW, kraw[1], vraw[1]);
 
   // one chunk's update: dot → PER-LANE online softmax → accumulate.
   // The dot's 8-lane reduce is the ONLY cross-lane op in the loop; each
@@ -144,17 +141,10 @@ __global__ __launch_bounds__(256, 2) void paged_attention_kernel(
     }
   };
 
-  for (int c = wave; c < nchunks; c += 2 * NW) {
-    short8_t kn[2][NV], vn[2][NV];
-    if (c + 2 * NW < nchunks) load_chunk(c + 2 * NW, kn[0], vn[0]);
-    if (c + 3 * NW < nchunks) load_chunk(c + 3 * NW, kn[1], vn[1]);
-    process(c, kraw[0], vraw[0]);
-    process(c + NW, kraw[1], vraw[1]);
-#pragma unroll
-    for (int vv = 0; vv < NV; vv++) {
-      kraw[0][vv] = kn[0][vv]; vraw[0][vv] = vn[0][vv];
-      kraw[1][vv] = kn[1][vv]; vraw[1][vv] = vn[1][vv];
-    }
+  for (int c = wave; c < nchunks; c += NW) {
+    short8_t kd[NV], vd[NV];
+    load_chunk(c, kd, vd);
+    process(c, kd, vd);
   }
 
   // ---- merge lane-local softmax state (once, not per chunk) ----
