@@ -64,7 +64,12 @@ def build_rag_app(cfg: Optional[RagConfig] = None, embedding=None,
                   manager: Optional[VectorStoreManager] = None) -> FastAPI:
     cfg = cfg or RagConfig()
     embedding = embedding or make_embedding(cfg)
-    manager = manager or VectorStoreManager(embedding)
+    if manager is None:
+        factory = None
+        if cfg.vector_db_type.lower() == "qdrant" and cfg.vector_db_url:
+            from .qdrant_store import make_index
+            factory = lambda dim, name: make_index(cfg, dim, name)  # noqa: E731
+        manager = VectorStoreManager(embedding, index_factory=factory)
     app = FastAPI(title="kaito-amd ragengine")
     app.state.manager = manager
     app.state.cfg = cfg

@@ -129,13 +129,16 @@ class FlatIndex:
 
 
 class VectorStoreIndex:
-    """One named index: docs + flat vector index + BM25 stats."""
+    """One named index: docs + vector index (flat local or qdrant remote)
+    + BM25 stats."""
 
-    def __init__(self, name: str, embedding, use_gpu: Optional[bool] = None):
+    def __init__(self, name: str, embedding, use_gpu: Optional[bool] = None,
+                 index_factory=None):
         self.name = name
         self.embedding = embedding
         self.docs: Dict[str, Document] = {}
-        self.flat = FlatIndex(embedding.dim, use_gpu)
+        self.flat = index_factory(embedding.dim, name) if index_factory \
+            else FlatIndex(embedding.dim, use_gpu)
         self.bm25 = BM25Index()
         self.lock = threading.RLock()
 
@@ -232,9 +235,11 @@ class VectorStoreIndex:
 class VectorStoreManager:
     """Named-index manager (reference: VectorStoreManager)."""
 
-    def __init__(self, embedding, use_gpu: Optional[bool] = None):
+    def __init__(self, embedding, use_gpu: Optional[bool] = None,
+                 index_factory=None):
         self.embedding = embedding
         self.use_gpu = use_gpu
+        self.index_factory = index_factory
         self.indexes: Dict[str, VectorStoreIndex] = {}
         self.lock = threading.RLock()
 
@@ -243,8 +248,8 @@ class VectorStoreManager:
             if name not in self.indexes:
                 if not create:
                     raise KeyError(f"index {name!r} not found")
-                self.indexes[name] = VectorStoreIndex(name, self.embedding,
-                                                      self.use_gpu)
+                self.indexes[name] = VectorStoreIndex(
+                    name, self.embedding, self.use_gpu, self.index_factory)
             return self.indexes[name]
 
     def list_indexes(self) -> List[str]:

@@ -191,3 +191,78 @@ def test_buffer_window_catches_split_matches(tmp_path):
         got += bw2.feed(text[i:i + 7])
     got += bw2.flush()
     assert got == text
+
+
+# ------------------------------------------------------------ qdrant backend
+def _fake_qdrant_app():
+    """Minimal in-process Qdrant REST double (collections/points API)."""
+    from fastapi import FastAPI
+    app = FastAPI()
+    store = {}
+
+    @app.get("/collections/{name}")
+    def getc(name: str):
+        from fastapi.responses import JSONResponse
+        if name in store:
+            return {"result": {}}
+        return JSONResponse({"status": "not found"}, status_code=404)
+
+    @app.put("/collections/{name}")
+    def putc(name: str, body: dict):
+        store[name] = {}
+        return {"result": True}
+
+    @app.put("/collections/{name}/points")
+    def upsert(name: str, body: dict):
+        for pt in body["points"]:
+            store[name][pt["id"]] = pt
+        return {"result": {}}
+
+    @app.post("/collections/{name}/points/delete")
+    def delete(name: str, body: dict):
+        for pid in body["points"]:
+            store[name].pop(pid, None)
+        return {"result": {}}
+
+    @app.post("/collections/{name}/points/search")
+    def search(name: str, body: dict):
+        q = np.array(body["vector"], dtype=np.float32)
+        hits = []
+        for pt in store[name].values():
+            v = np.array(pt["vector"], dtype=np.float32)
+            denom = (np.linalg.norm(q) * np.linalg.norm(v)) or 1.0
+            hits.append({"id": pt["id"], "score": float(q @ v / denom),
+                         "payload": pt["payload"]})
+        hits.sort(key=lambda h: -h["score"])
+        return {"result": hits[: body["limit"]]}
+
+    return app
+
+
+def test_qdrant_index_backend():
+    import socket
+    import threading
+    import time
+    import uvicorn
+    from kaito_amd.ragengine.qdrant_store import QdrantIndex
+    app = _fake_qdrant_app()
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    server = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=port,
+                                           log_level="error"))
+    t = threading.Thread(target=server.run, daemon=True)
+    t.start()
+    deadline = time.monotonic() + 10
+    while not server.started and time.monotonic() < deadline:
+        time.sleep(0.05)
+    idx = QdrantIndex(4, "kb", f"http://127.0.0.1:{port}")
+    idx.add("a", np.array([1, 0, 0, 0], dtype=np.float32))
+    idx.add("b", np.array([0, 1, 0, 0], dtype=np.float32))
+    idx.add("c", np.array([0.9, 0.1, 0, 0], dtype=np.float32))
+    hits = idx.search(np.array([1, 0, 0, 0], dtype=np.float32), 2)
+    assert [h[0] for h in hits] == ["a", "c"]
+    idx.remove("a")
+    hits = idx.search(np.array([1, 0, 0, 0], dtype=np.float32), 2)
+    assert hits[0][0] == "c"
+    assert len(idx) == 2
+    server.should_exit = True
