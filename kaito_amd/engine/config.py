@@ -36,6 +36,42 @@ class ModelConfig:
     num_experts: int = 0
     num_experts_per_tok: int = 2
     moe_intermediate_size: int = 0       # per-expert FFN width (0 → dense I)
+    # serving runtime: "native" → HIP engine (llama-architecture family);
+    # "transformers" → fallback runtime (reference: vLLM vs text-generation
+    # runtime split, supported_models.yaml `runtime:` field)
+    runtime: str = "native"
+
+    @classmethod
+    def from_hf_config(cls, cfg: dict, name: str = "") -> "ModelConfig":
+        """Build from a HuggingFace config.json dict (the reference's preset
+        generator derives BytesPerToken/size the same way,
+        presets/workspace/generator/generator.go:389,660)."""
+        h = cfg.get("hidden_size", 4096)
+        heads = cfg.get("num_attention_heads", 32)
+        arch = (cfg.get("architectures") or [""])[0]
+        native = any(a in arch for a in (
+            "Llama", "Mistral", "Qwen2", "Phi3", "Phi4", "Mixtral"))
+        return cls(
+            name=name or cfg.get("_name_or_path", "custom"),
+            hidden_size=h,
+            num_layers=cfg.get("num_hidden_layers", 32),
+            num_heads=heads,
+            num_kv_heads=cfg.get("num_key_value_heads", heads),
+            intermediate_size=cfg.get("intermediate_size", 4 * h),
+            vocab_size=cfg.get("vocab_size", 32000),
+            head_dim=cfg.get("head_dim", h // heads),
+            rope_theta=cfg.get("rope_theta", 10000.0),
+            max_position=cfg.get("max_position_embeddings", 8192),
+            rms_eps=cfg.get("rms_norm_eps", 1e-5),
+            tie_word_embeddings=cfg.get("tie_word_embeddings", False),
+            partial_rotary_factor=cfg.get("partial_rotary_factor", 1.0),
+            attention_bias=cfg.get("attention_bias", False),
+            num_experts=cfg.get("num_local_experts",
+                                cfg.get("n_routed_experts", 0)) or 0,
+            num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
+            moe_intermediate_size=cfg.get("moe_intermediate_size", 0),
+            runtime="native" if native else "transformers",
+        )
 
     @property
     def rotary_dim(self) -> int:

@@ -81,6 +81,137 @@ register(ModelConfig(
     rope_theta=1000000.0, max_position=32768, num_experts=8,
     num_experts_per_tok=2, moe_intermediate_size=14336))
 
+# ---- instruct aliases (reference preset names, supported_models.yaml) ------
+def _alias(new: str, base: str, **over) -> None:
+    import dataclasses
+    register(dataclasses.replace(_REGISTRY[base], name=new, **over))
+
+
+_alias("llama-3.1-8b-instruct", "llama-3.1-8b")
+_alias("llama-3.3-70b-instruct", "llama-3.3-70b")
+_alias("mistral-7b-instruct", "mistral-7b")
+_alias("deepseek-r1-distill-llama-8b", "llama-3.1-8b")
+
+# ---- remaining reference catalog (31 curated presets). Architectures are
+# public model specs; presets whose architecture falls outside the HIP
+# engine's llama-family support matrix (parallel-block falcon, gemma-3
+# norm/sliding-window, deepseek MLA, gpt-oss attention sinks) carry
+# runtime="transformers" — the same native/fallback runtime split the
+# reference makes between its vLLM and text-generation runtimes. ----------
+
+# Phi family (remaining)
+register(ModelConfig(
+    name="phi-3-mini-128k-instruct", hidden_size=3072, num_layers=32,
+    num_heads=32, num_kv_heads=32, intermediate_size=8192, vocab_size=32064,
+    head_dim=96, rope_theta=10000.0, max_position=131072))
+register(ModelConfig(
+    name="phi-3-medium-4k-instruct", hidden_size=5120, num_layers=40,
+    num_heads=40, num_kv_heads=10, intermediate_size=17920, vocab_size=32064,
+    head_dim=128, rope_theta=10000.0, max_position=4096))
+register(ModelConfig(
+    name="phi-3-medium-128k-instruct", hidden_size=5120, num_layers=40,
+    num_heads=40, num_kv_heads=10, intermediate_size=17920, vocab_size=32064,
+    head_dim=128, rope_theta=10000.0, max_position=131072))
+register(ModelConfig(
+    name="phi-3.5-mini-instruct", hidden_size=3072, num_layers=32,
+    num_heads=32, num_kv_heads=32, intermediate_size=8192, vocab_size=32064,
+    head_dim=96, rope_theta=10000.0, max_position=131072))
+register(ModelConfig(
+    name="phi-4", hidden_size=5120, num_layers=40, num_heads=40,
+    num_kv_heads=10, intermediate_size=17920, vocab_size=100352,
+    head_dim=128, rope_theta=250000.0, max_position=16384))
+register(ModelConfig(
+    name="phi-2", hidden_size=2560, num_layers=32, num_heads=32,
+    num_kv_heads=32, intermediate_size=10240, vocab_size=51200, head_dim=80,
+    rope_theta=10000.0, max_position=2048, partial_rotary_factor=0.4,
+    runtime="transformers"))  # parallel-block + GELU MLP (non-llama arch)
+
+# Mistral family (remaining); ministral-3 dims from published model cards
+register(ModelConfig(
+    name="ministral-3-3b-instruct", hidden_size=3072, num_layers=26,
+    num_heads=24, num_kv_heads=8, intermediate_size=8192, vocab_size=131072,
+    head_dim=128, rope_theta=1000000.0, max_position=131072))
+register(ModelConfig(
+    name="ministral-3-8b-instruct", hidden_size=4096, num_layers=34,
+    num_heads=32, num_kv_heads=8, intermediate_size=12288, vocab_size=131072,
+    head_dim=128, rope_theta=1000000.0, max_position=131072))
+register(ModelConfig(
+    name="ministral-3-14b-instruct", hidden_size=5120, num_layers=40,
+    num_heads=40, num_kv_heads=8, intermediate_size=16384, vocab_size=131072,
+    head_dim=128, rope_theta=1000000.0, max_position=131072))
+register(ModelConfig(
+    name="mistral-large-3-675b-instruct", hidden_size=7168, num_layers=61,
+    num_heads=128, num_kv_heads=128, intermediate_size=18432,
+    vocab_size=131072, head_dim=64, rope_theta=1000000.0,
+    max_position=131072, num_experts=256, num_experts_per_tok=8,
+    moe_intermediate_size=2048, runtime="transformers"))  # MLA-class MoE
+
+# Qwen family (remaining)
+register(ModelConfig(
+    name="qwen2.5-coder-7b-instruct", hidden_size=3584, num_layers=28,
+    num_heads=28, num_kv_heads=4, intermediate_size=18944, vocab_size=152064,
+    head_dim=128, rope_theta=1000000.0, max_position=32768,
+    attention_bias=True))
+register(ModelConfig(
+    name="qwen2.5-coder-32b-instruct", hidden_size=5120, num_layers=64,
+    num_heads=40, num_kv_heads=8, intermediate_size=27648, vocab_size=152064,
+    head_dim=128, rope_theta=1000000.0, max_position=32768,
+    attention_bias=True))
+register(ModelConfig(
+    name="deepseek-r1-distill-qwen-14b", hidden_size=5120, num_layers=48,
+    num_heads=40, num_kv_heads=8, intermediate_size=13824, vocab_size=152064,
+    head_dim=128, rope_theta=1000000.0, max_position=131072,
+    attention_bias=True))
+
+# Falcon family — multi-query attention + parallel attn/MLP block → fallback
+register(ModelConfig(
+    name="falcon-7b", hidden_size=4544, num_layers=32, num_heads=71,
+    num_kv_heads=1, intermediate_size=18176, vocab_size=65024, head_dim=64,
+    rope_theta=10000.0, max_position=2048, tie_word_embeddings=True,
+    runtime="transformers"))
+_alias("falcon-7b-instruct", "falcon-7b")
+register(ModelConfig(
+    name="falcon-40b", hidden_size=8192, num_layers=60, num_heads=128,
+    num_kv_heads=8, intermediate_size=32768, vocab_size=65024, head_dim=64,
+    rope_theta=10000.0, max_position=2048, tie_word_embeddings=True,
+    runtime="transformers"))
+_alias("falcon-40b-instruct", "falcon-40b")
+
+# Gemma-3 — RMSNorm(+1)/pre-post norms/sliding window → fallback runtime
+register(ModelConfig(
+    name="gemma-3-4b-instruct", hidden_size=2560, num_layers=34, num_heads=8,
+    num_kv_heads=4, intermediate_size=10240, vocab_size=262208, head_dim=256,
+    rope_theta=1000000.0, max_position=131072, tie_word_embeddings=True,
+    runtime="transformers"))
+register(ModelConfig(
+    name="gemma-3-27b-instruct", hidden_size=5376, num_layers=62,
+    num_heads=32, num_kv_heads=16, intermediate_size=21504,
+    vocab_size=262208, head_dim=128, rope_theta=1000000.0,
+    max_position=131072, tie_word_embeddings=True, runtime="transformers"))
+
+# gpt-oss MoE — attention sinks + sliding window → fallback runtime
+register(ModelConfig(
+    name="gpt-oss-20b", hidden_size=2880, num_layers=24, num_heads=64,
+    num_kv_heads=8, intermediate_size=2880, vocab_size=201088, head_dim=64,
+    rope_theta=150000.0, max_position=131072, num_experts=32,
+    num_experts_per_tok=4, moe_intermediate_size=2880,
+    runtime="transformers"))
+register(ModelConfig(
+    name="gpt-oss-120b", hidden_size=2880, num_layers=36, num_heads=64,
+    num_kv_heads=8, intermediate_size=2880, vocab_size=201088, head_dim=64,
+    rope_theta=150000.0, max_position=131072, num_experts=128,
+    num_experts_per_tok=4, moe_intermediate_size=2880,
+    runtime="transformers"))
+
+# DeepSeek V3/R1 — MLA attention + 256-expert MoE → fallback runtime
+register(ModelConfig(
+    name="deepseek-v3-0324", hidden_size=7168, num_layers=61, num_heads=128,
+    num_kv_heads=128, intermediate_size=18432, vocab_size=129280,
+    head_dim=64, rope_theta=10000.0, max_position=131072, num_experts=256,
+    num_experts_per_tok=8, moe_intermediate_size=2048,
+    runtime="transformers"))
+_alias("deepseek-r1-0528", "deepseek-v3-0324")
+
 # ---- tiny configs for tests ------------------------------------------------
 register(ModelConfig(
     name="tiny-llama-test", hidden_size=256, num_layers=2, num_heads=4,

@@ -76,8 +76,18 @@ def main(argv=None):
     from .tokenizer import load_tokenizer
     from . import metrics
 
-    init_parallel(tp_size=args.tensor_parallel_size)
     mc = get_model_config(args.model)
+    if mc.runtime == "transformers":
+        # non-llama-family architecture (falcon/gemma-3/deepseek-MLA/
+        # gpt-oss): serve via the fallback runtime — the reference's
+        # vLLM vs text-generation runtime split (supported_models.yaml
+        # `runtime: tfs` rows).
+        from .transformers_runtime import main as tfs_main
+        return tfs_main(["--model", args.model, "--port", str(args.port),
+                         "--host", args.host] +
+                        (["--weights-path", args.weights_path]
+                         if args.weights_path else []))
+    init_parallel(tp_size=args.tensor_parallel_size)
     max_len = None if str(args.max_model_len) == "auto" else int(args.max_model_len)
     cfg = EngineConfig(
         model=mc,

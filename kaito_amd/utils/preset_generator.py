@@ -73,3 +73,62 @@ def generate_preset_metadata(config_path: str,
         total_param_bytes=param_bytes,
         disk_storage_gib=disk_storage_gib(param_bytes),
         model_token_limit=limit)
+
+
+def catalog_row(model_dir: str, name: str = "",
+                runtime: str = "") -> dict:
+    """One model-catalog row (reference: model_catalog.yaml entries), from
+    a local model directory. Prefers the safetensors index's total_size
+    over the parameter-count estimate."""
+    from ..engine.config import ModelConfig
+    cfg_path = os.path.join(model_dir, "config.json")
+    with open(cfg_path) as f:
+        hf = json.load(f)
+    mc = ModelConfig.from_hf_config(hf, name=name)
+    size = None
+    idx = os.path.join(model_dir, "model.safetensors.index.json")
+    if os.path.exists(idx):
+        with open(idx) as f:
+            size = json.load(f).get("metadata", {}).get("total_size")
+    md = generate_preset_metadata(cfg_path, name=name or mc.name,
+                                  param_bytes=size)
+    return {
+        "name": md.name,
+        "runtime": runtime or mc.runtime,
+        "architecture": (hf.get("architectures") or ["unknown"])[0],
+        "hiddenSize": md.hidden_size,
+        "numLayers": md.num_layers,
+        "numKeyValueHeads": md.num_kv_heads,
+        "headDim": md.head_dim,
+        "bytesPerToken": md.bytes_per_token,
+        "modelTokenLimit": md.model_token_limit,
+        "totalFileSizeBytes": md.total_param_bytes,
+        "diskStorageRequirementGiB": md.disk_storage_gib,
+    }
+
+
+def main(argv=None):
+    """CLI: python -m kaito_amd.utils.preset_generator <model_dir>
+    [--name N] [--runtime native|transformers] [--append-to catalog.json]"""
+    import argparse
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("model_dir")
+    p.add_argument("--name", default="")
+    p.add_argument("--runtime", default="",
+                   choices=["", "native", "transformers"])
+    p.add_argument("--append-to", default=None)
+    args = p.parse_args(argv)
+    row = catalog_row(args.model_dir, args.name, args.runtime)
+    print(json.dumps(row, indent=2))
+    if args.append_to:
+        cat = []
+        if os.path.exists(args.append_to):
+            with open(args.append_to) as f:
+                cat = json.load(f)
+        cat = [r for r in cat if r["name"] != row["name"]] + [row]
+        with open(args.append_to, "w") as f:
+            json.dump(cat, f, indent=2)
+
+
+if __name__ == "__main__":
+    main()

@@ -286,3 +286,70 @@ def test_tuning_metrics_server():
     assert c.get("/health").json()["status"] == "ok"
     body = c.get("/metrics").text
     assert "tuning_cpu_percent" in body or body.strip() == ""
+
+
+def test_full_preset_catalog():
+    """The 31 curated reference presets (supported_models.yaml) all resolve
+    to ModelConfigs with sane planner inputs."""
+    from kaito_amd.models import get_model_config
+    reference_presets = [
+        "llama-3.1-8b-instruct", "llama-3.3-70b-instruct",
+        "deepseek-r1-0528", "deepseek-v3-0324",
+        "falcon-7b", "falcon-7b-instruct", "falcon-40b",
+        "falcon-40b-instruct",
+        "mistral-7b", "mistral-7b-instruct",
+        "ministral-3-3b-instruct", "ministral-3-8b-instruct",
+        "ministral-3-14b-instruct", "mistral-large-3-675b-instruct",
+        "phi-2", "phi-3-mini-4k-instruct", "phi-3-mini-128k-instruct",
+        "phi-3-medium-4k-instruct", "phi-3-medium-128k-instruct",
+        "phi-3.5-mini-instruct", "phi-4-mini-instruct", "phi-4",
+        "qwen2.5-coder-7b-instruct", "qwen2.5-coder-32b-instruct",
+        "deepseek-r1-distill-qwen-14b", "deepseek-r1-distill-llama-8b",
+        "gemma-3-4b-instruct", "gemma-3-27b-instruct",
+        "gpt-oss-20b", "gpt-oss-120b",
+    ]
+    for name in reference_presets:
+        mc = get_model_config(name)
+        assert mc.param_bytes() > 1 << 30, name
+        assert mc.kv_bytes_per_token() > 0, name
+        assert mc.runtime in ("native", "transformers"), name
+    # spot-check planner-relevant sizes (bf16 weights)
+    g = (1 << 30)
+    # falcon MLP is ungated (2 matrices); the 3-matrix formula
+    # overestimates — safe direction for node planning
+    assert 80 * g < get_model_config("falcon-40b").param_bytes() < 120 * g
+    assert 1150 * g < get_model_config("deepseek-v3-0324").param_bytes() \
+        < 1500 * g
+    # archs outside the HIP engine's llama-family matrix use the fallback
+    assert get_model_config("gemma-3-27b-instruct").runtime == "transformers"
+    assert get_model_config("phi-4").runtime == "native"
+
+
+def test_from_hf_config_and_catalog_row(tmp_path):
+    import json
+    from kaito_amd.engine.config import ModelConfig
+    from kaito_amd.utils.preset_generator import catalog_row, main as gen_main
+    hf = {"architectures": ["LlamaForCausalLM"], "hidden_size": 4096,
+          "num_hidden_layers": 32, "num_attention_heads": 32,
+          "num_key_value_heads": 8, "intermediate_size": 14336,
+          "vocab_size": 128256, "max_position_embeddings": 8192,
+          "rope_theta": 500000.0, "torch_dtype": "bfloat16"}
+    mc = ModelConfig.from_hf_config(hf, name="l3")
+    assert mc.runtime == "native" and mc.num_kv_heads == 8
+    assert mc.kv_bytes_per_token() == 131072
+    falcon = ModelConfig.from_hf_config(
+        {"architectures": ["FalconForCausalLM"], "hidden_size": 4544,
+         "num_attention_heads": 71, "num_hidden_layers": 32}, name="f7")
+    assert falcon.runtime == "transformers"
+    d = tmp_path / "m"
+    d.mkdir()
+    (d / "config.json").write_text(json.dumps(hf))
+    (d / "model.safetensors.index.json").write_text(
+        json.dumps({"metadata": {"total_size": 16 * (1 << 30)}}))
+    row = catalog_row(str(d), name="l3")
+    assert row["totalFileSizeBytes"] == 16 * (1 << 30)
+    assert row["bytesPerToken"] == 131072
+    assert row["diskStorageRequirementGiB"] == 90  # 16*2.5+48=88 → 90
+    cat = tmp_path / "catalog.json"
+    gen_main([str(d), "--name", "l3", "--append-to", str(cat)])
+    assert json.loads(cat.read_text())[0]["name"] == "l3"
