@@ -113,6 +113,12 @@ class EngineConfig:
     max_lora_rank: int = 64
     kv_offload: bool = False                # LMCache-style host KV cache
     kv_offload_bytes: Optional[int] = None  # None → 0.5 * available RAM
+    # mixed steps: decode every step, with a bounded prefill chunk run
+    # CONCURRENTLY on a side HIP stream (prefill MFMA GEMMs overlap the
+    # decode step's bandwidth-bound paged attention). Pure-prefill steps
+    # still use the full max_num_batched_tokens budget.
+    enable_mixed_batch: bool = True
+    mixed_prefill_tokens: int = 2048        # per-step overlap prefill budget
     device: str = "cuda"
     seed: int = 0
     # decode graph buckets (batch sizes to capture)

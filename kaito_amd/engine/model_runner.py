@@ -235,23 +235,29 @@ class ModelRunner:
             self.model.compute_logits(hidden)
             torch.cuda.synchronize()
         self._buf["seq_lens"].zero_()
-        # prefill shapes: chunked prefill pins M at max_num_batched_tokens
+        # prefill shapes: pure-prefill steps pin M at max_num_batched_tokens;
+        # mixed (overlapped) steps run chunks around mixed_prefill_tokens.
         # (the untuned default heuristic costs ~1/3 of bench time there)
-        M = self.cfg.max_num_batched_tokens
         dev = self.device
         mcfg = self.cfg.model
         from ..models.llama import AttnMetadata
-        meta = AttnMetadata(
-            is_prefill=True,
-            slot_mapping=torch.full((M,), -1, dtype=torch.long, device=dev),
-            cu_seqlens=torch.arange(0, M + 1, 256, dtype=torch.int32,
-                                    device=dev),
-            max_seqlen=256)
-        ids = torch.randint(0, mcfg.vocab_size, (M,), device=dev)
-        pos = torch.arange(M, device=dev) % 256
-        hidden = self.model(ids, pos, None, meta)
-        self.model.compute_logits(hidden[:64])
-        torch.cuda.synchronize()
+        mp = self.cfg.mixed_prefill_tokens
+        shapes = {self.cfg.max_num_batched_tokens, mp, 4096, 2000, 1024,
+                  512, 256}
+        for M in sorted(shapes, reverse=True):
+            step = min(M, 256)
+            cu = list(range(0, M, step)) + [M]   # covers a non-multiple tail
+            meta = AttnMetadata(
+                is_prefill=True,
+                slot_mapping=torch.full((M,), -1, dtype=torch.long,
+                                        device=dev),
+                cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+                max_seqlen=step)
+            ids = torch.randint(0, mcfg.vocab_size, (M,), device=dev)
+            pos = torch.arange(M, device=dev) % step
+            hidden = self.model(ids, pos, None, meta)
+            self.model.compute_logits(hidden[:64])
+            torch.cuda.synchronize()
         tunable.tuning_enable(False)
         logger.info("TunableOp tuned; results flush to %s at exit", dest)
 

@@ -88,14 +88,26 @@ class Scheduler:
             return batch
         return self._schedule_decode()
 
-    def _schedule_prefill(self) -> Optional[ScheduledBatch]:
+    def schedule_mixed(self, prefill_budget: int):
+        """Decode-priority mixed scheduling: a decode batch over all running
+        sequences PLUS a bounded prefill chunk the engine overlaps on a side
+        stream. Falls back to full-budget pure prefill when nothing is
+        decoding (startup burst)."""
+        decode = self._schedule_decode()
+        if decode is None:
+            return None, self._schedule_prefill()
+        return decode, self._schedule_prefill(budget=prefill_budget)
+
+    def _schedule_prefill(self, budget: Optional[int] = None
+                          ) -> Optional[ScheduledBatch]:
         """Chunked prefill: each step processes up to max_num_batched_tokens
         of prompt tokens; long prompts span multiple steps (context
         attention handles suffix chunks). New sequences are admitted after
         in-flight prefills continue."""
         if not self.waiting and not self.prefilling:
             return None
-        budget = self.cfg.max_num_batched_tokens
+        if budget is None:
+            budget = self.cfg.max_num_batched_tokens
         chunks: List[PrefillChunk] = []
 
         # 1. continue partially-prefilled sequences

@@ -288,3 +288,59 @@ def test_engine_stochastic_sampling_paths():
         assert len(seq.output_token_ids) == 6
         assert all(0 <= t < cfg.model.vocab_size
                    for t in seq.output_token_ids)
+
+
+def test_mixed_overlap_steps_match_classic():
+    """Mixed (decode + concurrent prefill chunk) steps must produce exactly
+    the same greedy outputs as the classic either/or stepping, including
+    while requests join mid-decode (the overlap scheduling changes WHEN a
+    prompt is prefilled, never what its sequence decodes to)."""
+    prompts = [[3, 14, 15, 92, 65], [35, 89, 79], list(range(40, 70)),
+               [7, 11, 13, 17], [21, 22, 23, 24, 25, 26]]
+    sp = SamplingParams(max_tokens=7, ignore_eos=True)
+
+    def run(mixed: bool):
+        cfg = _cfg(enable_mixed_batch=mixed, mixed_prefill_tokens=8,
+                   seed=7)
+        eng = LLMEngine(cfg)
+        ids = [eng.add_request(prompts[0], sp), eng.add_request(prompts[1], sp)]
+        eng.step()      # prefill burst
+        eng.step()      # decode (mixed: none waiting → pure decode)
+        ids.append(eng.add_request(prompts[2], sp))   # joins mid-decode
+        eng.step()
+        ids += [eng.add_request(p, sp) for p in prompts[3:]]
+        while eng.has_unfinished():
+            eng.step()
+        return [eng.seqs[i].output_token_ids for i in ids]
+
+    assert run(True) == run(False)
+
+
+def test_mixed_step_schedules_decode_and_prefill_together():
+    cfg = _cfg(mixed_prefill_tokens=16)
+    eng = LLMEngine(cfg)
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+    a = eng.add_request([3, 14, 15], sp)
+    eng.step()                      # prefill a (pure: nothing decoding)
+    b = eng.add_request([5, 6, 7, 8], sp)
+    d, p = eng.scheduler.schedule_mixed(cfg.mixed_prefill_tokens)
+    assert d is not None and not d.is_prefill
+    assert [s.seq_id for s in d.seqs] == [a]
+    assert p is not None and p.is_prefill
+    assert [s.seq_id for s in p.seqs] == [b]
+
+
+def test_mixed_preemption_recovers():
+    """Preemption inside mixed stepping: epoch guard drops stale pending
+    tokens and the victim re-runs to the same greedy output."""
+    cfg = _cfg(max_num_seqs=3, num_gpu_blocks=5, mixed_prefill_tokens=32)
+    eng = LLMEngine(cfg)
+    sp = SamplingParams(max_tokens=22, ignore_eos=True)
+    prompts = [[3, 14, 15, 92, 65], [35, 89, 79, 32], [11, 12, 13]]
+    ids = [eng.add_request(p, sp) for p in prompts]
+    while eng.has_unfinished():
+        eng.step()
+    assert max(s.epoch for s in eng.seqs.values()) > 0, "no preemption hit"
+    for i, p in zip(ids, prompts):
+        assert eng.seqs[i].output_token_ids == _naive_generate(
+            eng.runner.model, cfg, p, 22)
