@@ -32,6 +32,20 @@ class _PendingStep:
     host: Optional[torch.Tensor]         # pinned copy (GPU) or None (CPU)
     event: Optional[object]              # cuda Event or None
     index: Dict[int, int]                # seq_id -> batch index
+    # logprob parts: (host_vals [n,K+1], host_ids [n,K+1], start_row) per
+    # sampling call that produced logprobs (decode part / prefill part)
+    lp_parts: List[tuple] = None
+
+
+@dataclass
+class _PendingPrefill:
+    """A fire-and-forget prefill chunk in flight on the side stream."""
+    seqs: List[Sequence]                 # sampling (prompt-completing) seqs
+    epochs: List[int]
+    tokens: Optional[torch.Tensor]       # first sampled tokens (device)
+    host: Optional[torch.Tensor]         # pinned copy (GPU) or None
+    event: Optional[object]              # side-stream cuda Event or None
+    lp_part: Optional[tuple]             # (host_vals, host_ids) or None
 
 
 class LLMEngine:
@@ -59,6 +73,7 @@ class LLMEngine:
         self._next_id = itertools.count()
         self.seqs: Dict[int, Sequence] = {}
         self._pending: Optional[_PendingStep] = None
+        self._prefill_pend: Optional[_PendingPrefill] = None
         # counters for /metrics (names consumed by the benchmark probe)
         self.num_generation_tokens = 0
         self.num_prompt_tokens = 0
@@ -92,6 +107,10 @@ class LLMEngine:
         seq = self.seqs.pop(seq_id, None)
         if seq is None:
             return
+        pp = self._prefill_pend
+        if pp is not None and any(s2.seq_id == seq_id for s2 in pp.seqs) \
+            and pp.event is not None:
+            pp.event.synchronize()   # rare: in-flight prefill being aborted
         if seq.status == SeqStatus.RUNNING:
             self.scheduler.finish(seq)
         elif seq in self.scheduler.waiting:
@@ -100,7 +119,8 @@ class LLMEngine:
         seq.finish_reason = "abort"
 
     def has_unfinished(self) -> bool:
-        return self.scheduler.has_work() or self._pending is not None
+        return self.scheduler.has_work() or self._pending is not None \
+            or self._prefill_pend is not None
 
     # ------------------------------------------------------------- stepping
     #
@@ -133,6 +153,35 @@ class LLMEngine:
             self.kv_publisher.block_stored(blocks)
         return samp
 
+    def _sample_lp(self, logits, seqs):
+        """sampler.sample with logprob staging: returns (tokens, part) where
+        part = (host_vals, host_ids) copied async, or None."""
+        tokens, lpv, lpi = self.sampler.sample(logits, seqs,
+                                               return_logprobs=True)
+        part = None
+        if lpv is not None:
+            if lpv.is_cuda:
+                hv = torch.empty_like(lpv, device="cpu", pin_memory=True)
+                hi = torch.empty_like(lpi, device="cpu", pin_memory=True)
+                hv.copy_(lpv, non_blocking=True)
+                hi.copy_(lpi, non_blocking=True)
+                part = (hv, hi)
+            else:
+                part = (lpv, lpi)
+        return tokens, part
+
+    @staticmethod
+    def _lp_rows(parts, n):
+        """Expand pending lp_parts into one per-row list (None where the
+        row's step produced no logprobs)."""
+        rows = [None] * n
+        for hv, hi, start in parts or []:
+            v = hv.tolist()
+            i = hi.tolist()
+            for r in range(len(v)):
+                rows[start + r] = list(zip(i[r], v[r]))
+        return rows
+
     def _side_stream(self):
         s = getattr(self, "_side", None)
         if s is None:
@@ -142,10 +191,12 @@ class LLMEngine:
     @torch.no_grad()
     def _step_mixed(self) -> List[Sequence]:
         """Mixed step: decode over all running seqs on the main stream
-        (hipGraph), with a bounded prefill chunk launched CONCURRENTLY on a
-        side stream. Prefill's MFMA-bound GEMMs overlap decode's
-        bandwidth-bound paged attention — on the bench workload pure
-        prefill steps are ~25-30% of wall time, and this hides them.
+        (hipGraph), with a bounded prefill chunk launched FIRE-AND-FORGET
+        on a side stream. Prefill's MFMA-bound GEMMs overlap decode's
+        bandwidth-bound paged attention; the decode chain NEVER waits on
+        the side stream — prefill results resolve via a non-blocking
+        event.query() on a later step, and a prefilled sequence only joins
+        the decode set after its event fired (its KV writes are complete).
 
         Ordering: the side stream waits on an event recorded BEFORE this
         step's decode launch. That event orders the prefill after (a) the
@@ -153,21 +204,29 @@ class LLMEngine:
         been freed by a finish/preemption and re-allocated to a prefill
         seq this step — and (b) this step's prefix-restore H2D copies,
         without serializing it after this step's decode."""
-        d_batch, p_batch = self.scheduler.schedule_mixed(
-            self.cfg.mixed_prefill_tokens)
-        if d_batch is None and p_batch is None:
-            return self._resolve_pending()
         finished: List[Sequence] = []
+        finished += self._resolve_prefill(block=False)
+        allow_prefill = self._prefill_pend is None
+        d_batch, p_batch = self.scheduler.schedule_mixed(
+            self.cfg.mixed_prefill_tokens, allow_prefill)
+        if d_batch is None and p_batch is None:
+            if self._prefill_pend is not None:
+                finished += self._resolve_prefill(block=True)
+                return finished          # next step schedules the promotees
+            return finished + self._resolve_pending()
         if d_batch is None:
             # startup burst: classic synchronous full-budget prefill
+            finished += self._resolve_prefill(block=True)
             finished += self._resolve_pending()
             hidden = self.runner.execute_prefill(p_batch.chunks)
             samp = self._prefill_bookkeep(p_batch)
             if not samp:
                 return finished
-            tokens = self._sample_maybe_pp(hidden, samp)
+            tokens, lp = self._sample_prefill_pp(hidden, samp)
             finished += self._commit(samp, tokens.tolist(),
-                                     [s.epoch for s in samp])
+                                     [s.epoch for s in samp],
+                                     self._lp_rows([lp + (0,)], len(samp))
+                                     if lp is not None else None)
             return finished
 
         use_streams = p_batch is not None and self.runner.is_gpu
@@ -175,41 +234,16 @@ class LLMEngine:
             pre_ev = torch.cuda.Event()
             pre_ev.record()
 
+        if any(s.sampling.needs_history for s in d_batch.seqs):
+            # penalties need exact token history: drain the pipeline
+            finished += self._resolve_pending()
         pend = self._pending
         sampled = pend.tokens if pend is not None else None
         pending_map = (pend.tokens, pend.index) if pend is not None else None
         logits = self.runner.execute(d_batch, sampled, pending_map)
-        tokens = self.sampler.sample(logits, d_batch.seqs)
+        tokens, lp_d = self._sample_lp(logits, d_batch.seqs)
         for seq in d_batch.seqs:
             seq.sched_len = seq.sched_tokens + 1
-        seqs_all = list(d_batch.seqs)
-        parts = [tokens]
-
-        if p_batch is not None:
-            if use_streams:
-                side = self._side_stream()
-                side.wait_event(pre_ev)
-                with torch.cuda.stream(side):
-                    hidden = self.runner.execute_prefill(p_batch.chunks)
-                    samp_pre = p_batch.sampling_seqs
-                    tok_p = self._sample_maybe_pp(hidden, samp_pre) \
-                        if samp_pre else None
-                if tok_p is not None:
-                    ev = torch.cuda.Event()
-                    ev.record(side)
-                    torch.cuda.current_stream().wait_event(ev)
-                    tok_p.record_stream(torch.cuda.current_stream())
-            else:
-                hidden = self.runner.execute_prefill(p_batch.chunks)
-                samp_pre = p_batch.sampling_seqs
-                tok_p = self._sample_maybe_pp(hidden, samp_pre) \
-                    if samp_pre else None
-            samp = self._prefill_bookkeep(p_batch)
-            if tok_p is not None:
-                seqs_all += samp
-                parts.append(tok_p)
-
-        tokens = torch.cat(parts) if len(parts) > 1 else parts[0]
         host_copy = None
         if tokens.is_cuda:
             host_copy = self._pinned(len(tokens))
@@ -219,15 +253,80 @@ class LLMEngine:
         else:
             event = None
         new_pend = _PendingStep(
-            seqs=seqs_all,
-            epochs=[s.epoch for s in seqs_all],
+            seqs=list(d_batch.seqs),
+            epochs=[s.epoch for s in d_batch.seqs],
             tokens=tokens,
             host=host_copy,
             event=event,
-            index={s.seq_id: i for i, s in enumerate(seqs_all)})
+            index={s.seq_id: i for i, s in enumerate(d_batch.seqs)},
+            lp_parts=[(lp_d[0], lp_d[1], 0)] if lp_d is not None else [])
+
+        if p_batch is not None:
+            self._launch_prefill(p_batch, pre_ev if use_streams else None)
+
         finished += self._resolve_pending()
         self._pending = new_pend
         return finished
+
+    def _launch_prefill(self, p_batch, pre_ev) -> None:
+        """Fire-and-forget prefill chunk on the side stream."""
+        samp_pre = p_batch.sampling_seqs
+        if pre_ev is not None:
+            side = self._side_stream()
+            side.wait_event(pre_ev)
+            with torch.cuda.stream(side):
+                hidden = self.runner.execute_prefill(p_batch.chunks)
+                tok_p = lp_p = host = None
+                if samp_pre:
+                    logits_p = self.runner.model.compute_logits(hidden)
+                    tok_p, lp_p = self._sample_lp(logits_p, samp_pre)
+                    host = torch.empty(len(samp_pre), dtype=torch.long,
+                                       pin_memory=True)
+                    host.copy_(tok_p, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record(side)
+        else:
+            hidden = self.runner.execute_prefill(p_batch.chunks)
+            tok_p = lp_p = host = None
+            ev = None
+            if samp_pre:
+                logits_p = self.runner.model.compute_logits(hidden)
+                tok_p, lp_p = self._sample_lp(logits_p, samp_pre)
+        # bookkeeping at launch: advance chunks, count prompt tokens;
+        # PROMOTION to the decode set waits for the event (resolve)
+        self.scheduler.advance_prefill_chunks(p_batch)
+        for seq in samp_pre:
+            seq.sched_len = seq.num_prompt_tokens + 1
+            self.num_prompt_tokens += seq.num_prompt_tokens
+        if self.kv_publisher is not None and samp_pre:
+            blocks = [b for s2 in samp_pre for b in s2.block_table]
+            self.kv_publisher.block_stored(blocks)
+        self._prefill_pend = _PendingPrefill(
+            seqs=list(samp_pre),
+            epochs=[s.epoch for s in samp_pre],
+            tokens=tok_p, host=host, event=ev,
+            lp_part=lp_p)
+
+    def _resolve_prefill(self, block: bool) -> List[Sequence]:
+        """Resolve the in-flight prefill chunk if its event fired (or
+        always, when block=True): commit first tokens and promote the
+        sequences into the decode set."""
+        p = self._prefill_pend
+        if p is None:
+            return []
+        if p.event is not None:
+            if not block and not p.event.query():
+                return []
+            p.event.synchronize()
+            vals = p.host.tolist() if p.host is not None else []
+        else:
+            vals = p.tokens.tolist() if p.tokens is not None else []
+        self._prefill_pend = None
+        lps = self._lp_rows([p.lp_part + (0,)], len(p.seqs)) \
+            if p.lp_part is not None else None
+        out = self._commit(p.seqs, vals, p.epochs, lps)
+        self.scheduler.promote_prefilled(p.seqs)
+        return out
 
     @torch.no_grad()
     def _step_classic(self) -> List[Sequence]:
@@ -244,9 +343,11 @@ class LLMEngine:
             samp = self._prefill_bookkeep(batch)
             if not samp:
                 return finished          # all chunks partial: no sampling
-            tokens = self._sample_maybe_pp(hidden, samp)
+            tokens, lp = self._sample_prefill_pp(hidden, samp)
             finished += self._commit(samp, tokens.tolist(),
-                                     [s.epoch for s in samp])
+                                     [s.epoch for s in samp],
+                                     self._lp_rows([lp + (0,)], len(samp))
+                                     if lp is not None else None)
             return finished
 
         if self._pp:
@@ -290,6 +391,17 @@ class LLMEngine:
         finished += self._resolve_pending()
         self._pending = new_pend
         return finished
+
+    def _sample_prefill_pp(self, hidden, seqs):
+        """Prefill-completion sampling: logprobs on single-stage; PP
+        lockstep broadcasts token ids only (logprobs unsupported under
+        PP)."""
+        from ..parallel import state as ps
+        st = ps.get_state()
+        if st.pp_size == 1:
+            logits = self.runner.model.compute_logits(hidden)
+            return self._sample_lp(logits, seqs)
+        return self._sample_maybe_pp(hidden, seqs), None
 
     def _sample_maybe_pp(self, hidden_or_logits, seqs,
                          precomputed_logits: bool = False) -> torch.Tensor:
@@ -335,19 +447,27 @@ class LLMEngine:
             vals = p.host.tolist()
         else:
             vals = p.tokens.tolist()
-        return self._commit(p.seqs, vals, p.epochs)
+        lps = self._lp_rows(p.lp_parts, len(p.seqs)) if p.lp_parts else None
+        return self._commit(p.seqs, vals, p.epochs, lps)
 
     def flush(self) -> List[Sequence]:
-        """Drain the pipelined step (bench/end-of-stream)."""
-        return self._resolve_pending()
+        """Drain the pipelined step + in-flight prefill."""
+        return self._resolve_prefill(block=True) + self._resolve_pending()
 
     def _commit(self, seqs: List[Sequence], vals: List[int],
-                epochs: List[int]) -> List[Sequence]:
+                epochs: List[int],
+                lps: Optional[List] = None) -> List[Sequence]:
         finished: List[Sequence] = []
-        for seq, tok, ep in zip(seqs, vals, epochs):
+        for i, (seq, tok, ep) in enumerate(zip(seqs, vals, epochs)):
             if seq.epoch != ep or seq.status == SeqStatus.FINISHED:
                 continue  # preempted or already finished: drop stale token
             seq.append_token(int(tok))
+            k = seq.sampling.logprobs
+            if k and lps is not None and lps[i] is not None:
+                # row = top-K(batch max) pairs + the sampled token's own
+                # logprob as the last entry; trim to this seq's K
+                row = lps[i]
+                seq.output_logprobs.append(row[:k] + [row[-1]])
             self.num_generation_tokens += 1
             if seq.check_finished(self.eos_token_id):
                 if self.kv_offload is not None:

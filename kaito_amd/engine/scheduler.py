@@ -88,14 +88,18 @@ class Scheduler:
             return batch
         return self._schedule_decode()
 
-    def schedule_mixed(self, prefill_budget: int):
+    def schedule_mixed(self, prefill_budget: int,
+                       allow_prefill: bool = True):
         """Decode-priority mixed scheduling: a decode batch over all running
         sequences PLUS a bounded prefill chunk the engine overlaps on a side
-        stream. Falls back to full-budget pure prefill when nothing is
-        decoding (startup burst)."""
+        stream (allow_prefill gates it while a prior chunk is in flight).
+        Falls back to full-budget pure prefill when nothing is decoding
+        (startup burst)."""
         decode = self._schedule_decode()
         if decode is None:
             return None, self._schedule_prefill()
+        if not allow_prefill:
+            return decode, None
         return decode, self._schedule_prefill(budget=prefill_budget)
 
     def _schedule_prefill(self, budget: Optional[int] = None
@@ -110,11 +114,14 @@ class Scheduler:
             budget = self.cfg.max_num_batched_tokens
         chunks: List[PrefillChunk] = []
 
-        # 1. continue partially-prefilled sequences
+        # 1. continue partially-prefilled sequences (remaining == 0 means
+        # the seq's last chunk is in flight, awaiting promotion)
         for seq in list(self.prefilling):
             if budget <= 0:
                 break
             remaining = seq.num_prompt_tokens - seq.prefilled_len
+            if remaining <= 0:
+                continue
             n = min(remaining, budget)
             chunks.append(PrefillChunk(seq, seq.prefilled_len, n))
             budget -= n
@@ -152,11 +159,24 @@ class Scheduler:
 
     def finish_prefill_chunks(self, batch: ScheduledBatch) -> None:
         """Advance prefilled_len; move completed sequences to running."""
+        self.advance_prefill_chunks(batch)
+        self.promote_prefilled([c.seq for c in batch.chunks if c.completes])
+
+    def advance_prefill_chunks(self, batch: ScheduledBatch) -> None:
+        """Mark chunk tokens as (about to be) in the cache — called at
+        LAUNCH so the next step schedules the following chunk, not a
+        repeat."""
         for c in batch.chunks:
             c.seq.prefilled_len = c.start + c.length
-            if c.completes and c.seq in self.prefilling:
-                self.prefilling.remove(c.seq)
-                self.running.append(c.seq)
+
+    def promote_prefilled(self, seqs: List[Sequence]) -> None:
+        """Move fully-prefilled sequences into the decode set — called at
+        RESOLVE (fire-and-forget prefill: a seq may only decode after its
+        prefill KV writes are known complete on the GPU)."""
+        for s in seqs:
+            if s in self.prefilling:
+                self.prefilling.remove(s)
+                self.running.append(s)
 
     def _schedule_decode(self) -> Optional[ScheduledBatch]:
         if not self.running:

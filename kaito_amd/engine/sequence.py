@@ -16,6 +16,19 @@ class SamplingParams:
     stop_token_ids: tuple = ()
     ignore_eos: bool = False
     seed: Optional[int] = None
+    # OpenAI-style penalties over GENERATED tokens (presence/frequency) or
+    # prompt+generated (repetition). Using any forces synchronous sampling
+    # (exact token history; the pipelined step is drained first).
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0
+    # top-N logprobs returned per sampled token (None = off)
+    logprobs: Optional[int] = None
+
+    @property
+    def needs_history(self) -> bool:
+        return (self.presence_penalty != 0.0 or self.frequency_penalty != 0.0
+                or self.repetition_penalty != 1.0)
 
 
 class SeqStatus(enum.Enum):
@@ -46,6 +59,9 @@ class Sequence:
     # prompt tokens whose KV is in the cache (chunked prefill / prefix
     # restore); prompt fully prefilled when == num_prompt_tokens
     prefilled_len: int = 0
+    # per output token: [(token_id, logprob), ...] top-N, when
+    # sampling.logprobs is set
+    output_logprobs: List[list] = field(default_factory=list)
 
     @property
     def num_prompt_tokens(self) -> int:
