@@ -18,6 +18,11 @@ class RagConfig:
         self.remote_embedding_url = _env("REMOTE_EMBEDDING_URL")
         self.remote_embedding_access_secret = _env("REMOTE_EMBEDDING_ACCESS_SECRET")
         self.vector_db_type = _env("VECTOR_DB_TYPE", "faiss")
+        # faiss → local flat (GEMM + HIP top-k); ivf → local IVF-flat
+        # (coarse-quantized, nprobe lists); qdrant → remote server
+        self.ivf_nlist = int(_env("IVF_NLIST", "64"))
+        self.ivf_nprobe = int(_env("IVF_NPROBE", "8"))
+        self.ivf_min_train = int(_env("IVF_MIN_TRAIN", "256"))
         self.vector_db_url = _env("VECTOR_DB_URL")
         self.vector_db_access_secret = _env("VECTOR_DB_ACCESS_SECRET")
         self.llm_inference_url = _env("LLM_INFERENCE_URL",

@@ -66,9 +66,14 @@ def build_rag_app(cfg: Optional[RagConfig] = None, embedding=None,
     embedding = embedding or make_embedding(cfg)
     if manager is None:
         factory = None
-        if cfg.vector_db_type.lower() == "qdrant" and cfg.vector_db_url:
+        kind = cfg.vector_db_type.lower()
+        if kind == "qdrant" and cfg.vector_db_url:
             from .qdrant_store import make_index
             factory = lambda dim, name: make_index(cfg, dim, name)  # noqa: E731
+        elif kind == "ivf":
+            from .ivf import IVFFlatIndex
+            factory = lambda dim, name: IVFFlatIndex(  # noqa: E731
+                dim, cfg.ivf_nlist, cfg.ivf_nprobe, cfg.ivf_min_train)
         manager = VectorStoreManager(embedding, index_factory=factory)
     app = FastAPI(title="kaito-amd ragengine")
     app.state.manager = manager
