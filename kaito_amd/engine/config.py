@@ -111,6 +111,7 @@ class EngineConfig:
     enable_lora: bool = False
     max_loras: int = 8
     max_lora_rank: int = 64
+    enable_prefix_caching: bool = True      # block-hash APC (zero-copy)
     kv_offload: bool = False                # LMCache-style host KV cache
     kv_offload_bytes: Optional[int] = None  # None → 0.5 * available RAM
     # mixed steps: decode every step, with a bounded prefill chunk run

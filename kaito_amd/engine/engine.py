@@ -54,7 +54,12 @@ class LLMEngine:
         self.runner = ModelRunner(cfg).load_model(weights_path, cfg.seed)
         self.runner.setup_tunable()
         self.runner.profile_and_allocate_kv()
-        self.pool = BlockPool(self.runner.num_gpu_blocks, cfg.block_size)
+        if cfg.enable_prefix_caching:
+            from .block_pool import PrefixCachingPool
+            self.pool = PrefixCachingPool(self.runner.num_gpu_blocks,
+                                          cfg.block_size)
+        else:
+            self.pool = BlockPool(self.runner.num_gpu_blocks, cfg.block_size)
         self.kv_offload = None
         if cfg.kv_offload:
             from .kv_offload import KVOffloadManager
@@ -148,6 +153,7 @@ class LLMEngine:
         for seq in samp:
             seq.sched_len = seq.num_prompt_tokens + 1
             self.num_prompt_tokens += seq.num_prompt_tokens
+            self._register_prefix(seq)
         if self.kv_publisher is not None and samp:
             blocks = [b for s2 in samp for b in s2.block_table]
             self.kv_publisher.block_stored(blocks)
@@ -181,6 +187,14 @@ class LLMEngine:
             for r in range(len(v)):
                 rows[start + r] = list(zip(i[r], v[r]))
         return rows
+
+    def _register_prefix(self, seq) -> None:
+        """Publish a completed prompt's full blocks to the prefix cache."""
+        if hasattr(self.pool, "register_prefix") and seq.block_table:
+            bs = self.cfg.block_size
+            nfull = seq.num_prompt_tokens // bs
+            self.pool.register_prefix(seq.prompt_token_ids[:nfull * bs],
+                                      seq.block_table[:nfull])
 
     def _side_stream(self):
         s = getattr(self, "_side", None)
@@ -326,6 +340,8 @@ class LLMEngine:
             if p.lp_part is not None else None
         out = self._commit(p.seqs, vals, p.epochs, lps)
         self.scheduler.promote_prefilled(p.seqs)
+        for seq in p.seqs:
+            self._register_prefix(seq)
         return out
 
     @torch.no_grad()

@@ -134,9 +134,16 @@ class Scheduler:
             if not self.pool.can_allocate(need):
                 break
             self.waiting.popleft()
-            seq.block_table = self.pool.allocate(need)
+            shared = []
+            if hasattr(self.pool, "match_prefix"):
+                # automatic prefix caching: skip prefill for cached full
+                # prompt blocks (KV already resident, zero copies)
+                shared, covered = self.pool.match_prefix(
+                    seq.prompt_token_ids)
+                seq.prefilled_len = max(seq.prefilled_len, covered)
+            seq.block_table = shared + self.pool.allocate(need - len(shared))
             seq.status = SeqStatus.RUNNING
-            if self.restore_cb is not None:
+            if self.restore_cb is not None and seq.prefilled_len == 0:
                 seq.prefilled_len = self.restore_cb(seq)
             if seq.prefilled_len >= seq.num_prompt_tokens:
                 # full prefix-cache hit: straight to decode (the next decode

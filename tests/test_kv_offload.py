@@ -17,7 +17,10 @@ def _engine(**kw):
     base = dict(model=get_model_config("tiny-llama-test"), device="cpu",
                 max_num_seqs=8, num_gpu_blocks=64, enforce_eager=True,
                 max_model_len=128, kv_offload=True,
-                kv_offload_bytes=64 << 20)
+                kv_offload_bytes=64 << 20,
+                # offload-path tests: the block-hash prefix cache would
+                # intercept first (it takes priority at admission)
+                enable_prefix_caching=False)
     base.update(kw)
     return LLMEngine(EngineConfig(**base))
 
