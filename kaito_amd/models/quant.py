@@ -1,0 +1,115 @@
+"""W4A16 group quantization: pack/repack + the QuantLinear module.
+
+Covers the reference catalog's AWQ/GPTQ presets (SURVEY.md §2.3
+"quantized GEMM" row; supported_models.yaml qwen3-8b-awq etc.). Native
+layout is kernel-first (ops/csrc/w4a16.hip): qweight u32 [N, K/8] with 8
+consecutive K nibbles per word, fp32 scales/zeros [N, K/G], w = s*q - z.
+AWQ checkpoints repack into it at load time.
+"""
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from .. import ops
+
+# AWQ packs 8 nibbles along N in interleaved order (public awq format)
+AWQ_ORDER = (0, 2, 4, 6, 1, 3, 5, 7)
+
+
+def quantize_w4(weight: torch.Tensor, group: int = 128
+                ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Asymmetric 4-bit group quantization of [N, K] → native layout.
+    Returns (qweight u32 [N, K/8], scales f32 [N, K/G], zeros f32)."""
+    N, K = weight.shape
+    assert K % group == 0 and group % 8 == 0
+    w = weight.float().reshape(N, K // group, group)
+    wmax = w.amax(dim=-1)
+    wmin = w.amin(dim=-1)
+    scale = (wmax - wmin).clamp(min=1e-8) / 15.0
+    zq = (-wmin / scale).round().clamp(0, 15)
+    q = (w / scale.unsqueeze(-1) + zq.unsqueeze(-1)).round().clamp(0, 15)
+    q = q.reshape(N, K).to(torch.int64)
+    shifts = torch.arange(8, dtype=torch.int64) * 4
+    packed = (q.reshape(N, K // 8, 8) << shifts).sum(dim=-1)
+    qweight = packed.to(torch.int32)          # bit pattern == uint32
+    zeros = (scale * zq).float()              # z folded: w = s*q - z
+    return qweight, scale.float(), zeros
+
+
+def repack_awq(qweight_awq: torch.Tensor, qzeros_awq: torch.Tensor,
+               scales_awq: torch.Tensor, group: int = 128
+               ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Repack a public-AWQ checkpoint (qweight i32 [K, N/8] nibbles along N
+    in AWQ_ORDER; qzeros i32 [K/G, N/8]; scales [K/G, N]) into the native
+    layout."""
+    K, nw = qweight_awq.shape
+    N = nw * 8
+
+    def unpack_n(t: torch.Tensor) -> torch.Tensor:
+        # [R, N/8] i32 → [R, N] int in logical n order
+        cols = []
+        for j in AWQ_ORDER:
+            cols.append((t >> (4 * j)) & 0xF)
+        return torch.stack(cols, dim=-1).reshape(t.shape[0], N)
+
+    q = unpack_n(qweight_awq.long())            # [K, N]
+    zq = unpack_n(qzeros_awq.long())            # [K/G, N]
+    s = scales_awq.float()                      # [K/G, N]
+    qn = q.T.contiguous()                       # [N, K]
+    shifts = torch.arange(8, dtype=torch.int64) * 4
+    packed = (qn.reshape(N, K // 8, 8) << shifts).sum(dim=-1).to(torch.int32)
+    scales = s.T.contiguous()                   # [N, K/G]
+    zeros = (scales * zq.T.float())
+    return packed, scales, zeros
+
+
+class QuantLinear(torch.nn.Module):
+    """W4A16 linear: HIP GEMV for decode-sized M, dequant + MFMA GEMM
+    (hipBLASLt) beyond. Drop-in for a bias-free nn.Linear."""
+
+    GEMV_MAX_M = 32
+
+    def __init__(self, qweight, scales, zeros, group: int = 128):
+        super().__init__()
+        self.register_buffer("qweight", qweight)
+        self.register_buffer("scales", scales)
+        self.register_buffer("zeros", zeros)
+        self.group = group
+        self.out_features = qweight.size(0)
+        self.in_features = qweight.size(1) * 8
+
+    @classmethod
+    def from_float(cls, weight: torch.Tensor, group: int = 128):
+        return cls(*quantize_w4(weight, group), group)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        shape = x.shape[:-1]
+        x2 = x.reshape(-1, self.in_features)
+        if x2.size(0) <= self.GEMV_MAX_M:
+            y = ops.w4a16_gemv(x2, self.qweight, self.scales, self.zeros,
+                               self.group)
+        else:
+            w = ops.w4a16_dequant(self.qweight, self.scales, self.zeros,
+                                  self.group)
+            y = torch.nn.functional.linear(x2.to(w.dtype), w)
+        return y.reshape(*shape, self.out_features).to(x.dtype)
+
+
+def quantize_model_linears(model: torch.nn.Module, group: int = 128,
+                           suffixes=("qkv_proj", "o_proj", "gate_up_proj",
+                                     "down_proj")) -> int:
+    """Swap matching nn.Linear weights for QuantLinear (weight-only W4).
+    Returns the number of modules converted."""
+    n = 0
+    for parent in model.modules():
+        for name, child in list(parent.named_children()):
+            if any(name.endswith(sfx) for sfx in suffixes) and \
+                    isinstance(child, torch.nn.Linear) and \
+                    child.in_features % group == 0:
+                ql = QuantLinear.from_float(child.weight.data, group)
+                ql = ql.to(child.weight.device)
+                setattr(parent, name, ql)
+                n += 1
+    return n

@@ -157,3 +157,32 @@ def _build_tiles(cu_seqlens: torch.Tensor):
     dev = cu_seqlens.device
     return (torch.tensor(seqs, dtype=torch.int32, device=dev),
             torch.tensor(bases, dtype=torch.int32, device=dev))
+
+
+def w4a16_gemv(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
+               zeros: torch.Tensor, group: int) -> torch.Tensor:
+    """Group-quantized W4A16 linear, small M (decode): out = x @ W^T with
+    W dequantized on the fly inside the kernel."""
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty(x.size(0), qweight.size(0), dtype=x.dtype,
+                          device=x.device)
+        torch.ops.kaito.w4a16_gemv(out, x, qweight, scales, zeros, group)
+        return out
+    return torch_ref.w4a16_gemv(x, qweight, scales, zeros, group)
+
+
+def w4a16_dequant(qweight: torch.Tensor, scales: torch.Tensor,
+                  zeros: torch.Tensor, group: int,
+                  out: torch.Tensor = None) -> torch.Tensor:
+    """Dequantize packed 4-bit weights to bf16 [N, K] (feeds hipBLASLt
+    MFMA for large-M GEMMs)."""
+    if qweight.is_cuda:
+        _require_ext()
+        if out is None:
+            out = torch.empty(qweight.size(0), qweight.size(1) * 8,
+                              dtype=torch.bfloat16, device=qweight.device)
+        torch.ops.kaito.w4a16_dequant(out, qweight, scales, zeros, group)
+        return out
+    ref = torch_ref.w4a16_unpack(qweight, scales, zeros, group)
+    return ref.to(torch.bfloat16) if out is None else out.copy_(ref)

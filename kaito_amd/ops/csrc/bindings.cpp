@@ -32,6 +32,10 @@ void lora_expand(at::Tensor y, at::Tensor tmp, at::Tensor B, at::Tensor idx);
 void paged_read_bw(at::Tensor out, at::Tensor k_cache, at::Tensor v_cache,
                    at::Tensor block_tables, at::Tensor seq_lens,
                    int64_t mode);
+void w4a16_gemv(at::Tensor out, at::Tensor x, at::Tensor qweight,
+                at::Tensor scales, at::Tensor zeros, int64_t group);
+void w4a16_dequant(at::Tensor out, at::Tensor qweight, at::Tensor scales,
+                   at::Tensor zeros, int64_t group);
 }  // namespace kaito
 
 TORCH_LIBRARY(kaito, m) {
@@ -48,6 +52,8 @@ TORCH_LIBRARY(kaito, m) {
   m.def("lora_shrink(Tensor(a!) tmp, Tensor x, Tensor A, Tensor idx, float scale) -> ()");
   m.def("lora_expand(Tensor(a!) y, Tensor tmp, Tensor B, Tensor idx) -> ()");
   m.def("paged_read_bw(Tensor(a!) out, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, int mode) -> ()");
+  m.def("w4a16_gemv(Tensor(a!) out, Tensor x, Tensor qweight, Tensor scales, Tensor zeros, int group) -> ()");
+  m.def("w4a16_dequant(Tensor(a!) out, Tensor qweight, Tensor scales, Tensor zeros, int group) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
@@ -64,4 +70,6 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("lora_shrink", &kaito::lora_shrink);
   m.impl("lora_expand", &kaito::lora_expand);
   m.impl("paged_read_bw", &kaito::paged_read_bw);
+  m.impl("w4a16_gemv", &kaito::w4a16_gemv);
+  m.impl("w4a16_dequant", &kaito::w4a16_dequant);
 }

@@ -1,0 +1,140 @@
+// W4A16 group-quantized linear kernels (AWQ/GPTQ-class presets —
+// reference inventory row "quantized GEMM"; the catalog
+// carries qwen3-8b-awq etc.).
+//
+// Native layout (loaders repack AWQ checkpoints into this):
+//   qweight: uint32 [N, K/8]   8 consecutive K nibbles per word, row-major
+//   scales:  float  [N, K/G]   group size G (typ. 128), w = s*q - z
+//   zeros:   float  [N, K/G]   z folded as s*zq at repack time
+//
+// Two paths:
+//   w4a16_gemv    : M <= 32 decode shapes. Memory-bound on the 4-bit
+//                   weights (N*K/2 bytes vs 2*N*K for bf16 -> ~4x less
+//                   traffic than a bf16 GEMM at small M). One wave per
+//                   output row; lane takes 8-k packs round-robin so the
+//                   qweight reads coalesce per wave; x reads are short8.
+//   w4a16_dequant : tile dequant to bf16 scratch; large-M callers then run
+//                   the scratch through hipBLASLt MFMA (dequant is
+//                   O(N*K) against the GEMM's O(M*N*K) -> noise at M>=128).
+#include "common.h"
+#include <torch/library.h>
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace kaito {
+
+// ---------------------------------------------------------------- gemv
+// grid.x = N / ROWS, block = ROWS waves; each wave owns one output row.
+template <int M_TILE>
+__global__ __launch_bounds__(256, 4)
+void w4a16_gemv_kernel(short* __restrict__ out,          // [M, N] bf16
+                       const short* __restrict__ x,      // [M, K] bf16
+                       const uint32_t* __restrict__ qw,  // [N, K/8]
+                       const float* __restrict__ scales, // [N, K/G]
+                       const float* __restrict__ zeros,  // [N, K/G]
+                       int M, int N, int K, int group) {
+  const int wave = threadIdx.x >> 6;          // 4 waves per WG
+  const int lane = threadIdx.x & 63;
+  const int n = blockIdx.x * 4 + wave;
+  if (n >= N) return;
+  const int kw = K >> 3;                      // u32 words per row
+  const uint32_t* qrow = qw + (int64_t)n * kw;
+  const int gstride = K / group;
+
+  float acc[M_TILE];
+#pragma unroll
+  for (int m = 0; m < M_TILE; ++m) acc[m] = 0.f;
+
+  for (int w = lane; w < kw; w += 64) {
+    uint32_t q = qrow[w];
+    const int k0 = w << 3;
+    const int g = k0 / group;
+    const float s = scales[(int64_t)n * gstride + g];
+    const float z = zeros[(int64_t)n * gstride + g];
+    // unpack 8 nibbles once
+    float qv[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) qv[j] = (float)((q >> (4 * j)) & 0xF);
+#pragma unroll
+    for (int m = 0; m < M_TILE; ++m) {
+      if (m >= M) break;
+      short8_t xv = *reinterpret_cast<const short8_t*>(
+          x + (int64_t)m * K + k0);
+      float dot = 0.f, xs = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float xf = bf16_to_f32(xv[j]);
+        dot += xf * qv[j];
+        xs += xf;
+      }
+      acc[m] += s * dot - z * xs;
+    }
+  }
+#pragma unroll
+  for (int m = 0; m < M_TILE; ++m) {
+    if (m >= M) break;
+    const float r = wave_reduce_sum(acc[m]);
+    if (lane == 0) out[(int64_t)m * N + n] = f32_to_bf16(r);
+  }
+}
+
+// ---------------------------------------------------------------- dequant
+// out[N, K] bf16 <- dequantized weights. Elementwise, one u32 pack per
+// thread; writes are short8 (16B) so the store path is fully vectorized.
+__global__ __launch_bounds__(256)
+void w4a16_dequant_kernel(short* __restrict__ out,
+                          const uint32_t* __restrict__ qw,
+                          const float* __restrict__ scales,
+                          const float* __restrict__ zeros,
+                          int N, int K, int group) {
+  const int kw = K >> 3;
+  const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (int64_t)N * kw) return;
+  const int n = idx / kw;
+  const int w = idx % kw;
+  const int k0 = w << 3;
+  const int g = k0 / group;
+  const int gstride = K / group;
+  const float s = scales[(int64_t)n * gstride + g];
+  const float z = zeros[(int64_t)n * gstride + g];
+  const uint32_t q = qw[idx];
+  short8_t o;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    o[j] = f32_to_bf16(s * (float)((q >> (4 * j)) & 0xF) - z);
+  *reinterpret_cast<short8_t*>(out + (int64_t)n * K + k0) = o;
+}
+
+void w4a16_gemv(at::Tensor out, at::Tensor x, at::Tensor qweight,
+                at::Tensor scales, at::Tensor zeros, int64_t group) {
+  const int M = x.size(0), K = x.size(1), N = qweight.size(0);
+  TORCH_CHECK(M <= 32, "w4a16_gemv: M must be <= 32 (use dequant+GEMM)");
+  TORCH_CHECK(K % 8 == 0 && group % 8 == 0 && K % group == 0);
+  dim3 grid((N + 3) / 4), block(256);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto launch = [&](auto mt) {
+    w4a16_gemv_kernel<decltype(mt)::value><<<grid, block, 0, stream>>>(
+        (short*)out.data_ptr(), (const short*)x.data_ptr(),
+        (const uint32_t*)qweight.data_ptr(), scales.data_ptr<float>(),
+        zeros.data_ptr<float>(), M, N, K, (int)group);
+  };
+  if (M <= 1) launch(std::integral_constant<int, 1>{});
+  else if (M <= 4) launch(std::integral_constant<int, 4>{});
+  else if (M <= 8) launch(std::integral_constant<int, 8>{});
+  else if (M <= 16) launch(std::integral_constant<int, 16>{});
+  else launch(std::integral_constant<int, 32>{});
+}
+
+void w4a16_dequant(at::Tensor out, at::Tensor qweight, at::Tensor scales,
+                   at::Tensor zeros, int64_t group) {
+  const int N = qweight.size(0), K = out.size(1);
+  TORCH_CHECK(K % 8 == 0 && K % group == 0);
+  const int64_t total = (int64_t)N * (K >> 3);
+  dim3 grid((total + 255) / 256), block(256);
+  auto stream = at::hip::getCurrentHIPStream();
+  w4a16_dequant_kernel<<<grid, block, 0, stream>>>(
+      (short*)out.data_ptr(), (const uint32_t*)qweight.data_ptr(),
+      scales.data_ptr<float>(), zeros.data_ptr<float>(), N, K, (int)group);
+}
+
+}  // namespace kaito

@@ -145,3 +145,23 @@ def context_attention(q: torch.Tensor, k_cache: torch.Tensor,
         o = torch.einsum("hqk,hkd->hqd", p, vx)
         out[s0:s1] = o.transpose(0, 1).to(q.dtype)
     return out
+
+
+def w4a16_unpack(qweight: torch.Tensor, scales: torch.Tensor,
+                 zeros: torch.Tensor, group: int) -> torch.Tensor:
+    """Dequantize the native W4A16 layout (qweight u32 [N, K/8], 8
+    consecutive K nibbles per word; w = s*q - z) to float [N, K]."""
+    N, kw = qweight.shape
+    K = kw * 8
+    shifts = torch.arange(8, device=qweight.device, dtype=torch.long) * 4
+    q = (qweight.unsqueeze(-1).long() >> shifts) & 0xF      # [N, K/8, 8]
+    q = q.reshape(N, K).float()
+    s = scales.repeat_interleave(group, dim=1).float()       # [N, K]
+    z = zeros.repeat_interleave(group, dim=1).float()
+    return s * q - z
+
+
+def w4a16_gemv(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
+               zeros: torch.Tensor, group: int) -> torch.Tensor:
+    w = w4a16_unpack(qweight, scales, zeros, group)
+    return (x.float() @ w.T).to(x.dtype)

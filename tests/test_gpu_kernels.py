@@ -303,3 +303,41 @@ def test_context_attention_gpu(cfgs):
                               vc.cpu().float(), cu.cpu(), kvl.cpu(),
                               bt.cpu(), scale)
     _close(out, ref.to(DEV), atol=2e-2)
+
+
+# ------------------------------------------------------------------ W4A16
+def test_w4a16_gemv_matches_ref():
+    torch.manual_seed(11)
+    from kaito_amd.models.quant import quantize_w4
+    N, K, G = 512, 1024, 128
+    w = torch.randn(N, K)
+    qw, s, z = quantize_w4(w, G)
+    deq = R.w4a16_unpack(qw, s, z, G)
+    for M in (1, 3, 8, 17, 32):
+        x = _bf16(M, K)
+        y = ops.w4a16_gemv(x.contiguous(), qw.to(DEV), s.to(DEV), z.to(DEV), G)
+        expect = x.float() @ deq.to(DEV).T
+        _close(y, expect, atol=5e-2, rtol=5e-2)
+
+
+def test_w4a16_dequant_matches_ref():
+    torch.manual_seed(12)
+    from kaito_amd.models.quant import quantize_w4
+    N, K, G = 256, 2048, 128
+    w = torch.randn(N, K)
+    qw, s, z = quantize_w4(w, G)
+    out = ops.w4a16_dequant(qw.to(DEV), s.to(DEV), z.to(DEV), G)
+    _close(out, R.w4a16_unpack(qw, s, z, G).to(DEV), atol=1e-2, rtol=1e-2)
+
+
+def test_w4a16_quantlinear_gpu_both_paths():
+    torch.manual_seed(13)
+    from kaito_amd.models.quant import QuantLinear
+    lin = torch.nn.Linear(1024, 512, bias=False)
+    ql = QuantLinear.from_float(lin.weight.data, 128).to(DEV)
+    deq = R.w4a16_unpack(ql.qweight.cpu(), ql.scales.cpu(),
+                         ql.zeros.cpu(), 128).to(DEV)
+    x_small = _bf16(4, 1024)
+    x_large = _bf16(128, 1024)
+    _close(ql(x_small), x_small.float() @ deq.T, atol=5e-2, rtol=5e-2)
+    _close(ql(x_large), x_large.float() @ deq.T, atol=8e-2, rtol=8e-2)
