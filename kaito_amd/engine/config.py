@@ -36,6 +36,9 @@ class ModelConfig:
     num_experts: int = 0
     num_experts_per_tok: int = 2
     moe_intermediate_size: int = 0       # per-expert FFN width (0 → dense I)
+    # weight quantization: "" (bf16) | "w4a16"/"awq" (4-bit group-quantized
+    # linears via the HIP GEMV / dequant+MFMA kernels)
+    quant_method: str = ""
     # serving runtime: "native" → HIP engine (llama-architecture family);
     # "transformers" → fallback runtime (reference: vLLM vs text-generation
     # runtime split, supported_models.yaml `runtime:` field)

@@ -56,6 +56,10 @@ class ModelRunner:
             load_safetensors_weights(self.model, weights_path)
         else:
             self.model.random_init(seed)
+        if self.cfg.model.quant_method in ("awq", "w4a16"):
+            from ..models.quant import quantize_parallel_linears
+            n = quantize_parallel_linears(self.model)
+            logger.info("quantized %d linears to W4A16", n)
         self.model.init_rope(self.device, self.max_model_len)
         self.model.eval()
         return self

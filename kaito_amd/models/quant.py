@@ -31,7 +31,7 @@ def quantize_w4(weight: torch.Tensor, group: int = 128
     zq = (-wmin / scale).round().clamp(0, 15)
     q = (w / scale.unsqueeze(-1) + zq.unsqueeze(-1)).round().clamp(0, 15)
     q = q.reshape(N, K).to(torch.int64)
-    shifts = torch.arange(8, dtype=torch.int64) * 4
+    shifts = torch.arange(8, dtype=torch.int64, device=q.device) * 4
     packed = (q.reshape(N, K // 8, 8) << shifts).sum(dim=-1)
     qweight = packed.to(torch.int32)          # bit pattern == uint32
     zeros = (scale * zq).float()              # z folded: w = s*q - z
@@ -58,7 +58,7 @@ def repack_awq(qweight_awq: torch.Tensor, qzeros_awq: torch.Tensor,
     zq = unpack_n(qzeros_awq.long())            # [K/G, N]
     s = scales_awq.float()                      # [K/G, N]
     qn = q.T.contiguous()                       # [N, K]
-    shifts = torch.arange(8, dtype=torch.int64) * 4
+    shifts = torch.arange(8, dtype=torch.int64, device=qn.device) * 4
     packed = (qn.reshape(N, K // 8, 8) << shifts).sum(dim=-1).to(torch.int32)
     scales = s.T.contiguous()                   # [N, K/G]
     zeros = (scales * zq.T.float())
@@ -95,6 +95,21 @@ class QuantLinear(torch.nn.Module):
                                   self.group)
             y = torch.nn.functional.linear(x2.to(w.dtype), w)
         return y.reshape(*shape, self.out_features).to(x.dtype)
+
+
+def quantize_parallel_linears(model: torch.nn.Module, group: int = 128,
+                              suffixes=("qkv_proj", "o_proj",
+                                        "gate_up_proj", "down_proj")) -> int:
+    """Quantize the engine's TP-parallel linears in place (weight-only W4;
+    PTQ of whatever weights are loaded). Returns modules converted."""
+    n = 0
+    for name, mod in model.named_modules():
+        if any(name.endswith(sfx) for sfx in suffixes) and \
+                hasattr(mod, "quantize_") and \
+                mod.weight.numel() and mod.weight.size(1) % group == 0:
+            mod.quantize_(group)
+            n += 1
+    return n
 
 
 def quantize_model_linears(model: torch.nn.Module, group: int = 128,

@@ -14,7 +14,45 @@ from .state import get_state, tp_all_reduce
 from ..engine import lora as lora_mod
 
 
-class ColumnParallelLinear(nn.Module):
+class _QuantMixin:
+    """Weight-only W4A16 for the parallel linears: quantize_() packs the
+    rank-local weight shard into the native 4-bit layout and switches the
+    matmul to the HIP GEMV (decode M) / dequant+MFMA (large M) path. TP
+    comms, bias, and the LoRA hook are untouched — quantization is purely
+    the local GEMM."""
+
+    _quantized = False
+
+    def quantize_(self, group: int = 128) -> None:
+        from ..models.quant import quantize_w4, QuantLinear
+        w = self.weight.data
+        assert w.size(1) % group == 0, (w.shape, group)
+        qw, sc, z = quantize_w4(w.float(), group)
+        dev = w.device
+        self.register_buffer("qweight", qw.to(dev))
+        self.register_buffer("scales", sc.to(dev))
+        self.register_buffer("zeros", z.to(dev))
+        self.q_group = group
+        self.weight = nn.Parameter(torch.empty(0, dtype=w.dtype, device=dev),
+                                   requires_grad=False)  # drop bf16 copy
+        self._quantized = True
+        self._gemv_max_m = QuantLinear.GEMV_MAX_M
+
+    def _qmatmul(self, x: torch.Tensor) -> torch.Tensor:
+        from .. import ops
+        shape = x.shape[:-1]
+        x2 = x.reshape(-1, x.shape[-1])
+        if x2.size(0) <= self._gemv_max_m:
+            y = ops.w4a16_gemv(x2.contiguous(), self.qweight, self.scales,
+                               self.zeros, self.q_group)
+        else:
+            w = ops.w4a16_dequant(self.qweight, self.scales, self.zeros,
+                                  self.q_group)
+            y = F.linear(x2.to(w.dtype), w)
+        return y.reshape(*shape, -1).to(x.dtype)
+
+
+class ColumnParallelLinear(_QuantMixin, nn.Module):
     def __init__(self, in_features: int, out_features: int, bias: bool = False,
                  dtype: torch.dtype = torch.bfloat16):
         super().__init__()
@@ -30,11 +68,16 @@ class ColumnParallelLinear(nn.Module):
             requires_grad=False) if bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        out = F.linear(x, self.weight, self.bias)
+        if self._quantized:
+            out = self._qmatmul(x)
+            if self.bias is not None:
+                out = out + self.bias
+        else:
+            out = F.linear(x, self.weight, self.bias)
         return lora_mod.maybe_apply(self, x, out)
 
 
-class RowParallelLinear(nn.Module):
+class RowParallelLinear(_QuantMixin, nn.Module):
     def __init__(self, in_features: int, out_features: int, bias: bool = False,
                  dtype: torch.dtype = torch.bfloat16):
         super().__init__()
@@ -51,7 +94,7 @@ class RowParallelLinear(nn.Module):
             requires_grad=False) if bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        out = F.linear(x, self.weight)
+        out = self._qmatmul(x) if self._quantized else F.linear(x, self.weight)
         out = lora_mod.maybe_apply(self, x, out)
         out = tp_all_reduce(out)
         if self.bias is not None:

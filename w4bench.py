@@ -1,0 +1,20 @@
+import torch, time
+from kaito_amd.models.quant import QuantLinear
+torch.manual_seed(0)
+K, N = 4096, 6144   # llama-8b qkv-ish shape
+lin = torch.nn.Linear(K, N, bias=False).to("cuda", torch.bfloat16)
+ql = QuantLinear.from_float(lin.weight.data.float(), 128).to("cuda")
+def t(f, n=200):
+    for _ in range(20): f()
+    torch.cuda.synchronize(); t0=time.perf_counter()
+    for _ in range(n): f()
+    torch.cuda.synchronize(); return (time.perf_counter()-t0)/n*1e6
+for M in (1, 4, 8, 16, 32):
+    x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    us_q = t(lambda: ql(x))
+    us_b = t(lambda: lin(x))
+    gb = (N*K/2 + M*K*2 + M*N*2)/1e9
+    print(f"M={M:3d}  w4a16={us_q:7.1f}us ({gb/us_q*1e6:.2f} TB/s)  bf16={us_b:7.1f}us  speedup={us_b/us_q:.2f}x")
+M = 512
+x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+print(f"M=512 dequant+gemm={t(lambda: ql(x)):7.1f}us  bf16={t(lambda: lin(x)):7.1f}us")
