@@ -38,6 +38,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--swap-space", type=float, default=0.0)
     p.add_argument("--kv-cache-dtype", default="auto")
     p.add_argument("--enable-lora", action="store_true")
+    p.add_argument("--quantization", default="",
+                   choices=["", "w4a16", "awq"],
+                   help="weight-only 4-bit serving (HIP GEMV/dequant+MFMA)")
     p.add_argument("--enforce-eager", action="store_true")
     p.add_argument("--port", type=int, default=int(os.environ.get("PORT", 5000)))
     p.add_argument("--host", default="0.0.0.0")
@@ -77,6 +80,9 @@ def main(argv=None):
     from . import metrics
 
     mc = get_model_config(args.model)
+    if args.quantization:
+        import dataclasses
+        mc = dataclasses.replace(mc, quant_method=args.quantization)
     if mc.runtime == "transformers":
         # non-llama-family architecture (falcon/gemma-3/deepseek-MLA/
         # gpt-oss): serve via the fallback runtime — the reference's
@@ -87,7 +93,8 @@ def main(argv=None):
                          "--host", args.host] +
                         (["--weights-path", args.weights_path]
                          if args.weights_path else []))
-    init_parallel(tp_size=args.tensor_parallel_size)
+    init_parallel(tp_size=args.tensor_parallel_size,
+                  pp_size=args.pipeline_parallel_size)
     max_len = None if str(args.max_model_len) == "auto" else int(args.max_model_len)
     cfg = EngineConfig(
         model=mc,
@@ -97,6 +104,7 @@ def main(argv=None):
         gpu_memory_utilization=args.gpu_memory_utilization,
         tensor_parallel_size=args.tensor_parallel_size,
         enforce_eager=args.enforce_eager,
+        enable_lora=args.enable_lora,
     )
     metrics.MODEL_DOWNLOAD_PROGRESS.set(0.0)
     engine = LLMEngine(cfg, weights_path=args.weights_path)
