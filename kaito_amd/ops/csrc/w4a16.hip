@@ -33,39 +33,48 @@ void w4a16_gemv_kernel(short* __restrict__ out,          // [M, N] bf16
                        const float* __restrict__ scales, // [N, K/G]
                        const float* __restrict__ zeros,  // [N, K/G]
                        int M, int N, int K, int group) {
-  const int wave = threadIdx.x >> 6;          // 4 waves per WG
+  // Each wave owns one output row n. A lane takes a CONTIGUOUS 8-word
+  // (64-k) slice per tile so its slice sits inside ONE quant group:
+  // scale/zero load once per slice (vs per word), q loads are uint4
+  // (16B), x loads are short8 (16B). Weight traffic is then the pure
+  // N*K/2 bytes and the kernel runs at HBM rate for small M.
+  const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int n = blockIdx.x * 4 + wave;
   if (n >= N) return;
   const int kw = K >> 3;                      // u32 words per row
   const uint32_t* qrow = qw + (int64_t)n * kw;
   const int gstride = K / group;
+  const float* srow = scales + (int64_t)n * gstride;
+  const float* zrow = zeros + (int64_t)n * gstride;
 
   float acc[M_TILE];
 #pragma unroll
   for (int m = 0; m < M_TILE; ++m) acc[m] = 0.f;
 
-  for (int w = lane; w < kw; w += 64) {
-    uint32_t q = qrow[w];
-    const int k0 = w << 3;
-    const int g = k0 / group;
-    const float s = scales[(int64_t)n * gstride + g];
-    const float z = zeros[(int64_t)n * gstride + g];
-    // unpack 8 nibbles once
-    float qv[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) qv[j] = (float)((q >> (4 * j)) & 0xF);
+  typedef uint32_t uint4_t __attribute__((ext_vector_type(4)));
+  for (int t = lane * 8; t < kw; t += 64 * 8) {  // 8 contiguous words/lane
+    const int k0 = t << 3;                       // 64 k per slice
+    const float s = srow[k0 / group];
+    const float z = zrow[k0 / group];
+    const uint4_t qa = *reinterpret_cast<const uint4_t*>(qrow + t);
+    const uint4_t qb = *reinterpret_cast<const uint4_t*>(qrow + t + 4);
 #pragma unroll
     for (int m = 0; m < M_TILE; ++m) {
       if (m >= M) break;
-      short8_t xv = *reinterpret_cast<const short8_t*>(
-          x + (int64_t)m * K + k0);
+      const short8_t* xp =
+          reinterpret_cast<const short8_t*>(x + (int64_t)m * K + k0);
       float dot = 0.f, xs = 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float xf = bf16_to_f32(xv[j]);
-        dot += xf * qv[j];
-        xs += xf;
+      for (int w = 0; w < 8; ++w) {
+        const uint32_t q = (w < 4) ? qa[w] : qb[w - 4];
+        const short8_t xv = xp[w];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float xf = bf16_to_f32(xv[j]);
+          dot += xf * (float)((q >> (4 * j)) & 0xF);
+          xs += xf;
+        }
       }
       acc[m] += s * dot - z * xs;
     }
@@ -109,7 +118,8 @@ void w4a16_gemv(at::Tensor out, at::Tensor x, at::Tensor qweight,
                 at::Tensor scales, at::Tensor zeros, int64_t group) {
   const int M = x.size(0), K = x.size(1), N = qweight.size(0);
   TORCH_CHECK(M <= 32, "w4a16_gemv: M must be <= 32 (use dequant+GEMM)");
-  TORCH_CHECK(K % 8 == 0 && group % 8 == 0 && K % group == 0);
+  TORCH_CHECK(K % 64 == 0 && group % 64 == 0 && K % group == 0,
+              "w4a16_gemv: 64|K and 64|group required");
   dim3 grid((N + 3) / 4), block(256);
   auto stream = at::hip::getCurrentHIPStream();
   auto launch = [&](auto mt) {
