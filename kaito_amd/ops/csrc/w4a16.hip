@@ -59,25 +59,30 @@ void w4a16_gemv_kernel(short* __restrict__ out,          // [M, N] bf16
     const float z = zrow[k0 / group];
     const uint4_t qa = *reinterpret_cast<const uint4_t*>(qrow + t);
     const uint4_t qb = *reinterpret_cast<const uint4_t*>(qrow + t + 4);
+    float dotm[M_TILE], xsm[M_TILE];
 #pragma unroll
-    for (int m = 0; m < M_TILE; ++m) {
-      if (m >= M) break;
-      const short8_t* xp =
-          reinterpret_cast<const short8_t*>(x + (int64_t)m * K + k0);
-      float dot = 0.f, xs = 0.f;
+    for (int m = 0; m < M_TILE; ++m) { dotm[m] = 0.f; xsm[m] = 0.f; }
 #pragma unroll
-      for (int w = 0; w < 8; ++w) {
-        const uint32_t q = (w < 4) ? qa[w] : qb[w - 4];
-        const short8_t xv = xp[w];
+    for (int w = 0; w < 8; ++w) {
+      const uint32_t q = (w < 4) ? qa[w] : qb[w - 4];
+      float qv[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) qv[j] = (float)((q >> (4 * j)) & 0xF);
+#pragma unroll
+      for (int m = 0; m < M_TILE; ++m) {
+        if (m >= M) break;
+        const short8_t xv = *reinterpret_cast<const short8_t*>(
+            x + (int64_t)m * K + k0 + w * 8);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const float xf = bf16_to_f32(xv[j]);
-          dot += xf * (float)((q >> (4 * j)) & 0xF);
-          xs += xf;
+          dotm[m] += xf * qv[j];
+          xsm[m] += xf;
         }
       }
-      acc[m] += s * dot - z * xs;
     }
+#pragma unroll
+    for (int m = 0; m < M_TILE; ++m) acc[m] += s * dotm[m] - z * xsm[m];
   }
 #pragma unroll
   for (int m = 0; m < M_TILE; ++m) {
