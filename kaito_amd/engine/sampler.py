@@ -106,6 +106,18 @@ class Sampler:
 
         probs = torch.softmax(scaled, dim=-1)
         sampled = torch.multinomial(probs, 1, generator=self.gen).squeeze(1)
+        # per-request seeded generators (reproducible sampling); the
+        # per-row draw stays on-device so the call remains async
+        for i, sq in enumerate(seqs):
+            sp = sq.sampling
+            if sp.seed is not None and temps_l[i] > 0.0:
+                g = getattr(sq, "_sampler_gen", None)
+                if g is None:
+                    g = torch.Generator(device=probs.device)
+                    g.manual_seed(sp.seed)
+                    sq._sampler_gen = g
+                sampled[i] = torch.multinomial(probs[i:i + 1], 1,
+                                               generator=g)[0, 0]
         tokens = torch.where(greedy, logits.argmax(dim=-1), sampled)
         return self._with_logprobs(logits, tokens, seqs) \
             if return_logprobs else tokens

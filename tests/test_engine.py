@@ -484,3 +484,24 @@ def test_w4a16_quantized_engine_decodes():
     out2 = eng2.generate([[3, 14, 15, 92]],
                         SamplingParams(max_tokens=6, ignore_eos=True))[0]
     assert out2.output_token_ids == out.output_token_ids
+
+
+def test_seeded_sampling_reproducible():
+    """SamplingParams.seed: same seed → identical stochastic output even
+    across engine instances and batch compositions; different seed differs
+    (overwhelmingly likely over 24 temperature-1.0 draws)."""
+    cfg = _cfg()
+
+    def run(seed, extra=False):
+        eng = LLMEngine(cfg)
+        if extra:   # change batch composition
+            eng.add_request([9, 9, 9], SamplingParams(max_tokens=24,
+                                                      ignore_eos=True))
+        sid = eng.add_request([3, 14, 15], SamplingParams(
+            max_tokens=24, ignore_eos=True, temperature=1.0, seed=seed))
+        while eng.has_unfinished():
+            eng.step()
+        return eng.seqs[sid].output_token_ids
+
+    assert run(123) == run(123) == run(123, extra=True)
+    assert run(123) != run(321)
