@@ -69,7 +69,10 @@ class QuantLinear(torch.nn.Module):
     """W4A16 linear: HIP GEMV for decode-sized M, dequant + MFMA GEMM
     (hipBLASLt) beyond. Drop-in for a bias-free nn.Linear."""
 
-    GEMV_MAX_M = 32
+    # measured crossover on MI355X (profiles/r01_decode_profile.md):
+    # GEMV wins to M=4 (10-24us vs ~38us dequant+GEMM); beyond, tile
+    # dequant into the MFMA GEMM path wins
+    GEMV_MAX_M = 4
 
     def __init__(self, qweight, scales, zeros, group: int = 128):
         super().__init__()
