@@ -214,3 +214,29 @@ def test_stop_strings_truncate(client):
     assert data["choices"][0]["finish_reason"] == "stop"
     assert stop not in data["choices"][0]["text"]
     assert data["choices"][0]["text"] == full.split(stop)[0]
+
+
+def test_completions_logprobs_and_penalties(client):
+    c = client
+    if True:
+        r = c.post("/v1/completions", json={
+            "prompt": [3, 14, 15], "max_tokens": 4, "temperature": 0.0,
+            "ignore_eos": True, "logprobs": 2})
+        assert r.status_code == 200
+        lp = r.json()["choices"][0]["logprobs"]
+        assert len(lp["tokens"]) == 4
+        assert len(lp["token_logprobs"]) == 4
+        assert all(v is not None and v <= 0 for v in lp["token_logprobs"])
+        assert all(len(t) == 2 for t in lp["top_logprobs"])
+        r2 = c.post("/v1/completions", json={
+            "prompt": [3, 14, 15], "max_tokens": 6, "temperature": 0.0,
+            "ignore_eos": True, "frequency_penalty": 50.0})
+        assert r2.status_code == 200
+        r3 = c.post("/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 3, "temperature": 0, "ignore_eos": True,
+            "logprobs": True, "top_logprobs": 2})
+        assert r3.status_code == 200
+        content = r3.json()["choices"][0]["logprobs"]["content"]
+        assert len(content) == 3
+        assert all(len(e["top_logprobs"]) == 2 for e in content)
