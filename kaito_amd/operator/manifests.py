@@ -52,6 +52,16 @@ def generate_service(ws: Workspace, headless: bool = False) -> Dict[str, Any]:
             "spec": spec}
 
 
+def readiness_timeout_for(model: Optional[ModelConfig]) -> int:
+    """Size-scaled model readiness timeout (reference: metadata.go:54-59
+    scales the startup window with TotalSafeTensorFileSize): 10 min floor
+    plus ~12 s per GiB of weights (load + quantize + graph capture)."""
+    if model is None:
+        return 600
+    gib = model.param_bytes() / (1 << 30)
+    return max(600, int(gib * 12))
+
+
 def _probes(readiness_timeout_s: int = 600) -> Dict[str, Any]:
     """preset_inferences.go:403-422: startup probe sized by model readiness
     timeout, liveness/readiness on /health."""
@@ -148,7 +158,9 @@ def generate_statefulset(ws: Workspace, model: ModelConfig, gpu: GPUConfig,
     """manifests.go:135-200 GenerateStatefulSetManifest."""
     plan = plan or configure_parallelism(model, gpu)
     sel = workspace_selector(ws)
-    pod_spec = generate_inference_pod_spec(ws, model, gpu, plan, image)
+    pod_spec = generate_inference_pod_spec(
+        ws, model, gpu, plan, image,
+        readiness_timeout_s=readiness_timeout_for(model))
     return {
         "apiVersion": "apps/v1",
         "kind": "StatefulSet",

@@ -541,3 +541,14 @@ def test_operator_loop_reconciles_inferencesets(client):
     assert len(children) == 2
     loop.tick()   # reconciles the children to statefulsets
     assert client.get("StatefulSet", "default", "is-loop-0")
+
+
+def test_readiness_timeout_scales_with_model_size():
+    from kaito_amd.operator.manifests import readiness_timeout_for
+    from kaito_amd.models import get_model_config
+    small = readiness_timeout_for(get_model_config("tiny-llama-test"))
+    mid = readiness_timeout_for(get_model_config("llama-3-8b"))
+    big = readiness_timeout_for(get_model_config("llama-3-70b"))
+    assert small == 600                      # floor
+    assert mid == 600 or mid < big           # 8B near floor, 70B scaled
+    assert big > 1200                        # ~141 GiB * 12 s
