@@ -8,6 +8,8 @@ unchanged.
 """
 from __future__ import annotations
 
+import asyncio
+import dataclasses
 import json
 import time
 import uuid
@@ -63,14 +65,14 @@ class ChatCompletionRequest(BaseModel):
 
 def _sampling(max_tokens, temperature, top_p, top_k=0, ignore_eos=False,
               eos_id=None, logprobs=None, presence_penalty=0.0,
-              frequency_penalty=0.0, repetition_penalty=1.0):
+              frequency_penalty=0.0, repetition_penalty=1.0, seed=None):
     return SamplingParams(
         max_tokens=max_tokens or 16,
         temperature=temperature if temperature is not None else 1.0,
         top_p=top_p, top_k=top_k, ignore_eos=ignore_eos,
         logprobs=logprobs, presence_penalty=presence_penalty,
         frequency_penalty=frequency_penalty,
-        repetition_penalty=repetition_penalty)
+        repetition_penalty=repetition_penalty, seed=seed)
 
 
 def _lp_openai(lp_rows, toks, tokenizer):
@@ -154,41 +156,56 @@ def build_app(async_engine: AsyncLLMEngine, tokenizer, model_name: str,
                        req.ignore_eos, logprobs=req.logprobs,
                        presence_penalty=req.presence_penalty,
                        frequency_penalty=req.frequency_penalty,
-                       repetition_penalty=req.repetition_penalty)
+                       repetition_penalty=req.repetition_penalty,
+                       seed=req.seed)
         rid = f"cmpl-{uuid.uuid4().hex[:24]}"
         if req.stream:
+            if req.n != 1:
+                raise HTTPException(400, "stream with n>1 not supported")
             return StreamingResponse(
                 _stream_completion(rid, prompt_ids, sp),
                 media_type="text/event-stream")
-        toks: List[int] = []
-        lp_rows: List[list] = []
-        finish = "length"
         stops = _stop_list(req.stop)
-        text = ""
-        async for item in async_engine.generate(prompt_ids, sp):
-            if item.finished:
-                finish = item.finish_reason or "stop"
-            else:
-                toks.append(item.token_id)
-                if req.logprobs:
-                    lp_rows.append(item.logprobs or [])
-                if stops:
-                    text = tokenizer.decode(toks)
-                    cut, hit = _truncate_at_stop(text, stops)
-                    if hit:
-                        text, finish = cut, "stop"
-                        break
-        if not stops:
-            text = tokenizer.decode(toks)
+
+        async def one_choice(index: int):
+            # n>1: independent samples; the block-hash prefix cache makes
+            # the shared prompt KV zero-copy across choices
+            spi = sp if req.n == 1 else dataclasses.replace(
+                sp, seed=(sp.seed + index if sp.seed is not None else None))
+            toks: List[int] = []
+            lp_rows: List[list] = []
+            finish = "length"
+            text = ""
+            async for item in async_engine.generate(prompt_ids, spi):
+                if item.finished:
+                    finish = item.finish_reason or "stop"
+                else:
+                    toks.append(item.token_id)
+                    if req.logprobs:
+                        lp_rows.append(item.logprobs or [])
+                    if stops:
+                        text = tokenizer.decode(toks)
+                        cut, hit = _truncate_at_stop(text, stops)
+                        if hit:
+                            text, finish = cut, "stop"
+                            break
+            if not stops:
+                text = tokenizer.decode(toks)
+            return {"index": index, "text": text, "finish_reason": finish,
+                    "logprobs": _lp_openai(lp_rows, toks, tokenizer)
+                    if req.logprobs else None}, len(toks)
+
+        results = await asyncio.gather(*[one_choice(i)
+                                         for i in range(max(req.n, 1))])
+        choices = [r[0] for r in results]
+        ntoks = sum(r[1] for r in results)
         return {
             "id": rid, "object": "text_completion",
             "created": int(time.time()), "model": model_name,
-            "choices": [{"index": 0, "text": text, "finish_reason": finish,
-                         "logprobs": _lp_openai(lp_rows, toks, tokenizer)
-                         if req.logprobs else None}],
+            "choices": choices,
             "usage": {"prompt_tokens": len(prompt_ids),
-                      "completion_tokens": len(toks),
-                      "total_tokens": len(prompt_ids) + len(toks)}}
+                      "completion_tokens": ntoks,
+                      "total_tokens": len(prompt_ids) + ntoks}}
 
     async def _stream_completion(rid, prompt_ids, sp) -> AsyncIterator[str]:
         async for item in async_engine.generate(prompt_ids, sp):

@@ -240,3 +240,20 @@ def test_completions_logprobs_and_penalties(client):
         content = r3.json()["choices"][0]["logprobs"]["content"]
         assert len(content) == 3
         assert all(len(e["top_logprobs"]) == 2 for e in content)
+
+
+def test_completions_n_choices(client):
+    r = client.post("/v1/completions", json={
+        "prompt": [3, 14, 15], "max_tokens": 5, "temperature": 1.0,
+        "seed": 7, "ignore_eos": True, "n": 3})
+    assert r.status_code == 200
+    ch = r.json()["choices"]
+    assert [c["index"] for c in ch] == [0, 1, 2]
+    assert all(c["text"] for c in ch)
+    # seeded n-choices: per-index derived seeds → deterministic across calls
+    r2 = client.post("/v1/completions", json={
+        "prompt": [3, 14, 15], "max_tokens": 5, "temperature": 1.0,
+        "seed": 7, "ignore_eos": True, "n": 3})
+    assert [c["text"] for c in r2.json()["choices"]] == \
+        [c["text"] for c in ch]
+    assert r.json()["usage"]["completion_tokens"] == 15
