@@ -341,3 +341,27 @@ def test_w4a16_quantlinear_gpu_both_paths():
     x_large = _bf16(128, 1024)
     _close(ql(x_small), x_small.float() @ deq.T, atol=5e-2, rtol=5e-2)
     _close(ql(x_large), x_large.float() @ deq.T, atol=8e-2, rtol=8e-2)
+
+
+def test_engine_gpu_w4a16_decode():
+    """Quantized tiny engine on GPU: all linears on the 4-bit path, greedy
+    decode runs both the GEMV (bs<=4) and dequant+GEMM (larger prefill M)
+    kernels; output matches the same engine re-run (determinism) and has
+    full length."""
+    import dataclasses
+    from kaito_amd.engine import SamplingParams
+    from kaito_amd.models import get_model_config
+    mc = dataclasses.replace(get_model_config("tiny-llama-test"),
+                             quant_method="w4a16")
+
+    def run():
+        eng = _gpu_engine(model=mc, enforce_eager=True)
+        assert sum(1 for m in eng.runner.model.modules()
+                   if getattr(m, "_quantized", False)) == 4 * mc.num_layers
+        outs = eng.generate([list(range(30, 80))],   # 50-token prompt: M>4
+                            SamplingParams(max_tokens=8, ignore_eos=True))
+        return outs[0].output_token_ids
+
+    a = run()
+    b = run()
+    assert len(a) == 8 and a == b
