@@ -365,3 +365,26 @@ def test_engine_gpu_w4a16_decode():
     a = run()
     b = run()
     assert len(a) == 8 and a == b
+
+
+def test_engine_gpu_mixed_overlap_matches_classic():
+    """Fire-and-forget side-stream prefill on a real GPU: greedy outputs
+    must be identical to classic either/or stepping, including requests
+    joining mid-decode (exercises the cross-stream event ordering)."""
+    from kaito_amd.engine import SamplingParams
+    prompts = [list(range(10, 45)), list(range(50, 70)),
+               list(range(100, 140)), [7, 8, 9, 10, 11]]
+    sp = SamplingParams(max_tokens=10, ignore_eos=True)
+
+    def run(mixed):
+        eng = _gpu_engine(enforce_eager=True, enable_mixed_batch=mixed,
+                          mixed_prefill_tokens=24)
+        ids = [eng.add_request(prompts[0], sp)]
+        eng.step()
+        eng.step()
+        ids += [eng.add_request(p, sp) for p in prompts[1:]]
+        while eng.has_unfinished():
+            eng.step()
+        return [eng.seqs[i].output_token_ids for i in ids]
+
+    assert run(True) == run(False)
