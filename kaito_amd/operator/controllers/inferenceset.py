@@ -53,6 +53,8 @@ def _ws_obj(iset: InferenceSet, index: int, revision: str) -> Dict:
 
 class InferenceSetReconciler:
     def __init__(self, client: KubeClient, gateway_api: bool = False):
+        from ..expectations import ControllerExpectations
+        self.expectations = ControllerExpectations()
         self.client = client
         self.gateway_api = gateway_api  # gatewayAPIInferenceExtension gate
 
@@ -95,22 +97,34 @@ class InferenceSetReconciler:
         children = self._list_children(iset)
         desired = iset.spec.replicas
         res = ISReconcileResult()
+        key = f"{iset.namespace}/{iset.name}"
 
-        if len(children) > desired:
-            for obj in self.select_workspaces_to_delete(
-                    children, len(children) - desired, revision):
+        # cache-staleness guard (reference: ControllerExpectations,
+        # inferenceset_controller.go:336-386): while a previous reconcile's
+        # creates/deletes have not shown up in the store, do not scale
+        # again — a stale child list would double-create or over-delete.
+        if not self.expectations.satisfied(key):
+            res.requeue_after_s = 1.0
+        elif len(children) > desired:
+            victims = self.select_workspaces_to_delete(
+                children, len(children) - desired, revision)
+            self.expectations.expect_deletions(key, len(victims))
+            for obj in victims:
                 self.client.delete("Workspace", iset.namespace,
                                    obj["metadata"]["name"])
+                self.expectations.deletion_observed(key)
                 res.deleted += 1
         elif len(children) < desired:
             used = {int(o["metadata"]["name"].rsplit("-", 1)[1])
                     for o in children}
             idx = 0
+            self.expectations.expect_creations(key, desired - len(children))
             for _ in range(desired - len(children)):
                 while idx in used:
                     idx += 1
                 used.add(idx)
                 self.client.create(_ws_obj(iset, idx, revision))
+                self.expectations.creation_observed(key)
                 res.created += 1
 
         # upgrade: children on an old revision get the upgrade label

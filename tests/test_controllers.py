@@ -552,3 +552,52 @@ def test_readiness_timeout_scales_with_model_size():
     assert small == 600                      # floor
     assert mid == 600 or mid < big           # 8B near floor, 70B scaled
     assert big > 1200                        # ~141 GiB * 12 s
+
+
+def test_controller_expectations_guard():
+    import time as _t
+    from kaito_amd.operator import expectations as ex
+    e = ex.ControllerExpectations()
+    assert e.satisfied("a")                 # nothing pending
+    e.expect_creations("a", 2)
+    assert not e.satisfied("a")
+    e.creation_observed("a")
+    assert not e.satisfied("a")
+    e.creation_observed("a")
+    assert e.satisfied("a")                 # fulfilled → cleared
+    assert e.pending("a") is None
+    e.expect_deletions("b", 1)
+    assert not e.satisfied("b")
+    old = ex.EXPECTATION_TIMEOUT_S
+    try:
+        ex.EXPECTATION_TIMEOUT_S = 0.0      # expire immediately
+        _t.sleep(0.01)
+        assert e.satisfied("b")             # expired → unwedged
+    finally:
+        ex.EXPECTATION_TIMEOUT_S = old
+
+
+def test_inferenceset_expectations_block_stale_scale(monkeypatch):
+    """With an artificially stale cache (creations expected but not yet
+    observed), the reconciler must requeue instead of re-creating."""
+    from kaito_amd.operator.kubeclient import FakeKubeClient
+    from kaito_amd.operator.controllers.inferenceset import (
+        InferenceSetReconciler)
+    from kaito_amd.operator.main import inferenceset_from_obj
+    client = FakeKubeClient()
+    rec = InferenceSetReconciler(client)
+    iset = inferenceset_from_obj({
+        "metadata": {"name": "pool", "namespace": "default"},
+        "spec": {"replicas": 2, "workspaceTemplate": {
+            "resource": {"instanceType": "Standard_MI355X_v1"},
+            "inference": {"preset": {"name": "llama-3-8b"}}}}})
+    r1 = rec.reconcile(iset)
+    assert r1.created == 2
+    # simulate unobserved in-flight creations
+    rec.expectations.expect_creations("default/pool", 1)
+    r2 = rec.reconcile(iset)
+    assert r2.created == 0 and r2.deleted == 0
+    assert r2.requeue_after_s     # requeued, no scale action
+    rec.expectations.creation_observed("default/pool")
+    r3 = rec.reconcile(iset)
+    assert r3.created == 0            # already at desired; acts normally
