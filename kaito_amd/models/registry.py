@@ -19,11 +19,23 @@ def register(cfg: ModelConfig) -> ModelConfig:
     return cfg
 
 
-def get_model_config(name: str) -> ModelConfig:
+def get_model_config(name: str, weights_path: str = None) -> ModelConfig:
+    """Resolve a preset by name; unknown names fall back to dynamic
+    resolution from the weights directory's config.json (the reference's
+    generateHuggingFaceModel path, vllm_model.go:153 — best-effort models
+    outside the curated list)."""
     key = name.lower()
-    if key not in _REGISTRY:
-        raise KeyError(f"unknown model preset: {name}; known: {sorted(_REGISTRY)}")
-    return _REGISTRY[key]
+    if key in _REGISTRY:
+        return _REGISTRY[key]
+    if weights_path:
+        import json
+        import os
+        cfg_path = os.path.join(weights_path, "config.json")
+        if os.path.exists(cfg_path):
+            with open(cfg_path) as f:
+                cfg = ModelConfig.from_hf_config(json.load(f), name=name)
+            return register(cfg)
+    raise KeyError(f"unknown model preset: {name}; known: {sorted(_REGISTRY)}")
 
 
 def list_models():

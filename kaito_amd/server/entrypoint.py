@@ -79,7 +79,7 @@ def main(argv=None):
     from .tokenizer import load_tokenizer
     from . import metrics
 
-    mc = get_model_config(args.model)
+    mc = get_model_config(args.model, args.weights_path)
     if args.quantization:
         import dataclasses
         mc = dataclasses.replace(mc, quant_method=args.quantization)

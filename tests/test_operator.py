@@ -353,3 +353,25 @@ def test_from_hf_config_and_catalog_row(tmp_path):
     cat = tmp_path / "catalog.json"
     gen_main([str(d), "--name", "l3", "--append-to", str(cat)])
     assert json.loads(cat.read_text())[0]["name"] == "l3"
+
+
+def test_dynamic_model_resolution_from_config_json(tmp_path):
+    """Unknown preset + weights dir with config.json → ModelConfig built
+    dynamically (generateHuggingFaceModel analog)."""
+    import json
+    import pytest as _pt
+    from kaito_amd.models import get_model_config
+    with _pt.raises(KeyError):
+        get_model_config("my-custom-model-x")
+    d = tmp_path / "m"
+    d.mkdir()
+    (d / "config.json").write_text(json.dumps({
+        "architectures": ["MistralForCausalLM"], "hidden_size": 1024,
+        "num_hidden_layers": 4, "num_attention_heads": 16,
+        "num_key_value_heads": 4, "intermediate_size": 4096,
+        "vocab_size": 32000, "max_position_embeddings": 4096,
+        "rope_theta": 10000.0}))
+    mc = get_model_config("my-custom-model-x", str(d))
+    assert mc.hidden_size == 1024 and mc.runtime == "native"
+    # registered now: resolvable by name alone
+    assert get_model_config("my-custom-model-x").num_layers == 4
