@@ -131,7 +131,7 @@ class EngineConfig:
     # decode graph buckets (batch sizes to capture)
     graph_batch_sizes: tuple = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,
                                 192, 256, 320, 384, 448, 512, 640, 768,
-                                896, 1024)
+                                896, 1024, 1280, 1536, 2048)
 
     def max_blocks_per_seq(self, max_len: int) -> int:
         return (max_len + self.block_size - 1) // self.block_size
