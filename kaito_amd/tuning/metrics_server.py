@@ -59,3 +59,17 @@ if __name__ == "__main__":
     import uvicorn
     uvicorn.run(build_metrics_app(), host="0.0.0.0",
                 port=int(os.environ.get("METRICS_PORT", "8090")))
+
+
+def main(argv=None):
+    """Serve the tuning metrics sidecar (reference: metrics_server.py
+    uvicorn on :5000)."""
+    import argparse
+
+    import uvicorn
+    p = argparse.ArgumentParser()
+    p.add_argument("--port", type=int, default=5000)
+    p.add_argument("--host", default="0.0.0.0")
+    args = p.parse_args(argv)
+    uvicorn.run(build_metrics_app(), host=args.host, port=args.port,
+                log_level="warning")
