@@ -35,8 +35,13 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--max-model-len", default="auto")
     p.add_argument("--max-num-seqs", type=int, default=256)
     p.add_argument("--gpu-memory-utilization", type=float, default=0.90)
-    p.add_argument("--swap-space", type=float, default=0.0)
+    p.add_argument("--swap-space", type=float, default=0.0,
+                   help="GiB of host RAM for KV offload (LMCache analog)")
     p.add_argument("--kv-cache-dtype", default="auto")
+    p.add_argument("--enable-kv-events", action="store_true",
+                   help="publish BlockStored/BlockRemoved on :5557 "
+                        "(EPP KVCache-aware routing)")
+    p.add_argument("--kv-events-port", type=int, default=5557)
     p.add_argument("--enable-lora", action="store_true")
     p.add_argument("--quantization", default="",
                    choices=["", "w4a16", "awq"],
@@ -105,9 +110,18 @@ def main(argv=None):
         tensor_parallel_size=args.tensor_parallel_size,
         enforce_eager=args.enforce_eager,
         enable_lora=args.enable_lora,
+        kv_offload=args.swap_space > 0,
+        kv_offload_bytes=int(args.swap_space * (1 << 30)) or None,
     )
+    if args.kv_cache_dtype not in ("auto", "bf16", "bfloat16"):
+        raise SystemExit(f"kv-cache-dtype {args.kv_cache_dtype!r} not "
+                         "supported in this build (bf16 KV only; fp8 KV is "
+                         "on the roadmap)")
     metrics.MODEL_DOWNLOAD_PROGRESS.set(0.0)
     engine = LLMEngine(cfg, weights_path=args.weights_path)
+    if args.enable_kv_events:
+        from ..engine.kv_events import KVEventPublisher
+        engine.kv_publisher = KVEventPublisher(port=args.kv_events_port)
     metrics.MODEL_DOWNLOAD_PROGRESS.set(1.0)
     metrics.MODEL_DOWNLOAD_DONE.set(1)
     if not args.enforce_eager and torch.cuda.is_available():
