@@ -100,6 +100,9 @@ class AsyncLLMEngine:
                 continue
             metrics.REQUESTS_RUNNING.set(eng.scheduler.num_running)
             metrics.REQUESTS_WAITING.set(eng.scheduler.num_waiting)
+            hit = getattr(eng.pool, "hit_tokens", None)
+            if hit is not None:
+                metrics.PREFIX_CACHE_HIT_TOKENS.set(hit)
             # push fresh tokens to streams
             for sid, p in list(self._streams.items()):
                 seq = eng.seqs.get(sid)

@@ -33,6 +33,9 @@ REQUESTS_RUNNING = Gauge(
 REQUESTS_WAITING = Gauge(
     "vllm:num_requests_waiting", "Sequences queued for prefill",
     registry=REGISTRY)
+PREFIX_CACHE_HIT_TOKENS = Gauge(
+    "vllm:prefix_cache_hit_tokens", "Prompt tokens served from the "
+    "block-hash prefix cache (prefill skipped)", registry=REGISTRY)
 E2E_LATENCY = Counter(
     "vllm:e2e_request_latency_seconds", "Sum of request latencies",
     registry=REGISTRY)
