@@ -51,12 +51,26 @@ class ModelRunner:
 
     # ------------------------------------------------------------- setup
     def load_model(self, weights_path: Optional[str] = None, seed: int = 0):
+        awq_ckpt = False
         if weights_path:
+            import json
+            import os
+            cfg_json = os.path.join(weights_path, "config.json")
+            if os.path.exists(cfg_json):
+                with open(cfg_json) as f:
+                    qm = (json.load(f).get("quantization_config") or {}) \
+                        .get("quant_method", "")
+                awq_ckpt = qm == "awq"
             from ..models.loader import load_safetensors_weights
-            load_safetensors_weights(self.model, weights_path)
+            load_safetensors_weights(self.model, weights_path,
+                                     skip_projections=awq_ckpt)
         else:
             self.model.random_init(seed)
-        if self.cfg.model.quant_method in ("awq", "w4a16"):
+        if awq_ckpt:
+            from ..models.quant import load_awq_checkpoint
+            n = load_awq_checkpoint(self.model, weights_path)
+            logger.info("loaded %d AWQ-quantized linears", n)
+        elif self.cfg.model.quant_method in ("awq", "w4a16"):
             from ..models.quant import quantize_parallel_linears
             n = quantize_parallel_linears(self.model)
             logger.info("quantized %d linears to W4A16", n)

@@ -22,7 +22,8 @@ def _shard(t: torch.Tensor, dim: int) -> torch.Tensor:
     return t.narrow(dim, st.tp_rank * n, n).contiguous()
 
 
-def load_safetensors_weights(model, path: str) -> None:
+def load_safetensors_weights(model, path: str,
+                             skip_projections: bool = False) -> None:
     from safetensors.torch import safe_open
 
     cfg = model.cfg
@@ -53,17 +54,18 @@ def load_safetensors_weights(model, path: str) -> None:
         setp("lm_head.weight", _shard(get("lm_head.weight"), 0))
     for i in range(cfg.num_layers):
         pre = f"layers.{i}."
-        q = _shard(get(pre + "self_attn.q_proj.weight"), 0)
-        k = _shard(get(pre + "self_attn.k_proj.weight"), 0)
-        v = _shard(get(pre + "self_attn.v_proj.weight"), 0)
-        setp(pre + "self_attn.qkv_proj.weight", torch.cat([q, k, v], 0))
-        setp(pre + "self_attn.o_proj.weight",
-             _shard(get(pre + "self_attn.o_proj.weight"), 1))
-        g = _shard(get(pre + "mlp.gate_proj.weight"), 0)
-        u = _shard(get(pre + "mlp.up_proj.weight"), 0)
-        setp(pre + "mlp.gate_up_proj.weight", torch.cat([g, u], 0))
-        setp(pre + "mlp.down_proj.weight",
-             _shard(get(pre + "mlp.down_proj.weight"), 1))
+        if not skip_projections:   # AWQ checkpoints carry qweight instead
+            q = _shard(get(pre + "self_attn.q_proj.weight"), 0)
+            k = _shard(get(pre + "self_attn.k_proj.weight"), 0)
+            v = _shard(get(pre + "self_attn.v_proj.weight"), 0)
+            setp(pre + "self_attn.qkv_proj.weight", torch.cat([q, k, v], 0))
+            setp(pre + "self_attn.o_proj.weight",
+                 _shard(get(pre + "self_attn.o_proj.weight"), 1))
+            g = _shard(get(pre + "mlp.gate_proj.weight"), 0)
+            u = _shard(get(pre + "mlp.up_proj.weight"), 0)
+            setp(pre + "mlp.gate_up_proj.weight", torch.cat([g, u], 0))
+            setp(pre + "mlp.down_proj.weight",
+                 _shard(get(pre + "mlp.down_proj.weight"), 1))
         setp(pre + "input_layernorm", get(pre + "input_layernorm.weight"))
         setp(pre + "post_attention_layernorm",
              get(pre + "post_attention_layernorm.weight"))

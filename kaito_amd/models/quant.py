@@ -131,3 +131,78 @@ def quantize_model_linears(model: torch.nn.Module, group: int = 128,
                 setattr(parent, name, ql)
                 n += 1
     return n
+
+
+def load_awq_checkpoint(model: torch.nn.Module, path: str) -> int:
+    """Load a public-AWQ safetensors checkpoint into the engine's
+    (fused, TP-sharded) quantized linears. AWQ stores per-module
+    qweight i32 [K, N/8] / qzeros i32 [K/G, N/8] / scales [K/G, N]
+    (nibbles along N in AWQ_ORDER); we repack each projection into the
+    native layout, fuse q/k/v and gate/up along N, and shard for TP
+    (column-parallel along N rows, row-parallel along packed-K words).
+    Returns the number of modules loaded."""
+    import json
+    import os
+    from pathlib import Path
+    from safetensors.torch import safe_open
+    from ..parallel.state import get_state
+
+    with open(os.path.join(path, "config.json")) as f:
+        hf = json.load(f)
+    qc = hf.get("quantization_config", {})
+    group = qc.get("group_size", 128)
+    tensors = {}
+    for f in sorted(Path(path).glob("*.safetensors")):
+        with safe_open(str(f), framework="pt", device="cpu") as sf:
+            for k in sf.keys():
+                tensors[k] = sf.get_tensor(k)
+
+    def native(prefix):
+        def get(sfx):
+            for p in (f"model.{prefix}.{sfx}", f"{prefix}.{sfx}"):
+                if p in tensors:
+                    return tensors[p]
+            raise KeyError(f"{prefix}.{sfx}")
+        return repack_awq(get("qweight"), get("qzeros"),
+                          get("scales"), group)
+
+    st = get_state()
+    tp, rank = st.tp_size, st.tp_rank
+
+    def col_shard(t):                      # [N, ...] → rank's N rows
+        n = t.size(0) // tp
+        return t[rank * n:(rank + 1) * n].contiguous()
+
+    def row_shard(qw, sc, z):              # along K: packed words + groups
+        kw = qw.size(1) // tp
+        kg = sc.size(1) // tp
+        return (qw[:, rank * kw:(rank + 1) * kw].contiguous(),
+                sc[:, rank * kg:(rank + 1) * kg].contiguous(),
+                z[:, rank * kg:(rank + 1) * kg].contiguous())
+
+    n_loaded = 0
+    for name, mod in model.named_modules():
+        if not hasattr(mod, "quantize_from_packed"):
+            continue
+        lid = name.split(".")[1] if name.startswith("layers.") else None
+        if name.endswith("qkv_proj"):
+            parts = [native(f"layers.{lid}.self_attn.{p}")
+                     for p in ("q_proj", "k_proj", "v_proj")]
+            qw = torch.cat([col_shard(p[0]) for p in parts])
+            sc = torch.cat([col_shard(p[1]) for p in parts])
+            z = torch.cat([col_shard(p[2]) for p in parts])
+        elif name.endswith("gate_up_proj"):
+            parts = [native(f"layers.{lid}.mlp.{p}")
+                     for p in ("gate_proj", "up_proj")]
+            qw = torch.cat([col_shard(p[0]) for p in parts])
+            sc = torch.cat([col_shard(p[1]) for p in parts])
+            z = torch.cat([col_shard(p[2]) for p in parts])
+        elif name.endswith("o_proj"):
+            qw, sc, z = row_shard(*native(f"layers.{lid}.self_attn.o_proj"))
+        elif name.endswith("down_proj"):
+            qw, sc, z = row_shard(*native(f"layers.{lid}.mlp.down_proj"))
+        else:
+            continue
+        mod.quantize_from_packed(qw, sc, z, group)
+        n_loaded += 1
+    return n_loaded

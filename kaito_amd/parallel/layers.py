@@ -24,16 +24,24 @@ class _QuantMixin:
     _quantized = False
 
     def quantize_(self, group: int = 128) -> None:
-        from ..models.quant import quantize_w4, QuantLinear
+        from ..models.quant import quantize_w4
         w = self.weight.data
         assert w.size(1) % group == 0, (w.shape, group)
         qw, sc, z = quantize_w4(w.float(), group)
-        dev = w.device
-        self.register_buffer("qweight", qw.to(dev))
-        self.register_buffer("scales", sc.to(dev))
-        self.register_buffer("zeros", z.to(dev))
+        self.quantize_from_packed(qw, sc, z, group)
+
+    def quantize_from_packed(self, qweight, scales, zeros,
+                             group: int = 128) -> None:
+        """Install pre-packed 4-bit weights (native layout) — the AWQ
+        checkpoint load path."""
+        from ..models.quant import QuantLinear
+        dev = self.weight.device
+        dtype = self.weight.dtype
+        self.register_buffer("qweight", qweight.to(dev))
+        self.register_buffer("scales", scales.to(dev))
+        self.register_buffer("zeros", zeros.to(dev))
         self.q_group = group
-        self.weight = nn.Parameter(torch.empty(0, dtype=w.dtype, device=dev),
+        self.weight = nn.Parameter(torch.empty(0, dtype=dtype, device=dev),
                                    requires_grad=False)  # drop bf16 copy
         self._quantized = True
         self._gemv_max_m = QuantLinear.GEMV_MAX_M

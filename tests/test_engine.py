@@ -505,3 +505,113 @@ def test_seeded_sampling_reproducible():
 
     assert run(123) == run(123) == run(123, extra=True)
     assert run(123) != run(321)
+
+
+def test_awq_checkpoint_end_to_end(tmp_path):
+    """Quantize a reference engine's weights into a PUBLIC-AWQ-format
+    checkpoint on disk (qweight/qzeros/scales per split projection, AWQ
+    nibble order, quantization_config in config.json), then load it into
+    a fresh engine — decode must match the PTQ-quantized original
+    exactly (same 4-bit grid either way)."""
+    import json
+    import torch
+    from safetensors.torch import save_file
+    from kaito_amd.models.quant import AWQ_ORDER
+    mc = get_model_config("tiny-llama-test")
+    G = 64
+
+    base = LLMEngine(_cfg())          # reference weights (seed-fixed)
+    model = base.runner.model
+    tensors = {}
+    # non-quantized params (embeddings, norms, lm_head)
+    tensors["model.embed_tokens.weight"] = model.embed_tokens.weight.data.clone()
+    tensors["model.norm.weight"] = dict(model.named_parameters())["norm"].data.clone()
+    tensors["lm_head.weight"] = model.lm_head.weight.data.clone()
+
+    def to_awq(w):
+        # quantize [N, K] on the same 4-bit grid as quantize_w4, then emit
+        # AWQ layout: qweight i32 [K, N/8], qzeros i32 [K/G, N/8],
+        # scales [K/G, N]
+        N, K = w.shape
+        wf = w.float().reshape(N, K // G, G)
+        s = (wf.amax(-1) - wf.amin(-1)).clamp(min=1e-8) / 15.0
+        zq = (-wf.amin(-1) / s).round().clamp(0, 15)
+        q = (wf / s.unsqueeze(-1) + zq.unsqueeze(-1)).round().clamp(0, 15)
+        q = q.reshape(N, K).to(torch.int64).T.contiguous()      # [K, N]
+        zqT = zq.to(torch.int64).T.contiguous()                 # [K/G, N]
+        sT = s.T.contiguous()                                   # [K/G, N]
+
+        def pack_n(t):
+            out = torch.zeros(t.shape[0], N // 8, dtype=torch.int64)
+            for b in range(N // 8):
+                for pos, j in enumerate(AWQ_ORDER):
+                    out[:, b] |= t[:, b * 8 + pos] << (4 * j)
+            return out.to(torch.int32)
+
+        return pack_n(q), pack_n(zqT), sT
+
+    params = dict(model.named_parameters())
+    for i in range(mc.num_layers):
+        pre = f"layers.{i}."
+        qkv = params[pre + "self_attn.qkv_proj.weight"].data
+        nq = mc.num_heads * mc.head_dim
+        nk = mc.num_kv_heads * mc.head_dim
+        for nm, w in (("q_proj", qkv[:nq]), ("k_proj", qkv[nq:nq + nk]),
+                      ("v_proj", qkv[nq + nk:])):
+            a, z, s = to_awq(w)
+            t = f"model.{pre}self_attn.{nm}."
+            tensors[t + "qweight"], tensors[t + "qzeros"], \
+                tensors[t + "scales"] = a, z, s
+        a, z, s = to_awq(params[pre + "self_attn.o_proj.weight"].data)
+        t = f"model.{pre}self_attn.o_proj."
+        tensors[t + "qweight"], tensors[t + "qzeros"], \
+            tensors[t + "scales"] = a, z, s
+        gu = params[pre + "mlp.gate_up_proj.weight"].data
+        ii = mc.intermediate_size
+        for nm, w in (("gate_proj", gu[:ii]), ("up_proj", gu[ii:])):
+            a, z, s = to_awq(w)
+            t = f"model.{pre}mlp.{nm}."
+            tensors[t + "qweight"], tensors[t + "qzeros"], \
+                tensors[t + "scales"] = a, z, s
+        a, z, s = to_awq(params[pre + "mlp.down_proj.weight"].data)
+        t = f"model.{pre}mlp.down_proj."
+        tensors[t + "qweight"], tensors[t + "qzeros"], \
+            tensors[t + "scales"] = a, z, s
+        for ln in ("input_layernorm", "post_attention_layernorm"):
+            tensors[f"model.{pre}{ln}.weight"] = params[pre + ln].data.clone()
+
+    d = tmp_path / "awq"
+    d.mkdir()
+    save_file({k: v.contiguous() for k, v in tensors.items()},
+              str(d / "model.safetensors"))
+    (d / "config.json").write_text(json.dumps({
+        "architectures": ["LlamaForCausalLM"],
+        "quantization_config": {"quant_method": "awq", "group_size": G,
+                                "bits": 4}}))
+
+    eng = LLMEngine(_cfg(), weights_path=str(d))
+    assert sum(1 for m in eng.runner.model.modules()
+               if getattr(m, "_quantized", False)) == 4 * mc.num_layers
+    got = eng.generate([[3, 14, 15, 92]],
+                       SamplingParams(max_tokens=6, ignore_eos=True)
+                       )[0].output_token_ids
+    # oracle: PTQ the same weights in-process on the same grid
+    import dataclasses
+    mcq = dataclasses.replace(mc, quant_method="w4a16")
+    ptq = LLMEngine(_cfg(model=mcq))
+    from kaito_amd.models.quant import quantize_parallel_linears  # noqa: F401
+    # PTQ uses G=128 default — requantize at G=64 for exact grid match
+    from kaito_amd.models.quant import quantize_w4  # noqa: F401
+    ref = LLMEngine(_cfg())
+    n = 0
+    for name, mod in ref.runner.model.named_modules():
+        if hasattr(mod, "quantize_") and any(
+                name.endswith(sfx) for sfx in
+                ("qkv_proj", "o_proj", "gate_up_proj", "down_proj")):
+            mod.quantize_(G)
+            n += 1
+    assert n == 4 * mc.num_layers
+    expect = ref.generate([[3, 14, 15, 92]],
+                          SamplingParams(max_tokens=6, ignore_eos=True)
+                          )[0].output_token_ids
+    assert got == expect
