@@ -169,6 +169,7 @@ def run_sft(model, tokenizer, texts: List[str], cfg: Dict,
         if 0 < max_steps <= step:
             break
     return {"steps": step, "final_loss": losses[-1] if losses else None,
+            "first_loss": losses[0] if losses else None,
             "train_seconds": time.monotonic() - t0}
 
 
