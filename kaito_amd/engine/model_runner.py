@@ -284,6 +284,13 @@ class ModelRunner:
         shared memory pool is sized once)."""
         if not self.is_gpu or self.cfg.enforce_eager:
             return
+        if self.cfg.model.num_experts > 0:
+            # the expert-sorted grouped-GEMM MoE path reads per-expert
+            # counts on the host (data-dependent shapes) — illegal inside
+            # hipGraph capture; MoE decodes eagerly until the fused MoE
+            # kernel lands (docs/ROADMAP.md)
+            logger.info("MoE model: skipping decode graph capture")
+            return
         if not hasattr(self, "_buf"):
             self._init_decode_buffers()
         self._buf["seq_lens"].fill_(1)  # benign shapes for capture
