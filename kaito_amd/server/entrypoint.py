@@ -75,6 +75,28 @@ def main(argv=None):
     args = build_parser().parse_args(argv)
     merge_config_file(args, args.kaito_config_file)
 
+    if args.data_parallel_size > 1:
+        # tier-1 DP pod: one engine process per GPU behind a front-end
+        # proxy (planner.py:50 gives sub-144GiB models all 8 GPUs as
+        # independent TP=1 replicas)
+        from .dp_frontend import serve_dp
+        raw = list(argv if argv is not None else sys.argv[1:])
+        child = []
+        skip = False
+        for i, a in enumerate(raw):
+            if skip:
+                skip = False
+                continue
+            if a in ("--data-parallel-size", "--port"):
+                skip = True
+                continue
+            if a.startswith("--data-parallel-size=") or \
+                    a.startswith("--port="):
+                continue
+            child.append(a)
+        return serve_dp(child, args.data_parallel_size, args.host,
+                        args.port)
+
     import torch
     from ..engine import EngineConfig, LLMEngine
     from ..models import get_model_config
