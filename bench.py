@@ -36,6 +36,9 @@ def parse_args():
     p.add_argument("--in-tokens", type=int, default=200)
     p.add_argument("--out-tokens", type=int, default=200)
     p.add_argument("--eager", action="store_true")
+    p.add_argument("--quantization", default="", choices=["", "w4a16"],
+                   help="W4A16 weight-only serving (NOT the headline "
+                        "config; reported dtype reflects it)")
     p.add_argument("--mixed-steps", action="store_true",
                    help="enable mixed (overlapped prefill+decode) steps")
     p.add_argument("--tune-gemms", default=None, nargs="?", const="",
@@ -62,6 +65,9 @@ def main():
     ps.init_parallel(tp_size=args.tp)
 
     mc = get_model_config(args.model)
+    if args.quantization:
+        import dataclasses
+        mc = dataclasses.replace(mc, quant_method=args.quantization)
     max_len = args.in_tokens + args.out_tokens + 16
     cfg = EngineConfig(
         model=mc,
@@ -158,7 +164,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": ("w4a16" if args.quantization else "bf16"),
             "data": "synthetic",
             "ttft_p50_ms": round(ttft_p50, 1) if ttft_p50 is not None else None,
             "itl_p50_ms": round(statistics.median(step_ms), 2) if step_ms else None,
