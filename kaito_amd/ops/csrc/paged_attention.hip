@@ -48,15 +48,18 @@ __global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
   // 16B lane-load is CONTIGUOUS across the 8 dc-lanes (dense 128B/token
   // transactions; the naive dc*DL mapping strides 32B between lanes and
   // halves effective HBM bandwidth).
-  float qreg[G][DL];
+  // Q stays PACKED bf16: the QK dot runs on v_dot2c_f32_bf16 (2-wide
+  // bf16 dot-accumulate, one VALU op per dim pair — vs convert+FMA at 2
+  // ops/dim), and packed Q halves its VGPR footprint. bf16xbf16 products
+  // are exact in the f32 accumulator, so numerics match the f32 path up
+  // to summation order.
+  short8_t qreg[G][DL / 8];
 #pragma unroll
   for (int g = 0; g < G; g++) {
 #pragma unroll
     for (int vv = 0; vv < DL / 8; vv++) {
-      short8_t x = *reinterpret_cast<const short8_t*>(
+      qreg[g][vv] = *reinterpret_cast<const short8_t*>(
           q + (int64_t)seq * q_stride + (kvh * G + g) * D + vv * 64 + dc * 8);
-#pragma unroll
-      for (int j = 0; j < 8; j++) qreg[g][vv * 8 + j] = bf16_to_f32(x[j]);
     }
   }
 
@@ -103,20 +106,23 @@ __global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
     const bool valid_c = c < nchunks;
     if (!valid_c) return;
     const bool valid = (c * 8 + tg) < seq_len;
-    float kreg[DL];
-#pragma unroll
-    for (int vv = 0; vv < NV; vv++)
-#pragma unroll
-      for (int j = 0; j < 8; j++) kreg[vv * 8 + j] = bf16_to_f32(kd[vv][j]);
+    typedef __bf16 bf16x2_t __attribute__((ext_vector_type(2)));
     float s[G];
 #pragma unroll
     for (int g = 0; g < G; g++) {
       float p = 0.f;
 #pragma unroll
-      for (int j = 0; j < DL; j++) p += qreg[g][j] * kreg[j];
+      for (int vv = 0; vv < NV; vv++) {
+        const bf16x2_t* kp = reinterpret_cast<const bf16x2_t*>(&kd[vv]);
+        const bf16x2_t* qp = reinterpret_cast<const bf16x2_t*>(&qreg[g][vv]);
+#pragma unroll
+        for (int jj = 0; jj < 4; jj++)
+          p = __builtin_amdgcn_fdot2_f32_bf16(qp[jj], kp[jj], p, false);
+      }
       p = group_reduce_sum<8>(p) * scale;   // dot over 8 dc-lanes
       s[g] = valid ? p : -1e30f;
     }
+    float kreg[DL];
 #pragma unroll
     for (int vv = 0; vv < NV; vv++)
 #pragma unroll
