@@ -131,19 +131,23 @@ __global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
 #pragma unroll
     for (int g = 0; g < G; g++) {
       float p;
+      // float2 views: the rescale and accumulate lower to packed
+      // v_pk_mul_f32 / v_pk_fma_f32 (2 lanes per VALU op)
+      float2_t* a2 = reinterpret_cast<float2_t*>(acc[g]);
       if (s[g] > m[g]) {          // lane-local rescale (no cross-lane max)
         const float corr = __expf(m[g] - s[g]);
         l[g] *= corr;
 #pragma unroll
-        for (int j = 0; j < DL; j++) acc[g][j] *= corr;
+        for (int j = 0; j < DL / 2; j++) a2[j] *= corr;
         m[g] = s[g];
         p = 1.0f;
       } else {
         p = __expf(s[g] - m[g]);
       }
       l[g] += p;
+      const float2_t* k2 = reinterpret_cast<const float2_t*>(kreg);
 #pragma unroll
-      for (int j = 0; j < DL; j++) acc[g][j] += p * kreg[j];
+      for (int j = 0; j < DL / 2; j++) a2[j] += p * k2[j];
     }
   };
 
