@@ -151,6 +151,9 @@ __global__ __launch_bounds__(256, G <= 4 ? 3 : 2) void paged_attention_kernel(
     }
   };
 
+  // (re-tried a 2-deep load pipeline post-dot2: the double buffer pushes
+  // the kernel to 168 VGPR with 156 B/lane of spill inside the 3-wave
+  // bound — worse than just letting occupancy hide the latency)
   for (int c = wave; c < nchunks; c += NW) {
     short8_t kd[NV], vd[NV];
     load_chunk(c, kd, vd);
