@@ -111,20 +111,31 @@ def build_dp_app(ports: List[int], procs=None) -> FastAPI:
     @app.api_route("/{path:path}",
                    methods=["GET", "POST", "PUT", "DELETE"])
     async def proxy(path: str, request: Request):
-        i = pick()
-        outstanding[i] += 1
-        try:
-            body = await request.body()
-            r = await client.request(
-                request.method, f"http://127.0.0.1:{ports[i]}/{path}",
-                content=body,
-                headers={k: v for k, v in request.headers.items()
-                         if k.lower() not in ("host", "content-length")},
-                params=dict(request.query_params))
-            return Response(r.content, status_code=r.status_code,
-                            media_type=r.headers.get("content-type"))
-        finally:
-            outstanding[i] -= 1
+        body = await request.body()
+        headers = {k: v for k, v in request.headers.items()
+                   if k.lower() not in ("host", "content-length")}
+        params = dict(request.query_params)
+        last_exc = None
+        tried = set()
+        # a dead/unreachable replica must not take the endpoint down:
+        # retry the request on each remaining replica once
+        for _ in range(len(ports)):
+            i = min((j for j in range(len(ports)) if j not in tried),
+                    key=lambda j: outstanding[j])
+            tried.add(i)
+            outstanding[i] += 1
+            try:
+                r = await client.request(
+                    request.method, f"http://127.0.0.1:{ports[i]}/{path}",
+                    content=body, headers=headers, params=params)
+                return Response(r.content, status_code=r.status_code,
+                                media_type=r.headers.get("content-type"))
+            except httpx.TransportError as e:
+                last_exc = e
+            finally:
+                outstanding[i] -= 1
+        return Response(f"all replicas unreachable: {last_exc}",
+                        status_code=503)
     return app
 
 
