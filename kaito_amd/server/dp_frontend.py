@@ -14,7 +14,6 @@ probe and KEDA see whole-pod numbers.
 """
 from __future__ import annotations
 
-import asyncio
 import os
 import re
 import subprocess
@@ -80,9 +79,6 @@ def build_dp_app(ports: List[int], procs=None) -> FastAPI:
     outstanding = [0] * len(ports)
     client = httpx.AsyncClient(timeout=PROXY_TIMEOUT_S)
     app.state.outstanding = outstanding
-
-    def pick() -> int:
-        return min(range(len(ports)), key=lambda i: outstanding[i])
 
     @app.get("/health")
     async def health():
