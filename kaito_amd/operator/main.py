@@ -132,13 +132,9 @@ def main(argv=None):
     p.add_argument("--reconcile-interval", type=float, default=5.0)
     args = p.parse_args(argv)
     gates = parse_feature_gates(args.feature_gates)
-    client = FakeKubeClient()  # in-cluster impl wraps `kubernetes` pkg
-    try:
-        import kubernetes  # noqa: F401
-        logger.warning("kubernetes package found but in-cluster client is "
-                       "not wired in this build; using in-memory store")
-    except ImportError:
-        pass
+    from .kubeclient_incluster import make_kube_client
+    client = make_kube_client()
+    logger.info("kube client: %s", type(client).__name__)
     loop = OperatorLoop(client, args.cloud_provider, args.node_provisioner,
                         args.preset_image, gates)
     loop.run(args.reconcile_interval)

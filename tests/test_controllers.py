@@ -601,3 +601,15 @@ def test_inferenceset_expectations_block_stale_scale(monkeypatch):
     rec.expectations.creation_observed("default/pool")
     r3 = rec.reconcile(iset)
     assert r3.created == 0            # already at desired; acts normally
+
+
+def test_make_kube_client_falls_back_to_fake():
+    """Air-gapped image has no `kubernetes` package → factory returns the
+    in-memory fake; selector helper stays deterministic."""
+    from kaito_amd.operator.kubeclient import FakeKubeClient
+    from kaito_amd.operator.kubeclient_incluster import (_selector_str,
+                                                         make_kube_client)
+    c = make_kube_client()
+    assert isinstance(c, FakeKubeClient)
+    assert _selector_str({"b": "2", "a": "1"}) == "a=1,b=2"
+    assert _selector_str(None) is None
