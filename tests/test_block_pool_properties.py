@@ -84,3 +84,26 @@ def test_plain_pool_conservation(a, b):
     p.free(x)                         # second free clears the fork
     p.free(y)
     assert p.num_free == 12 and not p._refcount
+
+
+# ---- quantization properties (same file: fast property suite) ----------
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+from hypothesis import given as _given  # noqa: E402
+
+
+@settings(max_examples=50, deadline=None)
+@_given(st.integers(1, 8), st.integers(1, 4), st.integers(0, 2**31 - 1),
+        st.floats(0.01, 100.0))
+def test_w4_quantization_error_bounded(nmul, kmul, seed, scale):
+    """For any weight distribution/scale, 4-bit group dequant error stays
+    within half a quantization step everywhere."""
+    from kaito_amd.models.quant import quantize_w4
+    from kaito_amd.ops import torch_ref
+    g = torch.Generator().manual_seed(seed)
+    N, K, G = 8 * nmul, 64 * kmul, 64
+    w = torch.randn(N, K, generator=g) * scale
+    qw, s, z = quantize_w4(w, G)
+    deq = torch_ref.w4a16_unpack(qw, s, z, G)
+    step = s.repeat_interleave(G, dim=1)
+    assert ((deq - w).abs() <= step * 0.5 + 1e-5 * scale).all()
