@@ -107,3 +107,31 @@ def test_w4_quantization_error_bounded(nmul, kmul, seed, scale):
     deq = torch_ref.w4a16_unpack(qw, s, z, G)
     step = s.repeat_interleave(G, dim=1)
     assert ((deq - w).abs() <= step * 0.5 + 1e-5 * scale).all()
+
+
+@settings(max_examples=50, deadline=None)
+@_given(st.integers(0, 2**31 - 1), st.integers(1, 20), st.floats(0.2, 2.0))
+def test_sampler_topk_topp_support(seed, k, temp):
+    """With top_k set, the sampled token must come from the top-k logits;
+    with top_p, from the minimal nucleus prefix (plus the argmax row when
+    tied). Greedy rows always return the argmax."""
+    from kaito_amd.engine.sampler import Sampler
+    from kaito_amd.engine.sequence import Sequence, SamplingParams
+    g = torch.Generator().manual_seed(seed)
+    logits = torch.randn(3, 64, generator=g)
+    seqs = [Sequence(0, [1], SamplingParams(temperature=temp, top_k=k,
+                                            seed=seed)),
+            Sequence(1, [1], SamplingParams(temperature=0.0)),
+            Sequence(2, [1], SamplingParams(temperature=temp, top_p=0.5,
+                                            seed=seed))]
+    s = Sampler("cpu")
+    toks = s.sample(logits.clone(), seqs)
+    topk_ids = set(torch.topk(logits[0], min(k, 64)).indices.tolist())
+    assert int(toks[0]) in topk_ids
+    assert int(toks[1]) == int(logits[1].argmax())
+    # nucleus: token must be inside the smallest prefix with mass >= 0.5
+    probs = torch.softmax(logits[2] / temp, dim=-1)
+    sp, si = torch.sort(probs, descending=True)
+    cum = torch.cumsum(sp, 0)
+    ncut = int((cum < 0.5).sum()) + 1
+    assert int(toks[2]) in set(si[:ncut].tolist())
