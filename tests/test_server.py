@@ -257,3 +257,29 @@ def test_completions_n_choices(client):
     assert [c["text"] for c in r2.json()["choices"]] == \
         [c["text"] for c in ch]
     assert r.json()["usage"]["completion_tokens"] == 15
+
+
+def test_download_monitor_progress(tmp_path):
+    """Bytes-on-disk sampling against the safetensors index total
+    (reference: download-progress gauges, inference_api.py:265-365)."""
+    import json
+    from kaito_amd.server.download_monitor import (DownloadMonitor,
+                                                   bytes_on_disk,
+                                                   expected_total_bytes)
+    d = tmp_path / "w"
+    d.mkdir()
+    (d / "model.safetensors.index.json").write_text(
+        json.dumps({"metadata": {"total_size": 1000}}))
+    assert expected_total_bytes(str(d)) == 1000
+    mon = DownloadMonitor(str(d), interval_s=0.01)
+    assert mon.progress() == 0.0
+    (d / "a.safetensors").write_bytes(b"x" * 400)
+    assert bytes_on_disk(str(d)) == 400
+    assert mon.progress() == 0.4
+    (d / "b.safetensors").write_bytes(b"y" * 900)   # over-report clamps
+    assert mon.progress() == 1.0
+    mon.start()
+    time.sleep(0.05)
+    mon.stop(done=True)
+    from kaito_amd.server import metrics as M
+    assert M.MODEL_DOWNLOAD_DONE._value.get() == 1
