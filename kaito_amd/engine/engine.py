@@ -95,8 +95,18 @@ class LLMEngine:
                     seq_id: Optional[int] = None,
                     lora_name: Optional[str] = None) -> int:
         sid = seq_id if seq_id is not None else next(self._next_id)
-        seq = Sequence(sid, list(prompt_token_ids),
-                       sampling or SamplingParams())
+        sampling = sampling or SamplingParams()
+        # clamp generation to the model context: an unclamped request
+        # would outgrow the fixed per-seq block-table width
+        room = self.runner.max_model_len - len(prompt_token_ids)
+        if room <= 0:
+            raise ValueError(
+                f"prompt ({len(prompt_token_ids)} tokens) leaves no room "
+                f"in max_model_len={self.runner.max_model_len}")
+        if sampling.max_tokens > room:
+            import dataclasses
+            sampling = dataclasses.replace(sampling, max_tokens=room)
+        seq = Sequence(sid, list(prompt_token_ids), sampling)
         if lora_name is not None:
             mgr = self.runner.lora_manager
             if mgr is None:

@@ -615,3 +615,19 @@ def test_awq_checkpoint_end_to_end(tmp_path):
                           SamplingParams(max_tokens=6, ignore_eos=True)
                           )[0].output_token_ids
     assert got == expect
+
+
+def test_max_tokens_clamped_to_model_len():
+    """A request whose max_tokens would outgrow max_model_len is clamped
+    at admission (finish_reason length at the context edge) instead of
+    overrunning the fixed block-table width mid-decode."""
+    cfg = _cfg(max_model_len=48, num_gpu_blocks=512)
+    eng = LLMEngine(cfg)
+    prompt = list(range(10, 40))                   # 30 tokens
+    out = eng.generate([prompt], SamplingParams(max_tokens=10_000,
+                                                ignore_eos=True))[0]
+    assert len(out.output_token_ids) == 48 - 30    # clamped to room
+    assert out.finish_reason == "length"
+    import pytest as _pt
+    with _pt.raises(ValueError):
+        eng.add_request(list(range(48)), SamplingParams(max_tokens=4))
