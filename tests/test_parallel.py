@@ -210,3 +210,50 @@ def _body_pp_engine(rank, world):
 
 def test_pp_engine_world2():
     _spawn("_body_pp_engine", port=29617)
+
+
+def _body_quant_tp(rank, world):
+    """TP-sharded W4A16: column and row quantized linears must reproduce
+    the full (unsharded) dequantized matmul — exercises the packed-N row
+    slicing and packed-K word/group slicing used by the AWQ loader."""
+    from kaito_amd.models.quant import quantize_w4
+    from kaito_amd.ops import torch_ref
+    from kaito_amd.parallel.layers import (ColumnParallelLinear,
+                                           RowParallelLinear)
+    torch.manual_seed(1)
+    K, N, G = 256, 64, 64
+    w_full = torch.randn(N, K)
+    qw, s, z = quantize_w4(w_full, G)
+    deq = torch_ref.w4a16_unpack(qw, s, z, G)
+    x = torch.randn(3, K, dtype=torch.bfloat16)
+
+    # column parallel: shard packed rows along N
+    col = ColumnParallelLinear(K, N)
+    n = N // world
+    col.quantize_from_packed(qw[rank * n:(rank + 1) * n],
+                             s[rank * n:(rank + 1) * n],
+                             z[rank * n:(rank + 1) * n], G)
+    out = col(x).float()
+    expect = (x.float() @ deq.T)[:, rank * n:(rank + 1) * n]
+    assert torch.allclose(out, expect, atol=5e-2, rtol=5e-2), "q col"
+
+    # row parallel: shard packed K words + scale groups; all-reduce sums
+    w2 = torch.randn(N, K)                   # here K=in=256 → out=N
+    qw2, s2, z2 = quantize_w4(w2, G)
+    deq2 = torch_ref.w4a16_unpack(qw2, s2, z2, G)
+    row = RowParallelLinear(K, N)
+    kw = (K // 8) // world
+    kg = (K // G) // world
+    row.quantize_from_packed(
+        qw2[:, rank * kw:(rank + 1) * kw].contiguous(),
+        s2[:, rank * kg:(rank + 1) * kg].contiguous(),
+        z2[:, rank * kg:(rank + 1) * kg].contiguous(), G)
+    xs = torch.randn(3, K, dtype=torch.bfloat16)
+    kper = K // world
+    out2 = row(xs[:, rank * kper:(rank + 1) * kper]).float()
+    expect2 = xs.float() @ deq2.float().T
+    assert torch.allclose(out2, expect2, atol=0.25, rtol=5e-2), "q row"
+
+
+def test_tp_quantized_linears_world2():
+    _spawn("_body_quant_tp", port=29651)
