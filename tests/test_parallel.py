@@ -257,3 +257,33 @@ def _body_quant_tp(rank, world):
 
 def test_tp_quantized_linears_world2():
     _spawn("_body_quant_tp", port=29651)
+
+
+def _body_kv_p2p(rank, world):
+    """P2PGroupConnector over gloo: rank 0 (prefill role) sends a KV
+    payload to rank 1 (decode role) via torch.distributed send/recv — the
+    CPU stand-in for the RCCL-over-xGMI same-node P/D path."""
+    from kaito_amd.engine.kv_transfer import KVPayload, P2PGroupConnector
+    torch.manual_seed(3)
+    layers = [(torch.randn(2, 4, 16, 8, dtype=torch.bfloat16),
+               torch.randn(2, 4, 16, 8, dtype=torch.bfloat16))
+              for _ in range(3)]
+    if rank == 0:
+        conn = P2PGroupConnector(peer_rank=1)
+        conn.send(KVPayload("req-7", [1, 2, 3], layers, first_token=42))
+    else:
+        conn = P2PGroupConnector(peer_rank=0)
+        got = conn.recv()
+        assert got.request_id == "req-7"
+        assert got.token_ids == [1, 2, 3] and got.first_token == 42
+        assert len(got.layers) == 3
+        torch.manual_seed(3)
+        expect = [(torch.randn(2, 4, 16, 8, dtype=torch.bfloat16),
+                   torch.randn(2, 4, 16, 8, dtype=torch.bfloat16))
+                  for _ in range(3)]
+        for (k, v), (ek, ev) in zip(got.layers, expect):
+            assert torch.equal(k, ek) and torch.equal(v, ev)
+
+
+def test_kv_p2p_connector_world2():
+    _spawn("_body_kv_p2p", port=29671)
