@@ -1,5 +1,7 @@
 import argparse
-import dataclasses, time, torch
+import dataclasses, os, sys, time
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+import torch
 from kaito_amd.engine import EngineConfig, LLMEngine, SamplingParams
 from kaito_amd.models import get_model_config
 from kaito_amd.parallel.state import init_parallel
