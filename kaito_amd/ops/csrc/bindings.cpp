@@ -36,6 +36,11 @@ void w4a16_gemv(at::Tensor out, at::Tensor x, at::Tensor qweight,
                 at::Tensor scales, at::Tensor zeros, int64_t group);
 void w4a16_dequant(at::Tensor out, at::Tensor qweight, at::Tensor scales,
                    at::Tensor zeros, int64_t group);
+void allreduce_rmsnorm(at::Tensor out, at::Tensor ptrs, at::Tensor weight,
+                       double eps);
+at::Tensor ipc_handle(at::Tensor t);
+int64_t ipc_open(at::Tensor handle_bytes);
+void ipc_close(int64_t ptr);
 }  // namespace kaito
 
 TORCH_LIBRARY(kaito, m) {
@@ -54,6 +59,10 @@ TORCH_LIBRARY(kaito, m) {
   m.def("paged_read_bw(Tensor(a!) out, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, int mode) -> ()");
   m.def("w4a16_gemv(Tensor(a!) out, Tensor x, Tensor qweight, Tensor scales, Tensor zeros, int group) -> ()");
   m.def("w4a16_dequant(Tensor(a!) out, Tensor qweight, Tensor scales, Tensor zeros, int group) -> ()");
+  m.def("allreduce_rmsnorm(Tensor(a!) out, Tensor ptrs, Tensor weight, float eps) -> ()");
+  m.def("ipc_handle(Tensor t) -> Tensor");
+  m.def("ipc_open(Tensor handle_bytes) -> int");
+  m.def("ipc_close(int ptr) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
@@ -72,4 +81,8 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("paged_read_bw", &kaito::paged_read_bw);
   m.impl("w4a16_gemv", &kaito::w4a16_gemv);
   m.impl("w4a16_dequant", &kaito::w4a16_dequant);
+  m.impl("allreduce_rmsnorm", &kaito::allreduce_rmsnorm);
+  m.impl("ipc_handle", &kaito::ipc_handle);
+  m.impl("ipc_open", &kaito::ipc_open);
+  m.impl("ipc_close", &kaito::ipc_close);
 }
