@@ -287,3 +287,17 @@ def _body_kv_p2p(rank, world):
 
 def test_kv_p2p_connector_world2():
     _spawn("_body_kv_p2p", port=29671)
+
+
+def test_one_shot_fused_local_cpu_ref():
+    """Fused allreduce+RMSNorm reference math (CPU path of fused_local;
+    the HIP kernel is compared against this shape on GPU)."""
+    from kaito_amd.parallel.one_shot import fused_local
+    torch.manual_seed(5)
+    xs = [torch.randn(4, 128, dtype=torch.bfloat16) for _ in range(8)]
+    w = torch.randn(128, dtype=torch.bfloat16)
+    out = fused_local(xs, w, 1e-5)
+    acc = sum(x.float() for x in xs)
+    var = acc.pow(2).mean(-1, keepdim=True)
+    expect = (acc * torch.rsqrt(var + 1e-5) * w.float())
+    assert torch.allclose(out.float(), expect, atol=3e-2, rtol=3e-2)
