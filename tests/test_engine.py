@@ -631,3 +631,60 @@ def test_max_tokens_clamped_to_model_len():
     import pytest as _pt
     with _pt.raises(ValueError):
         eng.add_request(list(range(48)), SamplingParams(max_tokens=4))
+
+
+def test_abort_waiting_running_and_finished():
+    """abort() frees blocks and drops the request in every lifecycle
+    state: still waiting, mid-decode (running), and already finished —
+    remaining sequences keep decoding to the oracle."""
+    cfg = _cfg()
+    eng = LLMEngine(cfg)
+    sp = SamplingParams(max_tokens=8, ignore_eos=True)
+    keep = eng.add_request([3, 14, 15, 92, 65], sp)
+    victim_waiting = eng.add_request([1, 2, 3], sp)
+    eng.abort(victim_waiting)                       # still WAITING
+    eng.step()                                      # prefill keep
+    eng.step()
+    victim_running = eng.add_request([7, 8, 9], sp)
+    eng.step()                                      # victim prefills
+    eng.step()
+    free_before = eng.pool.num_free + len(getattr(eng.pool, "_cached", {}))
+    eng.abort(victim_running)                       # mid-decode
+    assert eng.pool.num_free + len(getattr(eng.pool, "_cached", {})) \
+        > free_before - 1                           # blocks returned
+    while eng.has_unfinished():
+        eng.step()
+    expect = _naive_generate(eng.runner.model, cfg, [3, 14, 15, 92, 65], 8)
+    assert eng.seqs[keep].output_token_ids == expect
+    eng.abort(keep)                                 # already finished: noop
+    assert victim_waiting not in eng.seqs
+    assert victim_running not in eng.seqs
+
+
+def test_async_engine_abort_on_consumer_exit():
+    """Breaking out of the async stream aborts the request in the engine
+    (reference semantics: disconnect cancels the vLLM request)."""
+    import asyncio
+    from kaito_amd.server.async_engine import AsyncLLMEngine
+    cfg = _cfg()
+    eng = LLMEngine(cfg)
+    aeng = AsyncLLMEngine(eng).start()
+    try:
+        async def consume_two():
+            n = 0
+            async for item in aeng.generate(
+                    [5, 6, 7], SamplingParams(max_tokens=64,
+                                              ignore_eos=True)):
+                n += 1
+                if n == 2:
+                    break                       # consumer walks away
+            return n
+
+        assert asyncio.run(consume_two()) == 2
+        import time as _t
+        deadline = _t.monotonic() + 5
+        while _t.monotonic() < deadline and eng.scheduler.has_work():
+            _t.sleep(0.05)
+        assert not eng.scheduler.has_work(), "request not aborted"
+    finally:
+        aeng.shutdown()
