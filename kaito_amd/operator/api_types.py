@@ -148,6 +148,9 @@ class Workspace:
     annotations: Dict[str, str] = field(default_factory=dict)
     labels: Dict[str, str] = field(default_factory=dict)
     status: WorkspaceStatus = field(default_factory=WorkspaceStatus)
+    # set when the API server marks the object for deletion
+    deletionTimestamp: Optional[str] = None
+    finalizers: List[str] = field(default_factory=list)
 
     def validate(self, sku_handler=None, known_presets=None) -> None:
         """Admission validation — the semantic checks from

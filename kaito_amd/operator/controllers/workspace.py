@@ -16,7 +16,8 @@ from ..api_types import (ANNOTATION_DISABLE_BENCHMARK,
                          ANNOTATION_WORKSPACE_HASH, COND_INFERENCE_READY,
                          COND_JOB_STARTED, COND_NODECLAIM_READY,
                          COND_NODES_READY, COND_RESOURCE_READY,
-                         COND_WORKSPACE_SUCCEEDED, Condition,
+                         COND_WORKSPACE_DELETING, COND_WORKSPACE_SUCCEEDED,
+                         Condition, FINALIZER_WORKSPACE,
                          LABEL_WORKSPACE_NAME, ValidationError, Workspace)
 from ..estimator import NodeEstimateRequest, estimate_node_count
 from ..kubeclient import KubeClient, NotFound
@@ -116,6 +117,13 @@ class WorkspaceReconciler:
     # ------------------------------------------------------------ reconcile
     def reconcile(self, ws: Workspace) -> ReconcileResult:
         conds = ws.status.conditions
+        if ws.deletionTimestamp:
+            # deletion flow (reference: Reconcile :117-127 →
+            # garbageCollectWorkspace): tear down children, then drop the
+            # finalizer so the API server can remove the object
+            return self._finalize(ws)
+        if FINALIZER_WORKSPACE not in ws.finalizers:
+            ws.finalizers.append(FINALIZER_WORKSPACE)
         try:
             ws.validate(sku_handler=None)
         except ValidationError as e:
@@ -162,6 +170,24 @@ class WorkspaceReconciler:
         return res
 
     # ------------------------------------------------------------ inference
+    def _finalize(self, ws: Workspace) -> ReconcileResult:
+        _cond(ws.status.conditions, COND_WORKSPACE_DELETING, True,
+              "Deleting", "tearing down child resources")
+        for kind in ("StatefulSet", "Job"):
+            try:
+                self.client.delete(kind, ws.namespace, ws.name)
+            except NotFound:
+                pass
+        for svc in (ws.name, f"{ws.name}-headless"):
+            try:
+                self.client.delete("Service", ws.namespace, svc)
+            except NotFound:
+                pass
+        self.provisioner.delete_nodes(ws)
+        if FINALIZER_WORKSPACE in ws.finalizers:
+            ws.finalizers.remove(FINALIZER_WORKSPACE)
+        return ReconcileResult()
+
     def _apply_inference(self, ws: Workspace, model, gpu,
                          target: int) -> ReconcileResult:
         conds = ws.status.conditions

@@ -40,6 +40,8 @@ def workspace_from_obj(obj: Dict) -> Workspace:
             count=res.get("count")),
         annotations=obj["metadata"].get("annotations", {}) or {},
         labels=obj["metadata"].get("labels", {}) or {},
+        deletionTimestamp=obj["metadata"].get("deletionTimestamp"),
+        finalizers=obj["metadata"].get("finalizers", []) or [],
     )
     inf = spec.get("inference")
     if inf:
