@@ -209,6 +209,8 @@ class InferenceSet:
     namespace: str = "default"
     spec: InferenceSetSpec = field(default_factory=InferenceSetSpec)
     status: InferenceSetStatus = field(default_factory=InferenceSetStatus)
+    deletionTimestamp: Optional[str] = None
+    finalizers: List[str] = field(default_factory=list)
 
     def validate(self) -> None:
         if self.spec.replicas < 0:

@@ -75,6 +75,29 @@ class InferenceSetReconciler:
     def _is_ready(ws_obj: Dict) -> bool:
         return ws_obj.get("status", {}).get("state") == "Running"
 
+    def _finalize(self, iset: InferenceSet) -> ISReconcileResult:
+        """Deletion flow: remove child Workspaces, then the finalizer
+        (reference: InferenceSetDeleting + finalizer drop)."""
+        from ..api_types import (COND_INFERENCESET_DELETING,
+                                 FINALIZER_INFERENCESET)
+        res = ISReconcileResult()
+        for obj in self._list_children(iset):
+            self.client.delete("Workspace", iset.namespace,
+                               obj["metadata"]["name"])
+            res.deleted += 1
+        found = False
+        for c in iset.status.conditions:
+            if c.type == COND_INFERENCESET_DELETING:
+                c.status = "True"
+                found = True
+        if not found:
+            iset.status.conditions.append(Condition(
+                COND_INFERENCESET_DELETING, "True", "Deleting"))
+        if FINALIZER_INFERENCESET in iset.finalizers:
+            iset.finalizers.remove(FINALIZER_INFERENCESET)
+        self.expectations.delete(f"{iset.namespace}/{iset.name}")
+        return res
+
     def select_workspaces_to_delete(self, children: List[Dict], excess: int,
                                     revision: str) -> List[Dict]:
         """Reference parity: selectWorkspacesToDelete
@@ -92,6 +115,11 @@ class InferenceSetReconciler:
         return sorted(children, key=sort_key)[:excess]
 
     def reconcile(self, iset: InferenceSet) -> ISReconcileResult:
+        if iset.deletionTimestamp:
+            return self._finalize(iset)
+        from ..api_types import FINALIZER_INFERENCESET
+        if FINALIZER_INFERENCESET not in iset.finalizers:
+            iset.finalizers.append(FINALIZER_INFERENCESET)
         iset.validate()
         revision = self._revision(iset)
         children = self._list_children(iset)

@@ -72,6 +72,8 @@ def inferenceset_from_obj(obj: Dict) -> InferenceSet:
     return InferenceSet(
         name=obj["metadata"]["name"],
         namespace=obj["metadata"].get("namespace", "default"),
+        deletionTimestamp=obj["metadata"].get("deletionTimestamp"),
+        finalizers=obj["metadata"].get("finalizers", []) or [],
         spec=InferenceSetSpec(
             replicas=spec.get("replicas", 1),
             workspaceTemplate=workspace_from_obj(tpl_obj),

@@ -658,3 +658,29 @@ def test_workspace_deletion_finalizer_flow():
         client.get("StatefulSet", "default", "ws1")
     with _pt.raises(NotFound):
         client.get("Service", "default", "ws1")
+
+
+def test_inferenceset_deletion_flow():
+    from kaito_amd.operator.api_types import (COND_INFERENCESET_DELETING,
+                                              FINALIZER_INFERENCESET)
+    from kaito_amd.operator.kubeclient import FakeKubeClient
+    from kaito_amd.operator.controllers.inferenceset import (
+        InferenceSetReconciler)
+    from kaito_amd.operator.main import inferenceset_from_obj
+    client = FakeKubeClient()
+    rec = InferenceSetReconciler(client)
+    iset = inferenceset_from_obj({
+        "metadata": {"name": "pool2", "namespace": "default"},
+        "spec": {"replicas": 2, "workspaceTemplate": {
+            "resource": {"instanceType": "Standard_MI355X_v1"},
+            "inference": {"preset": {"name": "llama-3-8b"}}}}})
+    rec.reconcile(iset)
+    assert FINALIZER_INFERENCESET in iset.finalizers
+    assert len(client.list("Workspace")) == 2
+    iset.deletionTimestamp = "2026-09-12T00:00:00Z"
+    r = rec.reconcile(iset)
+    assert r.deleted == 2
+    assert len(client.list("Workspace")) == 0
+    assert FINALIZER_INFERENCESET not in iset.finalizers
+    assert any(c.type == COND_INFERENCESET_DELETING and c.status == "True"
+               for c in iset.status.conditions)
