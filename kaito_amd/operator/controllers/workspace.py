@@ -186,6 +186,7 @@ class WorkspaceReconciler:
         self.provisioner.delete_nodes(ws)
         if FINALIZER_WORKSPACE in ws.finalizers:
             ws.finalizers.remove(FINALIZER_WORKSPACE)
+        self._push_status(ws)
         return ReconcileResult()
 
     def _apply_inference(self, ws: Workspace, model, gpu,
@@ -292,6 +293,9 @@ class WorkspaceReconciler:
             obj = self.client.get("Workspace", ws.namespace, ws.name)
         except NotFound:
             return
+        if obj["metadata"].get("finalizers", []) != ws.finalizers:
+            obj["metadata"]["finalizers"] = list(ws.finalizers)
+            obj = self.client.update(obj)
         obj["status"] = {
             "state": ws.status.state,
             "targetNodeCount": ws.status.targetNodeCount,
