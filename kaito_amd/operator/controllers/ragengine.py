@@ -126,6 +126,20 @@ class RAGEngineReconciler:
 
     def reconcile(self, rag: RAGEngine) -> bool:
         """Returns True when the service is ready."""
+        if getattr(rag, "deletionTimestamp", None):
+            # deletion flow: tear down Deployment/Service/ConfigMap and
+            # mark RAGEngineDeleting (reference condition_types.go)
+            from ..api_types import COND_RAGENGINE_DELETING
+            for kind in ("Deployment", "Service", "ConfigMap"):
+                try:
+                    self.client.delete(kind, rag.namespace, rag.name)
+                except NotFound:
+                    pass
+            conds = rag.status.setdefault("conditions", [])
+            if not any(c["type"] == COND_RAGENGINE_DELETING for c in conds):
+                conds.append({"type": COND_RAGENGINE_DELETING,
+                              "status": "True", "reason": "Deleting"})
+            return False
         rag.validate()
         cm = self._guardrails_configmap(rag)
         if cm is not None:
