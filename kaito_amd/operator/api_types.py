@@ -241,6 +241,7 @@ class RAGEngine:
     namespace: str = "default"
     spec: RAGEngineSpec = field(default_factory=RAGEngineSpec)
     status: Dict[str, Any] = field(default_factory=dict)
+    deletionTimestamp: Optional[str] = None
 
     def validate(self) -> None:
         emb = self.spec.embedding

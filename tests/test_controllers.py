@@ -684,3 +684,24 @@ def test_inferenceset_deletion_flow():
     assert FINALIZER_INFERENCESET not in iset.finalizers
     assert any(c.type == COND_INFERENCESET_DELETING and c.status == "True"
                for c in iset.status.conditions)
+
+
+def test_ragengine_deletion_flow(client):
+    from kaito_amd.operator.controllers.ragengine import RAGEngineReconciler
+    from kaito_amd.operator.kubeclient import NotFound
+    rag = at.RAGEngine("rag2", spec=at.RAGEngineSpec(
+        compute=at.ResourceSpec(instanceType=SKU),
+        embedding={"remote": {"url": "http://emb"}},
+        inferenceService={"url": "http://ws1/v1/chat/completions"}))
+    r = RAGEngineReconciler(client)
+    r.reconcile(rag)
+    assert client.get("Deployment", "default", "rag2")
+    rag.deletionTimestamp = "2026-09-12T00:00:00Z"
+    assert r.reconcile(rag) is False
+    import pytest as _pt
+    with _pt.raises(NotFound):
+        client.get("Deployment", "default", "rag2")
+    with _pt.raises(NotFound):
+        client.get("Service", "default", "rag2")
+    conds = {c["type"]: c["status"] for c in rag.status["conditions"]}
+    assert conds["RAGEngineDeleting"] == "True"
