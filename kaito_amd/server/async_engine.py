@@ -103,6 +103,9 @@ class AsyncLLMEngine:
             hit = getattr(eng.pool, "hit_tokens", None)
             if hit is not None:
                 metrics.PREFIX_CACHE_HIT_TOKENS.set(hit)
+            nb = eng.pool.num_blocks
+            if nb:
+                metrics.GPU_CACHE_USAGE.set(1.0 - eng.pool.num_free / nb)
             # push fresh tokens to streams
             for sid, p in list(self._streams.items()):
                 seq = eng.seqs.get(sid)

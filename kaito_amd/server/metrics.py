@@ -33,6 +33,9 @@ REQUESTS_RUNNING = Gauge(
 REQUESTS_WAITING = Gauge(
     "vllm:num_requests_waiting", "Sequences queued for prefill",
     registry=REGISTRY)
+GPU_CACHE_USAGE = Gauge(
+    "vllm:gpu_cache_usage_perc", "Fraction of KV blocks in use (EPP "
+    "kv-cache-utilization scorer input)", registry=REGISTRY)
 PREFIX_CACHE_HIT_TOKENS = Gauge(
     "vllm:prefix_cache_hit_tokens", "Prompt tokens served from the "
     "block-hash prefix cache (prefill skipped)", registry=REGISTRY)

@@ -283,3 +283,13 @@ def test_download_monitor_progress(tmp_path):
     mon.stop(done=True)
     from kaito_amd.server import metrics as M
     assert M.MODEL_DOWNLOAD_DONE._value.get() == 1
+
+
+def test_gpu_cache_usage_metric(client):
+    client.post("/v1/completions", json={"prompt": [3, 4, 5],
+                                         "max_tokens": 2,
+                                         "ignore_eos": True})
+    m = client.get("/metrics").text
+    line = [ln for ln in m.splitlines()
+            if ln.startswith("vllm:gpu_cache_usage_perc ")]
+    assert line and 0.0 <= float(line[0].split()[-1]) <= 1.0
