@@ -127,8 +127,15 @@ class AsyncLLMEngine:
                 self._emitted.pop(seq.seq_id, None)
                 if p is not None:
                     metrics.PROMPT_TOKENS.inc(seq.num_prompt_tokens)
-                    metrics.E2E_LATENCY.inc(
-                        (seq.finish_time or time.monotonic()) - seq.arrival_time)
+                    fin = seq.finish_time or time.monotonic()
+                    metrics.E2E_LATENCY.inc(fin - seq.arrival_time)
+                    if seq.first_token_time is not None:
+                        metrics.TTFT.observe(
+                            seq.first_token_time - seq.arrival_time)
+                        n_out = len(seq.output_token_ids)
+                        if n_out > 1:
+                            metrics.TPOT.observe(
+                                (fin - seq.first_token_time) / (n_out - 1))
                     self._push(p, StreamItem(-1, True, seq.finish_reason))
 
     @staticmethod

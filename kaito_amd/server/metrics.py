@@ -12,7 +12,7 @@ scrapes (SURVEY.md §8 contract appendix):
 """
 from __future__ import annotations
 
-from prometheus_client import (CollectorRegistry, Counter, Gauge,
+from prometheus_client import (CollectorRegistry, Counter, Gauge, Histogram,
                                generate_latest, CONTENT_TYPE_LATEST)
 
 REGISTRY = CollectorRegistry()
@@ -39,6 +39,15 @@ GPU_CACHE_USAGE = Gauge(
 PREFIX_CACHE_HIT_TOKENS = Gauge(
     "vllm:prefix_cache_hit_tokens", "Prompt tokens served from the "
     "block-hash prefix cache (prefill skipped)", registry=REGISTRY)
+TTFT = Histogram(
+    "vllm:time_to_first_token_seconds", "Request TTFT",
+    buckets=(0.05, 0.1, 0.25, 0.5, 1, 2, 5, 10, 30, 60),
+    registry=REGISTRY)
+TPOT = Histogram(
+    "vllm:time_per_output_token_seconds", "Mean inter-token latency "
+    "per request",
+    buckets=(0.005, 0.01, 0.02, 0.04, 0.08, 0.15, 0.3, 0.6),
+    registry=REGISTRY)
 E2E_LATENCY = Counter(
     "vllm:e2e_request_latency_seconds", "Sum of request latencies",
     registry=REGISTRY)

@@ -293,3 +293,15 @@ def test_gpu_cache_usage_metric(client):
     line = [ln for ln in m.splitlines()
             if ln.startswith("vllm:gpu_cache_usage_perc ")]
     assert line and 0.0 <= float(line[0].split()[-1]) <= 1.0
+
+
+def test_ttft_tpot_histograms(client):
+    client.post("/v1/completions", json={"prompt": [9, 8, 7],
+                                         "max_tokens": 4,
+                                         "ignore_eos": True})
+    m = client.get("/metrics").text
+    assert "vllm:time_to_first_token_seconds_bucket" in m
+    assert "vllm:time_per_output_token_seconds_bucket" in m
+    count = [ln for ln in m.splitlines()
+             if ln.startswith("vllm:time_to_first_token_seconds_count")]
+    assert count and float(count[0].split()[-1]) >= 1
