@@ -135,3 +135,31 @@ def test_sampler_topk_topp_support(seed, k, temp):
     cum = torch.cumsum(sp, 0)
     ncut = int((cum < 0.5).sum()) + 1
     assert int(toks[2]) in set(si[:ncut].tolist())
+
+
+@settings(max_examples=10, deadline=None)
+@_given(st.integers(0, 2**31 - 1), st.integers(8, 40),
+        st.sampled_from([True, False]))
+def test_engine_oracle_under_random_workloads(seed, budget, mixed):
+    """Capstone property: for random prompt sets, chunk budgets and both
+    stepping modes, every sequence's greedy output equals the standalone
+    full-recompute oracle."""
+    import numpy as np
+    from kaito_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from kaito_amd.models import get_model_config
+    from tests.test_engine import _naive_generate
+    rng = np.random.default_rng(seed)
+    cfg = EngineConfig(model=get_model_config("tiny-llama-test"),
+                       device="cpu", max_num_seqs=4, num_gpu_blocks=64,
+                       enforce_eager=True, max_model_len=128,
+                       max_num_batched_tokens=budget,
+                       enable_mixed_batch=mixed, mixed_prefill_tokens=16)
+    eng = LLMEngine(cfg)
+    prompts = [rng.integers(1, 500, rng.integers(3, 60)).tolist()
+               for _ in range(int(rng.integers(1, 5)))]
+    sp = SamplingParams(max_tokens=int(rng.integers(2, 8)),
+                        ignore_eos=True)
+    outs = eng.generate(prompts, sp)
+    for p, o in zip(prompts, outs):
+        assert o.output_token_ids == _naive_generate(
+            eng.runner.model, cfg, p, sp.max_tokens)
