@@ -171,6 +171,8 @@ def main():
             "config": {
                 "model": args.model,
                 "parallelism": f"dp{world}(tp={args.tp})",
+                # global_batch = concurrent sequences across all ranks
+                "global_batch": args.max_num_seqs * world // max(args.tp, 1),
                 "max_num_seqs": args.max_num_seqs,
                 "in_tokens": args.in_tokens,
                 "out_tokens": args.out_tokens,
