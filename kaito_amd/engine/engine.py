@@ -161,7 +161,7 @@ class LLMEngine:
         self.scheduler.finish_prefill_chunks(batch)
         samp = batch.sampling_seqs
         for seq in samp:
-            seq.sched_len = seq.num_prompt_tokens + 1
+            seq.sched_len = seq.num_context_tokens + 1
             self.num_prompt_tokens += seq.num_prompt_tokens
             self._register_prefix(seq)
         if self.kv_publisher is not None and samp:
@@ -320,7 +320,7 @@ class LLMEngine:
         # PROMOTION to the decode set waits for the event (resolve)
         self.scheduler.advance_prefill_chunks(p_batch)
         for seq in samp_pre:
-            seq.sched_len = seq.num_prompt_tokens + 1
+            seq.sched_len = seq.num_context_tokens + 1
             self.num_prompt_tokens += seq.num_prompt_tokens
         if self.kv_publisher is not None and samp_pre:
             blocks = [b for s2 in samp_pre for b in s2.block_table]

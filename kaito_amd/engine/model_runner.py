@@ -126,7 +126,7 @@ class ModelRunner:
         ids, pos, slots, cu = [], [], [], [0]
         fresh = all(c.start == 0 and c.completes for c in chunks)
         for c in chunks:
-            toks = c.seq.prompt_token_ids[c.start:c.start + c.length]
+            toks = c.seq.context_token_ids[c.start:c.start + c.length]
             ids.extend(toks)
             pos.extend(range(c.start, c.start + c.length))
             slots.extend(self._slot(c.seq, p)

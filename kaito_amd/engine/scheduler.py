@@ -30,7 +30,8 @@ class PrefillChunk:
 
     @property
     def completes(self) -> bool:
-        return self.start + self.length >= self.seq.num_prompt_tokens
+        # context (prompt + preserved output after a recompute preemption)
+        return self.start + self.length >= self.seq.num_context_tokens
 
 
 @dataclass
@@ -119,7 +120,7 @@ class Scheduler:
         for seq in list(self.prefilling):
             if budget <= 0:
                 break
-            remaining = seq.num_prompt_tokens - seq.prefilled_len
+            remaining = seq.num_context_tokens - seq.prefilled_len
             if remaining <= 0:
                 continue
             n = min(remaining, budget)
@@ -145,14 +146,14 @@ class Scheduler:
             seq.status = SeqStatus.RUNNING
             if self.restore_cb is not None and seq.prefilled_len == 0:
                 seq.prefilled_len = self.restore_cb(seq)
-            if seq.prefilled_len >= seq.num_prompt_tokens:
+            if seq.prefilled_len >= seq.num_context_tokens:
                 # full prefix-cache hit: straight to decode (the next decode
-                # step feeds the last prompt token over restored KV)
-                seq.sched_len = seq.num_prompt_tokens
+                # step feeds the last context token over restored KV)
+                seq.sched_len = seq.num_context_tokens
                 self.running.append(seq)
                 room -= 1
                 continue
-            n = min(seq.num_prompt_tokens - seq.prefilled_len, budget)
+            n = min(seq.num_context_tokens - seq.prefilled_len, budget)
             chunks.append(PrefillChunk(seq, seq.prefilled_len, n))
             self.prefilling.append(seq)
             budget -= n
@@ -205,8 +206,11 @@ class Scheduler:
                     victim = batch.pop()  # newest scheduled
                     self.running.remove(victim)
                     self.pool.free(victim.block_table)
+                    # recompute preemption (vLLM semantics): KEEP the
+                    # generated tokens — they were already streamed to the
+                    # client — and re-prefill prompt+generated as context
+                    # at re-admission (prefill paths use context_token_ids)
                     victim.block_table = []
-                    victim.output_token_ids = []
                     victim.sched_len = 0
                     victim.prefilled_len = 0
                     victim.epoch += 1

@@ -68,6 +68,21 @@ class Sequence:
         return len(self.prompt_token_ids)
 
     @property
+    def context_token_ids(self) -> List[int]:
+        """Tokens whose KV must be in the cache before the next decode:
+        prompt + already-generated output. After a recompute preemption the
+        generated tokens are preserved and re-prefilled as context (vLLM
+        recompute semantics), so streamed clients never see a divergent
+        re-generation of indices they already received."""
+        if not self.output_token_ids:
+            return self.prompt_token_ids
+        return self.prompt_token_ids + self.output_token_ids
+
+    @property
+    def num_context_tokens(self) -> int:
+        return len(self.prompt_token_ids) + len(self.output_token_ids)
+
+    @property
     def num_tokens(self) -> int:
         return len(self.prompt_token_ids) + len(self.output_token_ids)
 

@@ -14,8 +14,14 @@ import torch
 
 from .. import ops
 
-# AWQ packs 8 nibbles along N in interleaved order (public awq format)
+# Public AWQ packing (llm-awq/AutoAWQ): nibble position i of each i32 word
+# holds LOGICAL column AWQ_ORDER[i]. Unpacking to logical column order
+# therefore reads nibble positions in the INVERSE permutation,
+# AWQ_REVERSE_ORDER (logical col c lives at nibble AWQ_REVERSE_ORDER[c]) —
+# the same table vLLM uses. The permutation has order 3, so using the
+# forward table for unpacking scrambles weights within each 8-column group.
 AWQ_ORDER = (0, 2, 4, 6, 1, 3, 5, 7)
+AWQ_REVERSE_ORDER = (0, 4, 1, 5, 2, 6, 3, 7)
 
 
 def quantize_w4(weight: torch.Tensor, group: int = 128
@@ -48,9 +54,10 @@ def repack_awq(qweight_awq: torch.Tensor, qzeros_awq: torch.Tensor,
     N = nw * 8
 
     def unpack_n(t: torch.Tensor) -> torch.Tensor:
-        # [R, N/8] i32 → [R, N] int in logical n order
+        # [R, N/8] i32 → [R, N] int in logical n order: logical column c
+        # of each group is stored at nibble position AWQ_REVERSE_ORDER[c]
         cols = []
-        for j in AWQ_ORDER:
+        for j in AWQ_REVERSE_ORDER:
             cols.append((t >> (4 * j)) & 0xF)
         return torch.stack(cols, dim=-1).reshape(t.shape[0], N)
 

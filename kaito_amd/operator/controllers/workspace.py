@@ -100,8 +100,17 @@ class WorkspaceReconciler:
             from ..sku import MI355X
             cfg = GPUConfig(ws.resource.instanceType or "byo", 8, 288, MI355X,
                             "gfx950", 7)
-        if ws.resource.count:
-            cfg = cfg.scale_to_count(cfg.gpu_count)
+        # partition spec rescale (reference: ScaleGPUConfigToCount,
+        # pkg/sku/helpers.go:123-132): when the workspace requests a
+        # partitioned view (CPX), the estimator must see partition-sized
+        # devices, capped at the requested partitionCount per node
+        from ..partition import partitioned_gpu_config, validate_partition
+        prof = validate_partition(ws.resource.partition, cfg)
+        if prof is not None:
+            cfg = partitioned_gpu_config(cfg, prof)
+            pc = ws.resource.partition.partitionCount
+            if pc and pc < cfg.gpu_count:
+                cfg = cfg.scale_to_count(pc)
         return cfg
 
     def _spec_hash(self, ws: Workspace) -> str:

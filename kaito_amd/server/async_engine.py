@@ -36,6 +36,7 @@ class _Pending:
     out_q: asyncio.Queue
     loop: asyncio.AbstractEventLoop
     seq_id: Optional[int] = None
+    lora_name: Optional[str] = None
 
 
 class AsyncLLMEngine:
@@ -67,7 +68,15 @@ class AsyncLLMEngine:
                 p = self._submit.get_nowait()
             except queue.Empty:
                 return
-            sid = self.engine.add_request(p.prompt, p.sampling)
+            try:
+                sid = self.engine.add_request(p.prompt, p.sampling,
+                                              lora_name=p.lora_name)
+            except Exception as e:  # noqa: BLE001 — bad request (unknown
+                # LoRA, over-long prompt from a non-HTTP caller): fail ONLY
+                # this stream; a raise here would kill the engine thread
+                # and hang every in-flight and future request
+                self._push(p, StreamItem(-1, True, f"error: {e}"))
+                continue
             p.seq_id = sid
             self._streams[sid] = p
             self._emitted[sid] = 0
@@ -159,11 +168,12 @@ class AsyncLLMEngine:
         self._emitted.pop(seq_id, None)
         self.engine.abort(seq_id)
 
-    async def generate(self, prompt_ids: List[int], sampling: SamplingParams
+    async def generate(self, prompt_ids: List[int], sampling: SamplingParams,
+                       lora_name: Optional[str] = None
                        ) -> AsyncIterator[StreamItem]:
         loop = asyncio.get_running_loop()
         out_q: asyncio.Queue = asyncio.Queue()
-        p = _Pending(prompt_ids, sampling, out_q, loop)
+        p = _Pending(prompt_ids, sampling, out_q, loop, lora_name=lora_name)
         self._submit.put(p)
         self._wake.set()
         try:

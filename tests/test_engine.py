@@ -276,6 +276,37 @@ def test_engine_preemption_recovers_exact_outputs():
         assert eng.seqs[sid].output_token_ids == expect
 
 
+def test_preemption_never_rewrites_streamed_tokens():
+    """Recompute preemption must PRESERVE already-emitted tokens (vLLM
+    recompute semantics): with temperature>0, a preempted sequence that
+    restarted generation from scratch would re-sample a divergent
+    continuation for indices the client already received. Snapshot each
+    sequence's output after every step and require every snapshot to be
+    a prefix of the final output."""
+    cfg = _cfg(max_num_seqs=4, num_gpu_blocks=8, max_model_len=64)
+    eng = LLMEngine(cfg)
+    sp = SamplingParams(max_tokens=22, ignore_eos=True, temperature=1.0)
+    prompts = [list(range(10, 24)), list(range(30, 44)),
+               list(range(50, 64))]
+    ids = [eng.add_request(p, sp) for p in prompts]
+    snapshots = {sid: [] for sid in ids}
+    guard = 0
+    while eng.has_unfinished():
+        eng.step()
+        for sid in ids:
+            snapshots[sid].append(list(eng.seqs[sid].output_token_ids))
+        guard += 1
+        assert guard < 500
+    assert any(eng.seqs[sid].epoch > 0 for sid in ids), \
+        "test did not actually exercise preemption"
+    for sid in ids:
+        final = eng.seqs[sid].output_token_ids
+        assert len(final) == 22
+        for snap in snapshots[sid]:
+            assert final[:len(snap)] == snap, \
+                "a streamed prefix was rewritten after preemption"
+
+
 def test_engine_stochastic_sampling_paths():
     """temperature/top-k/top-p exercise the non-greedy sampler through the
     pipelined engine (finite, in-vocab tokens)."""
@@ -542,10 +573,11 @@ def test_awq_checkpoint_end_to_end(tmp_path):
         sT = s.T.contiguous()                                   # [K/G, N]
 
         def pack_n(t):
+            # real llm-awq packing loop: nibble position i of each word
+            # holds LOGICAL column AWQ_ORDER[i] of the 8-column group
             out = torch.zeros(t.shape[0], N // 8, dtype=torch.int64)
-            for b in range(N // 8):
-                for pos, j in enumerate(AWQ_ORDER):
-                    out[:, b] |= t[:, b * 8 + pos] << (4 * j)
+            for i in range(8):
+                out |= t[:, AWQ_ORDER[i]::8] << (4 * i)
             return out.to(torch.int32)
 
         return pack_n(q), pack_n(zqT), sT

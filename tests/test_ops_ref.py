@@ -158,11 +158,25 @@ def test_w4a16_quantize_roundtrip_accuracy():
     assert torch.allclose(y, x @ deq.T, atol=1e-3)
 
 
-def test_w4a16_awq_repack_matches_native():
-    """Packing a known q/zq/s set into public-AWQ layout and repacking
-    must reproduce the same dequantized weights."""
+def _pack_awq_public(t: "torch.Tensor", N: int) -> "torch.Tensor":
+    """Pack [R, N] logical nibbles into public-AWQ i32 [R, N/8] words,
+    reproducing the llm-awq/AutoAWQ packing loop exactly: nibble position
+    i of each word holds LOGICAL column AWQ_ORDER[i] of the 8-group
+    (llm-awq awq/quantize/qmodule.py pack order [0,2,4,6,1,3,5,7])."""
     import torch
-    from kaito_amd.models.quant import repack_awq, AWQ_ORDER
+    from kaito_amd.models.quant import AWQ_ORDER
+    out = torch.zeros(t.shape[0], N // 8, dtype=torch.int64)
+    for i in range(8):
+        out |= t[:, AWQ_ORDER[i]::8] << (4 * i)
+    return out.to(torch.int32)
+
+
+def test_w4a16_awq_repack_matches_native():
+    """Packing a known q/zq/s set into public-AWQ layout (with the real
+    llm-awq nibble order) and repacking must reproduce the same
+    dequantized weights."""
+    import torch
+    from kaito_amd.models.quant import repack_awq
     from kaito_amd.ops import torch_ref
     torch.manual_seed(1)
     K, N, G = 128, 32, 64
@@ -170,25 +184,41 @@ def test_w4a16_awq_repack_matches_native():
     zq = torch.randint(0, 16, (K // G, N), dtype=torch.int64)
     s = torch.rand(K // G, N) + 0.1
 
-    def pack_n(t):
-        out = torch.zeros(t.shape[0], N // 8, dtype=torch.int64)
-        for pos, j in enumerate(AWQ_ORDER):
-            out |= t[:, pos::8] << (4 * j)
-        return out.to(torch.int32)
-
-    # build AWQ-layout tensors column-block-wise: logical n = b*8 + pos
-    qa = torch.zeros(K, N // 8, dtype=torch.int64)
-    za = torch.zeros(K // G, N // 8, dtype=torch.int64)
-    for b in range(N // 8):
-        qa[:, b] = sum(q[:, b * 8 + pos] << (4 * j)
-                       for pos, j in enumerate(AWQ_ORDER))
-        za[:, b] = sum(zq[:, b * 8 + pos] << (4 * j)
-                       for pos, j in enumerate(AWQ_ORDER))
-    qn, sn, zn = repack_awq(qa.to(torch.int32), za.to(torch.int32), s, G)
+    qa = _pack_awq_public(q, N)
+    za = _pack_awq_public(zq, N)
+    qn, sn, zn = repack_awq(qa, za, s, G)
     deq = torch_ref.w4a16_unpack(qn, sn, zn, G)
     expect = (s.repeat_interleave(G, dim=0) *
               (q - zq.repeat_interleave(G, dim=0)).float()).T
     assert torch.allclose(deq, expect, atol=1e-5)
+
+
+def test_w4a16_awq_repack_not_identity_permutation():
+    """Guard against the masked-bug failure mode the round-1 advisor
+    found: packing with the FORWARD order and unpacking with the forward
+    order cancel out, hiding a scrambled load of real checkpoints. The
+    repack must be sensitive to the nibble order — a forward-order pack
+    (wrong convention) must NOT round-trip."""
+    import torch
+    from kaito_amd.models.quant import repack_awq, AWQ_ORDER
+    from kaito_amd.ops import torch_ref
+    torch.manual_seed(3)
+    K, N, G = 128, 32, 64
+    q = torch.randint(0, 16, (K, N), dtype=torch.int64)
+    zq = torch.zeros(K // G, N, dtype=torch.int64)
+    s = torch.ones(K // G, N)
+
+    def pack_wrong(t):
+        out = torch.zeros(t.shape[0], N // 8, dtype=torch.int64)
+        for i in range(8):
+            # wrong: nibble AWQ_ORDER[i] <- logical i (inverse convention)
+            out |= t[:, i::8] << (4 * AWQ_ORDER[i])
+        return out.to(torch.int32)
+
+    qn, sn, zn = repack_awq(pack_wrong(q), pack_wrong(zq), s, G)
+    deq = torch_ref.w4a16_unpack(qn, sn, zn, G)
+    expect = q.float().T
+    assert not torch.allclose(deq, expect, atol=1e-5)
 
 
 def test_w4a16_quantlinear_cpu_paths():
