@@ -116,23 +116,33 @@ def main():
     # The timed region must measure STEADY-STATE serving, independent of
     # the driver-chosen steps/warmup: a short run that starts timing at
     # t=0 only measures the initial prefill burst (the whole queue is
-    # prefilling, nothing is decoding yet). Step until >=90% of
-    # max_num_seqs are in the decode set, then hand over to the normal
+    # prefilling, nothing is decoding yet) — and a window that ends
+    # before the first requests finish measures pure decode with no
+    # refill prefills, which OVERSTATES serving throughput. Ramp until
+    # (a) >=90% of max_num_seqs are in the decode set AND (b) the finish/
+    # refill churn is established (>= max_num_seqs/4 requests finished,
+    # i.e. well past the first finish wave), then hand over to the normal
     # warmup/timed counters. Bounded by steps and wall-clock so a
     # misconfigured run still terminates.
     ramp_target = int(0.9 * args.max_num_seqs)
+    churn_target = max(args.max_num_seqs // 4, 1)
     ramp_t0 = time.perf_counter()
     ramp_steps = 0
-    while (len(eng.scheduler.running) < ramp_target
-           and ramp_steps < 5000
+    ramp_finished = 0
+    while (ramp_steps < 10000
            and time.perf_counter() - ramp_t0 < 300.0):
-        run_steps(1)
+        if (len(eng.scheduler.running) >= ramp_target
+                and ramp_finished >= churn_target):
+            break
+        ramp_finished += len(run_steps(1))
         ramp_steps += 1
     ramp_s = time.perf_counter() - ramp_t0
-    if rank == 0 and len(eng.scheduler.running) < ramp_target:
+    if rank == 0 and (len(eng.scheduler.running) < ramp_target
+                      or ramp_finished < churn_target):
         import sys
         print(f"[bench] WARNING: ramp ended at "
-              f"{len(eng.scheduler.running)}/{ramp_target} decoding after "
+              f"{len(eng.scheduler.running)}/{ramp_target} decoding, "
+              f"{ramp_finished}/{churn_target} finished, after "
               f"{ramp_steps} steps / {ramp_s:.1f}s", file=sys.stderr)
 
     # ---- warmup (decode-regime steps) ----

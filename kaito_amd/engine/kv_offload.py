@@ -10,9 +10,11 @@ Mechanics:
   * a resubmitted prompt whose FULL token sequence matches a cached chain
     restores host→GPU into freshly allocated blocks and skips prefill
     entirely — the engine then runs one decode step from the last prompt
-    token (its KV slot is rewritten with identical values).
-    Partial-prefix reuse (tail prefill over restored KV) needs the paged
-    context-prefill kernel — tracked for a later round.
+    token (its KV slot is rewritten with identical values);
+  * a prompt sharing only a PREFIX with a cached chain restores the
+    longest common prefix (restore_prefix) and the scheduler prefills
+    just the suffix as a chunked tail over the restored KV (the paged
+    context-prefill path, ops/csrc/context_attention.hip);
   * LRU eviction under a byte budget (default: 0.5 of available RAM,
     the reference's LMCACHE default utilization).
 """

@@ -118,7 +118,9 @@ def test_kv_event_bus_roundtrip():
                                             KVEventSubscriber, BLOCK_STORED)
     pub = KVEventPublisher(host="127.0.0.1", port=0)
     sub = KVEventSubscriber(port=pub.port)
-    time.sleep(0.2)  # let subscription register
+    deadline = time.monotonic() + 3
+    while pub._pub.num_subscriptions < 1 and time.monotonic() < deadline:
+        time.sleep(0.02)  # ZMQ slow-joiner: wait for the subscription
     pub.block_stored([1, 2, 3])
     pub.block_removed([2])
     pub.all_cleared()
@@ -145,7 +147,9 @@ def test_engine_publishes_kv_events():
     pub = KVEventPublisher(host="127.0.0.1", port=0)
     eng.kv_publisher = pub
     sub = KVEventSubscriber(port=pub.port)
-    time.sleep(0.2)
+    deadline = time.monotonic() + 3
+    while pub._pub.num_subscriptions < 1 and time.monotonic() < deadline:
+        time.sleep(0.02)
     eng.generate([[1, 2, 3, 4]], SamplingParams(max_tokens=3, ignore_eos=True))
     deadline = time.monotonic() + 3
     while len(sub.events) < 2 and time.monotonic() < deadline:
