@@ -63,6 +63,12 @@ def main():
     from kaito_amd.parallel import state as ps
 
     ps.init_parallel(tp_size=args.tp)
+    st = ps.get_state()
+    # TP replicates the request stream across the group: every rank of a
+    # TP group must enqueue IDENTICAL prompts (lockstep scheduling), and
+    # only one rank per group contributes to the token aggregate.
+    dp_rank = st.dp_rank
+    is_group_lead = st.tp_rank == 0
 
     mc = get_model_config(args.model)
     if args.quantization:
@@ -77,7 +83,7 @@ def main():
         tensor_parallel_size=args.tp,
         enforce_eager=args.eager,
         enable_mixed_batch=args.mixed_steps,
-        seed=1234 + rank,
+        seed=1234 + dp_rank,
     )
     eng = LLMEngine(cfg)
     if args.tune_gemms is not None:
@@ -87,7 +93,7 @@ def main():
         eng.capture_graphs()
 
     import numpy as np
-    rng = np.random.default_rng(42 + rank)
+    rng = np.random.default_rng(42 + dp_rank)
 
     def new_prompt():
         return rng.integers(10, mc.vocab_size - 10, args.in_tokens).tolist()
@@ -180,12 +186,15 @@ def main():
     ttfts = timed or everything
     ttft_p50 = statistics.median(ttfts) if ttfts else None
 
-    # aggregate across ranks: MAX(elapsed), SUM(tokens)
+    # aggregate across ranks: MAX(elapsed), SUM(tokens over DP replicas —
+    # every rank of a TP group counts the same tokens, so only the group
+    # lead contributes)
     if world > 1:
         import torch.distributed as dist
         dev = "cuda" if torch.cuda.is_available() else "cpu"
         te = torch.tensor([elapsed], device=dev)
-        tt = torch.tensor([float(tokens)], device=dev)
+        tt = torch.tensor([float(tokens if is_group_lead else 0)],
+                          device=dev)
         dist.all_reduce(te, op=dist.ReduceOp.MAX)
         dist.all_reduce(tt, op=dist.ReduceOp.SUM)
         elapsed = float(te.item())
@@ -213,7 +222,8 @@ def main():
             "itl_p50_ms": round(statistics.median(step_ms), 2) if step_ms else None,
             "config": {
                 "model": args.model,
-                "parallelism": f"dp{world}(tp={args.tp})",
+                "parallelism": f"dp{world // max(args.tp, 1)}"
+                               f"(tp={args.tp})",
                 # global_batch = concurrent sequences across all ranks
                 "global_batch": args.max_num_seqs * world // max(args.tp, 1),
                 "max_num_seqs": args.max_num_seqs,

@@ -126,6 +126,11 @@ class EngineConfig:
     # small-batch serving (prefill bursts stall decode there).
     enable_mixed_batch: bool = False
     mixed_prefill_tokens: int = 2048        # per-step overlap prefill budget
+    # fused one-shot allreduce+add+RMSNorm over xGMI for TP decode
+    # (parallel/one_shot.py). None → auto: on when tp>1 and pp==1 (the
+    # hipIpc peer mapping needs all TP ranks on one node). Batches larger
+    # than max_num_seqs fall back to the RCCL ring.
+    enable_one_shot_allreduce: Optional[bool] = None
     device: str = "cuda"
     seed: int = 0
     # decode graph buckets (batch sizes to capture)

@@ -51,6 +51,7 @@ class _PendingPrefill:
 class LLMEngine:
     def __init__(self, cfg: EngineConfig, weights_path: Optional[str] = None):
         self.cfg = cfg
+        self._maybe_enable_one_shot(cfg)
         self.runner = ModelRunner(cfg).load_model(weights_path, cfg.seed)
         self.runner.setup_tunable()
         self.runner.profile_and_allocate_kv()
@@ -84,6 +85,25 @@ class LLMEngine:
         self.num_prompt_tokens = 0
         # optional KV event bus (EPP KVCache-aware routing surface)
         self.kv_publisher = None
+
+    @staticmethod
+    def _maybe_enable_one_shot(cfg: EngineConfig) -> None:
+        """Activate the fused one-shot allreduce+RMSNorm group for TP
+        decode (xGMI hipIpc peer staging; gloo-emulated on CPU). pp>1 is
+        excluded: the stage-boundary residual fold has no norm call site
+        to absorb a deferred reduce."""
+        from ..parallel.state import get_state
+        st = get_state()
+        want = cfg.enable_one_shot_allreduce
+        if want is None:
+            want = st.tp_size > 1 and st.pp_size == 1
+        if not (want and st.tp_size > 1 and st.pp_size == 1):
+            return
+        from ..parallel import one_shot
+        if one_shot.active() is not None:
+            return
+        one_shot.activate(one_shot.make_group(
+            cfg.max_num_seqs, cfg.model.hidden_size, cfg.model.dtype))
 
     def capture_graphs(self):
         self.runner.capture_decode_graphs()

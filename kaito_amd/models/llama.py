@@ -68,6 +68,10 @@ class LlamaAttention(nn.Module):
             bias=cfg.attention_bias, dtype=cfg.dtype)
         self.o_proj = RowParallelLinear(
             cfg.num_heads * cfg.head_dim, h, bias=False, dtype=cfg.dtype)
+        # o_proj feeds post_attention_layernorm: with an active one-shot
+        # group its ring all-reduce is deferred into the fused
+        # allreduce+add+RMSNorm kernel (parallel/one_shot.py)
+        self.o_proj.fuse_norm = True
         self.q_size = self.num_heads * self.head_dim
         self.kv_size = self.num_kv_heads * self.head_dim
 
@@ -110,12 +114,32 @@ class LlamaMLP(nn.Module):
             cfg.hidden_size, 2 * cfg.intermediate_size, dtype=cfg.dtype)
         self.down_proj = RowParallelLinear(
             cfg.intermediate_size, cfg.hidden_size, dtype=cfg.dtype)
+        # down_proj feeds the next input_layernorm / final norm
+        self.down_proj.fuse_norm = True
         tp = get_state().tp_size
         self.inter_per_rank = cfg.intermediate_size // tp
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         gu = self.gate_up_proj(x)
         return self.down_proj(ops.silu_and_mul(gu))
+
+
+def _tp_fused_add_rms_norm(hidden, residual, weight, eps: float,
+                           deferred_producer: bool):
+    """fused_add_rms_norm that ALSO performs the TP reduction when the
+    producing RowParallelLinear deferred it to the one-shot group
+    (allreduce + residual-add + RMSNorm in ONE kernel over xGMI peer
+    staging — the fusion the reference disables on NVIDIA,
+    interface.go:439-446). Falls back to the already-reduced plain fused
+    norm when no group is active or the batch exceeds the staging
+    window (the RowParallel layer makes the same `defer` decision, so
+    the paths always agree)."""
+    if deferred_producer and get_state().tp_size > 1:
+        from ..parallel import one_shot
+        grp = one_shot.defer(hidden.reshape(-1, hidden.shape[-1]).shape[0])
+        if grp is not None:
+            return grp.allreduce_add_rmsnorm(hidden, residual, weight, eps)
+    return ops.fused_add_rms_norm(hidden, residual, weight, eps)
 
 
 class LlamaDecoderLayer(nn.Module):
@@ -138,11 +162,16 @@ class LlamaDecoderLayer(nn.Module):
             residual = hidden
             hidden = ops.rms_norm(hidden, self.input_layernorm, self.cfg.rms_eps)
         else:
-            hidden, residual = ops.fused_add_rms_norm(
-                hidden, residual, self.input_layernorm, self.cfg.rms_eps)
+            # `hidden` here is the PREVIOUS layer's mlp output — a
+            # deferred TP partial when a one-shot group is active and the
+            # mlp is the dense LlamaMLP (MoE's reduce is not deferred)
+            hidden, residual = _tp_fused_add_rms_norm(
+                hidden, residual, self.input_layernorm, self.cfg.rms_eps,
+                deferred_producer=self.cfg.num_experts == 0)
         hidden = self.self_attn(hidden, positions, kv_cache, meta, cos_sin)
-        hidden, residual = ops.fused_add_rms_norm(
-            hidden, residual, self.post_attention_layernorm, self.cfg.rms_eps)
+        hidden, residual = _tp_fused_add_rms_norm(
+            hidden, residual, self.post_attention_layernorm, self.cfg.rms_eps,
+            deferred_producer=True)   # o_proj always defer-capable
         hidden = self.mlp(hidden)
         return hidden, residual
 
@@ -192,6 +221,10 @@ class LlamaForCausalLM(nn.Module):
         the stage output hidden (+ residual folded in): intermediate stages
         return the value to SEND; the last stage returns final normed
         hidden."""
+        from ..parallel import one_shot
+        grp = one_shot.active()
+        if grp is not None:
+            grp.begin_step()   # deterministic staging flip for hipGraphs
         if self.is_first:
             hidden = self.embed_tokens(input_ids)
             residual = None
@@ -204,8 +237,9 @@ class LlamaForCausalLM(nn.Module):
             hidden, residual = layer(hidden, residual, positions, kv, meta,
                                      self.cos_sin_cache)
         if self.is_last:
-            hidden, _ = ops.fused_add_rms_norm(hidden, residual, self.norm,
-                                               self.cfg.rms_eps)
+            hidden, _ = _tp_fused_add_rms_norm(
+                hidden, residual, self.norm, self.cfg.rms_eps,
+                deferred_producer=self.cfg.num_experts == 0)
             return hidden
         # fold the residual stream so one tensor crosses the stage boundary
         return (hidden.float() + residual.float()).to(hidden.dtype)

@@ -93,14 +93,25 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
     torch_ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
 
 
+# decode-attention implementation: "sp" = split-phase (r02, K-stream →
+# exp pass → V-stream, no softmax chain in the memory loops), "fused" =
+# the r01 single-pass online-softmax kernel. Overridable for A/B runs.
+import os as _os
+_PA_IMPL = _os.environ.get("KAITO_PA_IMPL", "sp")
+
+
 def paged_attention(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
                     block_tables: torch.Tensor, seq_lens: torch.Tensor,
                     scale: float) -> torch.Tensor:
     if q.is_cuda:
         _require_ext()
         out = torch.empty_like(q)
-        torch.ops.kaito.paged_attention(out, q, k_cache, v_cache,
-                                        block_tables, seq_lens, scale)
+        if _PA_IMPL == "sp":
+            torch.ops.kaito.paged_attention_sp(out, q, k_cache, v_cache,
+                                               block_tables, seq_lens, scale)
+        else:
+            torch.ops.kaito.paged_attention(out, q, k_cache, v_cache,
+                                            block_tables, seq_lens, scale)
         return out
     return torch_ref.paged_attention(q, k_cache, v_cache, block_tables,
                                      seq_lens, scale)

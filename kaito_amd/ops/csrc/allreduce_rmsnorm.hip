@@ -64,6 +64,137 @@ void allreduce_rmsnorm(at::Tensor out, at::Tensor ptrs, at::Tensor weight,
       (const short*)weight.data_ptr(), (float)eps, N, H);
 }
 
+// ======================================================================
+// Graph-capturable one-shot all-reduce (+residual) + RMSNorm.
+//
+// The v1 kernel above needs host-side torch.distributed barriers around
+// it, which (a) cost ~100us of host latency per call and (b) make it
+// impossible to capture the decode step into a hipGraph. This variant
+// fuses the synchronization INTO the kernel (vLLM custom-allreduce
+// style): each rank owns an IPC-mapped signal page; block `row` of the
+// kernel bumps a device-memory epoch counter, release-stores the epoch
+// into every peer's signal slot [row][self] over xGMI, and acquire-spins
+// on its own slots until all peers arrive. Combined with DOUBLE-BUFFERED
+// staging on the Python side (a peer entering call k+1 has necessarily
+// finished reading call k's buffer — kernels on one stream serialize),
+// one in-kernel barrier per call suffices and no host sync is needed:
+// the whole decode step, collectives included, captures into a hipGraph.
+//
+// The epoch lives in device memory and is incremented BY the kernel, so
+// graph replays keep advancing it. Epoch comparison uses signed distance
+// so the uint32 wrap is harmless.
+__device__ __forceinline__ void one_shot_barrier(
+    const uint64_t* __restrict__ sig_ptrs, int rank, int N, int row,
+    uint32_t* __restrict__ counter) {
+  __shared__ uint32_t s_flag;
+  if (threadIdx.x == 0) s_flag = ++counter[row];  // single writer per row
+  __syncthreads();
+  const uint32_t flag = s_flag;
+  if ((int)threadIdx.x < N) {
+    uint32_t* peer = reinterpret_cast<uint32_t*>(sig_ptrs[threadIdx.x]);
+    __hip_atomic_store(&peer[row * 8 + rank], flag, __ATOMIC_RELEASE,
+                       __HIP_MEMORY_SCOPE_SYSTEM);
+    uint32_t* own = reinterpret_cast<uint32_t*>(sig_ptrs[rank]);
+    while ((int32_t)(__hip_atomic_load(&own[row * 8 + threadIdx.x],
+                                       __ATOMIC_ACQUIRE,
+                                       __HIP_MEMORY_SCOPE_SYSTEM) - flag) < 0) {
+      __builtin_amdgcn_s_sleep(1);
+    }
+  }
+  __syncthreads();
+}
+
+// one workgroup per row; vectorized short8 (8 bf16 / 16B per lane-step).
+__global__ __launch_bounds__(256)
+void one_shot_ar_rmsnorm_kernel(
+    short* __restrict__ out,             // [T, H]
+    short* __restrict__ residual,        // [T, H] in-place or nullptr
+    const uint64_t* __restrict__ ptrs,   // [N] staging base ptrs
+    const uint64_t* __restrict__ sig_ptrs,  // [N] signal base ptrs
+    uint32_t* __restrict__ counter,      // [max_rows] local epochs
+    const short* __restrict__ weight,    // [H]
+    float eps, int N, int H, int rank) {
+  const int row = blockIdx.x;
+  one_shot_barrier(sig_ptrs, rank, N, row, counter);
+
+  extern __shared__ float s_red[];
+  float ss = 0.f;
+  for (int h8 = threadIdx.x * 8; h8 < H; h8 += blockDim.x * 8) {
+    float v[8];
+#pragma unroll
+    for (int j = 0; j < 8; j++) v[j] = 0.f;
+    for (int r = 0; r < N; r++) {
+      const short8_t sv = *reinterpret_cast<const short8_t*>(
+          reinterpret_cast<const short*>(ptrs[r]) + (int64_t)row * H + h8);
+#pragma unroll
+      for (int j = 0; j < 8; j++) v[j] += bf16_to_f32(sv[j]);
+    }
+    short8_t stash;
+    if (residual != nullptr) {
+      short8_t* res = reinterpret_cast<short8_t*>(
+          residual + (int64_t)row * H + h8);
+      const short8_t rv = *res;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        // residual stream stays bf16 (matches fused_add_rms_norm)
+        stash[j] = f32_to_bf16(v[j] + bf16_to_f32(rv[j]));
+        const float rf = bf16_to_f32(stash[j]);
+        ss += rf * rf;
+      }
+      *res = stash;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        stash[j] = f32_to_bf16(v[j]);
+        const float rf = bf16_to_f32(stash[j]);
+        ss += rf * rf;
+      }
+      *reinterpret_cast<short8_t*>(out + (int64_t)row * H + h8) = stash;
+    }
+  }
+  ss = block_reduce_sum(ss, s_red);
+  __shared__ float s_scale;
+  if (threadIdx.x == 0) s_scale = rsqrtf(ss / H + eps);
+  __syncthreads();
+  const float scale = s_scale;
+  const short* src = residual != nullptr ? residual : out;
+  for (int h8 = threadIdx.x * 8; h8 < H; h8 += blockDim.x * 8) {
+    const short8_t rv = *reinterpret_cast<const short8_t*>(
+        src + (int64_t)row * H + h8);
+    const short8_t wv = *reinterpret_cast<const short8_t*>(weight + h8);
+    short8_t o;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      o[j] = f32_to_bf16(bf16_to_f32(rv[j]) * scale * bf16_to_f32(wv[j]));
+    *reinterpret_cast<short8_t*>(out + (int64_t)row * H + h8) = o;
+  }
+}
+
+void one_shot_ar_rmsnorm(at::Tensor out, at::Tensor residual,
+                         at::Tensor ptrs, at::Tensor sig_ptrs,
+                         at::Tensor counter, at::Tensor weight, double eps,
+                         int64_t rank) {
+  const int T = out.size(0), H = out.size(1), N = ptrs.size(0);
+  TORCH_CHECK(out.dtype() == at::kBFloat16 && out.is_contiguous());
+  TORCH_CHECK(H % (256 * 8) == 0 || H % 8 == 0, "H must be 8-aligned");
+  TORCH_CHECK(ptrs.dtype() == at::kLong && ptrs.is_cuda());
+  TORCH_CHECK(sig_ptrs.dtype() == at::kLong && sig_ptrs.is_cuda());
+  TORCH_CHECK(counter.dtype() == at::kInt && counter.size(0) >= T,
+              "epoch counter smaller than row count");
+  short* res = nullptr;
+  if (residual.numel() > 0) {
+    TORCH_CHECK(residual.is_contiguous() && residual.sizes() == out.sizes());
+    res = (short*)residual.data_ptr();
+  }
+  dim3 grid(T), block(256);
+  auto stream = at::hip::getCurrentHIPStream();
+  const int smem = 32 * sizeof(float);
+  one_shot_ar_rmsnorm_kernel<<<grid, block, smem, stream>>>(
+      (short*)out.data_ptr(), res, (const uint64_t*)ptrs.data_ptr(),
+      (const uint64_t*)sig_ptrs.data_ptr(), (uint32_t*)counter.data_ptr(),
+      (const short*)weight.data_ptr(), (float)eps, N, H, (int)rank);
+}
+
 // ---- hipIpc plumbing for cross-process peer mapping --------------------
 // (multi-GPU TP: each rank shares its buffer handle; peers open it and
 // pass the mapped pointer into allreduce_rmsnorm's ptrs array. Exchange

@@ -15,6 +15,9 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
 void paged_attention(at::Tensor out, at::Tensor query, at::Tensor k_cache,
                      at::Tensor v_cache, at::Tensor block_tables,
                      at::Tensor seq_lens, double scale);
+void paged_attention_sp(at::Tensor out, at::Tensor query, at::Tensor k_cache,
+                        at::Tensor v_cache, at::Tensor block_tables,
+                        at::Tensor seq_lens, double scale);
 void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
                        at::Tensor tile_seq, at::Tensor tile_qbase,
                        at::Tensor cu_seqlens, double scale);
@@ -38,6 +41,9 @@ void w4a16_dequant(at::Tensor out, at::Tensor qweight, at::Tensor scales,
                    at::Tensor zeros, int64_t group);
 void allreduce_rmsnorm(at::Tensor out, at::Tensor ptrs, at::Tensor weight,
                        double eps);
+void one_shot_ar_rmsnorm(at::Tensor out, at::Tensor residual, at::Tensor ptrs,
+                         at::Tensor sig_ptrs, at::Tensor counter,
+                         at::Tensor weight, double eps, int64_t rank);
 at::Tensor ipc_handle(at::Tensor t);
 int64_t ipc_open(at::Tensor handle_bytes);
 void ipc_close(int64_t ptr);
@@ -50,6 +56,7 @@ TORCH_LIBRARY(kaito, m) {
   m.def("silu_and_mul(Tensor(a!) out, Tensor x) -> ()");
   m.def("reshape_and_cache(Tensor k, Tensor v, Tensor(a!) k_cache, Tensor(b!) v_cache, Tensor slot_mapping) -> ()");
   m.def("paged_attention(Tensor(a!) out, Tensor query, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, float scale) -> ()");
+  m.def("paged_attention_sp(Tensor(a!) out, Tensor query, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, float scale) -> ()");
   m.def("prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens, float scale) -> ()");
   m.def("context_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens_q, Tensor kv_lens, Tensor block_tables, float scale) -> ()");
   m.def("mfma_tile_gemm(Tensor a, Tensor b) -> Tensor");
@@ -60,6 +67,7 @@ TORCH_LIBRARY(kaito, m) {
   m.def("w4a16_gemv(Tensor(a!) out, Tensor x, Tensor qweight, Tensor scales, Tensor zeros, int group) -> ()");
   m.def("w4a16_dequant(Tensor(a!) out, Tensor qweight, Tensor scales, Tensor zeros, int group) -> ()");
   m.def("allreduce_rmsnorm(Tensor(a!) out, Tensor ptrs, Tensor weight, float eps) -> ()");
+  m.def("one_shot_ar_rmsnorm(Tensor(a!) out, Tensor(b!) residual, Tensor ptrs, Tensor sig_ptrs, Tensor(c!) counter, Tensor weight, float eps, int rank) -> ()");
   m.def("ipc_handle(Tensor t) -> Tensor");
   m.def("ipc_open(Tensor handle_bytes) -> int");
   m.def("ipc_close(int ptr) -> ()");
@@ -72,6 +80,7 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("silu_and_mul", &kaito::silu_and_mul);
   m.impl("reshape_and_cache", &kaito::reshape_and_cache);
   m.impl("paged_attention", &kaito::paged_attention);
+  m.impl("paged_attention_sp", &kaito::paged_attention_sp);
   m.impl("prefill_attention", &kaito::prefill_attention);
   m.impl("context_attention", &kaito::context_attention);
   m.impl("mfma_tile_gemm", &kaito::mfma_tile_gemm);
@@ -82,6 +91,7 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("w4a16_gemv", &kaito::w4a16_gemv);
   m.impl("w4a16_dequant", &kaito::w4a16_dequant);
   m.impl("allreduce_rmsnorm", &kaito::allreduce_rmsnorm);
+  m.impl("one_shot_ar_rmsnorm", &kaito::one_shot_ar_rmsnorm);
   m.impl("ipc_handle", &kaito::ipc_handle);
   m.impl("ipc_open", &kaito::ipc_open);
   m.impl("ipc_close", &kaito::ipc_close);

@@ -86,6 +86,13 @@ class ColumnParallelLinear(_QuantMixin, nn.Module):
 
 
 class RowParallelLinear(_QuantMixin, nn.Module):
+    # set True by the model builder on linears whose output feeds a
+    # fused allreduce+RMSNorm call site (attention o_proj, MLP
+    # down_proj): when a one-shot group is active and the batch fits its
+    # staging window, forward returns the LOCAL partial and the fused
+    # kernel performs the reduction (parallel/one_shot.py).
+    fuse_norm = False
+
     def __init__(self, in_features: int, out_features: int, bias: bool = False,
                  dtype: torch.dtype = torch.bfloat16):
         super().__init__()
@@ -104,6 +111,12 @@ class RowParallelLinear(_QuantMixin, nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         out = self._qmatmul(x) if self._quantized else F.linear(x, self.weight)
         out = lora_mod.maybe_apply(self, x, out)
+        if self.fuse_norm and get_state().tp_size > 1:
+            from . import one_shot
+            if one_shot.defer(out.reshape(-1, out.shape[-1]).shape[0]):
+                # deferred: the following fused one-shot call reduces
+                assert self.bias is None
+                return out
         out = tp_all_reduce(out)
         if self.bias is not None:
             out = out + self.bias

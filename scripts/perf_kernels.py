@@ -45,9 +45,15 @@ def main():
     sl = torch.full((T,), L, dtype=torch.int32, device=dev)
     scale = 1.0 / math.sqrt(D)
 
-    t = bench(lambda: ops.paged_attention(q, kc, vc, bt, sl, scale))
+    out = torch.empty_like(q)
+    t = bench(lambda: torch.ops.kaito.paged_attention(
+        out, q, kc, vc, bt, sl, scale))
     bytes_moved = T * L * KH * D * 2 * 2  # K+V read
-    print(f"paged_attention bs={T} L={L}: {t*1e6:.1f} us  "
+    print(f"paged_attention    bs={T} L={L}: {t*1e6:.1f} us  "
+          f"{bytes_moved/t/1e12:.2f} TB/s")
+    t = bench(lambda: torch.ops.kaito.paged_attention_sp(
+        out, q, kc, vc, bt, sl, scale))
+    print(f"paged_attention_sp bs={T} L={L}: {t*1e6:.1f} us  "
           f"{bytes_moved/t/1e12:.2f} TB/s")
 
     # diagnostics: mode0 = pure read; mode1 = +dot+accumulate (no shfl)

@@ -129,6 +129,40 @@ def test_paged_attention(G, D, lens):
     _close(out, ref.to(DEV), atol=2e-2)
 
 
+@pytest.mark.parametrize("impl", ["paged_attention", "paged_attention_sp"])
+@pytest.mark.parametrize("G,D,lens", [
+    (4, 128, [1, 15, 16, 17, 400]),
+    (1, 128, [33, 256]),
+    (8, 128, [100]),
+    (3, 128, [57, 130]),
+    (4, 64, [77, 23]),
+    (4, 128, [1500]),       # multi-super-chunk path (sp: >SC*8*NW tokens)
+    (6, 128, [300]),
+])
+def test_paged_attention_impls(impl, G, D, lens):
+    """Both decode-attention implementations (r01 fused online-softmax and
+    r02 split-phase) vs the fp32 oracle."""
+    torch.manual_seed(42)
+    ops.load_extension()
+    KH = 2
+    QH = KH * G
+    BS, T = 16, len(lens)
+    max_blocks = (max(lens) + BS - 1) // BS
+    NB = T * max_blocks + 1
+    kc = _bf16(NB, KH, BS, D)
+    vc = _bf16(NB, KH, BS, D)
+    perm = torch.randperm(NB - 1)[: T * max_blocks].reshape(T, max_blocks) + 1
+    bt = perm.int().to(DEV)
+    q = _bf16(T, QH, D)
+    sl = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = torch.empty_like(q)
+    getattr(torch.ops.kaito, impl)(out, q, kc, vc, bt, sl, scale)
+    ref = R.paged_attention(q.cpu().float(), kc.cpu().float(), vc.cpu().float(),
+                            bt.cpu(), sl.cpu(), scale)
+    _close(out, ref.to(DEV), atol=2e-2)
+
+
 @pytest.mark.parametrize("lens", [[1], [5], [64], [65], [200, 200], [1, 333, 64, 17]])
 def test_prefill_attention(lens):
     torch.manual_seed(7)
@@ -404,6 +438,46 @@ def test_allreduce_rmsnorm_fused_kernel():
         var = acc.pow(2).mean(-1, keepdim=True)
         expect = acc * torch.rsqrt(var + 1e-5) * w.float()
         _close(out, expect, atol=3e-2, rtol=3e-2)
+
+
+def test_one_shot_ar_rmsnorm_kernel_single_rank():
+    """Graph-capturable one-shot kernel (in-kernel barrier + residual
+    fusion) at world=1: the barrier degenerates to self-signal; numerics
+    must match allreduce → bf16 residual add → RMSNorm. The multi-rank
+    IPC path runs on the driver's 8-GPU node."""
+    import kaito_amd.ops as O
+    from kaito_amd.ops import torch_ref
+    O.load_extension()
+    torch.manual_seed(33)
+    T, H = 64, 4096
+    for use_residual in (True, False):
+        x = _bf16(T, H)
+        res = _bf16(T, H) if use_residual else \
+            torch.empty(0, dtype=torch.bfloat16, device=DEV)
+        res_ref = res.clone()
+        w = _bf16(H, scale=0.5) + 1.0
+        staging = torch.empty(T, H, dtype=torch.bfloat16, device=DEV)
+        staging.copy_(x)
+        sig = torch.zeros(T * 8, dtype=torch.int32, device=DEV)
+        counter = torch.zeros(T, dtype=torch.int32, device=DEV)
+        ptrs = torch.tensor([staging.data_ptr()], dtype=torch.long,
+                            device=DEV)
+        sig_ptrs = torch.tensor([sig.data_ptr()], dtype=torch.long,
+                                device=DEV)
+        out = torch.empty_like(x)
+        # run twice: the device-side epoch counter must keep advancing
+        for _ in range(2):
+            r = res.clone() if use_residual else res
+            torch.ops.kaito.one_shot_ar_rmsnorm(
+                out, r, ptrs, sig_ptrs, counter, w, 1e-5, 0)
+        if use_residual:
+            expect, new_res = torch_ref.fused_add_rms_norm(
+                x.cpu(), res_ref.cpu(), w.cpu(), 1e-5)
+            _close(r, new_res.to(DEV), atol=3e-2, rtol=3e-2)
+        else:
+            expect = torch_ref.rms_norm(x.cpu(), w.cpu(), 1e-5)
+        _close(out, expect.to(DEV), atol=3e-2, rtol=3e-2)
+        assert counter.max().item() == 2  # device epoch advanced per call
 
 
 def test_ipc_handle_roundtrip_bytes():

@@ -124,6 +124,34 @@ def _body_llama_tp(rank, world):
     assert int(logits_tp.argmax()) == int(logits_ref.argmax())
 
 
+def _body_llama_tp_one_shot(rank, world):
+    """Same sharded-forward oracle as _body_llama_tp, but with the
+    one-shot fused allreduce+RMSNorm group ACTIVE (gloo-emulated): the
+    RowParallel layers defer their ring all-reduce into the fused norm
+    call sites, and the result must still match the tp=1 model. This
+    validates the defer wiring the 8-GPU hipIpc path relies on."""
+    from kaito_amd.parallel import one_shot
+    cfg = get_model_config("tiny-llama-test")
+    grp = one_shot.GlooEmulatedGroup(max_tokens=64, hidden=cfg.hidden_size)
+    one_shot.activate(grp)
+    # count deferred fused calls to prove the one-shot path actually ran
+    calls = {"n": 0}
+    orig = grp.allreduce_add_rmsnorm
+
+    def counting(*a, **k):
+        calls["n"] += 1
+        return orig(*a, **k)
+
+    grp.allreduce_add_rmsnorm = counting
+    try:
+        _body_llama_tp(rank, world)
+    finally:
+        one_shot.activate(None)
+    # 2 fused sites per layer (minus layer0 input norm) + final norm
+    assert calls["n"] >= 2 * cfg.num_layers, \
+        f"one-shot path not exercised ({calls['n']} calls)"
+
+
 # ---- tests -----------------------------------------------------------------
 def test_tp_parallel_linear_world2():
     _spawn("_body_linear", port=29611)
@@ -131,6 +159,10 @@ def test_tp_parallel_linear_world2():
 
 def test_tp_llama_forward_world2():
     _spawn("_body_llama_tp", port=29613)
+
+
+def test_tp_llama_forward_world2_one_shot_fused():
+    _spawn("_body_llama_tp_one_shot", port=29627)
 
 
 def test_vocab_parallel_embedding_single():
