@@ -119,7 +119,11 @@ void paged_attention_sp_kernel(
         kd[u] = *reinterpret_cast<const short8_t*>(
             k_cache + gb + u * (TPC * D));
       const int nu = min(CPB, n_ii - ii0);
-      for (int u = 0; u < nu; u++) {
+      // compile-time trip count: a runtime-bounded loop would make
+      // kd[u] a runtime-indexed register array → scratch (rule #20)
+#pragma unroll
+      for (int u = 0; u < CPB; u++) {
+        if (u >= nu) break;
         const bool valid = ((c0 + u) * TPC + tg) < seq_len;
         float s[G];
 #pragma unroll
@@ -187,7 +191,9 @@ void paged_attention_sp_kernel(
         vd[u] = *reinterpret_cast<const short8_t*>(
             v_cache + gb + u * (TPC * D));
       const int nu = min(CPB, n_ii - ii0);
-      for (int u = 0; u < nu; u++) {
+#pragma unroll
+      for (int u = 0; u < CPB; u++) {
+        if (u >= nu) break;
         // G probabilities for this token: one vector LDS broadcast
         float pg[G];
         if constexpr (G4 == 2) {
