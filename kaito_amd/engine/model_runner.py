@@ -284,12 +284,17 @@ class ModelRunner:
         shared memory pool is sized once)."""
         if not self.is_gpu or self.cfg.enforce_eager:
             return
-        if self.cfg.model.num_experts > 0:
-            # the expert-sorted grouped-GEMM MoE path reads per-expert
-            # counts on the host (data-dependent shapes) — illegal inside
-            # hipGraph capture; MoE decodes eagerly until the fused MoE
-            # kernel lands (docs/ROADMAP.md)
-            logger.info("MoE model: skipping decode graph capture")
+        m = self.cfg.model
+        tp = max(ps.get_state().tp_size, 1)
+        ie = m.moe_intermediate_size or m.intermediate_size
+        ie_local = ie if (tp > 1 and m.num_experts % tp == 0) or tp == 1 \
+            else ie // tp   # mirrors MoEMLP's EP-vs-IE sharding choice
+        if m.num_experts > 0 and (m.hidden_size % 64 != 0 or
+                                  ie_local % 64 != 0):
+            # odd shapes fall back to the per-expert torch loop, whose
+            # host-side segment reads cannot be captured
+            logger.info("MoE model with non-64-aligned dims: "
+                        "skipping decode graph capture")
             return
         if not hasattr(self, "_buf"):
             self._init_decode_buffers()

@@ -1,12 +1,23 @@
 """Mixture-of-experts FFN (Mixtral / Qwen-MoE / gpt-oss class).
 
 Replaces vLLM's fused-MoE CUDA path for the MoE presets in the reference
-catalog (SURVEY.md §2.3 note: gpt-oss-120b / Qwen MoE presets in
-supported_models.yaml). Round-1 implementation: top-k softmax gating +
-expert-sorted grouped GEMMs (torch/hipBLASLt); tokens are sorted by expert
-so each expert runs one contiguous GEMM (the standard moe-align layout a
-future fused HIP kernel will consume directly). EP/TP sharding of experts
-lands with the fused kernel.
+catalog (gpt-oss-120b / Qwen MoE presets in
+supported_models.yaml).
+
+Round 2: the GPU path is FUSED and host-sync-free — tokens are sorted by
+expert on the device and two HIP grouped-GEMM kernels (ops/csrc/moe.hip)
+consume (sorted_ids, offsets) directly: gather→gate/up GEMM→silu·mul,
+then down GEMM→weighted scatter-add. Worst-case tile grids make every
+shape static, so MoE decode steps capture into hipGraphs (the round-1
+per-expert torch loop read expert counts on the host each layer, forcing
+eager decode — docs/ROADMAP.md #3).
+
+Parallelism at tp>1:
+  * EP (expert parallel) when num_experts % tp == 0: each rank owns
+    E/tp experts; non-local tokens contribute nothing locally and the
+    rank outputs are summed by the existing TP all-reduce.
+  * TP fallback (IE sharding) otherwise: every rank runs all experts on
+    an intermediate-dim shard, same all-reduce.
 """
 from __future__ import annotations
 
@@ -15,56 +26,104 @@ import torch.nn as nn
 
 from .. import ops
 from ..engine.config import ModelConfig
-from ..parallel.state import get_state
+from ..parallel.state import get_state, tp_all_reduce
 
 
 class MoEMLP(nn.Module):
     def __init__(self, cfg: ModelConfig):
         super().__init__()
-        assert get_state().tp_size == 1, \
-            "MoE presets currently require tp=1 (expert parallelism lands " \
-            "with the fused HIP MoE kernel)"
+        st = get_state()
+        tp = st.tp_size
         self.cfg = cfg
         h = cfg.hidden_size
         ie = cfg.moe_intermediate_size or cfg.intermediate_size
         e = cfg.num_experts
         self.top_k = cfg.num_experts_per_tok
+        self.num_experts = e
+        if tp > 1 and e % tp == 0:
+            # expert parallelism: contiguous expert slice per rank
+            self.e_local = e // tp
+            self.e_base = st.tp_rank * self.e_local
+            self.ie_local = ie
+        else:
+            assert tp == 1 or ie % tp == 0, (ie, tp)
+            self.e_local = e
+            self.e_base = 0
+            self.ie_local = ie // tp
         self.gate = nn.Parameter(torch.empty(e, h, dtype=cfg.dtype),
                                  requires_grad=False)
-        # stacked expert weights: [E, 2*ie, h] (gate|up) and [E, h, ie]
+        # stacked expert weights (local shard):
+        # [E_l, 2*IE_l, h] (gate|up) and [E_l, h, IE_l]
         self.w_gate_up = nn.Parameter(
-            torch.empty(e, 2 * ie, h, dtype=cfg.dtype), requires_grad=False)
+            torch.empty(self.e_local, 2 * self.ie_local, h, dtype=cfg.dtype),
+            requires_grad=False)
         self.w_down = nn.Parameter(
-            torch.empty(e, h, ie, dtype=cfg.dtype), requires_grad=False)
+            torch.empty(self.e_local, h, self.ie_local, dtype=cfg.dtype),
+            requires_grad=False)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        T, H = x.shape
+    def _route(self, x: torch.Tensor):
+        """Device-only top-k routing + expert sort. Returns
+        (sorted_tok int32 [TK], gates f32 [TK], offsets int32 [E+1])."""
+        T = x.size(0)
         logits = torch.nn.functional.linear(x, self.gate)      # [T, E]
         probs = torch.softmax(logits.float(), dim=-1)
         topw, topi = torch.topk(probs, self.top_k, dim=-1)     # [T, K]
-        topw = (topw / topw.sum(-1, keepdim=True)).to(x.dtype)
-
-        # moe-align: flatten (token, k) pairs, sort by expert
+        topw = topw / topw.sum(-1, keepdim=True)
         flat_e = topi.reshape(-1)                              # [T*K]
-        flat_t = torch.arange(T, device=x.device).repeat_interleave(self.top_k)
+        flat_t = torch.arange(T, device=x.device,
+                              dtype=torch.int32).repeat_interleave(self.top_k)
         order = torch.argsort(flat_e, stable=True)
-        se, st_idx = flat_e[order], flat_t[order]
-        counts = torch.bincount(se, minlength=self.cfg.num_experts)
+        sorted_tok = flat_t[order]
+        gates = topw.reshape(-1)[order].float()
+        counts = torch.zeros(self.num_experts, device=x.device,
+                             dtype=torch.int32)
+        counts.scatter_add_(0, flat_e,
+                            torch.ones_like(flat_e, dtype=torch.int32))
+        offsets = torch.zeros(self.num_experts + 1, device=x.device,
+                              dtype=torch.int32)
+        torch.cumsum(counts, 0, out=offsets[1:])
+        return sorted_tok, gates, offsets
 
-        xs = x[st_idx]                                         # [T*K, H]
-        out_sorted = torch.empty_like(xs)
-        start = 0
-        for eid, n in enumerate(counts.tolist()):
-            if n == 0:
-                continue
-            sl = slice(start, start + n)
-            h1 = torch.nn.functional.linear(xs[sl], self.w_gate_up[eid])
-            act = ops.silu_and_mul(h1.contiguous())
-            out_sorted[sl] = torch.nn.functional.linear(act, self.w_down[eid])
-            start += n
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        T, H = x.shape
+        sorted_tok, gates, offsets = self._route(x)
+        if x.is_cuda and H % 64 == 0 and self.ie_local % 64 == 0:
+            out = self._forward_fused(x, sorted_tok, gates, offsets)
+        else:
+            out = self._forward_loop(x, sorted_tok, gates, offsets)
+        if get_state().tp_size > 1:
+            out = tp_all_reduce(out)
+        return out
 
-        # scatter-add back with gating weights
-        w_sorted = topw.reshape(-1)[order].unsqueeze(1)
+    def _forward_fused(self, x, sorted_tok, gates, offsets):
+        """Static-shape HIP grouped GEMMs (graph-capturable)."""
+        T, H = x.shape
+        TK = T * self.top_k
+        pad = (TK + 63) // 64 * 64   # staging reads whole 64-row tiles
+        act = torch.empty(pad, self.ie_local, dtype=x.dtype, device=x.device)
+        ops.moe_gate_silu(act, x, self.w_gate_up, sorted_tok, offsets,
+                          self.e_base, self.e_local)
+        out32 = torch.zeros(T, H, dtype=torch.float32, device=x.device)
+        ops.moe_down_scatter(out32, act, self.w_down, sorted_tok, gates,
+                             offsets, self.e_base, self.e_local)
+        return out32.to(x.dtype)
+
+    def _forward_loop(self, x, sorted_tok, gates, offsets):
+        """Per-expert torch GEMMs (CPU tests / odd shapes). Host sync on
+        the segment sizes — never used on the GPU hot path."""
+        off = offsets.tolist()
         out = torch.zeros_like(x)
-        out.index_add_(0, st_idx, out_sorted * w_sorted)
+        idx64 = sorted_tok.long()
+        for le in range(self.e_local):
+            e = self.e_base + le
+            s, t = off[e], off[e + 1]
+            if s == t:
+                continue
+            rows = idx64[s:t]
+            xs = x[rows]
+            h1 = torch.nn.functional.linear(xs, self.w_gate_up[le])
+            act = ops.silu_and_mul(h1.contiguous())
+            y = torch.nn.functional.linear(act, self.w_down[le])
+            out.index_add_(0, rows,
+                           (y.float() * gates[s:t, None]).to(out.dtype))
         return out

@@ -170,6 +170,28 @@ def _build_tiles(cu_seqlens: torch.Tensor):
             torch.tensor(bases, dtype=torch.int32, device=dev))
 
 
+def moe_gate_silu(act: torch.Tensor, x: torch.Tensor,
+                  w_gate_up: torch.Tensor, sorted_ids: torch.Tensor,
+                  offsets: torch.Tensor, e_base: int,
+                  n_local_experts: int) -> None:
+    """Fused MoE stage 1 (GPU only): gather rows by sorted token ids,
+    gate/up grouped GEMM per expert, silu·mul epilogue into `act`."""
+    _require_ext()
+    torch.ops.kaito.moe_gate_silu(act, x, w_gate_up, sorted_ids, offsets,
+                                  e_base, n_local_experts)
+
+
+def moe_down_scatter(out: torch.Tensor, act: torch.Tensor,
+                     w_down: torch.Tensor, sorted_ids: torch.Tensor,
+                     gates: torch.Tensor, offsets: torch.Tensor,
+                     e_base: int, n_local_experts: int) -> None:
+    """Fused MoE stage 2 (GPU only): down grouped GEMM + gated f32
+    atomic scatter into `out`."""
+    _require_ext()
+    torch.ops.kaito.moe_down_scatter(out, act, w_down, sorted_ids, gates,
+                                     offsets, e_base, n_local_experts)
+
+
 def w4a16_gemv(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
                zeros: torch.Tensor, group: int) -> torch.Tensor:
     """Group-quantized W4A16 linear, small M (decode): out = x @ W^T with

@@ -27,6 +27,13 @@ void context_attention(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor kv_lens, at::Tensor block_tables,
                        double scale);
 at::Tensor mfma_tile_gemm(at::Tensor a, at::Tensor b);
+void moe_gate_silu(at::Tensor act, at::Tensor x, at::Tensor w_gate_up,
+                   at::Tensor sorted_ids, at::Tensor offsets, int64_t e_base,
+                   int64_t n_local_experts);
+void moe_down_scatter(at::Tensor out, at::Tensor act, at::Tensor w_down,
+                      at::Tensor sorted_ids, at::Tensor gates,
+                      at::Tensor offsets, int64_t e_base,
+                      int64_t n_local_experts);
 void topk(at::Tensor out_vals, at::Tensor out_idx, at::Tensor scores,
           int64_t k);
 void lora_shrink(at::Tensor tmp, at::Tensor x, at::Tensor A, at::Tensor idx,
@@ -60,6 +67,8 @@ TORCH_LIBRARY(kaito, m) {
   m.def("prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens, float scale) -> ()");
   m.def("context_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens_q, Tensor kv_lens, Tensor block_tables, float scale) -> ()");
   m.def("mfma_tile_gemm(Tensor a, Tensor b) -> Tensor");
+  m.def("moe_gate_silu(Tensor(a!) act, Tensor x, Tensor w_gate_up, Tensor sorted_ids, Tensor offsets, int e_base, int n_local_experts) -> ()");
+  m.def("moe_down_scatter(Tensor(a!) out, Tensor act, Tensor w_down, Tensor sorted_ids, Tensor gates, Tensor offsets, int e_base, int n_local_experts) -> ()");
   m.def("topk(Tensor(a!) out_vals, Tensor(b!) out_idx, Tensor scores, int k) -> ()");
   m.def("lora_shrink(Tensor(a!) tmp, Tensor x, Tensor A, Tensor idx, float scale) -> ()");
   m.def("lora_expand(Tensor(a!) y, Tensor tmp, Tensor B, Tensor idx) -> ()");
@@ -84,6 +93,8 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("prefill_attention", &kaito::prefill_attention);
   m.impl("context_attention", &kaito::context_attention);
   m.impl("mfma_tile_gemm", &kaito::mfma_tile_gemm);
+  m.impl("moe_gate_silu", &kaito::moe_gate_silu);
+  m.impl("moe_down_scatter", &kaito::moe_down_scatter);
   m.impl("topk", &kaito::topk);
   m.impl("lora_shrink", &kaito::lora_shrink);
   m.impl("lora_expand", &kaito::lora_expand);

@@ -424,6 +424,71 @@ def test_engine_gpu_mixed_overlap_matches_classic():
     assert run(True) == run(False)
 
 
+# --------------------------------------------------------- fused MoE GEMMs
+@pytest.mark.parametrize("T,H,IE,E,K", [
+    (7, 256, 256, 4, 2),       # partial tiles, tiny experts
+    (130, 512, 448, 8, 2),     # multi-tile, uneven expert loads
+    (64, 256, 1024, 4, 4),     # wide intermediate, top-4
+])
+def test_moe_fused_grouped_gemms(T, H, IE, E, K):
+    """The host-sync-free MoE path (sorted ids + offsets → gather GEMM →
+    silu·mul → down GEMM → gated scatter) vs an fp32 per-expert loop."""
+    from kaito_amd.models.moe import MoEMLP
+    from kaito_amd.engine.config import ModelConfig
+    from kaito_amd.parallel.state import init_parallel
+    init_parallel(1)
+    torch.manual_seed(5)
+    cfg = ModelConfig(name="t", hidden_size=H, num_experts=E,
+                      num_experts_per_tok=K, moe_intermediate_size=IE)
+    m = MoEMLP(cfg).to(DEV)
+    with torch.no_grad():
+        for p in m.parameters():
+            p.normal_(0, 0.05)
+    x = _bf16(T, H, scale=0.5)
+    sorted_tok, gates, offsets = m._route(x)
+    out_fused = m._forward_fused(x, sorted_tok, gates, offsets)
+    # fp32 oracle on the same routing
+    off = offsets.tolist()
+    expect = torch.zeros(T, H, dtype=torch.float32, device=DEV)
+    idx = sorted_tok.long()
+    for e in range(E):
+        s, t = off[e], off[e + 1]
+        if s == t:
+            continue
+        xs = x[idx[s:t]].float()
+        h1 = xs @ m.w_gate_up[e].float().T
+        g, u = h1.chunk(2, dim=-1)
+        act = torch.nn.functional.silu(g) * u
+        y = act @ m.w_down[e].float().T
+        expect.index_add_(0, idx[s:t], y * gates[s:t, None])
+    _close(out_fused, expect, atol=5e-2, rtol=5e-2)
+
+
+def test_moe_engine_decode_uses_graphs_on_gpu():
+    """MoE decode must capture into hipGraphs (the round-1 blocker was
+    the host-side expert-count read) and still match the eager path."""
+    from kaito_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from kaito_amd.models import get_model_config
+    from kaito_amd.parallel.state import init_parallel
+    init_parallel(1)
+    cfg = EngineConfig(model=get_model_config("tiny-moe-test"),
+                       device=DEV, max_num_seqs=4, num_gpu_blocks=64,
+                       max_model_len=64, enforce_eager=False)
+    eng = LLMEngine(cfg)
+    eng.capture_graphs()
+    assert eng.runner._graphs, "MoE decode graphs were not captured"
+    prompt = [9, 8, 7, 6, 5]
+    out = eng.generate([prompt],
+                       SamplingParams(max_tokens=6, ignore_eos=True))
+    cfg2 = EngineConfig(model=get_model_config("tiny-moe-test"),
+                        device=DEV, max_num_seqs=4, num_gpu_blocks=64,
+                        max_model_len=64, enforce_eager=True)
+    eng2 = LLMEngine(cfg2)
+    out2 = eng2.generate([prompt],
+                         SamplingParams(max_tokens=6, ignore_eos=True))
+    assert out[0].output_token_ids == out2[0].output_token_ids
+
+
 # --------------------------------------------------- one-shot AR+RMSNorm
 def test_allreduce_rmsnorm_fused_kernel():
     """Fused one-shot allreduce+RMSNorm core: N peer buffers (local here;

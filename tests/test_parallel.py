@@ -152,6 +152,41 @@ def _body_llama_tp_one_shot(rank, world):
         f"one-shot path not exercised ({calls['n']} calls)"
 
 
+def _body_moe_ep(rank, world):
+    """Expert-parallel MoEMLP (E % tp == 0 → E/tp experts per rank, summed
+    by all-reduce) must match the tp=1 module on the same full weights."""
+    from kaito_amd.models.moe import MoEMLP
+    from kaito_amd.parallel import state as ps
+    cfg = get_model_config("tiny-moe-test")
+    st = ps.get_state()
+    torch.manual_seed(11)
+    # full-weight reference at tp=1
+    saved = st.tp_size, st.tp_rank
+    st.tp_size, st.tp_rank = 1, 0
+    ref = MoEMLP(cfg)
+    with torch.no_grad():
+        for p in ref.parameters():
+            p.normal_(0, 0.05)
+    st.tp_size, st.tp_rank = saved
+
+    ep = MoEMLP(cfg)
+    assert ep.e_local == cfg.num_experts // world, "EP sharding not chosen"
+    el = ep.e_local
+    with torch.no_grad():
+        ep.gate.copy_(ref.gate)
+        ep.w_gate_up.copy_(ref.w_gate_up[rank * el:(rank + 1) * el])
+        ep.w_down.copy_(ref.w_down[rank * el:(rank + 1) * el])
+
+    x = torch.randn(9, cfg.hidden_size, dtype=torch.bfloat16)
+    out = ep(x)
+    st.tp_size, st.tp_rank = 1, 0
+    expect = ref(x)
+    st.tp_size, st.tp_rank = saved
+    assert torch.allclose(out.float(), expect.float(), atol=0.05,
+                          rtol=0.05), \
+        f"EP mismatch {(out.float()-expect.float()).abs().max()}"
+
+
 # ---- tests -----------------------------------------------------------------
 def test_tp_parallel_linear_world2():
     _spawn("_body_linear", port=29611)
@@ -163,6 +198,10 @@ def test_tp_llama_forward_world2():
 
 def test_tp_llama_forward_world2_one_shot_fused():
     _spawn("_body_llama_tp_one_shot", port=29627)
+
+
+def test_moe_expert_parallel_world2():
+    _spawn("_body_moe_ep", port=29631)
 
 
 def test_vocab_parallel_embedding_single():
