@@ -54,9 +54,11 @@ KAITO_DEV void stage_tile_gathered(
     const int row = idx >> 3;
     const int c = (idx & 7) ^ (row & 7);   // pre-swizzled source chunk
     const short* src = src_base + (int64_t)row_ids[row] * row_stride + c * 8;
+    // segment = 64 lanes x 16 B = 512 shorts; HW writes lane i at
+    // base + i*16 (wave-uniform destination)
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)src,
-        (__attribute__((address_space(3))) unsigned int*)(lds + seg * 64),
+        (__attribute__((address_space(3))) unsigned int*)(lds + seg * 512),
         16, 0, 0);
   }
 }
@@ -74,7 +76,7 @@ KAITO_DEV void stage_tile_rows(
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)
             (src + (int64_t)row * row_stride + c * 8),
-        (__attribute__((address_space(3))) unsigned int*)(lds + seg * 64),
+        (__attribute__((address_space(3))) unsigned int*)(lds + seg * 512),
         16, 0, 0);
   }
 }
