@@ -21,7 +21,10 @@ from pathlib import Path
 
 import yaml
 
-OUT_DIR = Path(__file__).resolve().parent.parent / "config" / "crd" / "bases"
+_REPO = Path(__file__).resolve().parent.parent
+OUT_DIR = _REPO / "config" / "crd" / "bases"
+# Helm installs the same CRDs from the chart's crds/ directory
+CHART_CRD_DIR = _REPO / "charts" / "kaito-amd" / "crds"
 
 
 # ---------------------------------------------------------------- helpers
@@ -394,16 +397,17 @@ def main() -> int:
     args = p.parse_args()
     rc = 0
     OUT_DIR.mkdir(parents=True, exist_ok=True)
+    CHART_CRD_DIR.mkdir(parents=True, exist_ok=True)
     for fname, doc in build().items():
         text = HEADER + yaml.safe_dump(doc, sort_keys=False, width=78)
-        path = OUT_DIR / fname
-        if args.check:
-            if not path.exists() or path.read_text() != text:
-                print(f"DRIFT: {path}", file=sys.stderr)
-                rc = 1
-        else:
-            path.write_text(text)
-            print(f"wrote {path}")
+        for path in (OUT_DIR / fname, CHART_CRD_DIR / fname):
+            if args.check:
+                if not path.exists() or path.read_text() != text:
+                    print(f"DRIFT: {path}", file=sys.stderr)
+                    rc = 1
+            else:
+                path.write_text(text)
+                print(f"wrote {path}")
     return rc
 
 
