@@ -32,10 +32,53 @@ class ModelConfig:
     partial_rotary_factor: float = 1.0
     # attention bias (qwen1/2 style)
     attention_bias: bool = False
+    # ---- architecture variants (gemma / phi-2 / gpt-oss native support) ----
+    # MLP activation: "silu" (llama SwiGLU) | "gelu_tanh" (gemma GeGLU) |
+    # "gelu" (phi-2 plain-GELU, UNGATED fc1→act→fc2 MLP)
+    hidden_act: str = "silu"
+    # ungated MLP (phi-2/falcon): fc1 [I,H] → act → fc2 [H,I]
+    gated_mlp: bool = True
+    # norm layer: "rmsnorm" | "layernorm" (phi-2/falcon, with bias)
+    norm_type: str = "rmsnorm"
+    # gemma stores RMSNorm weights as w with y = x*(1+w); folded to (1+w)
+    # at load/init time so the kernel is unchanged
+    rms_norm_offset: bool = False
+    # gemma scales embeddings by sqrt(hidden_size)
+    embed_scale: float = 1.0
+    # phi-2/falcon parallel block: h += attn(ln(x)) + mlp(ln(x))
+    parallel_block: bool = False
+    # gemma-2/3 sandwich norms: extra pre/post feedforward + post-attn norms
+    sandwich_norms: bool = False
+    # per-head RMSNorm on q/k before RoPE (gemma-3, qwen3)
+    qk_norm: bool = False
+    # sliding-window attention: window size (0 = full); layer pattern
+    # "all" | "interleaved:N" (1 global every N layers — gemma-3 style,
+    # global layers are (i+1) % N == 0)
+    sliding_window: int = 0
+    sliding_window_pattern: str = "all"
+    # rope theta for LOCAL (sliding) layers when it differs (gemma-3:
+    # local 10k, global rope_theta); 0 → use rope_theta everywhere
+    rope_theta_local: float = 0.0
+    # attention sinks (gpt-oss): learned per-head logit folded into the
+    # softmax denominator
+    attn_sinks: bool = False
+    # final logit soft-capping (gemma-2): tanh cap value, 0 = off
+    final_logit_softcap: float = 0.0
     # mixture-of-experts (mixtral / qwen-moe / gpt-oss class); 0 = dense
     num_experts: int = 0
     num_experts_per_tok: int = 2
     moe_intermediate_size: int = 0       # per-expert FFN width (0 → dense I)
+    # expert activation: "silu" (SwiGLU) | "swiglu_oai" (gpt-oss clamped
+    # swiglu: (up+1) * gate*sigmoid(1.702*gate), clamp ±7)
+    moe_act: str = "silu"
+    moe_bias: bool = False               # per-expert gate_up/down biases
+    # routing: "softmax_topk" (mixtral: softmax→topk→renorm) |
+    # "topk_softmax" (gpt-oss: topk logits→softmax over the k)
+    moe_routing: str = "softmax_topk"
+    # attention scale override (padded-head models: phi-2 pads 80→128 for
+    # the D∈{64,128,256} kernels but keeps 80^-0.5 scaling; zero-padded
+    # dims contribute nothing, so the math is exact)
+    attn_scale: Optional[float] = None
     # weight quantization: "" (bf16) | "w4a16"/"awq" (4-bit group-quantized
     # linears via the HIP GEMV / dequant+MFMA kernels)
     quant_method: str = ""
@@ -80,6 +123,20 @@ class ModelConfig:
     def rotary_dim(self) -> int:
         r = int(self.head_dim * self.partial_rotary_factor)
         return r - (r % 2)
+
+    def layer_sliding_window(self, layer_idx: int) -> int:
+        """Effective attention window for a layer (0 = full). Pattern
+        "interleaved:N": one global layer every N (layers with
+        (i+1) % N == 0 are global — gemma-3 is interleaved:6, gpt-oss
+        alternates as interleaved:2)."""
+        if self.sliding_window <= 0:
+            return 0
+        if self.sliding_window_pattern == "all":
+            return self.sliding_window
+        if self.sliding_window_pattern.startswith("interleaved:"):
+            n = int(self.sliding_window_pattern.split(":", 1)[1])
+            return 0 if (layer_idx + 1) % n == 0 else self.sliding_window
+        raise ValueError(self.sliding_window_pattern)
 
     def kv_bytes_per_token(self, tp_size: int = 1) -> int:
         """Per-token KV cache bytes across all layers (per TP rank)."""

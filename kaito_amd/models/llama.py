@@ -41,12 +41,13 @@ class AttnMetadata:
     kv_lens: Optional[torch.Tensor] = None       # [B] int32 total ctx len
 
 
-def build_cos_sin_cache(cfg: ModelConfig, device, max_pos: Optional[int] = None
-                        ) -> torch.Tensor:
+def build_cos_sin_cache(cfg: ModelConfig, device, max_pos: Optional[int] = None,
+                        theta: Optional[float] = None) -> torch.Tensor:
     """[max_pos, rot_dim] f32 = [cos | sin] table."""
     rot = cfg.rotary_dim
     max_pos = max_pos or cfg.max_position
-    inv = 1.0 / (cfg.rope_theta ** (
+    theta = theta or cfg.rope_theta
+    inv = 1.0 / (theta ** (
         torch.arange(0, rot, 2, dtype=torch.float64, device=device) / rot))
     t = torch.arange(max_pos, dtype=torch.float64, device=device)
     freqs = torch.outer(t, inv)
@@ -54,14 +55,18 @@ def build_cos_sin_cache(cfg: ModelConfig, device, max_pos: Optional[int] = None
 
 
 class LlamaAttention(nn.Module):
-    def __init__(self, cfg: ModelConfig):
+    def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
         super().__init__()
         tp = get_state().tp_size
         self.cfg = cfg
         self.head_dim = cfg.head_dim
         self.num_heads = cfg.num_heads // tp
         self.num_kv_heads = max(cfg.num_kv_heads // tp, 1)
-        self.scale = 1.0 / math.sqrt(self.head_dim)
+        self.scale = cfg.attn_scale or 1.0 / math.sqrt(self.head_dim)
+        # sliding-window layers (gemma-3 / gpt-oss interleaved patterns)
+        self.window = cfg.layer_sliding_window(layer_idx)
+        # gemma-3: local (sliding) layers use a different rope theta
+        self.local_rope = cfg.rope_theta_local > 0 and self.window > 0
         h = cfg.hidden_size
         self.qkv_proj = ColumnParallelLinear(
             h, (cfg.num_heads + 2 * cfg.num_kv_heads) * cfg.head_dim,
@@ -70,19 +75,42 @@ class LlamaAttention(nn.Module):
             cfg.num_heads * cfg.head_dim, h, bias=False, dtype=cfg.dtype)
         # o_proj feeds post_attention_layernorm: with an active one-shot
         # group its ring all-reduce is deferred into the fused
-        # allreduce+add+RMSNorm kernel (parallel/one_shot.py)
-        self.o_proj.fuse_norm = True
+        # allreduce+add+RMSNorm kernel (parallel/one_shot.py); only the
+        # standard pre-norm block structure has that call site
+        self.o_proj.fuse_norm = (not cfg.parallel_block
+                                 and not cfg.sandwich_norms
+                                 and cfg.norm_type == "rmsnorm")
         self.q_size = self.num_heads * self.head_dim
         self.kv_size = self.num_kv_heads * self.head_dim
+        if cfg.qk_norm:   # per-head RMSNorm before RoPE (gemma-3, qwen3)
+            self.q_norm = nn.Parameter(
+                torch.empty(cfg.head_dim, dtype=cfg.dtype),
+                requires_grad=False)
+            self.k_norm = nn.Parameter(
+                torch.empty(cfg.head_dim, dtype=cfg.dtype),
+                requires_grad=False)
+        if cfg.attn_sinks:  # learned softmax sinks (gpt-oss)
+            self.sinks = nn.Parameter(
+                torch.empty(self.num_heads, dtype=torch.float32),
+                requires_grad=False)
 
     def forward(self, x: torch.Tensor, positions: torch.Tensor,
                 kv_cache: Optional[Tuple[torch.Tensor, torch.Tensor]],
-                meta: AttnMetadata, cos_sin: torch.Tensor) -> torch.Tensor:
+                meta: AttnMetadata, cos_sin: torch.Tensor,
+                cos_sin_local: Optional[torch.Tensor] = None) -> torch.Tensor:
         qkv = self.qkv_proj(x)
         # strided views into the fused qkv buffer — the HIP kernels take row
         # strides, so no .contiguous() copies on the hot path.
         q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
-        q, k = ops.rotary_embedding(positions, q, k, self.head_dim, cos_sin)
+        if self.cfg.qk_norm:
+            q = ops.rms_norm(
+                q.reshape(-1, self.head_dim).contiguous(), self.q_norm,
+                self.cfg.rms_eps).reshape(q.shape[0], -1)
+            k = ops.rms_norm(
+                k.reshape(-1, self.head_dim).contiguous(), self.k_norm,
+                self.cfg.rms_eps).reshape(k.shape[0], -1)
+        cs = cos_sin_local if self.local_rope else cos_sin
+        q, k = ops.rotary_embedding(positions, q, k, self.head_dim, cs)
         T = x.size(0)
         qh = q.unflatten(-1, (self.num_heads, self.head_dim))
         kh = k.unflatten(-1, (self.num_kv_heads, self.head_dim))
@@ -90,38 +118,58 @@ class LlamaAttention(nn.Module):
         if kv_cache is not None:
             ops.reshape_and_cache(kh, vh, kv_cache[0], kv_cache[1],
                                   meta.slot_mapping)
+        sinks = getattr(self, "sinks", None)
         if meta.is_prefill:
             if meta.kv_lens is not None:
                 # suffix-query attention over the paged cache (the suffix
                 # K/V was just written by reshape_and_cache above)
                 out = ops.context_attention(qh, kv_cache[0], kv_cache[1],
                                             meta.cu_seqlens, meta.kv_lens,
-                                            meta.block_tables, self.scale)
+                                            meta.block_tables, self.scale,
+                                            self.window, sinks)
             else:
                 out = ops.prefill_attention(qh, kh, vh, meta.cu_seqlens,
-                                            self.scale, meta.max_seqlen)
+                                            self.scale, meta.max_seqlen,
+                                            self.window, sinks)
         else:
             out = ops.paged_attention(qh, kv_cache[0], kv_cache[1],
                                       meta.block_tables, meta.seq_lens,
-                                      self.scale)
-        return self.o_proj(out.view(T, -1))
+                                      self.scale, self.window, sinks)
+        return self.o_proj(out.reshape(T, -1))
 
 
 class LlamaMLP(nn.Module):
     def __init__(self, cfg: ModelConfig):
         super().__init__()
-        self.gate_up_proj = ColumnParallelLinear(
-            cfg.hidden_size, 2 * cfg.intermediate_size, dtype=cfg.dtype)
-        self.down_proj = RowParallelLinear(
-            cfg.intermediate_size, cfg.hidden_size, dtype=cfg.dtype)
+        self.cfg = cfg
+        fuse = (not cfg.parallel_block and not cfg.sandwich_norms
+                and cfg.norm_type == "rmsnorm")
+        if cfg.gated_mlp:
+            self.gate_up_proj = ColumnParallelLinear(
+                cfg.hidden_size, 2 * cfg.intermediate_size, dtype=cfg.dtype)
+            self.down_proj = RowParallelLinear(
+                cfg.intermediate_size, cfg.hidden_size, dtype=cfg.dtype)
+        else:
+            # ungated (phi-2/falcon): fc1 → act → fc2, bias follows the
+            # model's attention-bias convention
+            self.up_proj = ColumnParallelLinear(
+                cfg.hidden_size, cfg.intermediate_size,
+                bias=cfg.attention_bias, dtype=cfg.dtype)
+            self.down_proj = RowParallelLinear(
+                cfg.intermediate_size, cfg.hidden_size,
+                bias=cfg.attention_bias, dtype=cfg.dtype)
         # down_proj feeds the next input_layernorm / final norm
-        self.down_proj.fuse_norm = True
+        self.down_proj.fuse_norm = fuse
         tp = get_state().tp_size
         self.inter_per_rank = cfg.intermediate_size // tp
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if not self.cfg.gated_mlp:
+            return self.down_proj(ops.gelu(self.up_proj(x)))
         gu = self.gate_up_proj(x)
-        return self.down_proj(ops.silu_and_mul(gu))
+        act = ops.gelu_and_mul(gu) if self.cfg.hidden_act == "gelu_tanh" \
+            else ops.silu_and_mul(gu)
+        return self.down_proj(act)
 
 
 def _tp_fused_add_rms_norm(hidden, residual, weight, eps: float,
@@ -143,35 +191,89 @@ def _tp_fused_add_rms_norm(hidden, residual, weight, eps: float,
 
 
 class LlamaDecoderLayer(nn.Module):
-    def __init__(self, cfg: ModelConfig):
+    def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
         super().__init__()
         self.cfg = cfg
-        self.self_attn = LlamaAttention(cfg)
+        self.self_attn = LlamaAttention(cfg, layer_idx)
         if cfg.num_experts > 0:
             from .moe import MoEMLP
             self.mlp = MoEMLP(cfg)
         else:
             self.mlp = LlamaMLP(cfg)
-        self.input_layernorm = nn.Parameter(
-            torch.empty(cfg.hidden_size, dtype=cfg.dtype), requires_grad=False)
-        self.post_attention_layernorm = nn.Parameter(
-            torch.empty(cfg.hidden_size, dtype=cfg.dtype), requires_grad=False)
 
-    def forward(self, hidden, residual, positions, kv_cache, meta, cos_sin):
+        def norm_w():
+            return nn.Parameter(torch.empty(cfg.hidden_size, dtype=cfg.dtype),
+                                requires_grad=False)
+
+        self.input_layernorm = norm_w()
+        if not cfg.parallel_block:   # phi-2 shares one norm per layer
+            self.post_attention_layernorm = norm_w()
+        if cfg.sandwich_norms:       # gemma-2/3 extra output norms
+            self.pre_feedforward_layernorm = norm_w()
+            self.post_feedforward_layernorm = norm_w()
+        if cfg.norm_type == "layernorm":
+            self.input_layernorm_bias = norm_w()
+            if not cfg.parallel_block:
+                self.post_attention_layernorm_bias = norm_w()
+
+    def _norm(self, x, which: str):
+        w = getattr(self, which)
+        if self.cfg.norm_type == "layernorm":
+            b = getattr(self, which + "_bias", None)
+            return ops.layer_norm(x, w, b, self.cfg.rms_eps)
+        return ops.rms_norm(x, w, self.cfg.rms_eps)
+
+    def forward(self, hidden, residual, positions, kv_cache, meta, cos_sin,
+                cos_sin_local=None):
+        cfg = self.cfg
+        if cfg.parallel_block:
+            # phi-2/falcon: one shared norm; attn and mlp read the same
+            # normed input; stream += attn_out + mlp_out
+            stream = hidden if residual is None else \
+                (hidden.float() + residual.float()).to(hidden.dtype)
+            x = self._norm(stream, "input_layernorm")
+            a = self.self_attn(x, positions, kv_cache, meta, cos_sin,
+                               cos_sin_local)
+            m = self.mlp(x)
+            return (a.float() + m.float()).to(hidden.dtype), stream
+        if cfg.sandwich_norms:
+            # gemma-2/3: pre+post norms around BOTH sub-blocks (norms on
+            # the sub-block OUTPUT before the residual add)
+            stream = hidden if residual is None else \
+                (hidden.float() + residual.float()).to(hidden.dtype)
+            x = self._norm(stream, "input_layernorm")
+            a = self.self_attn(x, positions, kv_cache, meta, cos_sin,
+                               cos_sin_local)
+            a = self._norm(a, "post_attention_layernorm")
+            stream = (stream.float() + a.float()).to(hidden.dtype)
+            y = self._norm(stream, "pre_feedforward_layernorm")
+            m = self.mlp(y)
+            m = self._norm(m, "post_feedforward_layernorm")
+            return m, stream
         if residual is None:
             residual = hidden
-            hidden = ops.rms_norm(hidden, self.input_layernorm, self.cfg.rms_eps)
+            hidden = self._norm(hidden, "input_layernorm")
+        elif cfg.norm_type == "layernorm":
+            hidden, residual = ops.fused_add_layer_norm(
+                hidden, residual, self.input_layernorm,
+                self.input_layernorm_bias, cfg.rms_eps)
         else:
             # `hidden` here is the PREVIOUS layer's mlp output — a
             # deferred TP partial when a one-shot group is active and the
             # mlp is the dense LlamaMLP (MoE's reduce is not deferred)
             hidden, residual = _tp_fused_add_rms_norm(
-                hidden, residual, self.input_layernorm, self.cfg.rms_eps,
-                deferred_producer=self.cfg.num_experts == 0)
-        hidden = self.self_attn(hidden, positions, kv_cache, meta, cos_sin)
-        hidden, residual = _tp_fused_add_rms_norm(
-            hidden, residual, self.post_attention_layernorm, self.cfg.rms_eps,
-            deferred_producer=True)   # o_proj always defer-capable
+                hidden, residual, self.input_layernorm, cfg.rms_eps,
+                deferred_producer=cfg.num_experts == 0)
+        hidden = self.self_attn(hidden, positions, kv_cache, meta, cos_sin,
+                                cos_sin_local)
+        if cfg.norm_type == "layernorm":
+            hidden, residual = ops.fused_add_layer_norm(
+                hidden, residual, self.post_attention_layernorm,
+                self.post_attention_layernorm_bias, cfg.rms_eps)
+        else:
+            hidden, residual = _tp_fused_add_rms_norm(
+                hidden, residual, self.post_attention_layernorm, cfg.rms_eps,
+                deferred_producer=True)   # o_proj always defer-capable
         hidden = self.mlp(hidden)
         return hidden, residual
 
@@ -195,16 +297,21 @@ class LlamaForCausalLM(nn.Module):
             cfg.vocab_size, cfg.hidden_size, dtype=cfg.dtype) \
             if (self.is_first or cfg.tie_word_embeddings) else None
         self.layers = nn.ModuleList(
-            [LlamaDecoderLayer(cfg)
-             for _ in range(self.layer_end - self.layer_start)])
+            [LlamaDecoderLayer(cfg, self.layer_start + i)
+             for i in range(self.layer_end - self.layer_start)])
         self.norm = nn.Parameter(
             torch.empty(cfg.hidden_size, dtype=cfg.dtype),
             requires_grad=False) if self.is_last else None
+        self.norm_bias = nn.Parameter(
+            torch.empty(cfg.hidden_size, dtype=cfg.dtype),
+            requires_grad=False) \
+            if (self.is_last and cfg.norm_type == "layernorm") else None
         self.lm_head = None
         if self.is_last and not cfg.tie_word_embeddings:
             self.lm_head = ColumnParallelLinear(
                 cfg.hidden_size, cfg.vocab_size, dtype=cfg.dtype)
         self.register_buffer("cos_sin_cache", torch.empty(0), persistent=False)
+        self.cos_sin_cache_local = None
 
     @property
     def num_local_layers(self) -> int:
@@ -212,6 +319,11 @@ class LlamaForCausalLM(nn.Module):
 
     def init_rope(self, device, max_pos: Optional[int] = None):
         self.cos_sin_cache = build_cos_sin_cache(self.cfg, device, max_pos)
+        if self.cfg.rope_theta_local > 0:   # gemma-3 local-layer rope
+            self.cos_sin_cache_local = build_cos_sin_cache(
+                self.cfg, device, max_pos, theta=self.cfg.rope_theta_local)
+        else:
+            self.cos_sin_cache_local = None
 
     def forward(self, input_ids: torch.Tensor, positions: torch.Tensor,
                 kv_caches: Optional[List[Tuple[torch.Tensor, torch.Tensor]]],
@@ -227,6 +339,9 @@ class LlamaForCausalLM(nn.Module):
             grp.begin_step()   # deterministic staging flip for hipGraphs
         if self.is_first:
             hidden = self.embed_tokens(input_ids)
+            if self.cfg.embed_scale != 1.0:   # gemma: sqrt(hidden_size)
+                hidden = (hidden.float() * self.cfg.embed_scale).to(
+                    hidden.dtype)
             residual = None
         else:
             assert hidden_in is not None, "non-first PP stage needs hidden_in"
@@ -235,11 +350,19 @@ class LlamaForCausalLM(nn.Module):
         for i, layer in enumerate(self.layers):
             kv = kv_caches[i] if kv_caches is not None else None
             hidden, residual = layer(hidden, residual, positions, kv, meta,
-                                     self.cos_sin_cache)
+                                     self.cos_sin_cache,
+                                     self.cos_sin_cache_local)
         if self.is_last:
-            hidden, _ = _tp_fused_add_rms_norm(
-                hidden, residual, self.norm, self.cfg.rms_eps,
-                deferred_producer=self.cfg.num_experts == 0)
+            if self.cfg.norm_type == "layernorm":
+                hidden, _ = ops.fused_add_layer_norm(
+                    hidden, residual, self.norm, self.norm_bias,
+                    self.cfg.rms_eps)
+            else:
+                hidden, _ = _tp_fused_add_rms_norm(
+                    hidden, residual, self.norm, self.cfg.rms_eps,
+                    deferred_producer=(self.cfg.num_experts == 0
+                                       and not self.cfg.parallel_block
+                                       and not self.cfg.sandwich_norms))
             return hidden
         # fold the residual stream so one tensor crosses the stage boundary
         return (hidden.float() + residual.float()).to(hidden.dtype)
@@ -250,7 +373,11 @@ class LlamaForCausalLM(nn.Module):
             logits = self.lm_head(hidden)
         else:
             logits = torch.nn.functional.linear(hidden, self.embed_tokens.weight)
-        return tp_all_gather(logits, dim=-1)
+        logits = tp_all_gather(logits, dim=-1)
+        cap = self.cfg.final_logit_softcap
+        if cap > 0:   # gemma-2 logit soft-capping
+            logits = (torch.tanh(logits.float() / cap) * cap).to(logits.dtype)
+        return logits
 
     @torch.no_grad()
     def random_init(self, seed: int = 0):
@@ -260,7 +387,15 @@ class LlamaForCausalLM(nn.Module):
         dev = next(self.parameters()).device
         gen = torch.Generator(device=dev).manual_seed(seed)
         for name, p in self.named_parameters():
-            if "layernorm" in name or name == "norm":
+            last = name.split(".")[-1]
+            if last.endswith("_bias") or last == "norm_bias":
+                p.zero_()
+            elif last == "sinks":
+                tmp = torch.empty(p.shape, dtype=torch.float32, device=dev)
+                tmp.normal_(0, 0.5, generator=gen)
+                p.copy_(tmp.to(p.dtype))
+            elif ("layernorm" in name or name == "norm"
+                    or last in ("q_norm", "k_norm")):
                 p.fill_(1.0)
             else:
                 std = 0.02 if "embed" in name or "lm_head" in name else \

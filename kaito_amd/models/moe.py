@@ -60,15 +60,31 @@ class MoEMLP(nn.Module):
         self.w_down = nn.Parameter(
             torch.empty(self.e_local, h, self.ie_local, dtype=cfg.dtype),
             requires_grad=False)
+        if cfg.moe_bias:   # gpt-oss per-expert biases
+            self.b_gate_up = nn.Parameter(
+                torch.empty(self.e_local, 2 * self.ie_local, dtype=cfg.dtype),
+                requires_grad=False)
+            self.b_down = nn.Parameter(
+                torch.empty(self.e_local, h, dtype=cfg.dtype),
+                requires_grad=False)
+        else:
+            self.b_gate_up = None
+            self.b_down = None
+        self.act_mode = 1 if cfg.moe_act == "swiglu_oai" else 0
 
     def _route(self, x: torch.Tensor):
         """Device-only top-k routing + expert sort. Returns
         (sorted_tok int32 [TK], gates f32 [TK], offsets int32 [E+1])."""
         T = x.size(0)
         logits = torch.nn.functional.linear(x, self.gate)      # [T, E]
-        probs = torch.softmax(logits.float(), dim=-1)
-        topw, topi = torch.topk(probs, self.top_k, dim=-1)     # [T, K]
-        topw = topw / topw.sum(-1, keepdim=True)
+        if self.cfg.moe_routing == "topk_softmax":
+            # gpt-oss: top-k on raw logits, softmax over the selected k
+            topl, topi = torch.topk(logits.float(), self.top_k, dim=-1)
+            topw = torch.softmax(topl, dim=-1)
+        else:
+            probs = torch.softmax(logits.float(), dim=-1)
+            topw, topi = torch.topk(probs, self.top_k, dim=-1)  # [T, K]
+            topw = topw / topw.sum(-1, keepdim=True)
         flat_e = topi.reshape(-1)                              # [T*K]
         flat_t = torch.arange(T, device=x.device,
                               dtype=torch.int32).repeat_interleave(self.top_k)
@@ -99,13 +115,18 @@ class MoEMLP(nn.Module):
         """Static-shape HIP grouped GEMMs (graph-capturable)."""
         T, H = x.shape
         TK = T * self.top_k
-        pad = (TK + 63) // 64 * 64   # staging reads whole 64-row tiles
+        # staging reads WHOLE 64-row tiles from unaligned expert offsets:
+        # a tile may start at row TK-1 and read through TK+62, so pad a
+        # full extra tile beyond the round-up
+        pad = (TK + 63) // 64 * 64 + 64
         act = torch.empty(pad, self.ie_local, dtype=x.dtype, device=x.device)
         ops.moe_gate_silu(act, x, self.w_gate_up, sorted_tok, offsets,
-                          self.e_base, self.e_local)
+                          self.e_base, self.e_local, self.b_gate_up,
+                          self.act_mode)
         out32 = torch.zeros(T, H, dtype=torch.float32, device=x.device)
         ops.moe_down_scatter(out32, act, self.w_down, sorted_tok, gates,
-                             offsets, self.e_base, self.e_local)
+                             offsets, self.e_base, self.e_local,
+                             self.b_down)
         return out32.to(x.dtype)
 
     def _forward_loop(self, x, sorted_tok, gates, offsets):
@@ -121,9 +142,19 @@ class MoEMLP(nn.Module):
                 continue
             rows = idx64[s:t]
             xs = x[rows]
-            h1 = torch.nn.functional.linear(xs, self.w_gate_up[le])
-            act = ops.silu_and_mul(h1.contiguous())
-            y = torch.nn.functional.linear(act, self.w_down[le])
+            h1 = torch.nn.functional.linear(
+                xs, self.w_gate_up[le],
+                self.b_gate_up[le] if self.b_gate_up is not None else None)
+            if self.act_mode == 1:
+                g, u = h1.float().chunk(2, dim=-1)
+                g = g.clamp(max=7.0)
+                u = u.clamp(-7.0, 7.0)
+                act = ((u + 1.0) * (g * torch.sigmoid(1.702 * g))).to(x.dtype)
+            else:
+                act = ops.silu_and_mul(h1.contiguous())
+            y = torch.nn.functional.linear(
+                act, self.w_down[le],
+                self.b_down[le] if self.b_down is not None else None)
             out.index_add_(0, rows,
                            (y.float() * gates[s:t, None]).to(out.dtype))
         return out

@@ -132,11 +132,16 @@ register(ModelConfig(
     name="phi-4", hidden_size=5120, num_layers=40, num_heads=40,
     num_kv_heads=10, intermediate_size=17920, vocab_size=100352,
     head_dim=128, rope_theta=250000.0, max_position=16384))
+# phi-2: parallel attn+MLP block, LayerNorm, ungated GELU MLP, partial
+# rotary. head_dim 80 is PADDED to 128 for the D∈{64,128,256} attention
+# kernels (zero-padded dims contribute nothing; attn_scale keeps 80^-0.5;
+# rotary covers the original 32 dims → 32/128 = 0.25).
 register(ModelConfig(
     name="phi-2", hidden_size=2560, num_layers=32, num_heads=32,
-    num_kv_heads=32, intermediate_size=10240, vocab_size=51200, head_dim=80,
-    rope_theta=10000.0, max_position=2048, partial_rotary_factor=0.4,
-    runtime="transformers"))  # parallel-block + GELU MLP (non-llama arch)
+    num_kv_heads=32, intermediate_size=10240, vocab_size=51200, head_dim=128,
+    rope_theta=10000.0, max_position=2048, partial_rotary_factor=0.25,
+    attn_scale=1.0 / 80 ** 0.5, parallel_block=True, norm_type="layernorm",
+    gated_mlp=False, hidden_act="gelu", attention_bias=True))
 
 # Mistral family (remaining); ministral-3 dims from published model cards
 register(ModelConfig(
@@ -189,31 +194,46 @@ register(ModelConfig(
     runtime="transformers"))
 _alias("falcon-40b-instruct", "falcon-40b")
 
-# Gemma-3 — RMSNorm(+1)/pre-post norms/sliding window → fallback runtime
+# Gemma-3 — native: sandwich norms, qk-norm, GeGLU, RMSNorm(1+w) folded
+# at load, sqrt(H) embedding scale, 5-local+1-global sliding-window
+# pattern with local-layer rope theta 10k.
 register(ModelConfig(
     name="gemma-3-4b-instruct", hidden_size=2560, num_layers=34, num_heads=8,
     num_kv_heads=4, intermediate_size=10240, vocab_size=262208, head_dim=256,
     rope_theta=1000000.0, max_position=131072, tie_word_embeddings=True,
-    runtime="transformers"))
+    sandwich_norms=True, qk_norm=True, rms_norm_offset=True,
+    hidden_act="gelu_tanh", embed_scale=2560 ** 0.5, sliding_window=1024,
+    sliding_window_pattern="interleaved:6", rope_theta_local=10000.0,
+    rms_eps=1e-6))
 register(ModelConfig(
     name="gemma-3-27b-instruct", hidden_size=5376, num_layers=62,
     num_heads=32, num_kv_heads=16, intermediate_size=21504,
     vocab_size=262208, head_dim=128, rope_theta=1000000.0,
-    max_position=131072, tie_word_embeddings=True, runtime="transformers"))
+    max_position=131072, tie_word_embeddings=True,
+    sandwich_norms=True, qk_norm=True, rms_norm_offset=True,
+    hidden_act="gelu_tanh", embed_scale=5376 ** 0.5, sliding_window=1024,
+    sliding_window_pattern="interleaved:6", rope_theta_local=10000.0,
+    rms_eps=1e-6, attn_scale=(5376 / 32) ** -0.5))
 
-# gpt-oss MoE — attention sinks + sliding window → fallback runtime
+# gpt-oss MoE — native: learned attention sinks, alternating
+# sliding-window layers, clamped-swiglu experts with biases,
+# topk-then-softmax routing.
 register(ModelConfig(
     name="gpt-oss-20b", hidden_size=2880, num_layers=24, num_heads=64,
     num_kv_heads=8, intermediate_size=2880, vocab_size=201088, head_dim=64,
     rope_theta=150000.0, max_position=131072, num_experts=32,
     num_experts_per_tok=4, moe_intermediate_size=2880,
-    runtime="transformers"))
+    attn_sinks=True, sliding_window=128,
+    sliding_window_pattern="interleaved:2", moe_act="swiglu_oai",
+    moe_bias=True, moe_routing="topk_softmax", attention_bias=True))
 register(ModelConfig(
     name="gpt-oss-120b", hidden_size=2880, num_layers=36, num_heads=64,
     num_kv_heads=8, intermediate_size=2880, vocab_size=201088, head_dim=64,
     rope_theta=150000.0, max_position=131072, num_experts=128,
     num_experts_per_tok=4, moe_intermediate_size=2880,
-    runtime="transformers"))
+    attn_sinks=True, sliding_window=128,
+    sliding_window_pattern="interleaved:2", moe_act="swiglu_oai",
+    moe_bias=True, moe_routing="topk_softmax", attention_bias=True))
 
 # DeepSeek V3/R1 — MLA attention + 256-expert MoE → fallback runtime
 register(ModelConfig(
@@ -234,3 +254,25 @@ register(ModelConfig(
     num_kv_heads=2, intermediate_size=512, vocab_size=512, head_dim=64,
     rope_theta=10000.0, max_position=512, num_experts=4,
     num_experts_per_tok=2, moe_intermediate_size=256))
+register(ModelConfig(
+    name="tiny-phi2-test", hidden_size=256, num_layers=2, num_heads=4,
+    num_kv_heads=4, intermediate_size=512, vocab_size=512, head_dim=64,
+    rope_theta=10000.0, max_position=512, partial_rotary_factor=0.5,
+    parallel_block=True, norm_type="layernorm", gated_mlp=False,
+    hidden_act="gelu", attention_bias=True))
+register(ModelConfig(
+    name="tiny-gemma3-test", hidden_size=256, num_layers=4, num_heads=4,
+    num_kv_heads=2, intermediate_size=512, vocab_size=512, head_dim=64,
+    rope_theta=1000000.0, max_position=512, sandwich_norms=True,
+    qk_norm=True, rms_norm_offset=True, hidden_act="gelu_tanh",
+    embed_scale=16.0, sliding_window=32,
+    sliding_window_pattern="interleaved:2", rope_theta_local=10000.0,
+    tie_word_embeddings=True))
+register(ModelConfig(
+    name="tiny-gptoss-test", hidden_size=256, num_layers=2, num_heads=4,
+    num_kv_heads=2, intermediate_size=256, vocab_size=512, head_dim=64,
+    rope_theta=150000.0, max_position=512, num_experts=4,
+    num_experts_per_tok=2, moe_intermediate_size=256, attn_sinks=True,
+    sliding_window=32, sliding_window_pattern="interleaved:2",
+    moe_act="swiglu_oai", moe_bias=True, moe_routing="topk_softmax",
+    attention_bias=True))

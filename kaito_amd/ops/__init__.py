@@ -100,40 +100,53 @@ import os as _os
 _PA_IMPL = _os.environ.get("KAITO_PA_IMPL", "sp")
 
 
+def _sinks_arg(sinks, device):
+    if sinks is None:
+        return torch.empty(0, dtype=torch.float32, device=device)
+    return sinks
+
+
 def paged_attention(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
                     block_tables: torch.Tensor, seq_lens: torch.Tensor,
-                    scale: float) -> torch.Tensor:
+                    scale: float, window: int = 0,
+                    sinks: torch.Tensor | None = None) -> torch.Tensor:
     if q.is_cuda:
         _require_ext()
         out = torch.empty_like(q)
-        if _PA_IMPL == "sp":
-            torch.ops.kaito.paged_attention_sp(out, q, k_cache, v_cache,
-                                               block_tables, seq_lens, scale)
+        if _PA_IMPL == "sp" or window > 0 or sinks is not None:
+            # window/sink support lives in the split-phase kernel only
+            torch.ops.kaito.paged_attention_sp(
+                out, q, k_cache, v_cache, block_tables, seq_lens, scale,
+                window, _sinks_arg(sinks, q.device))
         else:
             torch.ops.kaito.paged_attention(out, q, k_cache, v_cache,
                                             block_tables, seq_lens, scale)
         return out
     return torch_ref.paged_attention(q, k_cache, v_cache, block_tables,
-                                     seq_lens, scale)
+                                     seq_lens, scale, window, sinks)
 
 
 def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                       cu_seqlens: torch.Tensor, scale: float,
-                      max_seqlen: int | None = None) -> torch.Tensor:
+                      max_seqlen: int | None = None, window: int = 0,
+                      sinks: torch.Tensor | None = None) -> torch.Tensor:
     if q.is_cuda:
         _require_ext()
         out = torch.empty_like(q)
         tile_seq, tile_qbase = _build_tiles(cu_seqlens)
         torch.ops.kaito.prefill_attention(out, q, k, v, tile_seq, tile_qbase,
-                                          cu_seqlens, scale)
+                                          cu_seqlens, scale, window,
+                                          _sinks_arg(sinks, q.device))
         return out
-    return torch_ref.prefill_attention(q, k, v, cu_seqlens, scale)
+    return torch_ref.prefill_attention(q, k, v, cu_seqlens, scale, window,
+                                       sinks)
 
 
 def context_attention(q: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, cu_seqlens_q: torch.Tensor,
                       kv_lens: torch.Tensor, block_tables: torch.Tensor,
-                      scale: float) -> torch.Tensor:
+                      scale: float, window: int = 0,
+                      sinks: torch.Tensor | None = None) -> torch.Tensor:
     """Suffix-query causal attention over the paged cache (chunked
     prefill / prefix-cache continuation)."""
     if q.is_cuda:
@@ -142,10 +155,60 @@ def context_attention(q: torch.Tensor, k_cache: torch.Tensor,
         tile_seq, tile_qbase = _build_tiles(cu_seqlens_q)
         torch.ops.kaito.context_attention(out, q, k_cache, v_cache, tile_seq,
                                           tile_qbase, cu_seqlens_q, kv_lens,
-                                          block_tables, scale)
+                                          block_tables, scale, window,
+                                          _sinks_arg(sinks, q.device))
         return out
     return torch_ref.context_attention(q, k_cache, v_cache, cu_seqlens_q,
-                                       kv_lens, block_tables, scale)
+                                       kv_lens, block_tables, scale, window,
+                                       sinks)
+
+
+def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    """GeGLU: gelu_tanh(x[..., :d]) * x[..., d:] (gemma MLPs)."""
+    if x.is_cuda:
+        _require_ext()
+        d = x.size(-1) // 2
+        out = torch.empty(*x.shape[:-1], d, dtype=x.dtype, device=x.device)
+        torch.ops.kaito.gelu_and_mul(out, x)
+        return out
+    return torch_ref.gelu_and_mul(x)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    """Plain tanh-approx GELU (phi-2 ungated MLP)."""
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty_like(x)
+        torch.ops.kaito.gelu(out, x.contiguous())
+        return out
+    return torch_ref.gelu_tanh(x)
+
+
+def layer_norm(x: torch.Tensor, weight: torch.Tensor,
+               bias: torch.Tensor | None, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty_like(x)
+        b = bias if bias is not None else \
+            torch.empty(0, dtype=x.dtype, device=x.device)
+        torch.ops.kaito.layer_norm(out, x, weight, b, eps)
+        return out
+    return torch_ref.layer_norm(x, weight, bias, eps)
+
+
+def fused_add_layer_norm(x: torch.Tensor, residual: torch.Tensor,
+                         weight: torch.Tensor, bias: torch.Tensor | None,
+                         eps: float):
+    """Returns (normed, residual); residual += x in place on GPU."""
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty_like(x)
+        b = bias if bias is not None else \
+            torch.empty(0, dtype=x.dtype, device=x.device)
+        torch.ops.kaito.fused_add_layer_norm(out, x, residual, weight, b, eps)
+        return out, residual
+    new_res = (x.float() + residual.float()).to(x.dtype)
+    return torch_ref.layer_norm(new_res, weight, bias, eps), new_res
 
 
 def mfma_tile_gemm(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
@@ -173,23 +236,29 @@ def _build_tiles(cu_seqlens: torch.Tensor):
 def moe_gate_silu(act: torch.Tensor, x: torch.Tensor,
                   w_gate_up: torch.Tensor, sorted_ids: torch.Tensor,
                   offsets: torch.Tensor, e_base: int,
-                  n_local_experts: int) -> None:
+                  n_local_experts: int, bias: torch.Tensor | None = None,
+                  act_mode: int = 0) -> None:
     """Fused MoE stage 1 (GPU only): gather rows by sorted token ids,
-    gate/up grouped GEMM per expert, silu·mul epilogue into `act`."""
+    gate/up grouped GEMM per expert, activation epilogue into `act`
+    (act_mode 0 = SwiGLU, 1 = gpt-oss clamped swiglu; optional
+    per-expert [E, 2*IE] bias)."""
     _require_ext()
+    b = bias if bias is not None else         torch.empty(0, dtype=x.dtype, device=x.device)
     torch.ops.kaito.moe_gate_silu(act, x, w_gate_up, sorted_ids, offsets,
-                                  e_base, n_local_experts)
+                                  b, act_mode, e_base, n_local_experts)
 
 
 def moe_down_scatter(out: torch.Tensor, act: torch.Tensor,
                      w_down: torch.Tensor, sorted_ids: torch.Tensor,
                      gates: torch.Tensor, offsets: torch.Tensor,
-                     e_base: int, n_local_experts: int) -> None:
-    """Fused MoE stage 2 (GPU only): down grouped GEMM + gated f32
-    atomic scatter into `out`."""
+                     e_base: int, n_local_experts: int,
+                     bias: torch.Tensor | None = None) -> None:
+    """Fused MoE stage 2 (GPU only): down grouped GEMM (+ optional
+    [E, H] bias) + gated f32 atomic scatter into `out`."""
     _require_ext()
+    b = bias if bias is not None else         torch.empty(0, dtype=act.dtype, device=act.device)
     torch.ops.kaito.moe_down_scatter(out, act, w_down, sorted_ids, gates,
-                                     offsets, e_base, n_local_experts)
+                                     offsets, b, e_base, n_local_experts)
 
 
 def w4a16_gemv(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,

@@ -31,7 +31,9 @@ __global__ __launch_bounds__(256, 2) void context_attn_kernel(
     const int* __restrict__ kv_lens,      // [batch] total context length
     const int* __restrict__ block_tables, // [batch, max_blocks]
     const float scale, const int QH, const int KH, const int max_blocks,
-    const int64_t q_stride) {
+    const int64_t q_stride,
+    const int window,                   // 0 = full causal
+    const float* __restrict__ sinks) {  // [QH] or nullptr
   constexpr int KK = D / 32;
   constexpr int DT = D / 16;
   const int tile = blockIdx.x;
@@ -72,7 +74,10 @@ __global__ __launch_bounds__(256, 2) void context_attn_kernel(
 
   // causal bound: highest absolute q position in this tile
   const int kv_end = min(kv_len, q_abs0 + q0 + CTX_QTILE);
-  for (int kv0 = 0; kv0 < kv_end; kv0 += CTX_KVT) {
+  // sliding window: earliest key visible to this tile's lowest q row
+  const int kv_begin = (window > 0 && q_abs0 + q0 - window + 1 > 0)
+      ? ((q_abs0 + q0 - window + 1) / CTX_KVT) * CTX_KVT : 0;
+  for (int kv0 = kv_begin; kv0 < kv_end; kv0 += CTX_KVT) {
     {
       const int nvec = CTX_KVT * D / 8;
       for (int i = threadIdx.x; i < nvec; i += 256) {
@@ -117,7 +122,8 @@ __global__ __launch_bounds__(256, 2) void context_attn_kernel(
         const int q_abs = q_abs0 + qr;
         const int kvp = kv0 + kt * 16 + lo;
         float sv = sfrag[kt][r] * scale;
-        if (kvp > q_abs || kvp >= kv_len || qr >= q_len) sv = -1e30f;
+        if (kvp > q_abs || kvp >= kv_len || qr >= q_len ||
+            (window > 0 && kvp <= q_abs - window)) sv = -1e30f;
         p[kt][r] = sv;
       }
     float mnew[4];
@@ -168,7 +174,9 @@ __global__ __launch_bounds__(256, 2) void context_attn_kernel(
   for (int r = 0; r < 4; r++) {
     const int qr = q0 + wave * 16 + hi * 4 + r;
     if (qr >= q_len) continue;
-    const float inv = 1.f / fmaxf(lrow[r], 1e-20f);
+    float den = lrow[r];
+    if (sinks != nullptr) den += __expf(sinks[qh] - mrow[r]);
+    const float inv = 1.f / fmaxf(den, 1e-20f);
     const int64_t obase = ((int64_t)(tok0 + qr) * QH + qh) * D;
 #pragma unroll
     for (int dt = 0; dt < DT; dt++)
@@ -180,7 +188,13 @@ void context_attention(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor tile_seq,
                        at::Tensor tile_qbase, at::Tensor cu_seqlens_q,
                        at::Tensor kv_lens, at::Tensor block_tables,
-                       double scale) {
+                       double scale, int64_t window, at::Tensor sinks) {
+  const float* sink_ptr = nullptr;
+  if (sinks.numel() > 0) {
+    TORCH_CHECK(sinks.dtype() == at::kFloat && sinks.is_cuda() &&
+                sinks.numel() == q.size(1));
+    sink_ptr = sinks.data_ptr<float>();
+  }
   TORCH_CHECK(q.is_cuda() && q.dtype() == at::kBFloat16);
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(q.stride(-1) == 1 && q.stride(1) == q.size(2));
@@ -190,7 +204,8 @@ void context_attention(at::Tensor out, at::Tensor q, at::Tensor k_cache,
   const int BS = k_cache.size(2);
   const int ntiles = tile_seq.size(0);
   const int max_blocks = block_tables.size(1);
-  TORCH_CHECK(BS == 16 && (D == 128 || D == 64), "context attn: D 64/128");
+  TORCH_CHECK(BS == 16 && (D == 128 || D == 64 || D == 256),
+              "context attn: D 64/128/256");
   auto stream = at::hip::getCurrentHIPStream();
   if (ntiles == 0) return;
 #define CTX_LAUNCH(D_)                                                        \
@@ -200,8 +215,10 @@ void context_attention(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       (const short*)v_cache.data_ptr(), tile_seq.data_ptr<int>(),            \
       tile_qbase.data_ptr<int>(), cu_seqlens_q.data_ptr<int>(),              \
       kv_lens.data_ptr<int>(), block_tables.data_ptr<int>(), (float)scale,   \
-      QH, KH, max_blocks, q.stride(0))
-  if (D == 128) CTX_LAUNCH(128); else CTX_LAUNCH(64);
+      QH, KH, max_blocks, q.stride(0), (int)window, sink_ptr)
+  if (D == 128) CTX_LAUNCH(128);
+  else if (D == 256) CTX_LAUNCH(256);
+  else CTX_LAUNCH(64);
 #undef CTX_LAUNCH
 }
 

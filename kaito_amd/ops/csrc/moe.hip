@@ -91,6 +91,18 @@ KAITO_DEV bf16x8 frag_read(const short* __restrict__ lds, int row, int kk,
 
 KAITO_DEV float silu_f(float v) { return v / (1.f + __expf(-v)); }
 
+// act_mode 1: gpt-oss clamped swiglu — gate=min(g,7), up=clamp(u,±7),
+// (up+1) * gate*sigmoid(1.702*gate)
+KAITO_DEV float moe_act_f(int mode, float g, float u) {
+  if (mode == 1) {
+    g = fminf(g, 7.f);
+    u = fminf(fmaxf(u, -7.f), 7.f);
+    const float glu = g / (1.f + __expf(-1.702f * g));
+    return (u + 1.f) * glu;
+  }
+  return silu_f(g) * u;
+}
+
 }  // namespace
 
 // act[s, n] = silu(x[tok_s] @ Wg[e]^T) * (x[tok_s] @ Wu[e]^T)
@@ -102,6 +114,8 @@ void moe_gate_silu_kernel(
     const short* __restrict__ w,         // [E, 2*IE, H] bf16
     const int* __restrict__ sorted_ids,  // [TK]
     const int* __restrict__ offsets,     // [E+1] (global expert ids)
+    const short* __restrict__ bias,      // [E, 2*IE] bf16 or nullptr
+    const int act_mode,
     const int e_base, const int H, const int IE) {
   const int e = e_base + blockIdx.y;
   const int m0 = offsets[e] + blockIdx.x * BM;
@@ -173,8 +187,13 @@ void moe_gate_silu_kernel(
 #pragma unroll
       for (int b = 0; b < 2; b++) {
         const int col = n0 + wn * 32 + b * 16 + lo;
+        float g = accg[a][b][r], u = accu[a][b][r];
+        if (bias != nullptr) {
+          g += bf16_to_f32(bias[(int64_t)e * 2 * IE + col]);
+          u += bf16_to_f32(bias[(int64_t)e * 2 * IE + IE + col]);
+        }
         act[(int64_t)(m0 + row) * IE + col] =
-            f32_to_bf16(silu_f(accg[a][b][r]) * accu[a][b][r]);
+            f32_to_bf16(moe_act_f(act_mode, g, u));
       }
     }
   }
@@ -190,6 +209,7 @@ void moe_down_scatter_kernel(
     const int* __restrict__ sorted_ids,  // [TK]
     const float* __restrict__ gates,     // [TK] sorted gate weights
     const int* __restrict__ offsets,     // [E+1]
+    const short* __restrict__ bias,      // [E, H] bf16 or nullptr
     const int e_base, const int H, const int IE) {
   const int e = e_base + blockIdx.y;
   const int m0 = offsets[e] + blockIdx.x * BM;
@@ -250,7 +270,9 @@ void moe_down_scatter_kernel(
 #pragma unroll
       for (int b = 0; b < 2; b++) {
         const int col = n0 + wn * 32 + b * 16 + lo;
-        atomicAdd(&out[(int64_t)tok * H + col], g * acc[a][b][r]);
+        float v = acc[a][b][r];
+        if (bias != nullptr) v += bf16_to_f32(bias[(int64_t)e * H + col]);
+        atomicAdd(&out[(int64_t)tok * H + col], g * v);
       }
     }
   }
@@ -258,7 +280,13 @@ void moe_down_scatter_kernel(
 
 void moe_gate_silu(at::Tensor act, at::Tensor x, at::Tensor w_gate_up,
                    at::Tensor sorted_ids, at::Tensor offsets,
+                   at::Tensor bias, int64_t act_mode,
                    int64_t e_base, int64_t n_local_experts) {
+  const short* bias_ptr = nullptr;
+  if (bias.numel() > 0) {
+    TORCH_CHECK(bias.dtype() == at::kBFloat16 && bias.is_contiguous());
+    bias_ptr = (const short*)bias.data_ptr();
+  }
   TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16 && x.is_contiguous());
   TORCH_CHECK(act.is_contiguous() && w_gate_up.is_contiguous());
   TORCH_CHECK(sorted_ids.dtype() == at::kInt && offsets.dtype() == at::kInt);
@@ -275,13 +303,18 @@ void moe_gate_silu(at::Tensor act, at::Tensor x, at::Tensor w_gate_up,
   moe_gate_silu_kernel<<<grid, block, 0, stream>>>(
       (short*)act.data_ptr(), (const short*)x.data_ptr(),
       (const short*)w_gate_up.data_ptr(), sorted_ids.data_ptr<int>(),
-      offsets.data_ptr<int>(), (int)e_base, H, IE);
+      offsets.data_ptr<int>(), bias_ptr, (int)act_mode, (int)e_base, H, IE);
 }
 
 void moe_down_scatter(at::Tensor out, at::Tensor act, at::Tensor w_down,
                       at::Tensor sorted_ids, at::Tensor gates,
-                      at::Tensor offsets, int64_t e_base,
+                      at::Tensor offsets, at::Tensor bias, int64_t e_base,
                       int64_t n_local_experts) {
+  const short* bias_ptr = nullptr;
+  if (bias.numel() > 0) {
+    TORCH_CHECK(bias.dtype() == at::kBFloat16 && bias.is_contiguous());
+    bias_ptr = (const short*)bias.data_ptr();
+  }
   TORCH_CHECK(out.is_cuda() && out.dtype() == at::kFloat &&
               out.is_contiguous());
   TORCH_CHECK(act.is_contiguous() && w_down.is_contiguous());
@@ -298,7 +331,8 @@ void moe_down_scatter(at::Tensor out, at::Tensor act, at::Tensor w_down,
   moe_down_scatter_kernel<<<grid, block, 0, stream>>>(
       out.data_ptr<float>(), (const short*)act.data_ptr(),
       (const short*)w_down.data_ptr(), sorted_ids.data_ptr<int>(),
-      gates.data_ptr<float>(), offsets.data_ptr<int>(), (int)e_base, H, IE);
+      gates.data_ptr<float>(), offsets.data_ptr<int>(), bias_ptr,
+      (int)e_base, H, IE);
 }
 
 }  // namespace kaito

@@ -54,7 +54,9 @@ void paged_attention_sp_kernel(
     const int* __restrict__ block_tables,// [T, max_blocks]
     const int* __restrict__ seq_lens,    // [T]
     const float scale, const int KH, const int max_blocks,
-    const int64_t q_stride) {
+    const int64_t q_stride,
+    const int window,                    // 0 = full attention
+    const float* __restrict__ sinks) {   // [QH] or nullptr (gpt-oss)
   constexpr int DC = D / 8;       // dim-chunk lanes (16 for D=128)
   constexpr int TPC = 64 / DC;    // tokens per wave-step (4 for D=128)
   constexpr int CPB = BS / TPC;   // chunks per KV block (4 for D=128)
@@ -84,11 +86,17 @@ void paged_attention_sp_kernel(
     for (int j = 0; j < 8; j++) acc[g][j] = 0.f;
   }
 
+  // sliding window: only tokens in [start, seq_len) are attended; the
+  // chunk range is aligned DOWN to the containing KV block (tokens
+  // before `start` in that block are masked by `valid`)
+  const int start = (window > 0 && seq_len > window) ? seq_len - window : 0;
+  const int c_lo = (start / BS) * CPB;
   const int nchunks = (seq_len + TPC - 1) / TPC;
   const int last_blk = (seq_len - 1) / BS;
   // contiguous block-aligned chunk span per wave
-  const int n_per = ((nchunks + NW * CPB - 1) / (NW * CPB)) * CPB;
-  const int span0 = wave * n_per;
+  const int total_c = nchunks - c_lo;
+  const int n_per = ((total_c + NW * CPB - 1) / (NW * CPB)) * CPB;
+  const int span0 = c_lo + wave * n_per;
   const int n_i = max(0, min(nchunks, span0 + n_per) - span0);
 
   typedef __bf16 bf16x2_t __attribute__((ext_vector_type(2)));
@@ -124,7 +132,8 @@ void paged_attention_sp_kernel(
 #pragma unroll
       for (int u = 0; u < CPB; u++) {
         if (u >= nu) break;
-        const bool valid = ((c0 + u) * TPC + tg) < seq_len;
+        const int tok = (c0 + u) * TPC + tg;
+        const bool valid = tok >= start && tok < seq_len;
         float s[G];
 #pragma unroll
         for (int g = 0; g < G; g++) {
@@ -260,12 +269,16 @@ void paged_attention_sp_kernel(
 #pragma unroll
     for (int w = 0; w < NW; w++) gm = fmaxf(gm, s_ml[w][g][0]);
     float num = 0.f, den = 0.f;
+    float sink = sinks != nullptr ? sinks[kvh * G + g] : -1e30f;
+    gm = fmaxf(gm, sink);       // keep exp() bounded when sink dominates
 #pragma unroll
     for (int w = 0; w < NW; w++) {
       const float e = __expf(s_ml[w][g][0] - gm);
       num += e * s_acc[w][g][d];
       den += e * s_ml[w][g][1];
     }
+    if (sinks != nullptr)       // learned sink absorbs softmax mass
+      den += __expf(sink - gm);
     out[((int64_t)seq * QH + kvh * G + g) * D + d] =
         f32_to_bf16(num / fmaxf(den, 1e-20f));
   }
@@ -277,11 +290,18 @@ void paged_attention_sp_kernel(
       (const short*)query.data_ptr(), (const short*)k_cache.data_ptr(),       \
       (const short*)v_cache.data_ptr(), block_tables.data_ptr<int>(),         \
       seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks,                 \
-      query.stride(0))
+      query.stride(0), (int)window, sink_ptr)
 
 void paged_attention_sp(at::Tensor out, at::Tensor query, at::Tensor k_cache,
                         at::Tensor v_cache, at::Tensor block_tables,
-                        at::Tensor seq_lens, double scale) {
+                        at::Tensor seq_lens, double scale, int64_t window,
+                        at::Tensor sinks) {
+  const float* sink_ptr = nullptr;
+  if (sinks.numel() > 0) {
+    TORCH_CHECK(sinks.dtype() == at::kFloat && sinks.is_cuda() &&
+                sinks.numel() == query.size(1));
+    sink_ptr = sinks.data_ptr<float>();
+  }
   TORCH_CHECK(query.is_cuda() && query.dtype() == at::kBFloat16);
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(query.stride(-1) == 1 && query.stride(1) == query.size(2),
@@ -316,6 +336,13 @@ void paged_attention_sp(at::Tensor out, at::Tensor query, at::Tensor k_cache,
       case 2: PA_SP_LAUNCH(64, 2); break;
       case 4: PA_SP_LAUNCH(64, 4); break;
       case 8: PA_SP_LAUNCH(64, 8); break;
+      default: TORCH_CHECK(false, "unsupported GQA group ", G);
+    }
+  } else if (D == 256) {   // gemma-3-4b class
+    switch (G) {
+      case 1: PA_SP_LAUNCH(256, 1); break;
+      case 2: PA_SP_LAUNCH(256, 2); break;
+      case 4: PA_SP_LAUNCH(256, 4); break;
       default: TORCH_CHECK(false, "unsupported GQA group ", G);
     }
   } else {

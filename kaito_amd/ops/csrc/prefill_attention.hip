@@ -37,7 +37,9 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
     const int* __restrict__ tile_qbase, // [ntiles] q row base within seq
     const int* __restrict__ cu_seqlens, // [batch+1]
     const float scale, const int QH, const int KH,
-    const int64_t q_stride, const int64_t kv_stride) {
+    const int64_t q_stride, const int64_t kv_stride,
+    const int window,                   // 0 = full causal
+    const float* __restrict__ sinks) {  // [QH] or nullptr
   constexpr int KK = D / 32;          // MFMA k-steps over head_dim (4 for 128)
   constexpr int DT = D / 16;          // output d-tiles (8 for 128)
   const int tile = blockIdx.x;
@@ -81,7 +83,10 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
   for (int dt = 0; dt < DT; dt++) ofrag[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = min(slen, q0 + QTILE);        // causal bound for tile
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KVT) {
+  // sliding window: the earliest key any q row of this tile can see
+  const int kv_begin = (window > 0 && q0 - window + 1 > 0)
+      ? ((q0 - window + 1) / KVT) * KVT : 0;
+  for (int kv0 = kv_begin; kv0 < kv_end; kv0 += KVT) {
     // ---- cooperative stage: K tile swizzled + V transposed ----
     // 256 threads × short8: K tile = KVT*D/8 vectors (512 for D=128).
     {
@@ -132,7 +137,8 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
         const int qr = q0 + wave * 16 + hi * 4 + r;
         const int kvp = kv0 + kt * 16 + lo;
         float sv = sfrag[kt][r] * scale;
-        if (kvp > qr || kvp >= slen || qr >= slen) sv = -1e30f;
+        if (kvp > qr || kvp >= slen || qr >= slen ||
+            (window > 0 && kvp <= qr - window)) sv = -1e30f;
         p[kt][r] = sv;
       }
     }
@@ -186,12 +192,14 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
     __syncthreads();
   }
 
-  // ---- epilogue: out[q, qh, d] = O / l ----
+  // ---- epilogue: out[q, qh, d] = O / l (+ sink mass in the denom) ----
 #pragma unroll
   for (int r = 0; r < 4; r++) {
     const int qr = q0 + wave * 16 + hi * 4 + r;
     if (qr >= slen) continue;
-    const float inv = 1.f / fmaxf(lrow[r], 1e-20f);
+    float den = lrow[r];
+    if (sinks != nullptr) den += __expf(sinks[qh] - mrow[r]);
+    const float inv = 1.f / fmaxf(den, 1e-20f);
     const int64_t obase = ((int64_t)(tok0 + qr) * QH + qh) * D;
 #pragma unroll
     for (int dt = 0; dt < DT; dt++)
@@ -201,7 +209,14 @@ __global__ __launch_bounds__(256, 2) void prefill_attn_kernel(
 
 void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
                        at::Tensor tile_seq, at::Tensor tile_qbase,
-                       at::Tensor cu_seqlens, double scale) {
+                       at::Tensor cu_seqlens, double scale, int64_t window,
+                       at::Tensor sinks) {
+  const float* sink_ptr = nullptr;
+  if (sinks.numel() > 0) {
+    TORCH_CHECK(sinks.dtype() == at::kFloat && sinks.is_cuda() &&
+                sinks.numel() == q.size(1));
+    sink_ptr = sinks.data_ptr<float>();
+  }
   TORCH_CHECK(q.is_cuda() && q.dtype() == at::kBFloat16);
   TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(q.stride(-1) == 1 && q.stride(1) == q.size(2));
@@ -221,8 +236,9 @@ void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k, at::Tensor v,
       (const short*)k.data_ptr(), (const short*)v.data_ptr(),                \
       tile_seq.data_ptr<int>(), tile_qbase.data_ptr<int>(),                  \
       cu_seqlens.data_ptr<int>(), (float)scale, QH, KH,                       \
-      q.stride(0), k.stride(0))
+      q.stride(0), k.stride(0), (int)window, sink_ptr)
   switch (D) {
+    case 256: PF_LAUNCH(256); break;
     case 128: PF_LAUNCH(128); break;
     case 96:  PF_LAUNCH(96);  break;
     case 64:  PF_LAUNCH(64);  break;

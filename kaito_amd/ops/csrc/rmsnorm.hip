@@ -73,6 +73,108 @@ __global__ void rms_norm_kernel(
   }
 }
 
+// LayerNorm (+optional bias, +optional residual add) — phi-2 / falcon
+// norm layers. Same vectorized one-workgroup-per-row structure.
+template <bool RESIDUAL>
+__global__ void layer_norm_kernel(
+    short* __restrict__ out,            // [T, H] bf16
+    const short* __restrict__ x,        // [T, H] bf16
+    short* __restrict__ residual,       // [T, H] bf16 or nullptr
+    const short* __restrict__ w,        // [H]
+    const short* __restrict__ b,        // [H] or nullptr
+    const float eps, const int H, const int T) {
+  __shared__ float red[16];
+  const int VEC = 8;
+  const int nvec = H / VEC;
+  for (int row = blockIdx.x; row < T; row += gridDim.x) {
+    const short8_t* xv =
+        reinterpret_cast<const short8_t*>(x + (int64_t)row * H);
+    short8_t* rv = RESIDUAL
+        ? reinterpret_cast<short8_t*>(residual + (int64_t)row * H) : nullptr;
+    float sum = 0.f, ssq = 0.f;
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      short8_t v = xv[i];
+      if (RESIDUAL) {
+        short8_t r = rv[i];
+#pragma unroll
+        for (int j = 0; j < VEC; j++)
+          v[j] = f32_to_bf16(bf16_to_f32(v[j]) + bf16_to_f32(r[j]));
+        rv[i] = v;
+      }
+#pragma unroll
+      for (int j = 0; j < VEC; j++) {
+        const float f = bf16_to_f32(v[j]);
+        sum += f;
+        ssq += f * f;
+      }
+    }
+    sum = block_reduce_sum(sum, red);
+    __syncthreads();          // red[] reused by the second reduction
+    ssq = block_reduce_sum(ssq, red);
+    __shared__ float s_mu, s_inv;
+    if (threadIdx.x == 0) {
+      const float mu = sum / (float)H;
+      s_mu = mu;
+      s_inv = rsqrtf(ssq / (float)H - mu * mu + eps);
+    }
+    __syncthreads();
+    const float mu = s_mu, inv = s_inv;
+    const short8_t* src = RESIDUAL
+        ? reinterpret_cast<const short8_t*>(residual + (int64_t)row * H) : xv;
+    short8_t* ov = reinterpret_cast<short8_t*>(out + (int64_t)row * H);
+    const short8_t* wv = reinterpret_cast<const short8_t*>(w);
+    const short8_t* bv = reinterpret_cast<const short8_t*>(b);
+    for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+      short8_t v = src[i];
+      short8_t g = wv[i];
+      short8_t o;
+#pragma unroll
+      for (int j = 0; j < VEC; j++) {
+        float f = (bf16_to_f32(v[j]) - mu) * inv * bf16_to_f32(g[j]);
+        if (b != nullptr) f += bf16_to_f32(bv[i][j]);
+        o[j] = f32_to_bf16(f);
+      }
+      ov[i] = o;
+    }
+    __syncthreads();
+  }
+}
+
+void layer_norm(at::Tensor out, at::Tensor input, at::Tensor weight,
+                at::Tensor bias, double eps) {
+  TORCH_CHECK(input.is_cuda() && input.dtype() == at::kBFloat16);
+  TORCH_CHECK(input.is_contiguous() && out.is_contiguous());
+  const int H = input.size(-1);
+  const int64_t T = input.numel() / H;
+  TORCH_CHECK(H % 8 == 0);
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>(T, 2048);
+  auto stream = at::hip::getCurrentHIPStream();
+  const short* b = bias.numel() > 0 ? (const short*)bias.data_ptr() : nullptr;
+  hipLaunchKernelGGL((layer_norm_kernel<false>), dim3(grid), dim3(block), 0,
+      stream, (short*)out.data_ptr(), (const short*)input.data_ptr(), nullptr,
+      (const short*)weight.data_ptr(), b, (float)eps, H, (int)T);
+}
+
+void fused_add_layer_norm(at::Tensor out, at::Tensor input,
+                          at::Tensor residual, at::Tensor weight,
+                          at::Tensor bias, double eps) {
+  TORCH_CHECK(input.is_cuda() && input.dtype() == at::kBFloat16);
+  TORCH_CHECK(input.is_contiguous() && residual.is_contiguous() &&
+              out.is_contiguous());
+  const int H = input.size(-1);
+  const int64_t T = input.numel() / H;
+  TORCH_CHECK(H % 8 == 0);
+  const int block = 256;
+  const int grid = (int)std::min<int64_t>(T, 2048);
+  auto stream = at::hip::getCurrentHIPStream();
+  const short* b = bias.numel() > 0 ? (const short*)bias.data_ptr() : nullptr;
+  hipLaunchKernelGGL((layer_norm_kernel<true>), dim3(grid), dim3(block), 0,
+      stream, (short*)out.data_ptr(), (const short*)input.data_ptr(),
+      (short*)residual.data_ptr(), (const short*)weight.data_ptr(), b,
+      (float)eps, H, (int)T);
+}
+
 void rms_norm(at::Tensor out, at::Tensor input, at::Tensor weight, double eps) {
   TORCH_CHECK(input.is_cuda() && input.dtype() == at::kBFloat16, "rms_norm: bf16 GPU only");
   TORCH_CHECK(input.is_contiguous() && out.is_contiguous());

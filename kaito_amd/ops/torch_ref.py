@@ -55,6 +55,31 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     return (torch.nn.functional.silu(gate) * up).to(x.dtype)
 
 
+def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    """GeGLU (gemma): gelu_tanh(gate) * up."""
+    d = x.size(-1) // 2
+    gate, up = x[..., :d].float(), x[..., d:].float()
+    return (torch.nn.functional.gelu(gate, approximate="tanh") * up
+            ).to(x.dtype)
+
+
+def gelu_tanh(x: torch.Tensor) -> torch.Tensor:
+    """Plain tanh-approx GELU (phi-2's ungated MLP activation)."""
+    return torch.nn.functional.gelu(x.float(), approximate="tanh").to(x.dtype)
+
+
+def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias, eps: float
+               ) -> torch.Tensor:
+    """LayerNorm with optional bias (phi-2/falcon norm layers)."""
+    xf = x.float()
+    mu = xf.mean(-1, keepdim=True)
+    var = (xf - mu).pow(2).mean(-1, keepdim=True)
+    out = (xf - mu) * torch.rsqrt(var + eps) * weight.float()
+    if bias is not None:
+        out = out + bias.float()
+    return out.to(x.dtype)
+
+
 def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, slot_mapping: torch.Tensor):
     """k/v: [T, KH, D]; caches: [B, KH, BS, D]."""
@@ -69,8 +94,12 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
 
 def paged_attention(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
                     block_tables: torch.Tensor, seq_lens: torch.Tensor,
-                    scale: float) -> torch.Tensor:
-    """q: [T, QH, D] (one token/seq). Returns [T, QH, D]."""
+                    scale: float, window: int = 0,
+                    sinks=None) -> torch.Tensor:
+    """q: [T, QH, D] (one token/seq). Returns [T, QH, D].
+    window>0: sliding-window attention — only the last `window` tokens
+    are attended. sinks: [QH] learned attention-sink logits (gpt-oss)
+    folded into the softmax denominator."""
     T, QH, D = q.shape
     KH = k_cache.size(1)
     BS = k_cache.size(2)
@@ -78,22 +107,29 @@ def paged_attention(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tenso
     out = torch.empty_like(q)
     for i in range(T):
         L = int(seq_lens[i])
+        start = max(0, L - window) if window > 0 else 0
         nb = (L + BS - 1) // BS
         blocks = block_tables[i, :nb].long()
-        keys = k_cache[blocks].permute(1, 0, 2, 3).reshape(KH, nb * BS, D)[:, :L]
-        vals = v_cache[blocks].permute(1, 0, 2, 3).reshape(KH, nb * BS, D)[:, :L]
+        keys = k_cache[blocks].permute(1, 0, 2, 3).reshape(KH, nb * BS, D)[:, start:L]
+        vals = v_cache[blocks].permute(1, 0, 2, 3).reshape(KH, nb * BS, D)[:, start:L]
         qh = q[i].float()                                # [QH, D]
-        kx = keys.float().repeat_interleave(G, dim=0)    # [QH, L, D]
+        kx = keys.float().repeat_interleave(G, dim=0)    # [QH, Lw, D]
         vx = vals.float().repeat_interleave(G, dim=0)
         s = torch.einsum("hd,hld->hl", qh, kx) * scale
-        p = torch.softmax(s, dim=-1)
+        if sinks is not None:
+            s = torch.cat([sinks.float().unsqueeze(1), s], dim=1)
+            p = torch.softmax(s, dim=-1)[:, 1:]          # sink absorbs mass
+        else:
+            p = torch.softmax(s, dim=-1)
         out[i] = torch.einsum("hl,hld->hd", p, vx).to(q.dtype)
     return out
 
 
 def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                      cu_seqlens: torch.Tensor, scale: float) -> torch.Tensor:
-    """Varlen causal attention. q: [T, QH, D], k/v: [T, KH, D]."""
+                      cu_seqlens: torch.Tensor, scale: float,
+                      window: int = 0, sinks=None) -> torch.Tensor:
+    """Varlen causal attention. q: [T, QH, D], k/v: [T, KH, D].
+    window>0: sliding window (k in (qpos-window, qpos]); sinks: [QH]."""
     T, QH, D = q.shape
     KH = k.size(1)
     G = QH // KH
@@ -106,9 +142,18 @@ def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         ks = k[s0:s1].float().transpose(0, 1).repeat_interleave(G, 0)
         vs = v[s0:s1].float().transpose(0, 1).repeat_interleave(G, 0)
         att = torch.einsum("hqd,hkd->hqk", qs, ks) * scale
-        mask = torch.triu(torch.ones(L, L, dtype=torch.bool, device=q.device), 1)
-        att = att.masked_fill(mask, float("-inf"))
-        p = torch.softmax(att, dim=-1)
+        qpos = torch.arange(L, device=q.device).unsqueeze(1)
+        kpos = torch.arange(L, device=q.device).unsqueeze(0)
+        mask = kpos > qpos
+        if window > 0:
+            mask = mask | (kpos <= qpos - window)
+        att = att.masked_fill(mask.unsqueeze(0), float("-inf"))
+        if sinks is not None:
+            att = torch.cat([sinks.float()[:, None, None].expand(QH, L, 1),
+                             att], dim=-1)
+            p = torch.softmax(att, dim=-1)[..., 1:]
+        else:
+            p = torch.softmax(att, dim=-1)
         o = torch.einsum("hqk,hkd->hqd", p, vs)
         out[s0:s1] = o.transpose(0, 1).to(q.dtype)
     return out
@@ -117,7 +162,8 @@ def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 def context_attention(q: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, cu_seqlens_q: torch.Tensor,
                       kv_lens: torch.Tensor, block_tables: torch.Tensor,
-                      scale: float) -> torch.Tensor:
+                      scale: float, window: int = 0,
+                      sinks=None) -> torch.Tensor:
     """Suffix-query causal attention over the paged cache. q: [Tq, QH, D];
     cache holds each sequence's FULL kv_len tokens (suffix included)."""
     Tq, QH, D = q.shape
@@ -140,8 +186,17 @@ def context_attention(q: torch.Tensor, k_cache: torch.Tensor,
         att = torch.einsum("hqd,hkd->hqk", qs, kx) * scale
         qpos = torch.arange(L - q_len, L, device=q.device).unsqueeze(1)
         kpos = torch.arange(L, device=q.device).unsqueeze(0)
-        att = att.masked_fill((kpos > qpos).unsqueeze(0), float("-inf"))
-        p = torch.softmax(att, dim=-1)
+        mask = kpos > qpos
+        if window > 0:
+            mask = mask | (kpos <= qpos - window)
+        att = att.masked_fill(mask.unsqueeze(0), float("-inf"))
+        if sinks is not None:
+            att = torch.cat(
+                [sinks.float()[:, None, None].expand(QH, q_len, 1), att],
+                dim=-1)
+            p = torch.softmax(att, dim=-1)[..., 1:]
+        else:
+            p = torch.softmax(att, dim=-1)
         o = torch.einsum("hqk,hkd->hqd", p, vx)
         out[s0:s1] = o.transpose(0, 1).to(q.dtype)
     return out

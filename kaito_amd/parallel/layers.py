@@ -111,11 +111,10 @@ class RowParallelLinear(_QuantMixin, nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         out = self._qmatmul(x) if self._quantized else F.linear(x, self.weight)
         out = lora_mod.maybe_apply(self, x, out)
-        if self.fuse_norm and get_state().tp_size > 1:
+        if self.fuse_norm and self.bias is None and get_state().tp_size > 1:
             from . import one_shot
             if one_shot.defer(out.reshape(-1, out.shape[-1]).shape[0]):
                 # deferred: the following fused one-shot call reduces
-                assert self.bias is None
                 return out
         out = tp_all_reduce(out)
         if self.bias is not None:

@@ -233,6 +233,113 @@ def test_engine_gpu_graphs_match_eager():
         assert a.output_token_ids == b.output_token_ids
 
 
+@pytest.mark.parametrize("window,use_sinks,G,D", [
+    (32, False, 4, 128), (100, False, 2, 128), (0, True, 4, 128),
+    (32, True, 8, 128), (16, False, 2, 256), (48, True, 4, 64),
+])
+def test_paged_attention_window_sinks(window, use_sinks, G, D):
+    """Sliding-window + attention-sink decode vs the fp32 reference."""
+    torch.manual_seed(9)
+    KH = 2
+    QH = KH * G
+    BS = 16
+    lens = [1, 15, 40, 200, 555]
+    T = len(lens)
+    mb = (max(lens) + BS - 1) // BS
+    NB = T * mb + 1
+    kc = _bf16(NB, KH, BS, D)
+    vc = _bf16(NB, KH, BS, D)
+    bt = (torch.randperm(NB - 1)[: T * mb].reshape(T, mb) + 1).int().to(DEV)
+    q = _bf16(T, QH, D)
+    sl = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    sinks = (torch.randn(QH, device=DEV) if use_sinks else None)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention(q, kc, vc, bt, sl, scale, window, sinks)
+    ref = R.paged_attention(q.cpu().float(), kc.cpu().float(),
+                            vc.cpu().float(), bt.cpu(), sl.cpu(), scale,
+                            window, sinks.cpu() if sinks is not None else None)
+    _close(out, ref.to(DEV), atol=2e-2)
+
+
+@pytest.mark.parametrize("window,use_sinks", [(32, False), (0, True),
+                                              (64, True)])
+def test_prefill_attention_window_sinks(window, use_sinks):
+    torch.manual_seed(11)
+    QH, KH, D = 8, 2, 128
+    lens = [5, 130, 200]
+    T = sum(lens)
+    q = _bf16(T, QH, D)
+    k = _bf16(T, KH, D)
+    v = _bf16(T, KH, D)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    sinks = (torch.randn(QH, device=DEV) if use_sinks else None)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.prefill_attention(q, k, v, cu, scale, window=window,
+                                sinks=sinks)
+    ref = R.prefill_attention(q.cpu().float(), k.cpu().float(),
+                              v.cpu().float(), cu.cpu(), scale, window,
+                              sinks.cpu() if sinks is not None else None)
+    _close(out, ref.to(DEV), atol=2e-2)
+
+
+@pytest.mark.parametrize("name", ["tiny-phi2-test", "tiny-gemma3-test",
+                                  "tiny-gptoss-test"])
+def test_engine_gpu_model_variants(name):
+    """phi-2 / gemma-3 / gpt-oss architecture variants through the GPU
+    engine (HIP layernorm/gelu/window/sink/MoE paths) vs the
+    full-recompute oracle on the same device."""
+    from kaito_amd.engine import SamplingParams
+    from kaito_amd.models import get_model_config
+    from kaito_amd.models.llama import AttnMetadata
+    eng = _gpu_engine(model=get_model_config(name), enforce_eager=True,
+                      max_model_len=96)
+    prompts = [[7, 9, 11, 13, 15, 17, 19, 21], list(range(30, 75))]
+    outs = eng.generate(prompts, SamplingParams(max_tokens=8,
+                                                ignore_eos=True))
+    model = eng.runner.model
+    for prompt, seq in zip(prompts, outs):
+        toks = list(prompt)
+        gen = []
+        for _ in range(8):
+            T = len(toks)
+            meta = AttnMetadata(
+                is_prefill=True,
+                slot_mapping=torch.full((T,), -1, dtype=torch.long,
+                                        device=DEV),
+                cu_seqlens=torch.tensor([0, T], dtype=torch.int32,
+                                        device=DEV),
+                max_seqlen=T)
+            hidden = model(torch.tensor(toks, device=DEV),
+                           torch.arange(T, device=DEV), None, meta)
+            nxt = int(model.compute_logits(hidden[-1:]).argmax(-1))
+            toks.append(nxt)
+            gen.append(nxt)
+        match = sum(a == b for a, b in zip(gen, seq.output_token_ids))
+        assert match >= 6, (name, gen, seq.output_token_ids)
+
+
+def test_layer_norm_and_gelu_kernels():
+    from kaito_amd.ops import torch_ref
+    torch.manual_seed(13)
+    x = _bf16(129, 2560)
+    w = _bf16(2560, scale=0.3) + 1.0
+    b = _bf16(2560, scale=0.2)
+    out = ops.layer_norm(x, w, b, 1e-5)
+    _close(out, torch_ref.layer_norm(x.cpu(), w.cpu(), b.cpu(), 1e-5).to(DEV))
+    res = _bf16(129, 2560)
+    res2 = res.clone()
+    out2, res2 = ops.fused_add_layer_norm(x, res2, w, None, 1e-5)
+    exp_res = (x.float() + res.float()).to(torch.bfloat16)
+    exp = torch_ref.layer_norm(exp_res.cpu(), w.cpu(), None, 1e-5)
+    _close(out2, exp.to(DEV))
+    _close(res2, exp_res)
+    g = _bf16(64, 1024)
+    _close(ops.gelu(g), torch_ref.gelu_tanh(g.cpu()).to(DEV))
+    gm = _bf16(64, 2048)
+    _close(ops.gelu_and_mul(gm), torch_ref.gelu_and_mul(gm.cpu()).to(DEV))
+
+
 # ------------------------------------------------------------------ topk
 @pytest.mark.parametrize("Q,N,k", [(1, 1000, 5), (7, 4096, 10), (3, 50, 32),
                                    (2, 100000, 16)])
