@@ -109,6 +109,9 @@ class LlamaAttention(nn.Module):
             k = ops.rms_norm(
                 k.reshape(-1, self.head_dim).contiguous(), self.k_norm,
                 self.cfg.rms_eps).reshape(k.shape[0], -1)
+            # k is now a fresh contiguous tensor; reshape_and_cache needs
+            # k/v with EQUAL row strides, so v must match
+            v = v.contiguous()
         cs = cos_sin_local if self.local_rope else cos_sin
         q, k = ops.rotary_embedding(positions, q, k, self.head_dim, cs)
         T = x.size(0)

@@ -35,6 +35,10 @@ class IVFFlatIndex:
         self._flat = FlatIndex(dim, use_gpu)   # master storage + GPU scoring
         self.centroids: Optional[np.ndarray] = None     # [nlist, dim]
         self._assign: Dict[str, int] = {}               # doc_id → list id
+        # GPU-resident search state (vectors + centroids + list-sorted
+        # row order), rebuilt lazily after mutations — searches do NO
+        # per-query host↔device vector transfers and no Python gather
+        self._gpu = None
 
     # ---- FlatIndex surface -------------------------------------------------
     def __len__(self):
@@ -46,6 +50,7 @@ class IVFFlatIndex:
 
     def add(self, doc_id: str, vec: np.ndarray) -> None:
         self._flat.add(doc_id, vec)
+        self._gpu = None
         if self.centroids is not None:
             self._assign[doc_id] = int(np.argmax(self.centroids @ vec))
         elif len(self._flat) >= self.min_train:
@@ -54,6 +59,7 @@ class IVFFlatIndex:
     def remove(self, doc_id: str) -> None:
         self._flat.remove(doc_id)
         self._assign.pop(doc_id, None)
+        self._gpu = None
 
     def state(self):
         st = self._flat.state()
@@ -95,6 +101,7 @@ class IVFFlatIndex:
         self._reassign_all()
 
     def _reassign_all(self):
+        self._gpu = None
         st = self._flat.state()
         if not st["ids"]:
             self._assign = {}
@@ -102,11 +109,34 @@ class IVFFlatIndex:
         a = np.argmax(st["vecs"] @ self.centroids.T, axis=1)
         self._assign = {d: int(j) for d, j in zip(st["ids"], a)}
 
+    # ---- GPU-resident search state ----------------------------------------
+    def _gpu_state(self):
+        """(vecs [N,D], centroids [nlist,D], order [N] int64 rows sorted
+        by inverted list, offsets [nlist+1] host ints) — built once per
+        mutation epoch, reused by every search."""
+        if self._gpu is not None:
+            return self._gpu
+        import torch
+        st = self._flat.state()
+        dev = "cuda"
+        vecs = torch.from_numpy(np.ascontiguousarray(st["vecs"])).to(dev)
+        cent = torch.from_numpy(self.centroids).to(dev)
+        assign = np.array([self._assign.get(d, 0) for d in st["ids"]],
+                          dtype=np.int64)
+        order_np = np.argsort(assign, kind="stable")
+        counts = np.bincount(assign, minlength=self.nlist)
+        offsets = np.concatenate([[0], np.cumsum(counts)]).tolist()
+        order = torch.from_numpy(order_np).to(dev)
+        self._gpu = (vecs, cent, order, offsets)
+        return self._gpu
+
     # ---- search ------------------------------------------------------------
     def search(self, query: np.ndarray, top_k: int):
         if self.centroids is None:
             return self._flat.search(query, top_k)          # flat fallback
-        # coarse quantizer: [nlist] scores → nprobe lists
+        if self._flat.use_gpu:
+            return self._search_gpu(query, top_k)
+        # CPU fallback
         coarse = self.centroids @ query
         lists = set(np.argpartition(-coarse, self.nprobe - 1)
                     [:self.nprobe].tolist())
@@ -115,23 +145,41 @@ class IVFFlatIndex:
                 if self._assign.get(d, -1) in lists]
         if not cand:
             return self._flat.search(query, top_k)
-        if self._flat.use_gpu:
-            import torch
-            from .. import ops
-            dev = "cuda"
-            sub = torch.from_numpy(st["vecs"][cand]).to(dev)
-            q = torch.from_numpy(
-                np.ascontiguousarray(query[None, :])).to(dev)
-            scores = (q @ sub.T).float().contiguous()       # MFMA GEMM
-            k = min(top_k, len(cand), 32)
-            vals = torch.empty(1, k, dtype=torch.float32, device=dev)
-            idx = torch.empty(1, k, dtype=torch.int32, device=dev)
-            ops.load_extension()
-            torch.ops.kaito.topk(vals, idx, scores, k)      # HIP top-k
-            pairs = [(st["ids"][cand[i]], float(v))
-                     for i, v in zip(idx[0].cpu().tolist(),
-                                     vals[0].cpu().tolist())]
-            return pairs[:top_k]
         scores = st["vecs"][cand] @ query
         order = np.argsort(-scores)[:top_k]
         return [(st["ids"][cand[i]], float(scores[i])) for i in order]
+
+    def _search_gpu(self, query: np.ndarray, top_k: int):
+        """Fully GPU-resident IVF search: coarse centroid GEMM → HIP
+        top-k over lists → device gather of the probed lists' rows →
+        fine GEMM → HIP top-k. The only host↔device traffic per query is
+        the query vector up and (nprobe + top_k) scalars down."""
+        import torch
+        from .. import ops
+        ops.load_extension()
+        vecs, cent, order, offsets = self._gpu_state()
+        dev = vecs.device
+        q = torch.from_numpy(
+            np.ascontiguousarray(query[None, :].astype(np.float32))).to(dev)
+        # coarse: [1, nlist] scores → nprobe list ids (HIP top-k)
+        coarse = (q @ cent.T).contiguous()
+        npb = min(self.nprobe, self.nlist)
+        cvals = torch.empty(1, npb, dtype=torch.float32, device=dev)
+        cidx = torch.empty(1, npb, dtype=torch.int32, device=dev)
+        torch.ops.kaito.topk(cvals, cidx, coarse, npb)
+        lists = cidx[0].cpu().tolist()      # nprobe ints — the tiny D2H
+        segs = [order[offsets[j]:offsets[j + 1]] for j in lists
+                if offsets[j + 1] > offsets[j]]
+        if not segs:
+            return self._flat.search(query, top_k)
+        cand = torch.cat(segs)              # device gather indices
+        sub = vecs.index_select(0, cand)    # device row gather
+        scores = (q @ sub.T).contiguous()   # MFMA GEMM
+        k = min(top_k, cand.numel(), 32)
+        vals = torch.empty(1, k, dtype=torch.float32, device=dev)
+        idx = torch.empty(1, k, dtype=torch.int32, device=dev)
+        torch.ops.kaito.topk(vals, idx, scores, k)
+        rows = cand[idx[0].long()].cpu().tolist()
+        ids = self._flat.state()["ids"]
+        return [(ids[r], float(v))
+                for r, v in zip(rows, vals[0].cpu().tolist())][:top_k]
