@@ -51,8 +51,9 @@ def main():
     bytes_moved = T * L * KH * D * 2 * 2  # K+V read
     print(f"paged_attention    bs={T} L={L}: {t*1e6:.1f} us  "
           f"{bytes_moved/t/1e12:.2f} TB/s")
+    empty_sinks = torch.empty(0, dtype=torch.float32, device=dev)
     t = bench(lambda: torch.ops.kaito.paged_attention_sp(
-        out, q, kc, vc, bt, sl, scale))
+        out, q, kc, vc, bt, sl, scale, 0, empty_sinks))
     print(f"paged_attention_sp bs={T} L={L}: {t*1e6:.1f} us  "
           f"{bytes_moved/t/1e12:.2f} TB/s")
 

@@ -157,7 +157,12 @@ def test_paged_attention_impls(impl, G, D, lens):
     sl = torch.tensor(lens, dtype=torch.int32, device=DEV)
     scale = 1.0 / math.sqrt(D)
     out = torch.empty_like(q)
-    getattr(torch.ops.kaito, impl)(out, q, kc, vc, bt, sl, scale)
+    if impl == "paged_attention_sp":
+        empty_sinks = torch.empty(0, dtype=torch.float32, device=DEV)
+        torch.ops.kaito.paged_attention_sp(out, q, kc, vc, bt, sl, scale,
+                                           0, empty_sinks)
+    else:
+        torch.ops.kaito.paged_attention(out, q, kc, vc, bt, sl, scale)
     ref = R.paged_attention(q.cpu().float(), kc.cpu().float(), vc.cpu().float(),
                             bt.cpu(), sl.cpu(), scale)
     _close(out, ref.to(DEV), atol=2e-2)
