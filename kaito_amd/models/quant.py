@@ -100,6 +100,11 @@ class QuantLinear(torch.nn.Module):
         if x2.size(0) <= self.GEMV_MAX_M:
             y = ops.w4a16_gemv(x2, self.qweight, self.scales, self.zeros,
                                self.group)
+        elif x.is_cuda and self.out_features % 64 == 0 \
+                and self.in_features % 64 == 0:
+            # fused inline-dequant MFMA GEMM (no full-precision scratch)
+            y = ops.w4a16_gemm(x2, self.qweight, self.scales, self.zeros,
+                               self.group)
         else:
             w = ops.w4a16_dequant(self.qweight, self.scales, self.zeros,
                                   self.group)

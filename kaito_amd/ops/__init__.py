@@ -288,3 +288,18 @@ def w4a16_dequant(qweight: torch.Tensor, scales: torch.Tensor,
         return out
     ref = torch_ref.w4a16_unpack(qweight, scales, zeros, group)
     return ref.to(torch.bfloat16) if out is None else out.copy_(ref)
+
+
+def w4a16_gemm(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
+               zeros: torch.Tensor, group: int) -> torch.Tensor:
+    """Fused inline-dequant MFMA GEMM (GPU): y = x @ dequant(W)^T.
+    The 4-bit weights never round-trip through a full-precision scratch
+    buffer. CPU falls back to the unpack reference."""
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty(x.size(0), qweight.size(0), dtype=x.dtype,
+                          device=x.device)
+        torch.ops.kaito.w4a16_gemm(out, x.contiguous(), qweight, scales,
+                                   zeros, group)
+        return out
+    return torch_ref.w4a16_gemv(x, qweight, scales, zeros, group)

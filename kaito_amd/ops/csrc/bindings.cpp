@@ -55,6 +55,8 @@ void w4a16_gemv(at::Tensor out, at::Tensor x, at::Tensor qweight,
                 at::Tensor scales, at::Tensor zeros, int64_t group);
 void w4a16_dequant(at::Tensor out, at::Tensor qweight, at::Tensor scales,
                    at::Tensor zeros, int64_t group);
+void w4a16_gemm(at::Tensor out, at::Tensor x, at::Tensor qweight,
+                at::Tensor scales, at::Tensor zeros, int64_t group);
 void allreduce_rmsnorm(at::Tensor out, at::Tensor ptrs, at::Tensor weight,
                        double eps);
 void one_shot_ar_rmsnorm(at::Tensor out, at::Tensor residual, at::Tensor ptrs,
@@ -88,6 +90,7 @@ TORCH_LIBRARY(kaito, m) {
   m.def("paged_read_bw(Tensor(a!) out, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, int mode) -> ()");
   m.def("w4a16_gemv(Tensor(a!) out, Tensor x, Tensor qweight, Tensor scales, Tensor zeros, int group) -> ()");
   m.def("w4a16_dequant(Tensor(a!) out, Tensor qweight, Tensor scales, Tensor zeros, int group) -> ()");
+  m.def("w4a16_gemm(Tensor(a!) out, Tensor x, Tensor qweight, Tensor scales, Tensor zeros, int group) -> ()");
   m.def("allreduce_rmsnorm(Tensor(a!) out, Tensor ptrs, Tensor weight, float eps) -> ()");
   m.def("one_shot_ar_rmsnorm(Tensor(a!) out, Tensor(b!) residual, Tensor ptrs, Tensor sig_ptrs, Tensor(c!) counter, Tensor weight, float eps, int rank) -> ()");
   m.def("ipc_handle(Tensor t) -> Tensor");
@@ -118,6 +121,7 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("paged_read_bw", &kaito::paged_read_bw);
   m.impl("w4a16_gemv", &kaito::w4a16_gemv);
   m.impl("w4a16_dequant", &kaito::w4a16_dequant);
+  m.impl("w4a16_gemm", &kaito::w4a16_gemm);
   m.impl("allreduce_rmsnorm", &kaito::allreduce_rmsnorm);
   m.impl("one_shot_ar_rmsnorm", &kaito::one_shot_ar_rmsnorm);
   m.impl("ipc_handle", &kaito::ipc_handle);

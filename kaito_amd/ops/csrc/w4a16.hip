@@ -119,6 +119,136 @@ void w4a16_dequant_kernel(short* __restrict__ out,
   *reinterpret_cast<short8_t*>(out + (int64_t)n * K + k0) = o;
 }
 
+// ------------------------------------------------------- fused MFMA GEMM
+// Marlin-class mid/large-M path: C[M,N] = x[M,K]bf16 @ dequant(W)^T with
+// the dequant INLINE in the K-loop — the 4-bit weights go HBM → register
+// → (dequant) → swizzled LDS → MFMA, never touching a full-precision
+// scratch buffer (the round-1 path dequantized the whole weight to
+// global memory first: 5x the weight traffic, and it lost to bf16 at
+// saturation — docs/ROADMAP.md #2).
+//
+// Structure: 2-barrier 64x64x64 tile, 4 waves as 2x2, MFMA 16x16x32
+// (same skeleton as moe.hip). A staged with global_load_lds
+// (pre-swizzled source); B: each thread loads u32 packed words (one
+// word == one 16-B bf16 chunk after dequant), dequants with its group's
+// (s, z), and ds_writes the chunk at the XOR-swizzled slot.
+__global__ __launch_bounds__(256, 2)
+void w4a16_gemm_kernel(short* __restrict__ out,          // [M, N] bf16
+                       const short* __restrict__ x,      // [M, K] bf16
+                       const uint32_t* __restrict__ qw,  // [N, K/8]
+                       const float* __restrict__ scales, // [N, K/G]
+                       const float* __restrict__ zeros,
+                       int M, int N, int K, int group) {
+  constexpr int BM = 64, BN = 64, BK = 64;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wm = wave >> 1, wn = wave & 1;
+  const int lo = lane & 15, hi = lane >> 4;
+  const int kw = K >> 3;
+  const int gstride = K / group;
+
+  __shared__ short lds_a[BM * BK];
+  __shared__ short lds_b[BN * BK];
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int a = 0; a < 2; a++)
+#pragma unroll
+    for (int b = 0; b < 2; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- A tile via global_load_lds (rows clamped at the M tail) ----
+#pragma unroll
+    for (int i = 0; i < 2; i++) {
+      const int seg = i * 4 + wave;
+      const int idx = seg * 64 + lane;
+      const int row = idx >> 3;
+      const int c = (idx & 7) ^ (row & 7);
+      const int srow = min(m0 + row, M - 1);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)
+              (x + (int64_t)srow * K + k0 + c * 8),
+          (__attribute__((address_space(3))) unsigned int*)
+              (lds_a + seg * 512),
+          16, 0, 0);
+    }
+    // ---- B tile: packed u32 → inline dequant → swizzled ds_write ----
+#pragma unroll
+    for (int rep = 0; rep < 2; rep++) {
+      const int id = rep * 256 + threadIdx.x;     // 512 words per tile
+      const int row = id >> 3;                    // n row in tile
+      const int wslot = id & 7;                   // 8 words (64 k) / row
+      const int n = n0 + row;
+      const uint32_t q = qw[(int64_t)n * kw + (k0 >> 3) + wslot];
+      const int g = (k0 + wslot * 8) / group;
+      const float s = scales[(int64_t)n * gstride + g];
+      const float z = zeros[(int64_t)n * gstride + g];
+      short8_t dq;
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        dq[j] = f32_to_bf16(s * (float)((q >> (4 * j)) & 0xF) - z);
+      const int c = wslot ^ (row & 7);
+      *reinterpret_cast<short8_t*>(lds_b + row * BK + c * 8) = dq;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BK / 32; kk++) {
+      bf16x8 af[2], bf[2];
+#pragma unroll
+      for (int r = 0; r < 2; r++) {
+        {
+          const int row = wm * 32 + r * 16 + lo;
+          const int c = (kk * 4 + hi) ^ (row & 7);
+          af[r] = *reinterpret_cast<const bf16x8*>(lds_a + row * BK + c * 8);
+        }
+        {
+          const int row = wn * 32 + r * 16 + lo;
+          const int c = (kk * 4 + hi) ^ (row & 7);
+          bf[r] = *reinterpret_cast<const bf16x8*>(lds_b + row * BK + c * 8);
+        }
+      }
+#pragma unroll
+      for (int a = 0; a < 2; a++)
+#pragma unroll
+        for (int b = 0; b < 2; b++)
+          acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[a], bf[b], acc[a][b], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int a = 0; a < 2; a++) {
+    const int row_base = wm * 32 + a * 16 + hi * 4;
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      const int row = m0 + row_base + r;
+      if (row >= M) continue;
+#pragma unroll
+      for (int b = 0; b < 2; b++) {
+        const int col = n0 + wn * 32 + b * 16 + lo;
+        out[(int64_t)row * N + col] = f32_to_bf16(acc[a][b][r]);
+      }
+    }
+  }
+}
+
+void w4a16_gemm(at::Tensor out, at::Tensor x, at::Tensor qweight,
+                at::Tensor scales, at::Tensor zeros, int64_t group) {
+  const int M = x.size(0), K = x.size(1), N = qweight.size(0);
+  TORCH_CHECK(x.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(K % 64 == 0 && N % 64 == 0 && group % 64 == 0 &&
+              K % group == 0, "w4a16_gemm: 64|K, 64|N, 64|group required");
+  dim3 grid((M + 63) / 64, N / 64), block(256);
+  auto stream = at::hip::getCurrentHIPStream();
+  w4a16_gemm_kernel<<<grid, block, 0, stream>>>(
+      (short*)out.data_ptr(), (const short*)x.data_ptr(),
+      (const uint32_t*)qweight.data_ptr(), scales.data_ptr<float>(),
+      zeros.data_ptr<float>(), M, N, K, (int)group);
+}
+
 void w4a16_gemv(at::Tensor out, at::Tensor x, at::Tensor qweight,
                 at::Tensor scales, at::Tensor zeros, int64_t group) {
   const int M = x.size(0), K = x.size(1), N = qweight.size(0);

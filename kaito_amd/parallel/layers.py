@@ -53,6 +53,11 @@ class _QuantMixin:
         if x2.size(0) <= self._gemv_max_m:
             y = ops.w4a16_gemv(x2.contiguous(), self.qweight, self.scales,
                                self.zeros, self.q_group)
+        elif x.is_cuda and self.qweight.size(0) % 64 == 0 \
+                and x2.shape[-1] % 64 == 0:
+            # fused inline-dequant MFMA GEMM (no full-precision scratch)
+            y = ops.w4a16_gemm(x2.contiguous(), self.qweight, self.scales,
+                               self.zeros, self.q_group)
         else:
             w = ops.w4a16_dequant(self.qweight, self.scales, self.zeros,
                                   self.q_group)

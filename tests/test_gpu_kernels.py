@@ -476,6 +476,22 @@ def test_w4a16_dequant_matches_ref():
     _close(out, R.w4a16_unpack(qw, s, z, G).to(DEV), atol=1e-2, rtol=1e-2)
 
 
+@pytest.mark.parametrize("M,N,K,G", [(64, 256, 512, 128), (7, 128, 256, 64),
+                                     (200, 6144, 4096, 128)])
+def test_w4a16_fused_gemm_matches_ref(M, N, K, G):
+    """Inline-dequant MFMA GEMM vs the unpack+matmul fp32 reference."""
+    from kaito_amd.models.quant import quantize_w4
+    from kaito_amd.ops import torch_ref
+    torch.manual_seed(17)
+    w = torch.randn(N, K) * 0.1
+    qw, s, z = quantize_w4(w, G)
+    x = _bf16(M, K, scale=0.5)
+    out = ops.w4a16_gemm(x, qw.to(DEV), s.to(DEV), z.to(DEV), G)
+    deq = torch_ref.w4a16_unpack(qw, s, z, G)
+    expect = x.cpu().float() @ deq.T
+    _close(out, expect.to(DEV), atol=5e-2, rtol=5e-2)
+
+
 def test_w4a16_quantlinear_gpu_both_paths():
     torch.manual_seed(13)
     from kaito_amd.models.quant import QuantLinear
