@@ -201,10 +201,15 @@ class OperatorLoop:
             monitor_workspaces(self.client.list("Workspace"))
         return n
 
-    def run(self, interval_s: float = 5.0, max_ticks: Optional[int] = None):
+    def run(self, interval_s: float = 5.0, max_ticks: Optional[int] = None,
+            elector=None):
+        """Reconcile loop; with an elector, only the Lease holder
+        reconciles (controller-runtime leader-election semantics —
+        standby replicas keep renewing their candidacy)."""
         t = 0
         while max_ticks is None or t < max_ticks:
-            self.tick()
+            if elector is None or elector.try_acquire():
+                self.tick()
             t += 1
             time.sleep(interval_s)
 
@@ -221,15 +226,31 @@ def main(argv=None):
     p.add_argument("--controllers", default="all",
                    help="comma list (workspace,inferenceset,ragengine,"
                         "multirole,modelmirror) or 'all' (feature-gated)")
+    p.add_argument("--leader-elect", action="store_true", default=True)
+    p.add_argument("--no-leader-elect", dest="leader_elect",
+                   action="store_false")
+    p.add_argument("--bootstrap-webhook-cert", action="store_true",
+                   help="generate a self-signed serving cert and patch "
+                        "the ValidatingWebhookConfiguration caBundle")
     args = p.parse_args(argv)
     gates = parse_feature_gates(args.feature_gates)
     from .kubeclient_incluster import make_kube_client
     client = make_kube_client()
     logger.info("kube client: %s", type(client).__name__)
+    if args.bootstrap_webhook_cert:
+        from .leader import generate_self_signed_cert, patch_webhook_ca_bundle
+        crt, key, ca = generate_self_signed_cert()
+        patched = patch_webhook_ca_bundle(client, ca)
+        logger.info("webhook cert bootstrap: cert=%s patched=%s", crt,
+                    patched)
+    elector = None
+    if args.leader_elect:
+        from .leader import LeaderElector
+        elector = LeaderElector(client)
     loop = OperatorLoop(client, args.cloud_provider, args.node_provisioner,
                         args.preset_image, gates,
                         controllers=args.controllers)
-    loop.run(args.reconcile_interval)
+    loop.run(args.reconcile_interval, elector=elector)
 
 
 if __name__ == "__main__":

@@ -125,10 +125,128 @@ class KarpenterProvisioner(NodeProvisioner):
                                claim["metadata"]["name"])
 
 
+    # ---- NodePool / NodeClass management (karpenter/nodepool.go) ----
+    def ensure_node_pool(self, ws: Workspace, replicas: int = 1) -> Dict:
+        """Per-workspace NodePool with a zero disruption budget (nodes
+        are replaced only when the drift controller opens the budget —
+        reference karpenter/provisioner.go + nodepool.go)."""
+        name = f"{ws.name}-pool"
+        try:
+            return self.client.get("NodePool", ws.namespace, name)
+        except NotFound:
+            pool = {
+                "apiVersion": "karpenter.sh/v1", "kind": "NodePool",
+                "metadata": {"name": name, "namespace": ws.namespace,
+                             "labels": {LABEL_WORKSPACE_NAME: ws.name}},
+                "spec": {
+                    "disruption": {"budgets": [{"nodes": "0"}],
+                                   "consolidationPolicy": "WhenEmpty"},
+                    "limits": {"amd.com/gpu": str(replicas * 8)},
+                    "template": {"spec": {
+                        "nodeClassRef": {"name": self.node_class},
+                        "requirements": [{
+                            "key": "node.kubernetes.io/instance-type",
+                            "operator": "In",
+                            "values": [ws.resource.instanceType],
+                        }],
+                    }},
+                },
+            }
+            self.client.create(pool)
+            return pool
+
+    def set_drift_remediation(self, ws: Workspace, enabled: bool) -> None:
+        """Open/close the NodePool disruption budget (drift controller)."""
+        try:
+            np = self.client.get("NodePool", ws.namespace, f"{ws.name}-pool")
+        except NotFound:
+            return
+        np["spec"].setdefault("disruption", {})["budgets"] = [
+            {"nodes": "1" if enabled else "0"}]
+        self.client.update(np)
+
+
+class AzureGPUProvisioner(NodeProvisioner):
+    """azure-gpu-provisioner analog (reference
+    gpu-provisioner/gpu_provisioner.go, 283 L): one NodeClaim per node
+    with the Azure VM SKU requirement and kaito ownership labels; the
+    azure machine controller fulfils the claim. Differs from the
+    Karpenter path in claim shape (capacity-type on-demand, azure
+    node-class group) and in that no NodePool is managed."""
+
+    def __init__(self, client: KubeClient):
+        self.client = client
+
+    def _claim_name(self, ws: Workspace, i: int) -> str:
+        return f"{ws.name}-az-{i}"
+
+    def provision_nodes(self, ws, count):
+        created = []
+        for i in range(count):
+            name = self._claim_name(ws, i)
+            try:
+                self.client.get("NodeClaim", ws.namespace, name)
+            except NotFound:
+                self.client.create({
+                    "apiVersion": "karpenter.sh/v1",
+                    "kind": "NodeClaim",
+                    "metadata": {
+                        "name": name, "namespace": ws.namespace,
+                        "labels": {
+                            LABEL_WORKSPACE_NAME: ws.name,
+                            LABEL_WORKSPACE_NAMESPACE: ws.namespace,
+                            "karpenter.sh/capacity-type": "on-demand",
+                        },
+                        "annotations": {
+                            "kubernetes.azure.com/apiversion": "v1",
+                        },
+                    },
+                    "spec": {
+                        "nodeClassRef": {"group": "karpenter.azure.com",
+                                         "kind": "AKSNodeClass",
+                                         "name": "default"},
+                        "requirements": [
+                            {"key": "node.kubernetes.io/instance-type",
+                             "operator": "In",
+                             "values": [ws.resource.instanceType]},
+                            {"key": "karpenter.sh/capacity-type",
+                             "operator": "In", "values": ["on-demand"]},
+                        ],
+                        "resources": {"requests": {"amd.com/gpu": "1"}},
+                    },
+                    "status": {},
+                })
+                created.append(name)
+        return created
+
+    def ensure_nodes_ready(self, ws, count):
+        ready = []
+        for claim in self.client.list("NodeClaim", ws.namespace, {
+                LABEL_WORKSPACE_NAME: ws.name}):
+            node_name = claim.get("status", {}).get("nodeName")
+            if not node_name:
+                continue
+            try:
+                node = self.client.get("Node", "", node_name)
+            except NotFound:
+                continue
+            if _node_ready(node):
+                ready.append(node_name)
+        return ready[:count]
+
+    def delete_nodes(self, ws):
+        for claim in self.client.list("NodeClaim", ws.namespace, {
+                LABEL_WORKSPACE_NAME: ws.name}):
+            self.client.delete("NodeClaim", ws.namespace,
+                               claim["metadata"]["name"])
+
+
 def make_provisioner(kind: str, client: KubeClient, **kw) -> NodeProvisioner:
     """Reference parity: nodeprovision/manager/factory.go:121."""
     if kind in ("byo", "none"):
         return BYOProvisioner(client)
     if kind == "karpenter":
         return KarpenterProvisioner(client, **kw)
+    if kind in ("azure", "azure-gpu-provisioner", "gpuprovisioner"):
+        return AzureGPUProvisioner(client)
     raise ValueError(f"unknown provisioner {kind!r}")
