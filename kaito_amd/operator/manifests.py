@@ -119,23 +119,93 @@ def generate_inference_pod_spec(ws: Workspace, model: ModelConfig,
                         "configMap": {"name": ws.inference.config}})
         container["volumeMounts"].append(
             {"name": "config-volume", "mountPath": CONFIG_MOUNT})
-    init_containers = []
+
+    # ---- runtime-toolkit gate (the reference injects a cuda-toolkit
+    # provisioner init container for JIT models, :704-738; the MI355X
+    # analog asserts the node exposes gfx950 before the engine starts)
+    init_containers: List[Dict[str, Any]] = [{
+        "name": "rocm-runtime-check",
+        "image": image,
+        "command": ["/bin/sh", "-c",
+                    "rocminfo | grep -q gfx950 || "
+                    "{ echo 'node is not gfx950'; exit 1; }"],
+        "resources": {"limits": {GPU_RESOURCE: "1"}},
+    }]
+
+    # ---- adapters (+ per-adapter strength env, :886-956, :946-951)
     for ad in (ws.inference.adapters if ws.inference else []):
         nm = ad.source.get("name", "adapter")
         init_containers.append({
             "name": f"adapter-{nm}",
             "image": ad.source.get("image", ""),
             "command": ["/bin/sh", "-c",
+                        f"mkdir -p {ADAPTER_MOUNT}/{nm} && "
                         f"cp -r /data/* {ADAPTER_MOUNT}/{nm}/"],
             "volumeMounts": [{"name": "adapter-volume",
                               "mountPath": ADAPTER_MOUNT}],
         })
-    if init_containers:
+        if ad.strength:
+            container["env"].append({
+                "name": f"KAITO_ADAPTER_STRENGTH_{nm.upper().replace('-', '_')}",
+                "value": str(ad.strength)})
+    if len(init_containers) > 1:
         volumes.append({"name": "adapter-volume", "emptyDir": {}})
         container["volumeMounts"].append(
             {"name": "adapter-volume", "mountPath": ADAPTER_MOUNT})
+
+    # ---- local-weights NVMe cache + download monitor (:158-177; the
+    # entrypoint serves kaito_model_download_* gauges on /metrics while
+    # weights stream in — server/download_monitor.py)
+    from .api_types import (ANNOTATION_DISABLE_BENCHMARK,
+                            ANNOTATION_USE_LOCAL_WEIGHTS,
+                            LABEL_INFERENCE_ROLE)
+    use_local = ws.annotations.get(ANNOTATION_USE_LOCAL_WEIGHTS) == "true"
+    if use_local:
+        container["volumeMounts"].append(
+            {"name": "weights-cache", "mountPath": WEIGHTS_MOUNT})
+        container["env"] += [
+            {"name": "KAITO_WEIGHTS_PATH", "value": WEIGHTS_MOUNT},
+            {"name": "KAITO_DOWNLOAD_MONITOR", "value": "1"},
+        ]
+
+    # ---- benchmark startup probe (:455-480): one-shot saturation
+    # benchmark emitting KAITO_BENCHMARK_RESULT to the pod log, read
+    # back by the controller (controllers/workspace.py _ingest_benchmark)
+    if ws.annotations.get(ANNOTATION_DISABLE_BENCHMARK) != "true":
+        container["startupProbe"] = {
+            "exec": {"command": [
+                "python3", "-m", "kaito_amd.server.benchmark_entrypoint",
+                "--once"]},
+            "failureThreshold": max(readiness_timeout_s // 10, 6),
+            "periodSeconds": 10, "timeoutSeconds": 600,
+        }
+
+    # ---- P/D disaggregation (decode role: engine on :5001 behind the
+    # routing sidecar on :5000; KV-transfer side channel env —
+    # preset_inferences.go:1082-1152)
+    role = ws.labels.get(LABEL_INFERENCE_ROLE)
+    sidecars: List[Dict[str, Any]] = []
+    if role:
+        container["env"].append(
+            {"name": "KAITO_INFERENCE_ROLE", "value": role})
+        container["env"].append(
+            {"name": "KAITO_KV_TRANSFER_PORT", "value": "5600"})
+    if role == "decode":
+        container["env"].append(
+            {"name": "KAITO_INFERENCE_PORT", "value": "5001"})
+        container["ports"] = [
+            {"containerPort": 5001, "name": "engine"}]
+        sidecars.append({
+            "name": "routing-proxy",
+            "image": image,
+            "command": ["python3", "-m", "kaito_amd.server.dp_frontend",
+                        "--port", "5000", "--backends",
+                        "http://127.0.0.1:5001"],
+            "ports": [{"containerPort": INFERENCE_PORT, "name": "http"}],
+        })
+
     spec = {
-        "containers": [container],
+        "containers": [container] + sidecars,
         "initContainers": init_containers,
         "volumes": volumes,
         "tolerations": [
@@ -161,7 +231,7 @@ def generate_statefulset(ws: Workspace, model: ModelConfig, gpu: GPUConfig,
     pod_spec = generate_inference_pod_spec(
         ws, model, gpu, plan, image,
         readiness_timeout_s=readiness_timeout_for(model))
-    return {
+    sts = {
         "apiVersion": "apps/v1",
         "kind": "StatefulSet",
         "metadata": {"name": ws.name, "namespace": ws.namespace,
@@ -177,6 +247,22 @@ def generate_statefulset(ws: Workspace, model: ModelConfig, gpu: GPUConfig,
             },
         },
     }
+    # local NVMe weights cache: per-pod PVC template on the NVMe storage
+    # class (preset_inferences.go:158-177, :262-268) when the workspace
+    # opts into local weights
+    from .api_types import ANNOTATION_USE_LOCAL_WEIGHTS
+    if ws.annotations.get(ANNOTATION_USE_LOCAL_WEIGHTS) == "true":
+        size_gib = max(64, int(model.param_bytes() / (1 << 30) * 2.5) + 48)
+        sts["spec"]["volumeClaimTemplates"] = [{
+            "metadata": {"name": "weights-cache"},
+            "spec": {
+                "accessModes": ["ReadWriteOnce"],
+                "storageClassName": NVME_STORAGE_CLASS,
+                "resources": {"requests": {
+                    "storage": f"{(size_gib + 9) // 10 * 10}Gi"}},
+            },
+        }]
+    return sts
 
 
 def generate_tuning_job(ws: Workspace, model: ModelConfig, gpu: GPUConfig,

@@ -128,9 +128,12 @@ void moe_gate_silu_kernel(
   const int wm = wave >> 1, wn = wave & 1;   // 2x2 wave grid
   const int lo = lane & 15, hi = lane >> 4;
 
-  __shared__ short lds_a[BM * BK];
-  __shared__ short lds_bg[BN * BK];
-  __shared__ short lds_bu[BN * BK];
+  // double-buffered staging (guide §5.5 T3 "minimum 2-phase": issue the
+  // NEXT K-tile's global_load_lds before computing the current one, ONE
+  // barrier per tile — the loads stay in flight across the MFMA phase)
+  __shared__ short lds_a[2][BM * BK];
+  __shared__ short lds_bg[2][BN * BK];
+  __shared__ short lds_bu[2][BN * BK];
   __shared__ int s_rows[BM];
   for (int i = threadIdx.x; i < BM; i += 256)
     s_rows[i] = sorted_ids[m0 + min(i, m_rem - 1)];
@@ -148,19 +151,26 @@ void moe_gate_silu_kernel(
       accu[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
     }
 
+  stage_tile_gathered(x, H, s_rows, lds_a[0], wave, lane);
+  stage_tile_rows(wg, H, lds_bg[0], wave, lane);
+  stage_tile_rows(wu, H, lds_bu[0], wave, lane);
+  __syncthreads();
+  int cur = 0;
   for (int k0 = 0; k0 < H; k0 += BK) {
-    stage_tile_gathered(x + k0, H, s_rows, lds_a, wave, lane);
-    stage_tile_rows(wg + k0, H, lds_bg, wave, lane);
-    stage_tile_rows(wu + k0, H, lds_bu, wave, lane);
-    __syncthreads();
+    if (k0 + BK < H) {   // issue next tile BEFORE computing this one
+      stage_tile_gathered(x + k0 + BK, H, s_rows, lds_a[cur ^ 1], wave,
+                          lane);
+      stage_tile_rows(wg + k0 + BK, H, lds_bg[cur ^ 1], wave, lane);
+      stage_tile_rows(wu + k0 + BK, H, lds_bu[cur ^ 1], wave, lane);
+    }
 #pragma unroll
     for (int kk = 0; kk < BK / 32; kk++) {
       bf16x8 af[2], bg[2], bu[2];
 #pragma unroll
       for (int r = 0; r < 2; r++) {
-        af[r] = frag_read(lds_a, wm * 32 + r * 16 + lo, kk, hi);
-        bg[r] = frag_read(lds_bg, wn * 32 + r * 16 + lo, kk, hi);
-        bu[r] = frag_read(lds_bu, wn * 32 + r * 16 + lo, kk, hi);
+        af[r] = frag_read(lds_a[cur], wm * 32 + r * 16 + lo, kk, hi);
+        bg[r] = frag_read(lds_bg[cur], wn * 32 + r * 16 + lo, kk, hi);
+        bu[r] = frag_read(lds_bu[cur], wn * 32 + r * 16 + lo, kk, hi);
       }
 #pragma unroll
       for (int a = 0; a < 2; a++)
@@ -172,7 +182,8 @@ void moe_gate_silu_kernel(
               af[a], bu[b], accu[a][b], 0, 0, 0);
         }
     }
-    __syncthreads();
+    __syncthreads();   // next tile landed AND this buffer fully read
+    cur ^= 1;
   }
 
   // epilogue: silu(g)*u → act[m0+row][n0+col] (rows are SORTED order,
@@ -222,8 +233,8 @@ void moe_down_scatter_kernel(
   const int wm = wave >> 1, wn = wave & 1;
   const int lo = lane & 15, hi = lane >> 4;
 
-  __shared__ short lds_a[BM * BK];
-  __shared__ short lds_b[BN * BK];
+  __shared__ short lds_a[2][BM * BK];
+  __shared__ short lds_b[2][BN * BK];
 
   const short* wd = w + (int64_t)e * H * IE + (int64_t)n0 * IE;
 
@@ -233,17 +244,23 @@ void moe_down_scatter_kernel(
 #pragma unroll
     for (int b = 0; b < 2; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
+  stage_tile_rows(act + (int64_t)m0 * IE, IE, lds_a[0], wave, lane);
+  stage_tile_rows(wd, IE, lds_b[0], wave, lane);
+  __syncthreads();
+  int cur = 0;
   for (int k0 = 0; k0 < IE; k0 += BK) {
-    stage_tile_rows(act + (int64_t)m0 * IE + k0, IE, lds_a, wave, lane);
-    stage_tile_rows(wd + k0, IE, lds_b, wave, lane);
-    __syncthreads();
+    if (k0 + BK < IE) {
+      stage_tile_rows(act + (int64_t)m0 * IE + k0 + BK, IE,
+                      lds_a[cur ^ 1], wave, lane);
+      stage_tile_rows(wd + k0 + BK, IE, lds_b[cur ^ 1], wave, lane);
+    }
 #pragma unroll
     for (int kk = 0; kk < BK / 32; kk++) {
       bf16x8 af[2], bf[2];
 #pragma unroll
       for (int r = 0; r < 2; r++) {
-        af[r] = frag_read(lds_a, wm * 32 + r * 16 + lo, kk, hi);
-        bf[r] = frag_read(lds_b, wn * 32 + r * 16 + lo, kk, hi);
+        af[r] = frag_read(lds_a[cur], wm * 32 + r * 16 + lo, kk, hi);
+        bf[r] = frag_read(lds_b[cur], wn * 32 + r * 16 + lo, kk, hi);
       }
 #pragma unroll
       for (int a = 0; a < 2; a++)
@@ -253,6 +270,7 @@ void moe_down_scatter_kernel(
               af[a], bf[b], acc[a][b], 0, 0, 0);
     }
     __syncthreads();
+    cur ^= 1;
   }
 
   // NOTE: rows past m_end within this tile read garbage A rows — only

@@ -139,7 +139,17 @@ def main(argv=None):
     p.add_argument("--output-tokens", type=int, default=256)
     p.add_argument("--max-concurrency", type=int, default=512)
     p.add_argument("--health-timeout", type=float, default=3600)
+    p.add_argument("--once", action="store_true",
+                   help="startup-probe mode: run the benchmark exactly "
+                        "once per pod (marker file), succeed instantly "
+                        "on later probe invocations")
+    p.add_argument("--marker", default="/tmp/kaito-benchmark-done")
     args = p.parse_args(argv)
+
+    if args.once:
+        import os
+        if os.path.exists(args.marker):
+            return 0
 
     if not asyncio.run(wait_healthy(args.base_url, args.health_timeout)):
         print("engine never became healthy", file=sys.stderr)
@@ -156,6 +166,9 @@ def main(argv=None):
         args.base_url, args.duration, args.input_tokens, args.output_tokens,
         args.max_concurrency))
     emit(RESULT_TAG, result)
+    if args.once:
+        with open(args.marker, "w") as f:
+            f.write("done\n")
     return 0
 
 

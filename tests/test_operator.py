@@ -183,10 +183,47 @@ def test_statefulset_golden_8b():
     c = pod["containers"][0]
     assert c["resources"]["limits"]["amd.com/gpu"] == "8"
     assert any(e["name"] == "HSA_ENABLE_IPC_MODE_LEGACY" for e in c["env"])
-    assert c["startupProbe"]["httpGet"]["path"] == "/health"
+    # benchmark-on-by-default: the startup probe is the one-shot
+    # saturation benchmark exec (reference preset_inferences.go:455-480)
+    assert c["startupProbe"]["exec"]["command"][2] == \
+        "kaito_amd.server.benchmark_entrypoint"
+    assert "--once" in c["startupProbe"]["exec"]["command"]
+    assert c["livenessProbe"]["httpGet"]["path"] == "/health"
     assert pod["nodeSelector"]["node.kubernetes.io/instance-type"] == \
         "Standard_ND96isr_MI355X_v1"
     assert {"name": "dshm", "mountPath": "/dev/shm"} in c["volumeMounts"]
+    # runtime-toolkit analog: gfx950 gate init container
+    assert pod["initContainers"][0]["name"] == "rocm-runtime-check"
+
+
+def test_statefulset_modifiers_local_weights_and_roles():
+    import kaito_amd.operator.api_types as at2
+    ws = _ws()
+    ws.annotations[at2.ANNOTATION_USE_LOCAL_WEIGHTS] = "true"
+    ws.annotations[at2.ANNOTATION_DISABLE_BENCHMARK] = "true"
+    ws.labels[at2.LABEL_INFERENCE_ROLE] = "decode"
+    ws.inference.adapters = [at.AdapterSpec(
+        source={"name": "fr", "image": "r/a:1"}, strength="0.7")]
+    mc = get_model_config("llama-3-8b")
+    ss = mf.generate_statefulset(ws, mc, _gpu(), image="kaito/engine:v1")
+    pod = ss["spec"]["template"]["spec"]
+    c = pod["containers"][0]
+    # local weights: NVMe PVC template + mount + download monitor env
+    assert ss["spec"]["volumeClaimTemplates"][0]["spec"][
+        "storageClassName"] == "kaito-local-nvme-disk"
+    assert any(m["mountPath"] == "/workspace/weights"
+               for m in c["volumeMounts"])
+    env = {e["name"]: e.get("value") for e in c["env"]}
+    assert env.get("KAITO_DOWNLOAD_MONITOR") == "1"
+    # adapter strength env (preset_inferences.go:946-951)
+    assert env.get("KAITO_ADAPTER_STRENGTH_FR") == "0.7"
+    # disable-benchmark: startup probe stays httpGet
+    assert "httpGet" in c["startupProbe"]
+    # decode role: engine on :5001 + routing sidecar on :5000
+    assert env.get("KAITO_INFERENCE_ROLE") == "decode"
+    assert c["ports"][0]["containerPort"] == 5001
+    assert pod["containers"][1]["name"] == "routing-proxy"
+    assert pod["containers"][1]["ports"][0]["containerPort"] == 5000
 
 
 def test_statefulset_multinode_70b():

@@ -1,3 +1,4 @@
+import sys, os; sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import torch, time
 from kaito_amd.models.quant import QuantLinear
 torch.manual_seed(0)
