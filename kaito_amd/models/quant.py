@@ -101,8 +101,11 @@ class QuantLinear(torch.nn.Module):
             y = ops.w4a16_gemv(x2, self.qweight, self.scales, self.zeros,
                                self.group)
         elif x.is_cuda and self.out_features % 64 == 0 \
-                and self.in_features % 64 == 0:
-            # fused inline-dequant MFMA GEMM (no full-precision scratch)
+                and self.in_features % 64 == 0 and ops.W4A16_FUSED_GEMM:
+            # fused inline-dequant MFMA GEMM — measured SLOWER than
+            # dequant+hipBLASLt at M>=64 (Tensile ~1.1 PF vs this 64^2
+            # tile's 403 TF, tools/bench_w4a16.py); kept env-gated
+            # (KAITO_W4A16_FUSED=1) for further schedule work
             y = ops.w4a16_gemm(x2, self.qweight, self.scales, self.zeros,
                                self.group)
         else:

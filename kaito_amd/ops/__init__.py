@@ -98,6 +98,9 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
 # the r01 single-pass online-softmax kernel. Overridable for A/B runs.
 import os as _os
 _PA_IMPL = _os.environ.get("KAITO_PA_IMPL", "sp")
+# fused W4A16 GEMM for M>4: measured slower than dequant+hipBLASLt on
+# MI355X (tools/bench_w4a16.py) — off unless explicitly enabled
+W4A16_FUSED_GEMM = _os.environ.get("KAITO_W4A16_FUSED", "0") == "1"
 
 
 def _sinks_arg(sinks, device):

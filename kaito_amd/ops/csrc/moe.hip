@@ -107,7 +107,7 @@ KAITO_DEV float moe_act_f(int mode, float g, float u) {
 
 // act[s, n] = silu(x[tok_s] @ Wg[e]^T) * (x[tok_s] @ Wu[e]^T)
 // grid: (max_m_tiles, n_local_experts, IE/BN); block 256.
-__global__ __launch_bounds__(256, 2)
+__global__ __launch_bounds__(256, 3)
 void moe_gate_silu_kernel(
     short* __restrict__ act,             // [TK, IE] bf16
     const short* __restrict__ x,         // [T, H] bf16
@@ -212,7 +212,7 @@ void moe_gate_silu_kernel(
 
 // out[tok_s, n] += gate_s * (act[s] @ Wd[e]^T); out is f32, atomics.
 // grid: (max_m_tiles, n_local_experts, H/BN); block 256.
-__global__ __launch_bounds__(256, 2)
+__global__ __launch_bounds__(256, 4)
 void moe_down_scatter_kernel(
     float* __restrict__ out,             // [T, H] f32 (pre-zeroed)
     const short* __restrict__ act,       // [TK, IE] bf16

@@ -54,8 +54,9 @@ class _QuantMixin:
             y = ops.w4a16_gemv(x2.contiguous(), self.qweight, self.scales,
                                self.zeros, self.q_group)
         elif x.is_cuda and self.qweight.size(0) % 64 == 0 \
-                and x2.shape[-1] % 64 == 0:
-            # fused inline-dequant MFMA GEMM (no full-precision scratch)
+                and x2.shape[-1] % 64 == 0 and ops.W4A16_FUSED_GEMM:
+            # env-gated: see models/quant.py (dequant+hipBLASLt measured
+            # faster at mid/large M)
             y = ops.w4a16_gemm(x2.contiguous(), self.qweight, self.scales,
                                self.zeros, self.q_group)
         else:

@@ -8,6 +8,7 @@ streams flow back through per-request asyncio queues.
 from __future__ import annotations
 
 import asyncio
+import os
 import queue
 import threading
 import time
@@ -92,7 +93,13 @@ class AsyncLLMEngine:
 
     def _loop(self):
         eng = self.engine
+        hb = None
+        if os.environ.get("KAITO_HEARTBEAT", "1") != "0":
+            from .heartbeat import Heartbeat
+            hb = Heartbeat(int(os.environ.get("RANK", "0")))
         while not self._stop:
+            if hb is not None:
+                hb.beat()   # per-rank liveness (multi_node_health_check)
             self._drain_submissions()
             if not eng.has_unfinished():
                 metrics.REQUESTS_RUNNING.set(0)

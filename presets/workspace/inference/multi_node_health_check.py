@@ -1,8 +1,13 @@
 """Multi-node liveness/readiness probes — parity with the reference's
-presets/workspace/inference/vllm/multi-node-health-check.py:21-50:
-liveness fails when the distributed worker group lost a member (our
-torchrun-based bootstrap: the rendezvous store heartbeat); readiness is
-the leader's /health."""
+presets/workspace/inference/vllm/multi-node-health-check.py:21-50.
+
+Readiness: the leader's /health.
+Liveness: PER-RANK heartbeats (kaito_amd.server.heartbeat): every local
+engine rank must have refreshed its own file recently AND its recorded
+pid must be alive — the analog of the reference's Ray actor-death scan.
+A hung rank fails its own heartbeat; a dead rank fails the pid check;
+no surviving process can mask either (the round-1 shared-mtime file
+could)."""
 import os
 import sys
 import urllib.request
@@ -22,14 +27,16 @@ def main():
     leader = os.environ.get("KAITO_LEADER_HOST", "127.0.0.1")
     if mode == "readiness":
         ok = leader_health(leader)
-    else:  # liveness: engine process heartbeat file updated by the runner
-        hb = os.environ.get("KAITO_HEARTBEAT_FILE", "/tmp/kaito_heartbeat")
-        import time
-        try:
-            ok = time.time() - os.path.getmtime(hb) < 120
-        except OSError:
-            # no heartbeat file yet: fall back to leader health
-            ok = leader_health(leader)
+    else:  # liveness
+        from kaito_amd.server.heartbeat import check_all
+        n = int(os.environ.get("LOCAL_WORLD_SIZE",
+                               os.environ.get("KAITO_LOCAL_RANKS", "1")))
+        node = int(os.environ.get("GROUP_RANK",
+                                  os.environ.get("POD_INDEX", "0") or "0"))
+        ranks = [node * n + i for i in range(n)]   # this node's GLOBAL ranks
+        ok, detail = check_all(n, ranks=ranks)
+        if not ok:
+            print(f"liveness: {detail}", file=sys.stderr)
     sys.exit(0 if ok else 1)
 
 
