@@ -39,6 +39,9 @@ def parse_args():
     p.add_argument("--quantization", default="", choices=["", "w4a16"],
                    help="W4A16 weight-only serving (NOT the headline "
                         "config; reported dtype reflects it)")
+    p.add_argument("--kv-dtype", default="auto", choices=["auto", "fp8"],
+                   help="KV cache dtype (fp8 = OCP e4m3; NOT the headline "
+                        "config)")
     p.add_argument("--mixed-steps", action="store_true",
                    help="enable mixed (overlapped prefill+decode) steps")
     p.add_argument("--tune-gemms", default=None, nargs="?", const="",
@@ -83,6 +86,7 @@ def main():
         tensor_parallel_size=args.tp,
         enforce_eager=args.eager,
         enable_mixed_batch=args.mixed_steps,
+        kv_cache_dtype=args.kv_dtype,
         seed=1234 + dp_rank,
     )
     eng = LLMEngine(cfg)

@@ -161,6 +161,10 @@ class ModelConfig:
 class EngineConfig:
     model: ModelConfig = field(default_factory=ModelConfig)
     block_size: int = 16
+    # KV cache dtype: "auto"/"bf16" | "fp8" (OCP e4m3, GPU only —
+    # halves attention bytes and doubles KV capacity; vLLM's
+    # --kv-cache-dtype fp8 analog, unscaled)
+    kv_cache_dtype: str = "auto"
     max_num_seqs: int = 1024
     max_num_batched_tokens: int = 8192      # prefill token budget per step
     max_model_len: Optional[int] = None     # None → "auto": fit KV budget

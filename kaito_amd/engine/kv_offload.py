@@ -59,7 +59,8 @@ class KVOffloadManager:
     def _seq_bytes(self, num_tokens: int) -> int:
         k0 = self.kv_caches[0][0]
         kvh, _, d = k0.shape[1], k0.shape[2], k0.shape[3]
-        return len(self.kv_caches) * 2 * num_tokens * kvh * d * 2
+        return (len(self.kv_caches) * 2 * num_tokens * kvh * d
+                * k0.element_size())
 
     def _gather_tokens(self, block_table: List[int], num_tokens: int):
         """Copy the first num_tokens of a sequence's KV to host tensors

@@ -84,7 +84,12 @@ class ModelRunner:
         m = cfg.model
         tp = max(cfg.tensor_parallel_size, 1)
         kvh = max(m.num_kv_heads // tp, 1)
-        block_bytes = 2 * m.num_layers * kvh * cfg.block_size * m.head_dim * 2
+        fp8 = cfg.kv_cache_dtype == "fp8"
+        if fp8 and not self.is_gpu:
+            raise ValueError("kv_cache_dtype=fp8 needs a GPU")
+        elem = 1 if fp8 else 2
+        block_bytes = 2 * m.num_layers * kvh * cfg.block_size * m.head_dim \
+            * elem
         if cfg.num_gpu_blocks is not None:
             n_blocks = cfg.num_gpu_blocks
         elif self.is_gpu:
@@ -99,9 +104,10 @@ class ModelRunner:
         n_blocks = min(n_blocks, cap)
         self.num_gpu_blocks = n_blocks
         kvs = []
+        kv_dtype = torch.uint8 if fp8 else m.dtype
         for _ in range(self.model.num_local_layers):
             k = torch.zeros(n_blocks, kvh, cfg.block_size, m.head_dim,
-                            dtype=m.dtype, device=self.device)
+                            dtype=kv_dtype, device=self.device)
             v = torch.zeros_like(k)
             kvs.append((k, v))
         self.kv_caches = kvs

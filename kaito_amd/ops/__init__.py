@@ -116,7 +116,8 @@ def paged_attention(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tenso
     if q.is_cuda:
         _require_ext()
         out = torch.empty_like(q)
-        if _PA_IMPL == "sp" or window > 0 or sinks is not None:
+        if _PA_IMPL == "sp" or window > 0 or sinks is not None \
+                or k_cache.dtype == torch.uint8:
             # window/sink support lives in the split-phase kernel only
             torch.ops.kaito.paged_attention_sp(
                 out, q, k_cache, v_cache, block_tables, seq_lens, scale,

@@ -10,11 +10,14 @@
 
 namespace kaito {
 
+// FP8 = true: cache stores OCP e4m3 bytes (halves decode attention
+// bytes + doubles KV capacity; unscaled like vLLM's fp8 default).
+template <bool FP8>
 __global__ void reshape_and_cache_kernel(
     const short* __restrict__ k,       // [T, KH*D]
     const short* __restrict__ v,       // [T, KH*D]
-    short* __restrict__ k_cache,       // [B, KH, BS, D]
-    short* __restrict__ v_cache,
+    void* __restrict__ k_cache,        // [B, KH, BS, D] bf16 | fp8
+    void* __restrict__ v_cache,
     const int64_t* __restrict__ slots, // [T] = block*BS + off ; <0 = skip
     const int T, const int KH, const int D, const int BS,
     const int64_t kv_stride) {
@@ -31,8 +34,24 @@ __global__ void reshape_and_cache_kernel(
     const int h = (i * 8) / D;
     const int d = (i * 8) % D;
     const int64_t dst = (((block * KH + h) * BS + off) * D + d) / 8;
-    reinterpret_cast<short8_t*>(k_cache)[dst] = kv[i];
-    reinterpret_cast<short8_t*>(v_cache)[dst] = vv[i];
+    if constexpr (FP8) {
+      const short8_t kr = kv[i], vr = vv[i];
+      u8x8_t ko, vo;
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        const uint32_t kw = f32x2_to_fp8x2(bf16_to_f32(kr[2 * j]),
+                                           bf16_to_f32(kr[2 * j + 1]));
+        const uint32_t vw = f32x2_to_fp8x2(bf16_to_f32(vr[2 * j]),
+                                           bf16_to_f32(vr[2 * j + 1]));
+        ko[2 * j] = kw & 0xFF;     ko[2 * j + 1] = (kw >> 8) & 0xFF;
+        vo[2 * j] = vw & 0xFF;     vo[2 * j + 1] = (vw >> 8) & 0xFF;
+      }
+      reinterpret_cast<u8x8_t*>(k_cache)[dst] = ko;
+      reinterpret_cast<u8x8_t*>(v_cache)[dst] = vo;
+    } else {
+      reinterpret_cast<short8_t*>(k_cache)[dst] = kv[i];
+      reinterpret_cast<short8_t*>(v_cache)[dst] = vv[i];
+    }
   }
 }
 
@@ -52,10 +71,19 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
   auto stream = at::hip::getCurrentHIPStream();
   const int block = std::min(256, KH * D / 8);
   if (T == 0) return;
-  hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(T), dim3(block), 0, stream,
-      (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-      (short*)k_cache.data_ptr(), (short*)v_cache.data_ptr(),
-      slot_mapping.data_ptr<int64_t>(), T, KH, D, BS, k.stride(0));
+  if (k_cache.dtype() == at::kByte) {
+    hipLaunchKernelGGL((reshape_and_cache_kernel<true>), dim3(T),
+        dim3(block), 0, stream,
+        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+        k_cache.data_ptr(), v_cache.data_ptr(),
+        slot_mapping.data_ptr<int64_t>(), T, KH, D, BS, k.stride(0));
+  } else {
+    hipLaunchKernelGGL((reshape_and_cache_kernel<false>), dim3(T),
+        dim3(block), 0, stream,
+        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+        k_cache.data_ptr(), v_cache.data_ptr(),
+        slot_mapping.data_ptr<int64_t>(), T, KH, D, BS, k.stride(0));
+  }
 }
 
 }  // namespace kaito

@@ -31,6 +31,33 @@ KAITO_DEV float bf16_to_f32(short u) {
   return v.f;
 }
 
+// ---- fp8 (OCP e4m3fn — gfx950 native converts; NOT the MI300 fnuz) ----
+typedef uint8_t u8x8_t __attribute__((ext_vector_type(8)));   // 8 fp8 = 8B
+typedef float   f32x2_t __attribute__((ext_vector_type(2)));
+
+// pack two f32 into two e4m3 bytes (low half of the returned word)
+KAITO_DEV uint32_t f32x2_to_fp8x2(float a, float b) {
+  return (uint32_t)__builtin_amdgcn_cvt_pk_fp8_f32(a, b, 0, false);
+}
+
+// unpack bytes [0:1] (SEL=false) or [2:3] (SEL=true) of a word to 2 f32
+// (the builtin's word-select must be an immediate)
+template <bool SEL>
+KAITO_DEV f32x2_t fp8x2_to_f32x2(uint32_t w) {
+  return __builtin_amdgcn_cvt_pk_f32_fp8(w, SEL);
+}
+
+KAITO_DEV void fp8x8_to_f32(const u8x8_t v, float* __restrict__ out) {
+  const uint32_t* w = reinterpret_cast<const uint32_t*>(&v);
+#pragma unroll
+  for (int s = 0; s < 2; s++) {
+    const f32x2_t lo = fp8x2_to_f32x2<false>(w[s]);
+    const f32x2_t hi = fp8x2_to_f32x2<true>(w[s]);
+    out[4 * s + 0] = lo[0]; out[4 * s + 1] = lo[1];
+    out[4 * s + 2] = hi[0]; out[4 * s + 3] = hi[1];
+  }
+}
+
 KAITO_DEV short f32_to_bf16(float f) {
   union { float f; uint32_t i; } v;
   v.f = f;

@@ -19,12 +19,12 @@ constexpr int CTX_QTILE = 64;
 constexpr int CTX_KVT = 32;
 constexpr int CTX_PAD = 40;
 
-template <int D, int BS>
+template <int D, int BS, bool FP8 = false>
 __global__ __launch_bounds__(256, 2) void context_attn_kernel(
     short* __restrict__ out,            // [Tq, QH, D]
     const short* __restrict__ q,        // [Tq, QH, D] (suffix tokens)
-    const short* __restrict__ k_cache,  // [B, KH, BS, D]
-    const short* __restrict__ v_cache,
+    const void* __restrict__ k_cache,   // [B, KH, BS, D] bf16 | e4m3
+    const void* __restrict__ v_cache,
     const int* __restrict__ tile_seq,
     const int* __restrict__ tile_qbase,   // within-suffix q row base
     const int* __restrict__ cu_seqlens_q, // [batch+1] suffix lens prefix-sum
@@ -87,11 +87,27 @@ __global__ __launch_bounds__(256, 2) void context_attn_kernel(
         const int blk = bt[kvp / BS];
         const int64_t src =
             (((int64_t)blk * KH + kvh) * BS + (kvp % BS)) * D + col;
-        short8_t kd = *reinterpret_cast<const short8_t*>(k_cache + src);
+        short8_t kd, vd;
+        if constexpr (FP8) {   // e4m3 cache → bf16 LDS; body unchanged
+          float kf[8], vf[8];
+          fp8x8_to_f32(reinterpret_cast<const u8x8_t*>(k_cache)[src / 8],
+                       kf);
+          fp8x8_to_f32(reinterpret_cast<const u8x8_t*>(v_cache)[src / 8],
+                       vf);
+#pragma unroll
+          for (int j = 0; j < 8; j++) {
+            kd[j] = f32_to_bf16(kf[j]);
+            vd[j] = f32_to_bf16(vf[j]);
+          }
+        } else {
+          kd = *reinterpret_cast<const short8_t*>(
+              (const short*)k_cache + src);
+          vd = *reinterpret_cast<const short8_t*>(
+              (const short*)v_cache + src);
+        }
         const int bo = col * 2;
         const int swz = bo ^ ((row & 7) << 4);
         *reinterpret_cast<short8_t*>(&lds_k[row * D + swz / 2]) = kd;
-        short8_t vd = *reinterpret_cast<const short8_t*>(v_cache + src);
 #pragma unroll
         for (int j = 0; j < 8; j++) lds_vt[(col + j) * CTX_PAD + row] = vd[j];
       }
@@ -209,13 +225,26 @@ void context_attention(at::Tensor out, at::Tensor q, at::Tensor k_cache,
   auto stream = at::hip::getCurrentHIPStream();
   if (ntiles == 0) return;
 #define CTX_LAUNCH(D_)                                                        \
-  hipLaunchKernelGGL((context_attn_kernel<D_, 16>), dim3(ntiles, QH),         \
-      dim3(256), 0, stream, (short*)out.data_ptr(),                          \
-      (const short*)q.data_ptr(), (const short*)k_cache.data_ptr(),          \
-      (const short*)v_cache.data_ptr(), tile_seq.data_ptr<int>(),            \
-      tile_qbase.data_ptr<int>(), cu_seqlens_q.data_ptr<int>(),              \
-      kv_lens.data_ptr<int>(), block_tables.data_ptr<int>(), (float)scale,   \
-      QH, KH, max_blocks, q.stride(0), (int)window, sink_ptr)
+  do {                                                                       \
+    if (k_cache.dtype() == at::kByte)                                        \
+      hipLaunchKernelGGL((context_attn_kernel<D_, 16, true>),                 \
+          dim3(ntiles, QH), dim3(256), 0, stream, (short*)out.data_ptr(),    \
+          (const short*)q.data_ptr(), k_cache.data_ptr(),                    \
+          v_cache.data_ptr(), tile_seq.data_ptr<int>(),                      \
+          tile_qbase.data_ptr<int>(), cu_seqlens_q.data_ptr<int>(),          \
+          kv_lens.data_ptr<int>(), block_tables.data_ptr<int>(),             \
+          (float)scale, QH, KH, max_blocks, q.stride(0), (int)window,        \
+          sink_ptr);                                                         \
+    else                                                                     \
+      hipLaunchKernelGGL((context_attn_kernel<D_, 16, false>),                \
+          dim3(ntiles, QH), dim3(256), 0, stream, (short*)out.data_ptr(),    \
+          (const short*)q.data_ptr(), k_cache.data_ptr(),                    \
+          v_cache.data_ptr(), tile_seq.data_ptr<int>(),                      \
+          tile_qbase.data_ptr<int>(), cu_seqlens_q.data_ptr<int>(),          \
+          kv_lens.data_ptr<int>(), block_tables.data_ptr<int>(),             \
+          (float)scale, QH, KH, max_blocks, q.stride(0), (int)window,        \
+          sink_ptr);                                                         \
+  } while (0)
   if (D == 128) CTX_LAUNCH(128);
   else if (D == 256) CTX_LAUNCH(256);
   else CTX_LAUNCH(64);

@@ -37,18 +37,19 @@ namespace kaito {
 
 namespace {
 
-constexpr int BM = 64, BN = 64, BK = 64;
-constexpr int NW = 4;  // waves (2x2 over the 64x64 tile)
+constexpr int BM = 64, BK = 64;
+constexpr int NW = 4;  // waves (2x2 over the BM x BN tile)
 
+template <int ROWS = 64>
 KAITO_DEV void stage_tile_gathered(
     const short* __restrict__ src_base,  // element base (row stride given)
     int64_t row_stride,                  // elements per source row
-    const int* __restrict__ row_ids,     // BM source rows (-1 → row 0)
-    short* __restrict__ lds,             // [BM][BK] linear
+    const int* __restrict__ row_ids,     // ROWS source rows
+    short* __restrict__ lds,             // [ROWS][BK] linear
     int wave, int lane) {
-  // 64x64 bf16 tile = 8 KiB = 8 wave-segments of 64 lanes x 16 B.
+  // ROWSx64 bf16 tile = ROWS/8 wave-segments of 64 lanes x 16 B.
 #pragma unroll
-  for (int i = 0; i < 2; i++) {
+  for (int i = 0; i < ROWS / 32; i++) {
     const int seg = i * NW + wave;
     const int idx = seg * 64 + lane;
     const int row = idx >> 3;
@@ -63,12 +64,13 @@ KAITO_DEV void stage_tile_gathered(
   }
 }
 
+template <int ROWS = 64>
 KAITO_DEV void stage_tile_rows(
     const short* __restrict__ src,       // first row, element base
     int64_t row_stride,
     short* __restrict__ lds, int wave, int lane) {
 #pragma unroll
-  for (int i = 0; i < 2; i++) {
+  for (int i = 0; i < ROWS / 32; i++) {
     const int seg = i * NW + wave;
     const int idx = seg * 64 + lane;
     const int row = idx >> 3;
@@ -107,6 +109,7 @@ KAITO_DEV float moe_act_f(int mode, float g, float u) {
 
 // act[s, n] = silu(x[tok_s] @ Wg[e]^T) * (x[tok_s] @ Wu[e]^T)
 // grid: (max_m_tiles, n_local_experts, IE/BN); block 256.
+template <int BN>
 __global__ __launch_bounds__(256, 3)
 void moe_gate_silu_kernel(
     short* __restrict__ act,             // [TK, IE] bf16
@@ -127,13 +130,15 @@ void moe_gate_silu_kernel(
   const int lane = threadIdx.x & 63;
   const int wm = wave >> 1, wn = wave & 1;   // 2x2 wave grid
   const int lo = lane & 15, hi = lane >> 4;
+  constexpr int NR = BN / 32;   // 16-col frags per wave along N
 
-  // double-buffered staging (guide §5.5 T3 "minimum 2-phase": issue the
-  // NEXT K-tile's global_load_lds before computing the current one, ONE
-  // barrier per tile — the loads stay in flight across the MFMA phase)
-  __shared__ short lds_a[2][BM * BK];
-  __shared__ short lds_bg[2][BN * BK];
-  __shared__ short lds_bu[2][BN * BK];
+  // single-buffered 2-barrier staging: an explicit double-buffer
+  // measured +0% here (occupancy already hides the staging latency —
+  // these kernels run at ~5 TB/s weight streaming, tools/
+  // bench_moe_kernels.py) and the saved LDS buys 2x the resident blocks
+  __shared__ short lds_a[BM * BK];
+  __shared__ short lds_bg[BN * BK];
+  __shared__ short lds_bu[BN * BK];
   __shared__ int s_rows[BM];
   for (int i = threadIdx.x; i < BM; i += 256)
     s_rows[i] = sorted_ids[m0 + min(i, m_rem - 1)];
@@ -142,48 +147,42 @@ void moe_gate_silu_kernel(
   const short* wg = w + (int64_t)e * (2 * IE) * H + (int64_t)n0 * H;
   const short* wu = wg + (int64_t)IE * H;
 
-  f32x4 accg[2][2], accu[2][2];
+  f32x4 accg[2][NR], accu[2][NR];
 #pragma unroll
   for (int a = 0; a < 2; a++)
 #pragma unroll
-    for (int b = 0; b < 2; b++) {
+    for (int b = 0; b < NR; b++) {
       accg[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
       accu[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
     }
 
-  stage_tile_gathered(x, H, s_rows, lds_a[0], wave, lane);
-  stage_tile_rows(wg, H, lds_bg[0], wave, lane);
-  stage_tile_rows(wu, H, lds_bu[0], wave, lane);
-  __syncthreads();
-  int cur = 0;
   for (int k0 = 0; k0 < H; k0 += BK) {
-    if (k0 + BK < H) {   // issue next tile BEFORE computing this one
-      stage_tile_gathered(x + k0 + BK, H, s_rows, lds_a[cur ^ 1], wave,
-                          lane);
-      stage_tile_rows(wg + k0 + BK, H, lds_bg[cur ^ 1], wave, lane);
-      stage_tile_rows(wu + k0 + BK, H, lds_bu[cur ^ 1], wave, lane);
-    }
+    stage_tile_gathered<BM>(x + k0, H, s_rows, lds_a, wave, lane);
+    stage_tile_rows<BN>(wg + k0, H, lds_bg, wave, lane);
+    stage_tile_rows<BN>(wu + k0, H, lds_bu, wave, lane);
+    __syncthreads();
 #pragma unroll
     for (int kk = 0; kk < BK / 32; kk++) {
-      bf16x8 af[2], bg[2], bu[2];
+      bf16x8 af[2], bg[NR], bu[NR];
 #pragma unroll
-      for (int r = 0; r < 2; r++) {
-        af[r] = frag_read(lds_a[cur], wm * 32 + r * 16 + lo, kk, hi);
-        bg[r] = frag_read(lds_bg[cur], wn * 32 + r * 16 + lo, kk, hi);
-        bu[r] = frag_read(lds_bu[cur], wn * 32 + r * 16 + lo, kk, hi);
+      for (int r = 0; r < 2; r++)
+        af[r] = frag_read(lds_a, wm * 32 + r * 16 + lo, kk, hi);
+#pragma unroll
+      for (int b = 0; b < NR; b++) {
+        bg[b] = frag_read(lds_bg, wn * (BN / 2) + b * 16 + lo, kk, hi);
+        bu[b] = frag_read(lds_bu, wn * (BN / 2) + b * 16 + lo, kk, hi);
       }
 #pragma unroll
       for (int a = 0; a < 2; a++)
 #pragma unroll
-        for (int b = 0; b < 2; b++) {
+        for (int b = 0; b < NR; b++) {
           accg[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[a], bg[b], accg[a][b], 0, 0, 0);
           accu[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[a], bu[b], accu[a][b], 0, 0, 0);
         }
     }
-    __syncthreads();   // next tile landed AND this buffer fully read
-    cur ^= 1;
+    __syncthreads();
   }
 
   // epilogue: silu(g)*u → act[m0+row][n0+col] (rows are SORTED order,
@@ -196,8 +195,8 @@ void moe_gate_silu_kernel(
       const int row = row_base + r;
       if (row >= m_rem) continue;
 #pragma unroll
-      for (int b = 0; b < 2; b++) {
-        const int col = n0 + wn * 32 + b * 16 + lo;
+      for (int b = 0; b < NR; b++) {
+        const int col = n0 + wn * (BN / 2) + b * 16 + lo;
         float g = accg[a][b][r], u = accu[a][b][r];
         if (bias != nullptr) {
           g += bf16_to_f32(bias[(int64_t)e * 2 * IE + col]);
@@ -212,7 +211,8 @@ void moe_gate_silu_kernel(
 
 // out[tok_s, n] += gate_s * (act[s] @ Wd[e]^T); out is f32, atomics.
 // grid: (max_m_tiles, n_local_experts, H/BN); block 256.
-__global__ __launch_bounds__(256, 4)
+template <int BN>
+__global__ __launch_bounds__(256, 3)
 void moe_down_scatter_kernel(
     float* __restrict__ out,             // [T, H] f32 (pre-zeroed)
     const short* __restrict__ act,       // [TK, IE] bf16
@@ -232,45 +232,41 @@ void moe_down_scatter_kernel(
   const int lane = threadIdx.x & 63;
   const int wm = wave >> 1, wn = wave & 1;
   const int lo = lane & 15, hi = lane >> 4;
+  constexpr int NR = BN / 32;
 
-  __shared__ short lds_a[2][BM * BK];
-  __shared__ short lds_b[2][BN * BK];
+  __shared__ short lds_a[BM * BK];
+  __shared__ short lds_b[BN * BK];
 
   const short* wd = w + (int64_t)e * H * IE + (int64_t)n0 * IE;
 
-  f32x4 acc[2][2];
+  f32x4 acc[2][NR];
 #pragma unroll
   for (int a = 0; a < 2; a++)
 #pragma unroll
-    for (int b = 0; b < 2; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int b = 0; b < NR; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  stage_tile_rows(act + (int64_t)m0 * IE, IE, lds_a[0], wave, lane);
-  stage_tile_rows(wd, IE, lds_b[0], wave, lane);
-  __syncthreads();
-  int cur = 0;
   for (int k0 = 0; k0 < IE; k0 += BK) {
-    if (k0 + BK < IE) {
-      stage_tile_rows(act + (int64_t)m0 * IE + k0 + BK, IE,
-                      lds_a[cur ^ 1], wave, lane);
-      stage_tile_rows(wd + k0 + BK, IE, lds_b[cur ^ 1], wave, lane);
-    }
+    stage_tile_rows<BM>(act + (int64_t)m0 * IE + k0, IE, lds_a, wave,
+                        lane);
+    stage_tile_rows<BN>(wd + k0, IE, lds_b, wave, lane);
+    __syncthreads();
 #pragma unroll
     for (int kk = 0; kk < BK / 32; kk++) {
-      bf16x8 af[2], bf[2];
+      bf16x8 af[2], bf[NR];
 #pragma unroll
-      for (int r = 0; r < 2; r++) {
-        af[r] = frag_read(lds_a[cur], wm * 32 + r * 16 + lo, kk, hi);
-        bf[r] = frag_read(lds_b[cur], wn * 32 + r * 16 + lo, kk, hi);
-      }
+      for (int r = 0; r < 2; r++)
+        af[r] = frag_read(lds_a, wm * 32 + r * 16 + lo, kk, hi);
+#pragma unroll
+      for (int b = 0; b < NR; b++)
+        bf[b] = frag_read(lds_b, wn * (BN / 2) + b * 16 + lo, kk, hi);
 #pragma unroll
       for (int a = 0; a < 2; a++)
 #pragma unroll
-        for (int b = 0; b < 2; b++)
+        for (int b = 0; b < NR; b++)
           acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[a], bf[b], acc[a][b], 0, 0, 0);
     }
     __syncthreads();
-    cur ^= 1;
   }
 
   // NOTE: rows past m_end within this tile read garbage A rows — only
@@ -286,8 +282,8 @@ void moe_down_scatter_kernel(
       const int tok = sorted_ids[m0 + row];
       const float g = gates[m0 + row];
 #pragma unroll
-      for (int b = 0; b < 2; b++) {
-        const int col = n0 + wn * 32 + b * 16 + lo;
+      for (int b = 0; b < NR; b++) {
+        const int col = n0 + wn * (BN / 2) + b * 16 + lo;
         float v = acc[a][b][r];
         if (bias != nullptr) v += bf16_to_f32(bias[(int64_t)e * H + col]);
         atomicAdd(&out[(int64_t)tok * H + col], g * v);
@@ -311,17 +307,28 @@ void moe_gate_silu(at::Tensor act, at::Tensor x, at::Tensor w_gate_up,
   const int TK = act.size(0);
   const int IE = act.size(1);
   const int H = x.size(1);
-  TORCH_CHECK(H % BK == 0 && IE % BN == 0,
+  TORCH_CHECK(H % BK == 0 && IE % 64 == 0,
               "H/IE must be multiples of 64 for the fused MoE path");
   TORCH_CHECK(w_gate_up.size(1) == 2 * IE && w_gate_up.size(2) == H);
   const int mtiles = (TK + BM - 1) / BM;
   if (mtiles == 0 || n_local_experts == 0) return;
-  dim3 grid(mtiles, n_local_experts, IE / BN), block(256);
   auto stream = at::hip::getCurrentHIPStream();
-  moe_gate_silu_kernel<<<grid, block, 0, stream>>>(
-      (short*)act.data_ptr(), (const short*)x.data_ptr(),
-      (const short*)w_gate_up.data_ptr(), sorted_ids.data_ptr<int>(),
-      offsets.data_ptr<int>(), bias_ptr, (int)act_mode, (int)e_base, H, IE);
+  // BN=128 when it divides IE: halves the A-tile re-reads across
+  // n-tiles (the prefill-regime cost) at the same weight traffic
+  const int BN = IE % 128 == 0 ? 128 : 64;
+  dim3 grid(mtiles, n_local_experts, IE / BN), block(256);
+  if (BN == 128)
+    moe_gate_silu_kernel<128><<<grid, block, 0, stream>>>(
+        (short*)act.data_ptr(), (const short*)x.data_ptr(),
+        (const short*)w_gate_up.data_ptr(), sorted_ids.data_ptr<int>(),
+        offsets.data_ptr<int>(), bias_ptr, (int)act_mode, (int)e_base, H,
+        IE);
+  else
+    moe_gate_silu_kernel<64><<<grid, block, 0, stream>>>(
+        (short*)act.data_ptr(), (const short*)x.data_ptr(),
+        (const short*)w_gate_up.data_ptr(), sorted_ids.data_ptr<int>(),
+        offsets.data_ptr<int>(), bias_ptr, (int)act_mode, (int)e_base, H,
+        IE);
 }
 
 void moe_down_scatter(at::Tensor out, at::Tensor act, at::Tensor w_down,
@@ -340,17 +347,25 @@ void moe_down_scatter(at::Tensor out, at::Tensor act, at::Tensor w_down,
   const int TK = act.size(0);
   const int IE = act.size(1);
   const int H = out.size(1);
-  TORCH_CHECK(H % BN == 0 && IE % BK == 0);
+  TORCH_CHECK(H % 64 == 0 && IE % BK == 0);
   TORCH_CHECK(w_down.size(1) == H && w_down.size(2) == IE);
   const int mtiles = (TK + BM - 1) / BM;
   if (mtiles == 0 || n_local_experts == 0) return;
-  dim3 grid(mtiles, n_local_experts, H / BN), block(256);
   auto stream = at::hip::getCurrentHIPStream();
-  moe_down_scatter_kernel<<<grid, block, 0, stream>>>(
-      out.data_ptr<float>(), (const short*)act.data_ptr(),
-      (const short*)w_down.data_ptr(), sorted_ids.data_ptr<int>(),
-      gates.data_ptr<float>(), offsets.data_ptr<int>(), bias_ptr,
-      (int)e_base, H, IE);
+  const int BN = H % 128 == 0 ? 128 : 64;
+  dim3 grid(mtiles, n_local_experts, H / BN), block(256);
+  if (BN == 128)
+    moe_down_scatter_kernel<128><<<grid, block, 0, stream>>>(
+        out.data_ptr<float>(), (const short*)act.data_ptr(),
+        (const short*)w_down.data_ptr(), sorted_ids.data_ptr<int>(),
+        gates.data_ptr<float>(), offsets.data_ptr<int>(), bias_ptr,
+        (int)e_base, H, IE);
+  else
+    moe_down_scatter_kernel<64><<<grid, block, 0, stream>>>(
+        out.data_ptr<float>(), (const short*)act.data_ptr(),
+        (const short*)w_down.data_ptr(), sorted_ids.data_ptr<int>(),
+        gates.data_ptr<float>(), offsets.data_ptr<int>(), bias_ptr,
+        (int)e_base, H, IE);
 }
 
 }  // namespace kaito

@@ -44,13 +44,13 @@
 
 namespace kaito {
 
-template <int D, int G, int BS>
+template <int D, int G, int BS, bool FP8 = false>
 __global__ __launch_bounds__(256, G <= 4 ? 4 : 3)
 void paged_attention_sp_kernel(
     short* __restrict__ out,             // [T, QH, D] bf16
     const short* __restrict__ q,         // [T, QH, D] bf16
-    const short* __restrict__ k_cache,   // [B, KH, BS, D]
-    const short* __restrict__ v_cache,
+    const void* __restrict__ k_cache,    // [B, KH, BS, D] bf16 | e4m3
+    const void* __restrict__ v_cache,
     const int* __restrict__ block_tables,// [T, max_blocks]
     const int* __restrict__ seq_lens,    // [T]
     const float scale, const int KH, const int max_blocks,
@@ -107,12 +107,20 @@ void paged_attention_sp_kernel(
     // ---------------- phase A: stream K, raw scores to LDS -------------
     // Q is (re)loaded per super-chunk so its registers die before
     // phase C (Q and the accumulator never coexist). Lane dc holds dims
-    // dc*8..dc*8+8 — one contiguous 16B load per (head, lane).
+    // dc*8..dc*8+8 — one contiguous 16B load per (head, lane). The fp8
+    // cache path pre-converts Q to f32 (the dot runs on fmaf over the
+    // hardware fp8→f32 pair converts instead of v_dot2 bf16).
     short8_t qreg[G];
+    float qf[FP8 ? G : 1][FP8 ? 8 : 1];
 #pragma unroll
-    for (int g = 0; g < G; g++)
+    for (int g = 0; g < G; g++) {
       qreg[g] = *reinterpret_cast<const short8_t*>(
           q + (int64_t)seq * q_stride + (kvh * G + g) * D + dc * 8);
+      if constexpr (FP8) {
+#pragma unroll
+        for (int j = 0; j < 8; j++) qf[g][j] = bf16_to_f32(qreg[g][j]);
+      }
+    }
     float mloc[G];
 #pragma unroll
     for (int g = 0; g < G; g++) mloc[g] = -1e30f;
@@ -121,11 +129,17 @@ void paged_attention_sp_kernel(
       const int blk = bt[min(c0 / CPB, last_blk)];
       // token u*TPC+tg of the block; every intra-block offset is valid
       const int64_t gb = (((int64_t)blk * KH + kvh) * BS + tg) * D + dc * 8;
-      short8_t kd[CPB];
+      short8_t kd[FP8 ? 1 : CPB];
+      u8x8_t kd8[FP8 ? CPB : 1];
 #pragma unroll
-      for (int u = 0; u < CPB; u++)
-        kd[u] = *reinterpret_cast<const short8_t*>(
-            k_cache + gb + u * (TPC * D));
+      for (int u = 0; u < CPB; u++) {
+        if constexpr (FP8)
+          kd8[u] = reinterpret_cast<const u8x8_t*>(k_cache)[
+              (gb + u * (TPC * D)) / 8];
+        else
+          kd[u] = *reinterpret_cast<const short8_t*>(
+              (const short*)k_cache + gb + u * (TPC * D));
+      }
       const int nu = min(CPB, n_ii - ii0);
       // compile-time trip count: a runtime-bounded loop would make
       // kd[u] a runtime-indexed register array → scratch (rule #20)
@@ -135,14 +149,21 @@ void paged_attention_sp_kernel(
         const int tok = (c0 + u) * TPC + tg;
         const bool valid = tok >= start && tok < seq_len;
         float s[G];
+        float kf[FP8 ? 8 : 1];
+        if constexpr (FP8) fp8x8_to_f32(kd8[u], kf);
 #pragma unroll
         for (int g = 0; g < G; g++) {
           float p = 0.f;
-          const bf16x2_t* kp = reinterpret_cast<const bf16x2_t*>(&kd[u]);
-          const bf16x2_t* qp = reinterpret_cast<const bf16x2_t*>(&qreg[g]);
+          if constexpr (FP8) {
 #pragma unroll
-          for (int jj = 0; jj < 4; jj++)
-            p = __builtin_amdgcn_fdot2_f32_bf16(qp[jj], kp[jj], p, false);
+            for (int j = 0; j < 8; j++) p = fmaf(qf[g][j], kf[j], p);
+          } else {
+            const bf16x2_t* kp = reinterpret_cast<const bf16x2_t*>(&kd[u]);
+            const bf16x2_t* qp = reinterpret_cast<const bf16x2_t*>(&qreg[g]);
+#pragma unroll
+            for (int jj = 0; jj < 4; jj++)
+              p = __builtin_amdgcn_fdot2_f32_bf16(qp[jj], kp[jj], p, false);
+          }
           p = group_reduce_sum<DC>(p) * scale;   // dot over DC dim-lanes
           s[g] = valid ? p : -1e30f;
           mloc[g] = fmaxf(mloc[g], s[g]);
@@ -194,11 +215,17 @@ void paged_attention_sp_kernel(
       const int c0 = span0 + s0 + ii0;
       const int blk = bt[min(c0 / CPB, last_blk)];
       const int64_t gb = (((int64_t)blk * KH + kvh) * BS + tg) * D + dc * 8;
-      short8_t vd[CPB];
+      short8_t vd[FP8 ? 1 : CPB];
+      u8x8_t vd8[FP8 ? CPB : 1];
 #pragma unroll
-      for (int u = 0; u < CPB; u++)
-        vd[u] = *reinterpret_cast<const short8_t*>(
-            v_cache + gb + u * (TPC * D));
+      for (int u = 0; u < CPB; u++) {
+        if constexpr (FP8)
+          vd8[u] = reinterpret_cast<const u8x8_t*>(v_cache)[
+              (gb + u * (TPC * D)) / 8];
+        else
+          vd[u] = *reinterpret_cast<const short8_t*>(
+              (const short*)v_cache + gb + u * (TPC * D));
+      }
       const int nu = min(CPB, n_ii - ii0);
 #pragma unroll
       for (int u = 0; u < CPB; u++) {
@@ -224,8 +251,12 @@ void paged_attention_sp_kernel(
           for (int g = 0; g < G; g++) pg[g] = g < 4 ? pv0[g] : pv1[g - 4];
         }
         float vreg[8];
+        if constexpr (FP8) {
+          fp8x8_to_f32(vd8[u], vreg);
+        } else {
 #pragma unroll
-        for (int j = 0; j < 8; j++) vreg[j] = bf16_to_f32(vd[u][j]);
+          for (int j = 0; j < 8; j++) vreg[j] = bf16_to_f32(vd[u][j]);
+        }
         const float2_t* v2 = reinterpret_cast<const float2_t*>(vreg);
 #pragma unroll
         for (int g = 0; g < G; g++) {
@@ -285,12 +316,22 @@ void paged_attention_sp_kernel(
 }
 
 #define PA_SP_LAUNCH(D_, G_)                                                   \
-  hipLaunchKernelGGL((paged_attention_sp_kernel<D_, G_, 16>), dim3(T, KH),     \
-      dim3(256), 0, stream, (short*)out.data_ptr(),                           \
-      (const short*)query.data_ptr(), (const short*)k_cache.data_ptr(),       \
-      (const short*)v_cache.data_ptr(), block_tables.data_ptr<int>(),         \
-      seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks,                 \
-      query.stride(0), (int)window, sink_ptr)
+  do {                                                                        \
+    if (fp8)                                                                  \
+      hipLaunchKernelGGL((paged_attention_sp_kernel<D_, G_, 16, true>),        \
+          dim3(T, KH), dim3(256), 0, stream, (short*)out.data_ptr(),          \
+          (const short*)query.data_ptr(), k_cache.data_ptr(),                 \
+          v_cache.data_ptr(), block_tables.data_ptr<int>(),                   \
+          seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks,             \
+          query.stride(0), (int)window, sink_ptr);                            \
+    else                                                                      \
+      hipLaunchKernelGGL((paged_attention_sp_kernel<D_, G_, 16, false>),       \
+          dim3(T, KH), dim3(256), 0, stream, (short*)out.data_ptr(),          \
+          (const short*)query.data_ptr(), k_cache.data_ptr(),                 \
+          v_cache.data_ptr(), block_tables.data_ptr<int>(),                   \
+          seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks,             \
+          query.stride(0), (int)window, sink_ptr);                            \
+  } while (0)
 
 void paged_attention_sp(at::Tensor out, at::Tensor query, at::Tensor k_cache,
                         at::Tensor v_cache, at::Tensor block_tables,
@@ -307,6 +348,7 @@ void paged_attention_sp(at::Tensor out, at::Tensor query, at::Tensor k_cache,
   TORCH_CHECK(query.stride(-1) == 1 && query.stride(1) == query.size(2),
               "query must be [T, QH, D] with contiguous heads");
   TORCH_CHECK(block_tables.dtype() == at::kInt && seq_lens.dtype() == at::kInt);
+  const bool fp8 = k_cache.dtype() == at::kByte;   // OCP e4m3 KV cache
   const int T = query.size(0);
   const int QH = query.size(1);
   const int D = query.size(2);

@@ -134,11 +134,13 @@ def main(argv=None):
         enable_lora=args.enable_lora,
         kv_offload=args.swap_space > 0,
         kv_offload_bytes=int(args.swap_space * (1 << 30)) or None,
+        kv_cache_dtype=("fp8" if args.kv_cache_dtype in
+                        ("fp8", "fp8_e4m3") else "auto"),
     )
-    if args.kv_cache_dtype not in ("auto", "bf16", "bfloat16"):
+    if args.kv_cache_dtype not in ("auto", "bf16", "bfloat16", "fp8",
+                                   "fp8_e4m3"):
         raise SystemExit(f"kv-cache-dtype {args.kv_cache_dtype!r} not "
-                         "supported in this build (bf16 KV only; fp8 KV is "
-                         "on the roadmap)")
+                         "supported (auto/bf16/fp8)")
     metrics.MODEL_DOWNLOAD_PROGRESS.set(0.0)
     engine = LLMEngine(cfg, weights_path=args.weights_path)
     if args.enable_kv_events:

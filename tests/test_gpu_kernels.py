@@ -266,6 +266,53 @@ def test_paged_attention_window_sinks(window, use_sinks, G, D):
     _close(out, ref.to(DEV), atol=2e-2)
 
 
+def test_fp8_kv_cache_decode_matches_dequant_ref():
+    """fp8 (e4m3) KV cache: the decode kernel must match the fp32
+    reference computed on the DEQUANTIZED cache (the only error source
+    is the fp8 storage quantization itself)."""
+    torch.manual_seed(21)
+    ops.load_extension()
+    KH, G, D, BS = 2, 4, 128, 16
+    QH = KH * G
+    lens = [7, 130, 400]
+    T = len(lens)
+    mb = (max(lens) + BS - 1) // BS
+    NB = T * mb + 1
+    kc_bf = _bf16(NB, KH, BS, D, scale=0.5)
+    vc_bf = _bf16(NB, KH, BS, D, scale=0.5)
+    kc8 = kc_bf.float().to(torch.float8_e4m3fn).view(torch.uint8)
+    vc8 = vc_bf.float().to(torch.float8_e4m3fn).view(torch.uint8)
+    bt = (torch.randperm(NB - 1)[: T * mb].reshape(T, mb) + 1).int().to(DEV)
+    q = _bf16(T, QH, D)
+    sl = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention(q, kc8, vc8, bt, sl, scale)
+    deq_k = kc8.view(torch.float8_e4m3fn).float().cpu()
+    deq_v = vc8.view(torch.float8_e4m3fn).float().cpu()
+    ref = R.paged_attention(q.cpu().float(), deq_k, deq_v, bt.cpu(),
+                            sl.cpu(), scale)
+    _close(out, ref.to(DEV), atol=3e-2, rtol=3e-2)
+
+
+def test_fp8_kv_cache_engine_end_to_end():
+    """Engine with kv_cache_dtype=fp8: reshape_and_cache writes e4m3,
+    decode + chunked context prefill read it; greedy outputs must
+    mostly agree with the bf16-cache engine (fp8 quantization noise can
+    flip late tokens)."""
+    from kaito_amd.engine import SamplingParams
+    prompts = [list(range(10, 40)), [5, 6, 7], list(range(90, 140))]
+    sp = SamplingParams(max_tokens=8, ignore_eos=True)
+    ref_eng = _gpu_engine(enforce_eager=True, seed=4)
+    ref = ref_eng.generate(prompts, sp)
+    fp8_eng = _gpu_engine(enforce_eager=True, seed=4, kv_cache_dtype="fp8")
+    assert fp8_eng.runner.kv_caches[0][0].dtype == torch.uint8
+    got = fp8_eng.generate(prompts, sp)
+    for a, b in zip(ref, got):
+        match = sum(x == y for x, y in zip(a.output_token_ids,
+                                           b.output_token_ids))
+        assert match >= 6, (a.output_token_ids, b.output_token_ids)
+
+
 @pytest.mark.parametrize("window,use_sinks", [(32, False), (0, True),
                                               (64, True)])
 def test_prefill_attention_window_sinks(window, use_sinks):

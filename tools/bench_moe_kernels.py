@@ -4,8 +4,10 @@ import torch
 from kaito_amd import ops
 ops.load_extension()
 dev = "cuda"
-# mixtral-8x7b decode shape: T=256 tokens, top2 -> TK=512, E=8
-T, K_TOP, E, H, IE = 256, 2, 8, 4096, 14336
+import sys as _s
+# mixtral-8x7b shapes: decode T=256 (TK=512) or prefill T=8192
+T = int(_s.argv[1]) if len(_s.argv) > 1 else 256
+K_TOP, E, H, IE = 2, 8, 4096, 14336
 TK = T * K_TOP
 torch.manual_seed(0)
 x = (torch.randn(T, H, device=dev) * 0.1).to(torch.bfloat16)
