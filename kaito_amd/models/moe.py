@@ -100,10 +100,19 @@ class MoEMLP(nn.Module):
         torch.cumsum(counts, 0, out=offsets[1:])
         return sorted_tok, gates, offsets
 
+    # fused kernels win at decode-sized batches (weight-streaming
+    # bound, graph-capturable); per-expert hipBLASLt GEMMs win at
+    # prefill sizes (compute-bound, Tensile ~2x our 2-barrier tile —
+    # tools/bench_moe_kernels.py: TK=16384 fused 5.8ms vs loop 3.0ms).
+    # Prefill steps run eagerly, so the loop's host-side segment reads
+    # are harmless there.
+    FUSED_MAX_TOKENS = 4096
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, H = x.shape
         sorted_tok, gates, offsets = self._route(x)
-        if x.is_cuda and H % 64 == 0 and self.ie_local % 64 == 0:
+        if x.is_cuda and H % 64 == 0 and self.ie_local % 64 == 0 and \
+                T * self.top_k <= self.FUSED_MAX_TOKENS:
             out = self._forward_fused(x, sorted_tok, gates, offsets)
         else:
             out = self._forward_loop(x, sorted_tok, gates, offsets)
