@@ -236,6 +236,7 @@ def main():
                 "seq_len": args.in_tokens + args.out_tokens,
                 "kv_blocks": eng.runner.num_gpu_blocks,
                 "decode_graphs": not args.eager,
+                "kv_cache_dtype": args.kv_dtype,
                 "ramp_steps": ramp_steps,
                 "ramp_s": round(ramp_s, 2),
                 "decoding_at_t0": len(eng.scheduler.running),
