@@ -283,6 +283,89 @@ def test_pp_engine_world2():
     _spawn("_body_pp_engine", port=29617)
 
 
+def _body_tp_pp_engine(rank, world):
+    """Combined tier-3 topology at world=4 (tp=2 x pp=2): the engine must
+    reproduce the single-rank greedy output exactly — TP all-reduces
+    inside each stage, activations over PP send/recv between stages,
+    sampled tokens broadcast from the last stage."""
+    import torch
+    from kaito_amd.engine import EngineConfig, LLMEngine, SamplingParams
+    from kaito_amd.models.llama import AttnMetadata, LlamaForCausalLM
+    from kaito_amd.parallel import state as ps
+
+    ps.init_parallel(tp_size=2, pp_size=2, backend="gloo")
+    st = ps.get_state()
+    assert st.tp_size == 2 and st.pp_size == 2
+    tp_rank = st.tp_rank
+
+    cfg_model = get_model_config("tiny-llama-test")
+    saved = (st.tp_size, st.tp_rank, st.pp_size, st.pp_rank)
+    st.tp_size, st.tp_rank, st.pp_size, st.pp_rank = 1, 0, 1, 0
+    ref = LlamaForCausalLM(cfg_model).random_init(13)
+    ref.init_rope("cpu")
+    st.tp_size, st.tp_rank, st.pp_size, st.pp_rank = saved
+
+    cfg = EngineConfig(model=cfg_model, device="cpu", max_num_seqs=4,
+                       num_gpu_blocks=64, enforce_eager=True,
+                       max_model_len=96, tensor_parallel_size=2,
+                       enable_one_shot_allreduce=False)
+    eng = LLMEngine(cfg)
+    model = eng.runner.model
+    rp = dict(ref.named_parameters())
+    qh = cfg_model.num_heads * cfg_model.head_dim
+    kvh = cfg_model.num_kv_heads * cfg_model.head_dim
+    ii = cfg_model.intermediate_size
+    with torch.no_grad():
+        for name, p in model.named_parameters():
+            if name.startswith("layers."):
+                idx = int(name.split(".")[1]) + model.layer_start
+                src = rp["layers.%d.%s" % (idx, name.split(".", 2)[2])]
+            else:
+                src = rp[name]
+            if p.shape == src.shape:
+                p.copy_(src)
+                continue
+            if "qkv_proj" in name:
+                q, k, v = src.split([qh, kvh, kvh], dim=0)
+                p.copy_(torch.cat([q.chunk(2, 0)[tp_rank],
+                                   k.chunk(2, 0)[tp_rank],
+                                   v.chunk(2, 0)[tp_rank]], 0))
+            elif "gate_up" in name:
+                g, u = src.chunk(2, dim=0)
+                p.copy_(torch.cat([g.chunk(2, 0)[tp_rank],
+                                   u.chunk(2, 0)[tp_rank]], 0))
+            elif any(t in name for t in ("lm_head", "embed_tokens")):
+                p.copy_(src.chunk(2, 0)[tp_rank])
+            elif "o_proj" in name or "down_proj" in name:
+                p.copy_(src.chunk(2, 1)[tp_rank])
+            else:
+                raise AssertionError(f"unhandled {name}")
+
+    prompt = [3, 14, 15, 92, 65, 35, 89]
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+    got = eng.generate([prompt], sp)[0].output_token_ids
+
+    if rank == 0:
+        toks = list(prompt)
+        for _ in range(6):
+            T = len(toks)
+            meta = AttnMetadata(
+                is_prefill=True,
+                slot_mapping=torch.full((T,), -1, dtype=torch.long),
+                cu_seqlens=torch.tensor([0, T], dtype=torch.int32),
+                max_seqlen=T)
+            st.tp_size, st.tp_rank, st.pp_size, st.pp_rank = 1, 0, 1, 0
+            h = ref(torch.tensor(toks), torch.arange(T), None, meta)
+            nxt = int(ref.compute_logits(h[-1:]).argmax(-1))
+            st.tp_size, st.tp_rank, st.pp_size, st.pp_rank = saved
+            toks.append(nxt)
+        assert got == toks[len(prompt):], (got, toks[len(prompt):])
+
+
+def test_tp_pp_engine_world4():
+    _spawn("_body_tp_pp_engine", world=4, port=29641)
+
+
 def _body_quant_tp(rank, world):
     """TP-sharded W4A16: column and row quantized linears must reproduce
     the full (unsharded) dequantized matmul — exercises the packed-N row
