@@ -32,7 +32,10 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=250)
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--tp", type=int, default=1)
-    p.add_argument("--max-num-seqs", type=int, default=1024)
+    # 2048 concurrent seqs: measured 50,986 tok/s vs 33,541 at 1024
+    # (Llama-3-8B bf16; 288 GB HBM3E holds the KV with half the card to
+    # spare — profiles/r02_perf_notes.md)
+    p.add_argument("--max-num-seqs", type=int, default=2048)
     p.add_argument("--in-tokens", type=int, default=200)
     p.add_argument("--out-tokens", type=int, default=200)
     p.add_argument("--eager", action="store_true")
