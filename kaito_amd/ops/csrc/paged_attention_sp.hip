@@ -38,14 +38,24 @@
 // (same as paged_attention.hip; replaces the vLLM decode attention the
 // reference delegates to — SURVEY.md §2.3).
 #include "common.h"
+#include <cstdlib>
 #include <torch/library.h>
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
 
 namespace kaito {
 
-template <int D, int G, int BS, bool FP8 = false>
-__global__ __launch_bounds__(256, G <= 4 ? 4 : 3)
+// OCC = target workgroups/CU (4 ⇒ 4 waves/SIMD at ≤128 VGPRs).
+// Measured (tools/bench_decode_attn.py --sweep-occ): OCC=3 == OCC=4
+// exactly and OCC=5 spills — the kernel is NOT wave-count-limited, so
+// at OCC=3 there are ~170 VGPRs of free headroom. UB (blocks per
+// streaming iteration) spends that headroom on memory-level
+// parallelism: UB=2 keeps 8 independent 16 B loads in flight per lane
+// (two KV blocks) instead of 4. Variants selectable at runtime via
+// KAITO_PA_SP_OCC / KAITO_PA_SP_UB for A/B on real hardware.
+template <int D, int G, int BS, bool FP8 = false, int OCC = (G <= 4 ? 4 : 3),
+          int UB = 1>
+__global__ __launch_bounds__(256, OCC)
 void paged_attention_sp_kernel(
     short* __restrict__ out,             // [T, QH, D] bf16
     const short* __restrict__ q,         // [T, QH, D] bf16
@@ -124,27 +134,31 @@ void paged_attention_sp_kernel(
     float mloc[G];
 #pragma unroll
     for (int g = 0; g < G; g++) mloc[g] = -1e30f;
-    for (int ii0 = 0; ii0 < n_ii; ii0 += CPB) {
+    for (int ii0 = 0; ii0 < n_ii; ii0 += UB * CPB) {
       const int c0 = span0 + s0 + ii0;           // multiple of CPB
-      const int blk = bt[min(c0 / CPB, last_blk)];
-      // token u*TPC+tg of the block; every intra-block offset is valid
-      const int64_t gb = (((int64_t)blk * KH + kvh) * BS + tg) * D + dc * 8;
-      short8_t kd[FP8 ? 1 : CPB];
-      u8x8_t kd8[FP8 ? CPB : 1];
+      // UB block bases; clamped indices keep every offset valid
+      int64_t gb[UB];
 #pragma unroll
-      for (int u = 0; u < CPB; u++) {
+      for (int b = 0; b < UB; b++) {
+        const int blk = bt[min(c0 / CPB + b, last_blk)];
+        gb[b] = (((int64_t)blk * KH + kvh) * BS + tg) * D + dc * 8;
+      }
+      short8_t kd[FP8 ? 1 : UB * CPB];
+      u8x8_t kd8[FP8 ? UB * CPB : 1];
+#pragma unroll
+      for (int u = 0; u < UB * CPB; u++) {
         if constexpr (FP8)
           kd8[u] = reinterpret_cast<const u8x8_t*>(k_cache)[
-              (gb + u * (TPC * D)) / 8];
+              (gb[u / CPB] + (u % CPB) * (TPC * D)) / 8];
         else
           kd[u] = *reinterpret_cast<const short8_t*>(
-              (const short*)k_cache + gb + u * (TPC * D));
+              (const short*)k_cache + gb[u / CPB] + (u % CPB) * (TPC * D));
       }
-      const int nu = min(CPB, n_ii - ii0);
+      const int nu = min(UB * CPB, n_ii - ii0);
       // compile-time trip count: a runtime-bounded loop would make
       // kd[u] a runtime-indexed register array → scratch (rule #20)
 #pragma unroll
-      for (int u = 0; u < CPB; u++) {
+      for (int u = 0; u < UB * CPB; u++) {
         if (u >= nu) break;
         const int tok = (c0 + u) * TPC + tg;
         const bool valid = tok >= start && tok < seq_len;
@@ -211,24 +225,28 @@ void paged_attention_sp_kernel(
     }
 
     // ---------------- phase C: stream V, pure FMA accumulate -----------
-    for (int ii0 = 0; ii0 < n_ii; ii0 += CPB) {
+    for (int ii0 = 0; ii0 < n_ii; ii0 += UB * CPB) {
       const int c0 = span0 + s0 + ii0;
-      const int blk = bt[min(c0 / CPB, last_blk)];
-      const int64_t gb = (((int64_t)blk * KH + kvh) * BS + tg) * D + dc * 8;
-      short8_t vd[FP8 ? 1 : CPB];
-      u8x8_t vd8[FP8 ? CPB : 1];
+      int64_t gb[UB];
 #pragma unroll
-      for (int u = 0; u < CPB; u++) {
+      for (int b = 0; b < UB; b++) {
+        const int blk = bt[min(c0 / CPB + b, last_blk)];
+        gb[b] = (((int64_t)blk * KH + kvh) * BS + tg) * D + dc * 8;
+      }
+      short8_t vd[FP8 ? 1 : UB * CPB];
+      u8x8_t vd8[FP8 ? UB * CPB : 1];
+#pragma unroll
+      for (int u = 0; u < UB * CPB; u++) {
         if constexpr (FP8)
           vd8[u] = reinterpret_cast<const u8x8_t*>(v_cache)[
-              (gb + u * (TPC * D)) / 8];
+              (gb[u / CPB] + (u % CPB) * (TPC * D)) / 8];
         else
           vd[u] = *reinterpret_cast<const short8_t*>(
-              (const short*)v_cache + gb + u * (TPC * D));
+              (const short*)v_cache + gb[u / CPB] + (u % CPB) * (TPC * D));
       }
-      const int nu = min(CPB, n_ii - ii0);
+      const int nu = min(UB * CPB, n_ii - ii0);
 #pragma unroll
-      for (int u = 0; u < CPB; u++) {
+      for (int u = 0; u < UB * CPB; u++) {
         if (u >= nu) break;
         // G probabilities for this token: one vector LDS broadcast
         float pg[G];
@@ -315,22 +333,49 @@ void paged_attention_sp_kernel(
   }
 }
 
+// KAITO_PA_SP_OCC=3|4|5 selects the occupancy variant (default: the
+// template default, i.e. 4 for G<=4). Only instantiated for bf16
+// caches; the fp8 path always uses the default.
+static int pa_sp_occ_env() {
+  static int occ = []() {
+    const char* e = getenv("KAITO_PA_SP_OCC");
+    return e ? atoi(e) : 0;
+  }();
+  return occ;
+}
+
+static int pa_sp_ub_env() {
+  static int ub = []() {
+    const char* e = getenv("KAITO_PA_SP_UB");
+    return e ? atoi(e) : 0;
+  }();
+  return ub;
+}
+
+#define PA_SP_ARGS                                                            \
+  dim3(T, KH), dim3(256), 0, stream, (short*)out.data_ptr(),                  \
+      (const short*)query.data_ptr(), k_cache.data_ptr(),                     \
+      v_cache.data_ptr(), block_tables.data_ptr<int>(),                       \
+      seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks,                 \
+      query.stride(0), (int)window, sink_ptr
+
 #define PA_SP_LAUNCH(D_, G_)                                                   \
   do {                                                                        \
     if (fp8)                                                                  \
       hipLaunchKernelGGL((paged_attention_sp_kernel<D_, G_, 16, true>),        \
-          dim3(T, KH), dim3(256), 0, stream, (short*)out.data_ptr(),          \
-          (const short*)query.data_ptr(), k_cache.data_ptr(),                 \
-          v_cache.data_ptr(), block_tables.data_ptr<int>(),                   \
-          seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks,             \
-          query.stride(0), (int)window, sink_ptr);                            \
+          PA_SP_ARGS);                                                        \
+    else if (G_ <= 4 && pa_sp_ub_env() == 2)                                  \
+      hipLaunchKernelGGL((paged_attention_sp_kernel<D_, G_, 16, false, 3, 2>), \
+          PA_SP_ARGS);                                                        \
+    else if (G_ <= 4 && pa_sp_occ_env() == 5)                                 \
+      hipLaunchKernelGGL((paged_attention_sp_kernel<D_, G_, 16, false, 5>),    \
+          PA_SP_ARGS);                                                        \
+    else if (G_ <= 4 && pa_sp_occ_env() == 3)                                 \
+      hipLaunchKernelGGL((paged_attention_sp_kernel<D_, G_, 16, false, 3>),    \
+          PA_SP_ARGS);                                                        \
     else                                                                      \
       hipLaunchKernelGGL((paged_attention_sp_kernel<D_, G_, 16, false>),       \
-          dim3(T, KH), dim3(256), 0, stream, (short*)out.data_ptr(),          \
-          (const short*)query.data_ptr(), k_cache.data_ptr(),                 \
-          v_cache.data_ptr(), block_tables.data_ptr<int>(),                   \
-          seq_lens.data_ptr<int>(), (float)scale, KH, max_blocks,             \
-          query.stride(0), (int)window, sink_ptr);                            \
+          PA_SP_ARGS);                                                        \
   } while (0)
 
 void paged_attention_sp(at::Tensor out, at::Tensor query, at::Tensor k_cache,
