@@ -79,6 +79,34 @@ class ModelConfig:
     # the D∈{64,128,256} kernels but keeps 80^-0.5 scaling; zero-padded
     # dims contribute nothing, so the math is exact)
     attn_scale: Optional[float] = None
+    # ---- MLA (multi-head latent attention — DeepSeek V2/V3/R1) ----------
+    # kv_lora_rank > 0 enables MLA: the paged cache stores ONE compressed
+    # row per token (c_kv[kv_lora_rank] ‖ k_rope[qk_rope_head_dim]) shared
+    # by every head; decode runs in the absorbed space (q_nope·W_uk
+    # folded into q) over that latent cache (ops/csrc/mla_attention.hip)
+    kv_lora_rank: int = 0
+    q_lora_rank: int = 0                 # 0 = direct q projection (V2-Lite)
+    qk_nope_head_dim: int = 128
+    qk_rope_head_dim: int = 64
+    v_head_dim: int = 128
+    # ---- DeepSeek MoE extensions ----------------------------------------
+    # routing adds "noaux_tc" (V3: sigmoid scores + e_score_correction_bias
+    # for selection, group-limited top-k, ORIGINAL sigmoid scores as
+    # weights) to softmax_topk/topk_softmax
+    moe_norm_topk: bool = True           # renormalize selected weights
+    routed_scaling_factor: float = 1.0   # V3: 2.5
+    n_group: int = 1                     # expert groups (V3: 8)
+    topk_group: int = 1                  # groups kept per token (V3: 4)
+    n_shared_experts: int = 0            # always-on experts (width = n*moe_ie)
+    first_k_dense: int = 0               # leading dense (non-MoE) layers
+    # ---- YaRN rope scaling (DeepSeek long-context) ----------------------
+    rope_scaling_type: str = ""          # "" | "yarn"
+    rope_factor: float = 1.0
+    rope_orig_max_position: int = 0      # original_max_position_embeddings
+    rope_beta_fast: float = 32.0
+    rope_beta_slow: float = 1.0
+    rope_mscale: float = 1.0
+    rope_mscale_all_dim: float = 0.0
     # weight quantization: "" (bf16) | "w4a16"/"awq" (4-bit group-quantized
     # linears via the HIP GEMV / dequant+MFMA kernels)
     quant_method: str = ""
@@ -121,6 +149,8 @@ class ModelConfig:
 
     @property
     def rotary_dim(self) -> int:
+        if self.is_mla:          # rope applies to the decoupled rope dims
+            return self.qk_rope_head_dim
         r = int(self.head_dim * self.partial_rotary_factor)
         return r - (r % 2)
 
@@ -138,8 +168,22 @@ class ModelConfig:
             return 0 if (layer_idx + 1) % n == 0 else self.sliding_window
         raise ValueError(self.sliding_window_pattern)
 
+    @property
+    def is_mla(self) -> bool:
+        return self.kv_lora_rank > 0
+
+    @property
+    def kv_cache_row(self) -> int:
+        """Per-token latent cache width for MLA models."""
+        return self.kv_lora_rank + self.qk_rope_head_dim
+
     def kv_bytes_per_token(self, tp_size: int = 1) -> int:
         """Per-token KV cache bytes across all layers (per TP rank)."""
+        if self.is_mla:
+            # one compressed latent row shared by all heads, REPLICATED
+            # across TP ranks (MQA-like; the xGMI win is skipping the
+            # per-head cache entirely)
+            return self.num_layers * self.kv_cache_row * 2
         kvh = max(self.num_kv_heads // tp_size, 1)
         return 2 * self.num_layers * kvh * self.head_dim * 2  # k+v, bf16
 

@@ -87,9 +87,17 @@ class ModelRunner:
         fp8 = cfg.kv_cache_dtype == "fp8"
         if fp8 and not self.is_gpu:
             raise ValueError("kv_cache_dtype=fp8 needs a GPU")
+        if fp8 and m.is_mla:
+            raise ValueError("fp8 KV cache not supported for MLA models")
         elem = 1 if fp8 else 2
-        block_bytes = 2 * m.num_layers * kvh * cfg.block_size * m.head_dim \
-            * elem
+        if m.is_mla:
+            # ONE compressed latent row per token (c_kv ‖ k_rope), shared
+            # by all heads and replicated across TP ranks (models/mla.py)
+            block_bytes = m.num_layers * cfg.block_size * m.kv_cache_row \
+                * elem
+        else:
+            block_bytes = 2 * m.num_layers * kvh * cfg.block_size \
+                * m.head_dim * elem
         if cfg.num_gpu_blocks is not None:
             n_blocks = cfg.num_gpu_blocks
         elif self.is_gpu:
@@ -106,6 +114,12 @@ class ModelRunner:
         kvs = []
         kv_dtype = torch.uint8 if fp8 else m.dtype
         for _ in range(self.model.num_local_layers):
+            if m.is_mla:
+                # aliased pair: the latent cache IS both k and v
+                c = torch.zeros(n_blocks, cfg.block_size, m.kv_cache_row,
+                                dtype=kv_dtype, device=self.device)
+                kvs.append((c, c))
+                continue
             k = torch.zeros(n_blocks, kvh, cfg.block_size, m.head_dim,
                             dtype=kv_dtype, device=self.device)
             v = torch.zeros_like(k)

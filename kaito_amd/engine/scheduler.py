@@ -65,6 +65,11 @@ class Scheduler:
         self.running: List[Sequence] = []
         # called at admission: restore_cb(seq) -> covered prefix tokens
         self.restore_cb = restore_cb
+        # MLA models prefill prompts WHOLE (no context-attention suffix
+        # path over the latent cache): an admission that exceeds the
+        # step's remaining token budget waits for the next step instead
+        # of being chunked (models/mla.py)
+        self.whole_prompt_only = cfg.model.is_mla
 
     # ---- queue state (serves the rate limiter / metrics) ----
     @property
@@ -153,7 +158,15 @@ class Scheduler:
                 self.running.append(seq)
                 room -= 1
                 continue
-            n = min(seq.num_context_tokens - seq.prefilled_len, budget)
+            remaining = seq.num_context_tokens - seq.prefilled_len
+            if self.whole_prompt_only and remaining > budget:
+                # put it back; it needs a step with a larger free budget
+                self.waiting.appendleft(seq)
+                self.pool.free(seq.block_table)
+                seq.block_table = []
+                seq.status = SeqStatus.WAITING
+                break
+            n = min(remaining, budget)
             chunks.append(PrefillChunk(seq, seq.prefilled_len, n))
             self.prefilling.append(seq)
             budget -= n

@@ -43,15 +43,40 @@ class AttnMetadata:
 
 def build_cos_sin_cache(cfg: ModelConfig, device, max_pos: Optional[int] = None,
                         theta: Optional[float] = None) -> torch.Tensor:
-    """[max_pos, rot_dim] f32 = [cos | sin] table."""
+    """[max_pos, rot_dim] f32 = [cos | sin] table. Supports YaRN scaling
+    (DeepSeek long-context: NTK-by-parts frequency interpolation with the
+    beta_fast/beta_slow ramp + mscale on cos/sin)."""
     rot = cfg.rotary_dim
     max_pos = max_pos or cfg.max_position
     theta = theta or cfg.rope_theta
-    inv = 1.0 / (theta ** (
-        torch.arange(0, rot, 2, dtype=torch.float64, device=device) / rot))
+    idx = torch.arange(0, rot, 2, dtype=torch.float64, device=device)
+    inv = 1.0 / (theta ** (idx / rot))
+    mscale = 1.0
+    if cfg.rope_scaling_type == "yarn" and cfg.rope_factor > 1.0:
+        from .mla import yarn_mscale
+        orig = cfg.rope_orig_max_position or cfg.max_position
+        inv_interp = inv / cfg.rope_factor
+
+        def corr_dim(n_rot: float) -> float:
+            return (rot * math.log(orig / (n_rot * 2 * math.pi))
+                    / (2 * math.log(theta)))
+
+        low = max(math.floor(corr_dim(cfg.rope_beta_fast)), 0)
+        high = min(math.ceil(corr_dim(cfg.rope_beta_slow)), rot // 2 - 1)
+        ramp = ((torch.arange(rot // 2, dtype=torch.float64, device=device)
+                 - low) / max(high - low, 1e-3)).clamp(0.0, 1.0)
+        extrap_mask = 1.0 - ramp        # low dims extrapolate, high interp
+        inv = inv * extrap_mask + inv_interp * (1.0 - extrap_mask)
+        # HF deepseek: cos/sin scaled by mscale(factor, mscale) /
+        # mscale(factor, mscale_all_dim) (net 1.0 when both equal, e.g.
+        # V3); the all_dim part goes into the softmax scale (mla.py)
+        mscale = yarn_mscale(cfg.rope_factor, cfg.rope_mscale)
+        if cfg.rope_mscale_all_dim > 0:
+            mscale /= yarn_mscale(cfg.rope_factor, cfg.rope_mscale_all_dim)
     t = torch.arange(max_pos, dtype=torch.float64, device=device)
     freqs = torch.outer(t, inv)
-    return torch.cat([freqs.cos(), freqs.sin()], dim=-1).float().contiguous()
+    return (torch.cat([freqs.cos(), freqs.sin()], dim=-1)
+            * mscale).float().contiguous()
 
 
 class LlamaAttention(nn.Module):
@@ -145,8 +170,12 @@ class LlamaMLP(nn.Module):
     def __init__(self, cfg: ModelConfig):
         super().__init__()
         self.cfg = cfg
+        # num_experts>0 ⇒ the model mixes dense + MoE layers (deepseek
+        # first_k_dense); the next layer's norm assumes a NON-deferred
+        # producer there (LlamaDecoderLayer deferred_producer flag), so
+        # dense layers in MoE models must not defer either
         fuse = (not cfg.parallel_block and not cfg.sandwich_norms
-                and cfg.norm_type == "rmsnorm")
+                and cfg.norm_type == "rmsnorm" and cfg.num_experts == 0)
         if cfg.gated_mlp:
             self.gate_up_proj = ColumnParallelLinear(
                 cfg.hidden_size, 2 * cfg.intermediate_size, dtype=cfg.dtype)
@@ -197,8 +226,13 @@ class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
         super().__init__()
         self.cfg = cfg
-        self.self_attn = LlamaAttention(cfg, layer_idx)
-        if cfg.num_experts > 0:
+        if cfg.is_mla:
+            from .mla import MLAAttention
+            self.self_attn = MLAAttention(cfg, layer_idx)
+        else:
+            self.self_attn = LlamaAttention(cfg, layer_idx)
+        # deepseek: the first k layers are dense even in MoE models
+        if cfg.num_experts > 0 and layer_idx >= cfg.first_k_dense:
             from .moe import MoEMLP
             self.mlp = MoEMLP(cfg)
         else:

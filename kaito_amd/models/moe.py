@@ -71,20 +71,62 @@ class MoEMLP(nn.Module):
             self.b_gate_up = None
             self.b_down = None
         self.act_mode = 1 if cfg.moe_act == "swiglu_oai" else 0
+        # deepseek-v3 noaux_tc routing: per-expert selection bias
+        if cfg.moe_routing == "noaux_tc":
+            self.e_score_correction_bias = nn.Parameter(
+                torch.empty(e, dtype=torch.float32), requires_grad=False)
+        # deepseek shared experts: always-on dense SwiGLU of width
+        # n_shared * moe_ie, IE-sharded over TP; its down output is a
+        # TP-partial like the routed output, so both ride ONE all-reduce
+        if cfg.n_shared_experts > 0:
+            sie = cfg.n_shared_experts * ie
+            assert tp == 1 or sie % tp == 0, (sie, tp)
+            self.sie_local = sie // tp
+            self.w_shared_gate_up = nn.Parameter(
+                torch.empty(2 * self.sie_local, h, dtype=cfg.dtype),
+                requires_grad=False)
+            self.w_shared_down = nn.Parameter(
+                torch.empty(h, self.sie_local, dtype=cfg.dtype),
+                requires_grad=False)
+        else:
+            self.w_shared_gate_up = None
 
     def _route(self, x: torch.Tensor):
         """Device-only top-k routing + expert sort. Returns
         (sorted_tok int32 [TK], gates f32 [TK], offsets int32 [E+1])."""
         T = x.size(0)
+        cfg = self.cfg
         logits = torch.nn.functional.linear(x, self.gate)      # [T, E]
-        if self.cfg.moe_routing == "topk_softmax":
+        if cfg.moe_routing == "noaux_tc":
+            # deepseek-v3: sigmoid scores; SELECTION uses scores+bias with
+            # group-limited top-k; WEIGHTS are the original sigmoid scores
+            scores = torch.sigmoid(logits.float())
+            sel = scores + self.e_score_correction_bias
+            if cfg.n_group > 1:
+                gs = sel.view(T, cfg.n_group, -1)
+                group_score = gs.topk(min(2, gs.size(-1)), dim=-1)[0].sum(-1)
+                gi = group_score.topk(cfg.topk_group, dim=-1)[1]  # [T, kg]
+                mask = torch.zeros(T, cfg.n_group, device=x.device,
+                                   dtype=torch.bool)
+                mask.scatter_(1, gi, True)
+                sel = sel.masked_fill(
+                    ~mask.unsqueeze(-1).expand_as(gs).reshape(T, -1),
+                    float("-inf"))
+            topi = sel.topk(self.top_k, dim=-1)[1]             # [T, K]
+            topw = scores.gather(1, topi)
+            if cfg.moe_norm_topk:
+                topw = topw / (topw.sum(-1, keepdim=True) + 1e-20)
+            topw = topw * cfg.routed_scaling_factor
+        elif cfg.moe_routing == "topk_softmax":
             # gpt-oss: top-k on raw logits, softmax over the selected k
             topl, topi = torch.topk(logits.float(), self.top_k, dim=-1)
             topw = torch.softmax(topl, dim=-1)
         else:
             probs = torch.softmax(logits.float(), dim=-1)
             topw, topi = torch.topk(probs, self.top_k, dim=-1)  # [T, K]
-            topw = topw / topw.sum(-1, keepdim=True)
+            if cfg.moe_norm_topk:
+                topw = topw / topw.sum(-1, keepdim=True)
+            topw = topw * cfg.routed_scaling_factor
         flat_e = topi.reshape(-1)                              # [T*K]
         flat_t = torch.arange(T, device=x.device,
                               dtype=torch.int32).repeat_interleave(self.top_k)
@@ -116,6 +158,10 @@ class MoEMLP(nn.Module):
             out = self._forward_fused(x, sorted_tok, gates, offsets)
         else:
             out = self._forward_loop(x, sorted_tok, gates, offsets)
+        if self.w_shared_gate_up is not None:   # deepseek shared experts
+            h1 = torch.nn.functional.linear(x, self.w_shared_gate_up)
+            out = out + torch.nn.functional.linear(
+                ops.silu_and_mul(h1), self.w_shared_down)
         if get_state().tp_size > 1:
             out = tp_all_reduce(out)
         return out

@@ -235,14 +235,39 @@ register(ModelConfig(
     sliding_window_pattern="interleaved:2", moe_act="swiglu_oai",
     moe_bias=True, moe_routing="topk_softmax", attention_bias=True))
 
-# DeepSeek V3/R1 — MLA attention + 256-expert MoE → fallback runtime
+# DeepSeek V3/R1 — native MLA (models/mla.py + ops/csrc/
+# mla_attention.hip): compressed 576-dim latent KV cache, absorbed
+# decode, noaux_tc sigmoid routing with group-limited top-k, 1 shared
+# expert, first 3 layers dense, yarn rope. 671B total params → TP=8
+# (16 heads/rank == one decode-kernel head tile).
 register(ModelConfig(
     name="deepseek-v3-0324", hidden_size=7168, num_layers=61, num_heads=128,
-    num_kv_heads=128, intermediate_size=18432, vocab_size=129280,
-    head_dim=64, rope_theta=10000.0, max_position=131072, num_experts=256,
+    num_kv_heads=1, intermediate_size=18432, vocab_size=129280,
+    head_dim=192, rope_theta=10000.0, max_position=131072, num_experts=256,
     num_experts_per_tok=8, moe_intermediate_size=2048,
-    runtime="transformers"))
+    kv_lora_rank=512, q_lora_rank=1536, qk_nope_head_dim=128,
+    qk_rope_head_dim=64, v_head_dim=128,
+    moe_routing="noaux_tc", moe_norm_topk=True, routed_scaling_factor=2.5,
+    n_group=8, topk_group=4, n_shared_experts=1, first_k_dense=3,
+    rope_scaling_type="yarn", rope_factor=40.0, rope_orig_max_position=4096,
+    rope_mscale=1.0, rope_mscale_all_dim=1.0))
 _alias("deepseek-r1-0528", "deepseek-v3-0324")
+
+# DeepSeek V2-Lite — 15.7B MLA MoE that FITS one MI355X; direct q
+# projection (no q-lora), 64 routed + 2 shared experts, softmax greedy
+# routing without top-k renorm
+register(ModelConfig(
+    name="deepseek-v2-lite", hidden_size=2048, num_layers=27, num_heads=16,
+    num_kv_heads=1, intermediate_size=10944, vocab_size=102400,
+    head_dim=192, rope_theta=10000.0, max_position=163840, num_experts=64,
+    num_experts_per_tok=6, moe_intermediate_size=1408,
+    kv_lora_rank=512, q_lora_rank=0, qk_nope_head_dim=128,
+    qk_rope_head_dim=64, v_head_dim=128,
+    moe_routing="softmax_topk", moe_norm_topk=False,
+    routed_scaling_factor=1.0, n_shared_experts=2, first_k_dense=1,
+    rope_scaling_type="yarn", rope_factor=40.0, rope_orig_max_position=4096,
+    rope_mscale=0.707, rope_mscale_all_dim=0.707, rms_eps=1e-6))
+_alias("deepseek-v2-lite-chat", "deepseek-v2-lite")
 
 # ---- tiny configs for tests ------------------------------------------------
 register(ModelConfig(
@@ -268,6 +293,20 @@ register(ModelConfig(
     embed_scale=16.0, sliding_window=32,
     sliding_window_pattern="interleaved:2", rope_theta_local=10000.0,
     tie_word_embeddings=True))
+register(ModelConfig(
+    # full deepseek feature set at kernel-real latent dims (r=512/rope=64
+    # so the GPU decode kernel path is exercised): MLA with q-lora,
+    # noaux_tc routing, shared expert, first layer dense, yarn rope
+    name="tiny-deepseek-test", hidden_size=256, num_layers=2, num_heads=4,
+    num_kv_heads=1, intermediate_size=512, vocab_size=512, head_dim=192,
+    rope_theta=10000.0, max_position=512, num_experts=4,
+    num_experts_per_tok=2, moe_intermediate_size=128,
+    kv_lora_rank=512, q_lora_rank=64, qk_nope_head_dim=128,
+    qk_rope_head_dim=64, v_head_dim=128,
+    moe_routing="noaux_tc", moe_norm_topk=True, routed_scaling_factor=2.5,
+    n_group=2, topk_group=1, n_shared_experts=1, first_k_dense=1,
+    rope_scaling_type="yarn", rope_factor=8.0, rope_orig_max_position=64,
+    rope_mscale=1.0, rope_mscale_all_dim=1.0))
 register(ModelConfig(
     name="tiny-gptoss-test", hidden_size=256, num_layers=2, num_heads=4,
     num_kv_heads=2, intermediate_size=256, vocab_size=512, head_dim=64,

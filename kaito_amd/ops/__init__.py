@@ -130,6 +130,21 @@ def paged_attention(q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tenso
                                      seq_lens, scale, window, sinks)
 
 
+def mla_decode(q: torch.Tensor, cache: torch.Tensor,
+               block_tables: torch.Tensor, seq_lens: torch.Tensor,
+               scale: float, r: int) -> torch.Tensor:
+    """Absorbed MLA decode over the compressed latent cache (DeepSeek).
+    q: [T, H, r+rope]; cache: [NB, BS, r+rope] bf16. Returns [T, H, r]."""
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty(q.size(0), q.size(1), r, dtype=q.dtype,
+                          device=q.device)
+        torch.ops.kaito.mla_decode(out, q, cache, block_tables, seq_lens,
+                                   scale)
+        return out
+    return torch_ref.mla_decode(q, cache, block_tables, seq_lens, scale, r)
+
+
 def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                       cu_seqlens: torch.Tensor, scale: float,
                       max_seqlen: int | None = None, window: int = 0,

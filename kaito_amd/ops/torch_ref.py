@@ -220,3 +220,23 @@ def w4a16_gemv(x: torch.Tensor, qweight: torch.Tensor, scales: torch.Tensor,
                zeros: torch.Tensor, group: int) -> torch.Tensor:
     w = w4a16_unpack(qweight, scales, zeros, group)
     return (x.float() @ w.T).to(x.dtype)
+
+
+def mla_decode(q: torch.Tensor, cache: torch.Tensor,
+               block_tables: torch.Tensor, seq_lens: torch.Tensor,
+               scale: float, r: int) -> torch.Tensor:
+    """Absorbed MLA decode reference (fp32 math). q: [T, H, r+rope] latent
+    queries; cache: [NB, BS, r+rope] (c_kv || k_rope per token). Scores are
+    the full latent dot; output is the latent-space V accumulation
+    (first r dims). Mirrors ops/csrc/mla_attention.hip."""
+    T, H, DT = q.shape
+    NB, BS, _ = cache.shape
+    out = torch.empty(T, H, r, dtype=q.dtype, device=q.device)
+    for i in range(T):
+        L = int(seq_lens[i])
+        nb = (L + BS - 1) // BS
+        rows = cache[block_tables[i, :nb].long()].reshape(-1, DT)[:L].float()
+        s = q[i].float() @ rows.t() * scale          # [H, L]
+        p = torch.softmax(s, dim=-1)
+        out[i] = (p @ rows[:, :r]).to(out.dtype)
+    return out

@@ -224,11 +224,12 @@ def test_engine_model_variants_match_oracle(preset_kw):
 
 
 @pytest.mark.parametrize("model", ["tiny-phi2-test", "tiny-gemma3-test",
-                                   "tiny-gptoss-test"])
+                                   "tiny-gptoss-test", "tiny-deepseek-test"])
 def test_engine_model_variants_match_oracle(model):
     """Architecture variants (phi-2 parallel block + LayerNorm + ungated
     GELU; gemma-3 sandwich norms + qk-norm + GeGLU + sliding window +
-    local rope; gpt-oss sinks + sliding window + clamped-swiglu MoE)
+    local rope; gpt-oss sinks + sliding window + clamped-swiglu MoE;
+    deepseek MLA latent cache + q-lora + noaux_tc MoE + yarn rope)
     through the full paged engine must reproduce the full-recompute
     oracle."""
     cfg = _cfg(model=get_model_config(model), max_model_len=96)
