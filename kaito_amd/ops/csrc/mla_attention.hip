@@ -1,0 +1,225 @@
+// MLA (multi-head latent attention) DECODE kernel for gfx950 (MI355X).
+//
+// DeepSeek V2/V3/R1 absorbed decode (models/mla.py): one compressed
+// latent row per cached token — c_kv[R=512] ‖ k_rope[P=64], 576 bf16 —
+// shared by EVERY head. Replaces the vLLM MLA backend the reference
+// delegates to for the deepseek presets (SURVEY.md §2.3,
+// supported_models.yaml deepseek-v3-0324 / deepseek-r1-0528).
+//
+// Shape: scores[h][t] = q_latent[h]·cache[t] (the full 576-dot — rope
+// term included by construction), out[h] = Σ_t softmax·c_kv[t][0:512].
+// Compared with GQA decode (paged_attention_sp.hip) the arithmetic
+// intensity is ~16 heads × 2 dots per byte, so the kernel is organized
+// around HBM-byte REUSE, not pure streaming:
+//
+//   * grid (seq, head-tile of 16); each 256-thread workgroup stages one
+//     16-token KV block (18 KB) into LDS cooperatively, then all 16
+//     heads consume it — every HBM byte is read ONCE per tile. At
+//     deepseek-v3 TP=8 each rank holds exactly 16 heads = one tile;
+//     V2-Lite (16 heads) is one tile at TP=1.
+//   * score phase: lane = (head&3)<<4 | slice, 4 heads per wave; each
+//     lane dots a 36-dim slice (9 × b64 LDS reads, bf16 v_dot2) and the
+//     16 slice-lanes shfl-reduce. The 4 same-slice lanes read the same
+//     LDS address → broadcast, conflict-free.
+//   * online softmax per block: per-head running (M, l) and the
+//     accumulator rescale, amortized over 16 heads × 512 dims (the
+//     per-block rescale is ~6% of the accumulate FLOPs — cheap here,
+//     unlike the GQA kernel where it forced the split-phase design).
+//   * accumulate phase: thread (head = tid>>4, dim-slice = tid&15)
+//     carries 32 f32 accumulator registers; p broadcasts from LDS.
+//   * all LDS handoffs except the staging barrier are intra-wave
+//     (score→exp→accumulate stay inside the head's own wave), so each
+//     block iteration needs only TWO __syncthreads.
+//
+// Cache layout: [num_blocks, BLOCK_SIZE=16, R+P] bf16 (the aliased
+// (c, c) pair from model_runner.profile_and_allocate_kv).
+//
+// Known limit: grid = T × ceil(H/16) — below ~512 concurrent sequences
+// the launch underfills 256 CUs; a flash-decoding token-split tier is
+// the designated fix (docs/ROADMAP.md).
+#include "common.h"
+#include <torch/library.h>
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace kaito {
+
+typedef short short4_t __attribute__((ext_vector_type(4)));
+
+template <int R, int P, int BS>
+__global__ __launch_bounds__(256, 4)
+void mla_decode_kernel(
+    short* __restrict__ out,            // [T, H, R] bf16
+    const short* __restrict__ q,        // [T, H, R+P] bf16
+    const short* __restrict__ cache,    // [NB, BS, R+P] bf16
+    const int* __restrict__ block_tables,  // [T, max_blocks]
+    const int* __restrict__ seq_lens,   // [T]
+    const float scale, const int H, const int max_blocks) {
+  constexpr int DT = R + P;             // 576
+  constexpr int SL = DT / 16;           // 36 dims per score slice
+  constexpr int AD = R / 16;            // 32 dims per accum slice
+  constexpr int HT = 16;                // heads per workgroup
+  static_assert(DT % 16 == 0 && R % 16 == 0 && SL % 4 == 0 && AD % 8 == 0);
+
+  const int seq = blockIdx.x;
+  const int h0 = blockIdx.y * HT;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int seq_len = seq_lens[seq];
+  const int nblocks = (seq_len + BS - 1) / BS;
+  const int* bt = block_tables + (int64_t)seq * max_blocks;
+
+  // score-phase role: 4 heads per wave, 16 dim-slices per head
+  const int sc_head = wave * 4 + (lane >> 4);     // 0..15 == tid>>4
+  const int slice = lane & 15;
+  // accum/exp-phase role (same head): dim/token slice
+  const int ac_h = tid >> 4;                      // == sc_head
+  const int ac_s = tid & 15;
+
+  __shared__ short s_kv[BS * DT];                 // 18 KB staged block
+  __shared__ float s_p[BS][HT];                   // scores→probs
+  __shared__ float s_M[HT], s_l[HT], s_corr[HT];
+
+  if (tid < HT) { s_M[tid] = -1e30f; s_l[tid] = 0.f; }
+
+  // q slice (36 bf16 = 9 x b64) for the score phase; zero for padded
+  // heads so their scores stay finite (their output is never written)
+  short4_t qv[SL / 4];
+  const bool head_ok = h0 + sc_head < H;
+  {
+    const short* qp = q + ((int64_t)seq * H + (h0 + sc_head)) * DT
+        + slice * SL;
+#pragma unroll
+    for (int j = 0; j < SL / 4; j++)
+      qv[j] = head_ok ? *reinterpret_cast<const short4_t*>(qp + j * 4)
+                      : short4_t{0, 0, 0, 0};
+  }
+
+  float acc[AD];
+#pragma unroll
+  for (int j = 0; j < AD; j++) acc[j] = 0.f;
+
+  typedef __bf16 bf16x2_t __attribute__((ext_vector_type(2)));
+
+  for (int b = 0; b < nblocks; b++) {
+    const int blk = bt[b];
+    // ---- stage one KV block: 1152 x b128, 256 threads ----
+    {
+      const short8_t* src = reinterpret_cast<const short8_t*>(
+          cache + (int64_t)blk * (BS * DT));
+      short8_t* dst = reinterpret_cast<short8_t*>(s_kv);
+#pragma unroll
+      for (int c = 0; c < (BS * DT / 8 + 255) / 256; c++) {
+        const int idx = c * 256 + tid;
+        if (idx < BS * DT / 8) dst[idx] = src[idx];
+      }
+    }
+    __syncthreads();
+
+    // ---- scores for the 16 tokens (intra-wave handoff to exp) ----
+    for (int t = 0; t < BS; t++) {
+      const short* kp = s_kv + t * DT + slice * SL;
+      float part = 0.f;
+#pragma unroll
+      for (int j = 0; j < SL / 4; j++) {
+        const short4_t kv4 = *reinterpret_cast<const short4_t*>(kp + j * 4);
+        const bf16x2_t* k2 = reinterpret_cast<const bf16x2_t*>(&kv4);
+        const bf16x2_t* q2 = reinterpret_cast<const bf16x2_t*>(&qv[j]);
+        part = __builtin_amdgcn_fdot2_f32_bf16(q2[0], k2[0], part, false);
+        part = __builtin_amdgcn_fdot2_f32_bf16(q2[1], k2[1], part, false);
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        part += __shfl_xor(part, off, 64);
+      if (slice == 0) {
+        const bool valid = b * BS + t < seq_len;
+        s_p[t][sc_head] = valid ? part * scale : -1e30f;
+      }
+    }
+
+    // ---- online softmax update (thread (h, t); intra-wave) ----
+    {
+      const int t = ac_s;
+      float s = s_p[t][ac_h];
+      float bm = s;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        bm = fmaxf(bm, __shfl_xor(bm, off, 64));
+      const float Mold = s_M[ac_h];
+      const float Mnew = fmaxf(Mold, bm);
+      const float p = s > -1e29f ? __expf(s - Mnew) : 0.f;
+      float bsum = p;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        bsum += __shfl_xor(bsum, off, 64);
+      s_p[t][ac_h] = p;
+      if (t == 0) {
+        const float corr = Mnew > Mold ? __expf(Mold - Mnew) : 1.f;
+        s_corr[ac_h] = corr;
+        s_M[ac_h] = Mnew;
+        s_l[ac_h] = s_l[ac_h] * corr + bsum;
+      }
+    }
+
+    // ---- accumulate c_kv into the latent-space output ----
+    {
+      const float corr = s_corr[ac_h];
+#pragma unroll
+      for (int j = 0; j < AD; j++) acc[j] *= corr;
+      for (int t = 0; t < BS; t++) {
+        const float p = s_p[t][ac_h];
+        const short* vp = s_kv + t * DT + ac_s * AD;
+#pragma unroll
+        for (int c = 0; c < AD / 8; c++) {
+          const short8_t v8 = *reinterpret_cast<const short8_t*>(vp + c * 8);
+#pragma unroll
+          for (int j = 0; j < 8; j++)
+            acc[c * 8 + j] = fmaf(p, bf16_to_f32(v8[j]), acc[c * 8 + j]);
+        }
+      }
+    }
+    __syncthreads();   // s_kv free for the next block's staging
+  }
+
+  // ---- epilogue: normalize + write [T, H, R] ----
+  if (h0 + ac_h < H) {
+    const float inv = 1.f / fmaxf(s_l[ac_h], 1e-20f);
+    short* op = out + ((int64_t)seq * H + (h0 + ac_h)) * R + ac_s * AD;
+#pragma unroll
+    for (int j = 0; j < AD; j++) op[j] = f32_to_bf16(acc[j] * inv);
+  }
+}
+
+void mla_decode(at::Tensor out, at::Tensor q, at::Tensor cache,
+                at::Tensor block_tables, at::Tensor seq_lens, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == at::kBFloat16 && q.is_contiguous());
+  TORCH_CHECK(cache.dtype() == at::kBFloat16 && cache.is_contiguous());
+  TORCH_CHECK(out.is_contiguous() && out.dtype() == at::kBFloat16);
+  TORCH_CHECK(block_tables.dtype() == at::kInt
+              && seq_lens.dtype() == at::kInt);
+  const int T = q.size(0);
+  const int H = q.size(1);
+  const int DT = q.size(2);
+  const int BS = cache.size(1);
+  const int R = out.size(2);
+  TORCH_CHECK(cache.size(2) == DT && out.size(0) == T && out.size(1) == H);
+  TORCH_CHECK(BS == 16, "block_size must be 16");
+  const int max_blocks = block_tables.size(1);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (T == 0) return;
+  const dim3 grid(T, (H + 15) / 16);
+  if (R == 512 && DT == 576) {
+    hipLaunchKernelGGL((mla_decode_kernel<512, 64, 16>), grid, dim3(256), 0,
+                       stream, (short*)out.data_ptr(),
+                       (const short*)q.data_ptr(),
+                       (const short*)cache.data_ptr(),
+                       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
+                       (float)scale, H, max_blocks);
+  } else {
+    TORCH_CHECK(false, "unsupported MLA dims r=", R, " r+rope=", DT,
+                " (deepseek family is 512/576)");
+  }
+}
+
+}  // namespace kaito

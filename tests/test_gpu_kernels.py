@@ -344,12 +344,40 @@ def test_prefill_attention_window_sinks(window, use_sinks):
     _close(out, ref.to(DEV), atol=2e-2)
 
 
+def test_mla_decode_kernel_matches_ref():
+    """Absorbed MLA decode (ops/csrc/mla_attention.hip) vs the fp32
+    reference over the compressed latent cache — deepseek dims
+    (r=512, rope=64), ragged lengths, 16-head tile + padded-head tile."""
+    torch.manual_seed(7)
+    R_, P_, BS = 512, 64, 16
+    DT = R_ + P_
+    for T, H, lens in [(5, 16, [200, 33, 7, 390, 64]),
+                       (3, 4, [100, 17, 255]),       # padded head tile
+                       (2, 32, [48, 312])]:          # two head tiles
+        nb = [(x + BS - 1) // BS for x in lens]
+        NB = sum(nb) + 1
+        cache = _bf16(NB, BS, DT)
+        bt = torch.zeros(T, max(nb), dtype=torch.int32, device=DEV)
+        nxt = 1
+        for i in range(T):
+            bt[i, :nb[i]] = torch.arange(nxt, nxt + nb[i],
+                                         dtype=torch.int32, device=DEV)
+            nxt += nb[i]
+        sl = torch.tensor(lens, dtype=torch.int32, device=DEV)
+        q = _bf16(T, H, DT, scale=0.3)
+        scale = 192 ** -0.5
+        out = ops.mla_decode(q, cache, bt, sl, scale, R_)
+        ref = R.mla_decode(q.cpu().float(), cache.cpu().float(), bt.cpu(),
+                           sl.cpu(), scale, R_)
+        _close(out, ref.to(DEV), atol=2e-2)
+
+
 @pytest.mark.parametrize("name", ["tiny-phi2-test", "tiny-gemma3-test",
-                                  "tiny-gptoss-test"])
+                                  "tiny-gptoss-test", "tiny-deepseek-test"])
 def test_engine_gpu_model_variants(name):
-    """phi-2 / gemma-3 / gpt-oss architecture variants through the GPU
-    engine (HIP layernorm/gelu/window/sink/MoE paths) vs the
-    full-recompute oracle on the same device."""
+    """phi-2 / gemma-3 / gpt-oss / deepseek-MLA architecture variants
+    through the GPU engine (HIP layernorm/gelu/window/sink/MoE/latent-
+    cache paths) vs the full-recompute oracle on the same device."""
     from kaito_amd.engine import SamplingParams
     from kaito_amd.models import get_model_config
     from kaito_amd.models.llama import AttnMetadata

@@ -357,13 +357,14 @@ def test_full_preset_catalog():
     assert 80 * g < get_model_config("falcon-40b").param_bytes() < 120 * g
     assert 1150 * g < get_model_config("deepseek-v3-0324").param_bytes() \
         < 1500 * g
-    # round 2: gemma-3 / phi-2 / gpt-oss run NATIVELY on the HIP engine
-    # (sliding window, sandwich norms, sinks, clamped-swiglu MoE);
-    # MLA-class archs (deepseek) remain on the fallback runtime
+    # round 2: gemma-3 / phi-2 / gpt-oss AND deepseek (native MLA,
+    # models/mla.py) run on the HIP engine
     assert get_model_config("gemma-3-27b-instruct").runtime == "native"
     assert get_model_config("phi-2").runtime == "native"
     assert get_model_config("gpt-oss-120b").runtime == "native"
-    assert get_model_config("deepseek-v3-0324").runtime == "transformers"
+    assert get_model_config("deepseek-v3-0324").runtime == "native"
+    assert get_model_config("deepseek-v3-0324").is_mla
+    assert get_model_config("deepseek-v2-lite").is_mla
     assert get_model_config("phi-4").runtime == "native"
 
 
