@@ -94,8 +94,86 @@ def load_safetensors_weights(model, path: str,
             return pre + "dense"
         return pre + base
 
+    # deepseek stores rope dims INTERLEAVED (HF apply_rotary permutes
+    # inside the model); our rotate-half kernel wants the half-split
+    # layout, so permute those output ROWS once at load — exact
+    def deinterleave_pe(t, n_heads=1):
+        d = t.shape[0] // n_heads
+        x = t.reshape(n_heads, d // 2, 2, -1)
+        return torch.cat([x[:, :, 0], x[:, :, 1]],
+                         dim=1).reshape(t.shape[0], -1)
+
+    def load_mla_attn(pre):
+        a = pre + "self_attn."
+        nope, rope, r = (cfg.qk_nope_head_dim, cfg.qk_rope_head_dim,
+                         cfg.kv_lora_rank)
+        if cfg.q_lora_rank > 0:
+            setp(a + "q_a_proj", get(a + "q_a_proj.weight"))
+            setp(a + "q_a_layernorm", get(a + "q_a_layernorm.weight"))
+            qb = get(a + "q_b_proj.weight")
+        else:
+            qb = get(a + "q_proj.weight")
+        # per-head [nope | rope] rows; de-interleave the rope part
+        qb = qb.reshape(cfg.num_heads, nope + rope, -1)
+        qb = torch.cat([qb[:, :nope],
+                        deinterleave_pe(
+                            qb[:, nope:].reshape(-1, qb.shape[-1]),
+                            cfg.num_heads).reshape(cfg.num_heads, rope, -1)],
+                       dim=1).reshape(cfg.num_heads * (nope + rope), -1)
+        tgt = "q_b_proj" if cfg.q_lora_rank > 0 else "q_proj"
+        setp(a + tgt + ".weight", _shard(qb, 0))
+        kva = get(a + "kv_a_proj_with_mqa.weight")
+        kva = torch.cat([kva[:r], deinterleave_pe(kva[r:])], 0)
+        setp(a + "kv_a_proj_with_mqa", kva)
+        setp(a + "kv_a_layernorm", get(a + "kv_a_layernorm.weight"))
+        kvb = get(a + "kv_b_proj.weight").reshape(
+            cfg.num_heads, nope + cfg.v_head_dim, r)
+        setp(a + "w_kc", _shard(kvb[:, :nope], 0))
+        setp(a + "w_vc", _shard(kvb[:, nope:].transpose(1, 2), 0))
+        setp(a + "o_proj.weight", _shard(get(a + "o_proj.weight"), 1))
+
+    # deepseek MoE: mlp.gate (+ e_score_correction_bias), per-expert
+    # gate/up/down, optional shared_experts
+    def load_deepseek_moe(pre, i):
+        setp(pre + "mlp.gate", get(pre + "mlp.gate.weight"))
+        if has(pre + "mlp.gate.e_score_correction_bias"):
+            setp(pre + "mlp.e_score_correction_bias",
+                 get(pre + "mlp.gate.e_score_correction_bias"))
+        moe = params[pre + "mlp.w_gate_up"]
+        e_base = getattr(model.layers[i].mlp, "e_base", 0)
+        with torch.no_grad():
+            for le in range(moe.shape[0]):
+                ex = f"{pre}mlp.experts.{e_base + le}."
+                params[pre + "mlp.w_gate_up"][le].copy_(torch.cat(
+                    [get(ex + "gate_proj.weight"),
+                     get(ex + "up_proj.weight")], 0).to(moe.dtype))
+                params[pre + "mlp.w_down"][le].copy_(
+                    get(ex + "down_proj.weight").to(moe.dtype))
+        if cfg.n_shared_experts > 0:
+            sh = pre + "mlp.shared_experts."
+            setp(pre + "mlp.w_shared_gate_up", _shard(torch.cat(
+                [get(sh + "gate_proj.weight"),
+                 get(sh + "up_proj.weight")], 0), 0))
+            setp(pre + "mlp.w_shared_down",
+                 _shard(get(sh + "down_proj.weight"), 1))
+
     for i in range(cfg.num_layers):
         pre = f"layers.{i}."
+        if cfg.is_mla and not skip_projections:
+            load_mla_attn(pre)
+            if i < cfg.first_k_dense or cfg.num_experts == 0:
+                g = _shard(get(pre + "mlp.gate_proj.weight"), 0)
+                u = _shard(get(pre + "mlp.up_proj.weight"), 0)
+                setp(pre + "mlp.gate_up_proj.weight", torch.cat([g, u], 0))
+                setp(pre + "mlp.down_proj.weight",
+                     _shard(get(pre + "mlp.down_proj.weight"), 1))
+            else:
+                load_deepseek_moe(pre, i)
+            setp(pre + "input_layernorm",
+                 norm_w(pre + "input_layernorm.weight"))
+            setp(pre + "post_attention_layernorm",
+                 norm_w(pre + "post_attention_layernorm.weight"))
+            continue
         if not skip_projections:   # AWQ checkpoints carry qweight instead
             q = pad_heads(get(attn_name("q_proj", i) + ".weight"),
                           cfg.num_heads)

@@ -149,3 +149,91 @@ def test_mla_preemption_recompute_coherent():
     outs2 = eng2.generate(prompts, sp)
     for a, b in zip(outs, outs2):
         assert a.output_token_ids == b.output_token_ids
+
+
+def test_deepseek_checkpoint_roundtrip(tmp_path):
+    """Export a random-init tiny-deepseek engine's weights into an
+    HF-DeepSeek-named safetensors checkpoint (re-INTERLEAVING the rope
+    rows the way real checkpoints store them, merging w_kc/w_vc back
+    into kv_b_proj), then load into a fresh engine: greedy decode must
+    match exactly. Validates the loader mapping bidirectionally,
+    including the rope de-interleave (models/loader.py load_mla_attn)."""
+    from safetensors.torch import save_file
+    mc = get_model_config("tiny-deepseek-test")
+    base = LLMEngine(_cfg())
+    model = base.runner.model
+    params = dict(model.named_parameters())
+    H, NOPE, PE, R_ = (mc.num_heads, mc.qk_nope_head_dim,
+                       mc.qk_rope_head_dim, mc.kv_lora_rank)
+
+    def interleave_pe(t, n_heads=1):
+        # inverse of the loader's deinterleave: half-split -> interleaved
+        d = t.shape[0] // n_heads
+        x = t.reshape(n_heads, 2, d // 2, -1)
+        return x.transpose(1, 2).reshape(t.shape[0], -1)
+
+    tensors = {
+        "model.embed_tokens.weight": params["embed_tokens.weight"].data,
+        "model.norm.weight": params["norm"].data,
+        "lm_head.weight": params["lm_head.weight"].data,
+    }
+    for i in range(mc.num_layers):
+        pre = f"layers.{i}."
+        a = pre + "self_attn."
+        o = "model." + a
+        tensors[o + "q_a_proj.weight"] = params[a + "q_a_proj"].data
+        tensors[o + "q_a_layernorm.weight"] = params[a + "q_a_layernorm"].data
+        qb = params[a + "q_b_proj.weight"].data.reshape(H, NOPE + PE, -1)
+        qb_il = torch.cat(
+            [qb[:, :NOPE],
+             interleave_pe(qb[:, NOPE:].reshape(H * PE, -1), H
+                           ).reshape(H, PE, -1)], 1)
+        tensors[o + "q_b_proj.weight"] = qb_il.reshape(H * (NOPE + PE), -1)
+        kva = params[a + "kv_a_proj_with_mqa"].data
+        tensors[o + "kv_a_proj_with_mqa.weight"] = torch.cat(
+            [kva[:R_], interleave_pe(kva[R_:])], 0)
+        tensors[o + "kv_a_layernorm.weight"] = params[a + "kv_a_layernorm"].data
+        kvb = torch.cat([params[a + "w_kc"].data,
+                         params[a + "w_vc"].data.transpose(1, 2)], 1)
+        tensors[o + "kv_b_proj.weight"] = kvb.reshape(-1, R_)
+        tensors[o + "o_proj.weight"] = params[a + "o_proj.weight"].data
+        for ln in ("input_layernorm", "post_attention_layernorm"):
+            tensors[f"model.{pre}{ln}.weight"] = params[pre + ln].data
+        if i < mc.first_k_dense:
+            gu = params[pre + "mlp.gate_up_proj.weight"].data
+            ii = mc.intermediate_size
+            tensors[f"model.{pre}mlp.gate_proj.weight"] = gu[:ii]
+            tensors[f"model.{pre}mlp.up_proj.weight"] = gu[ii:]
+            tensors[f"model.{pre}mlp.down_proj.weight"] = \
+                params[pre + "mlp.down_proj.weight"].data
+        else:
+            tensors[f"model.{pre}mlp.gate.weight"] = params[pre + "mlp.gate"].data
+            tensors[f"model.{pre}mlp.gate.e_score_correction_bias"] = \
+                params[pre + "mlp.e_score_correction_bias"].data
+            for le in range(mc.num_experts):
+                ex = f"model.{pre}mlp.experts.{le}."
+                w = params[pre + "mlp.w_gate_up"].data[le]
+                ie = mc.moe_intermediate_size
+                tensors[ex + "gate_proj.weight"] = w[:ie]
+                tensors[ex + "up_proj.weight"] = w[ie:]
+                tensors[ex + "down_proj.weight"] = \
+                    params[pre + "mlp.w_down"].data[le]
+            sh = f"model.{pre}mlp.shared_experts."
+            sgu = params[pre + "mlp.w_shared_gate_up"].data
+            sie = mc.n_shared_experts * mc.moe_intermediate_size
+            tensors[sh + "gate_proj.weight"] = sgu[:sie]
+            tensors[sh + "up_proj.weight"] = sgu[sie:]
+            tensors[sh + "down_proj.weight"] = \
+                params[pre + "mlp.w_shared_down"].data
+
+    d = tmp_path / "ckpt"
+    d.mkdir()
+    save_file({k: v.contiguous().clone() for k, v in tensors.items()},
+              str(d / "model.safetensors"))
+
+    prompt = [5, 9, 13, 17, 21]
+    sp = SamplingParams(max_tokens=8, ignore_eos=True)
+    want = base.generate([prompt], sp)[0].output_token_ids
+    eng = LLMEngine(_cfg(), weights_path=str(d))
+    got = eng.generate([prompt], sp)[0].output_token_ids
+    assert got == want
