@@ -88,12 +88,15 @@ void mla_decode_kernel(
   short4_t qv[SL / 4];
   const bool head_ok = h0 + sc_head < H;
   {
-    const short* qp = q + ((int64_t)seq * H + (h0 + sc_head)) * DT
-        + slice * SL;
+    // clamp the ROW (not just the value) so a potentially if-converted
+    // load never dereferences past the q tensor for padded head tiles
+    const int hq = min(h0 + sc_head, H - 1);
+    const short* qp = q + ((int64_t)seq * H + hq) * DT + slice * SL;
 #pragma unroll
-    for (int j = 0; j < SL / 4; j++)
-      qv[j] = head_ok ? *reinterpret_cast<const short4_t*>(qp + j * 4)
-                      : short4_t{0, 0, 0, 0};
+    for (int j = 0; j < SL / 4; j++) {
+      qv[j] = *reinterpret_cast<const short4_t*>(qp + j * 4);
+      if (!head_ok) qv[j] = short4_t{0, 0, 0, 0};
+    }
   }
 
   float acc[AD];
