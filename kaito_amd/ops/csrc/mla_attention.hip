@@ -79,9 +79,11 @@ void mla_decode_kernel(
 
   __shared__ short s_kv[BS * DT];                 // 18 KB staged block
   __shared__ float s_p[BS][HT];                   // scores→probs
-  __shared__ float s_M[HT], s_l[HT], s_corr[HT];
 
-  if (tid < HT) { s_M[tid] = -1e30f; s_l[tid] = 0.f; }
+  // per-head online-softmax state, held REDUNDANTLY in registers by all
+  // 16 lanes of the head's thread group (every lane derives identical
+  // values from the same shfl reductions) — no cross-lane LDS handoff
+  float M = -1e30f, l = 0.f;
 
   // q slice (36 bf16 = 9 x b64) for the score phase; zero for padded
   // heads so their scores stay finite (their output is never written)
@@ -141,7 +143,8 @@ void mla_decode_kernel(
       }
     }
 
-    // ---- online softmax update (thread (h, t); intra-wave) ----
+    // ---- online softmax update (thread (h, t); state in registers) ----
+    float corr;
     {
       const int t = ac_s;
       float s = s_p[t][ac_h];
@@ -149,25 +152,20 @@ void mla_decode_kernel(
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1)
         bm = fmaxf(bm, __shfl_xor(bm, off, 64));
-      const float Mold = s_M[ac_h];
-      const float Mnew = fmaxf(Mold, bm);
+      const float Mnew = fmaxf(M, bm);
       const float p = s > -1e29f ? __expf(s - Mnew) : 0.f;
       float bsum = p;
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1)
         bsum += __shfl_xor(bsum, off, 64);
       s_p[t][ac_h] = p;
-      if (t == 0) {
-        const float corr = Mnew > Mold ? __expf(Mold - Mnew) : 1.f;
-        s_corr[ac_h] = corr;
-        s_M[ac_h] = Mnew;
-        s_l[ac_h] = s_l[ac_h] * corr + bsum;
-      }
+      corr = Mnew > M ? __expf(M - Mnew) : 1.f;
+      l = l * corr + bsum;
+      M = Mnew;
     }
 
     // ---- accumulate c_kv into the latent-space output ----
     {
-      const float corr = s_corr[ac_h];
 #pragma unroll
       for (int j = 0; j < AD; j++) acc[j] *= corr;
       for (int t = 0; t < BS; t++) {
@@ -187,7 +185,7 @@ void mla_decode_kernel(
 
   // ---- epilogue: normalize + write [T, H, R] ----
   if (h0 + ac_h < H) {
-    const float inv = 1.f / fmaxf(s_l[ac_h], 1e-20f);
+    const float inv = 1.f / fmaxf(l, 1e-20f);
     short* op = out + ((int64_t)seq * H + (h0 + ac_h)) * R + ac_s * AD;
 #pragma unroll
     for (int j = 0; j < AD; j++) op[j] = f32_to_bf16(acc[j] * inv);
