@@ -115,8 +115,13 @@ class ModelRunner:
         kv_dtype = torch.uint8 if fp8 else m.dtype
         for _ in range(self.model.num_local_layers):
             if m.is_mla:
-                # aliased pair: the latent cache IS both k and v
-                c = torch.zeros(n_blocks, cfg.block_size, m.kv_cache_row,
+                # aliased pair: the latent cache IS both k and v. One
+                # spare block at the end absorbs writes for PADDING slots
+                # (-1): the graph-safe index_copy_ cache write redirects
+                # them there (models/mla.py) the way the reshape_and_cache
+                # kernel skips negatives; the pool never hands it out and
+                # no block table references it.
+                c = torch.zeros(n_blocks + 1, cfg.block_size, m.kv_cache_row,
                                 dtype=kv_dtype, device=self.device)
                 kvs.append((c, c))
                 continue

@@ -151,9 +151,15 @@ class MLAAttention(nn.Module):
 
         latent = torch.cat([c_kv, k_pe], dim=-1)        # [T, r+rope]
         if kv_cache is not None:
-            cache = kv_cache[0]                         # [NB, BS, r+rope]
-            cache.view(-1, cache.size(-1)).index_copy_(
-                0, meta.slot_mapping, latent.to(cache.dtype))
+            cache = kv_cache[0]                    # [NB+1, BS, r+rope]
+            rows = cache.view(-1, cache.size(-1))
+            # padding slots are -1 (graph buckets): redirect them to the
+            # spare trash block the allocator reserves at the end —
+            # index_copy_ has no skip semantics and -1 would fault
+            slots = meta.slot_mapping
+            slots = torch.where(
+                slots < 0, torch.full_like(slots, rows.size(0) - 1), slots)
+            rows.index_copy_(0, slots, latent.to(cache.dtype))
 
         if meta.is_prefill:
             if meta.kv_lens is not None:
