@@ -147,8 +147,11 @@ class MoEMLP(nn.Module):
     # prefill sizes (compute-bound, Tensile ~2x our 2-barrier tile —
     # tools/bench_moe_kernels.py: TK=16384 fused 5.8ms vs loop 3.0ms).
     # Prefill steps run eagerly, so the loop's host-side segment reads
-    # are harmless there.
-    FUSED_MAX_TOKENS = 4096
+    # are harmless there. KAITO_MOE_FUSED_MAX overrides (high-top-k MoE
+    # like deepseek k=6 crosses TK=4096 at decode bs>682, where losing
+    # hipGraph capture costs more than the tile-efficiency gap).
+    import os as _os
+    FUSED_MAX_TOKENS = int(_os.environ.get("KAITO_MOE_FUSED_MAX", "4096"))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, H = x.shape
