@@ -156,8 +156,13 @@ class MoEMLP(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, H = x.shape
         sorted_tok, gates, offsets = self._route(x)
-        if x.is_cuda and H % 64 == 0 and self.ie_local % 64 == 0 and \
-                T * self.top_k <= self.FUSED_MAX_TOKENS:
+        fused_shapes = x.is_cuda and H % 64 == 0 and self.ie_local % 64 == 0
+        # under hipGraph capture the fused kernels are mandatory (the
+        # per-expert loop reads segment sizes on the host): deepseek k=6
+        # crosses TK=4096 at decode bs>682 and measured 30.9k vs timeout
+        # at bs=768 (profiles/r02_perf_notes.md)
+        if fused_shapes and (T * self.top_k <= self.FUSED_MAX_TOKENS
+                             or torch.cuda.is_current_stream_capturing()):
             out = self._forward_fused(x, sorted_tok, gates, offsets)
         else:
             out = self._forward_loop(x, sorted_tok, gates, offsets)
