@@ -187,6 +187,73 @@ def _body_moe_ep(rank, world):
         f"EP mismatch {(out.float()-expect.float()).abs().max()}"
 
 
+def _body_mla_tp(rank, world):
+    """TP=2 deepseek (MLA heads split, latent projections replicated,
+    EP experts + IE-sharded shared expert) must match the tp=1 model."""
+    from kaito_amd.models.llama import LlamaForCausalLM, AttnMetadata
+    from kaito_amd.parallel import state as ps
+    cfg = get_model_config("tiny-deepseek-test")
+    st = ps.get_state()
+    saved = st.tp_size, st.tp_rank
+    st.tp_size, st.tp_rank = 1, 0
+    ref = LlamaForCausalLM(cfg).random_init(13)
+    ref.init_rope("cpu")
+    st.tp_size, st.tp_rank = saved
+
+    tp = LlamaForCausalLM(cfg)
+    tp.init_rope("cpu")
+    rp = dict(ref.named_parameters())
+    H = cfg.num_heads
+    Hl = H // world
+    qk = cfg.qk_nope_head_dim + cfg.qk_rope_head_dim
+
+    def shard0(t):
+        n = t.shape[0] // world
+        return t[rank * n:(rank + 1) * n]
+
+    def shard1(t):
+        n = t.shape[1] // world
+        return t[:, rank * n:(rank + 1) * n]
+
+    with torch.no_grad():
+        for name, p in tp.named_parameters():
+            r = rp[name]
+            if p.shape == r.shape:
+                p.copy_(r)
+            elif "q_b_proj" in name or name.endswith("embed_tokens.weight")                     or "lm_head" in name:
+                p.copy_(shard0(r))
+            elif name.endswith("w_kc") or name.endswith("w_vc"):
+                p.copy_(shard0(r))
+            elif "o_proj" in name or name.endswith("mlp.down_proj.weight")                     or name.endswith("w_shared_down"):
+                p.copy_(shard1(r))
+            elif name.endswith("mlp.gate_up_proj.weight"):
+                ii = r.shape[0] // 2
+                p.copy_(torch.cat([shard0(r[:ii]), shard0(r[ii:])], 0))
+            elif name.endswith("w_shared_gate_up"):
+                sie = r.shape[0] // 2
+                p.copy_(torch.cat([shard0(r[:sie]), shard0(r[sie:])], 0))
+            elif name.endswith("w_gate_up") or name.endswith("w_down"):
+                el = r.shape[0] // world   # expert-parallel slice
+                p.copy_(r[rank * el:(rank + 1) * el])
+            else:
+                raise AssertionError(f"unhandled shard for {name} "
+                                     f"{p.shape} vs {r.shape}")
+
+    toks = torch.tensor([7, 9, 11, 13, 15])
+    pos = torch.arange(5)
+    meta = AttnMetadata(
+        is_prefill=True,
+        slot_mapping=torch.full((5,), -1, dtype=torch.long),
+        cu_seqlens=torch.tensor([0, 5], dtype=torch.int32), max_seqlen=5)
+    out = tp(toks, pos, None, meta)
+    logits = tp.compute_logits(out[-1:])
+    st.tp_size, st.tp_rank = 1, 0
+    ref_logits = ref.compute_logits(ref(toks, pos, None, meta)[-1:])
+    st.tp_size, st.tp_rank = saved
+    assert torch.allclose(logits.float(), ref_logits.float(), atol=0.1,
+                          rtol=0.05),         f"MLA TP mismatch {(logits.float()-ref_logits.float()).abs().max()}"
+
+
 # ---- tests -----------------------------------------------------------------
 def test_tp_parallel_linear_world2():
     _spawn("_body_linear", port=29611)
@@ -202,6 +269,10 @@ def test_tp_llama_forward_world2_one_shot_fused():
 
 def test_moe_expert_parallel_world2():
     _spawn("_body_moe_ep", port=29631)
+
+
+def test_mla_deepseek_tp_world2():
+    _spawn("_body_mla_tp", port=29661)
 
 
 def test_vocab_parallel_embedding_single():
