@@ -149,17 +149,22 @@ class MLAAttention(nn.Module):
             positions, q_pe.view(T, -1), k_pe, self.rope, cos_sin)
         q_pe = q_pe.view(T, self.num_heads, self.rope)
 
-        latent = torch.cat([c_kv, k_pe], dim=-1)        # [T, r+rope]
         if kv_cache is not None:
             cache = kv_cache[0]                    # [NB+1, BS, r+rope]
-            rows = cache.view(-1, cache.size(-1))
-            # padding slots are -1 (graph buckets): redirect them to the
-            # spare trash block the allocator reserves at the end —
-            # index_copy_ has no skip semantics and -1 would fault
-            slots = meta.slot_mapping
-            slots = torch.where(
-                slots < 0, torch.full_like(slots, rows.size(0) - 1), slots)
-            rows.index_copy_(0, slots, latent.to(cache.dtype))
+            if cache.is_cuda:
+                # one fused scatter (skips -1 padding slots in-kernel);
+                # replaces a where+cat+index_copy_ chain that cost ~7%
+                # of a serving run (profiles/r02_mla_kernels.md)
+                ops.mla_cache_write(cache, c_kv, k_pe, meta.slot_mapping)
+            else:
+                rows = cache.view(-1, cache.size(-1))
+                slots = meta.slot_mapping
+                slots = torch.where(
+                    slots < 0, torch.full_like(slots, rows.size(0) - 1),
+                    slots)
+                rows.index_copy_(
+                    0, slots,
+                    torch.cat([c_kv, k_pe], dim=-1).to(cache.dtype))
 
         if meta.is_prefill:
             if meta.kv_lens is not None:
@@ -174,15 +179,17 @@ class MLAAttention(nn.Module):
     # ---- absorbed decode over the latent cache --------------------------
     def _decode(self, q_nope, q_pe, cache, meta) -> torch.Tensor:
         T = q_nope.size(0)
-        # q_c[h] = q_nope[h] @ w_kc[h]: [Hl, T, nope] x [Hl, nope, r]
-        q_c = torch.bmm(q_nope.transpose(0, 1).contiguous(), self.w_kc)
+        # q_c[h] = q_nope[h] @ w_kc[h]: [Hl, T, nope] x [Hl, nope, r].
+        # strided views feed hipBLASLt batched GEMM directly (no
+        # .contiguous() copies — profiles/r02_mla_kernels.md overhead)
+        q_c = torch.bmm(q_nope.transpose(0, 1), self.w_kc)
         q_full = torch.cat(
             [q_c.transpose(0, 1), q_pe], dim=-1).contiguous()  # [T,Hl,r+rope]
         out_c = ops.mla_decode(q_full, cache, meta.block_tables,
                                meta.seq_lens, self.scale, self.r)
         # latent -> per-head v: [Hl, T, r] x [Hl, r, v]
-        o = torch.bmm(out_c.transpose(0, 1).contiguous(), self.w_vc)
-        return o.transpose(0, 1).contiguous()           # [T, Hl, v]
+        o = torch.bmm(out_c.transpose(0, 1), self.w_vc)
+        return o.transpose(0, 1).reshape(T, -1)         # one copy at most
 
     # ---- non-absorbed whole-prompt prefill ------------------------------
     def _prefill(self, q_nope, q_pe, c_kv, k_pe, meta) -> torch.Tensor:

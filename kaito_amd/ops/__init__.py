@@ -145,6 +145,14 @@ def mla_decode(q: torch.Tensor, cache: torch.Tensor,
     return torch_ref.mla_decode(q, cache, block_tables, seq_lens, scale, r)
 
 
+def mla_cache_write(cache: torch.Tensor, c_kv: torch.Tensor,
+                    k_pe: torch.Tensor, slots: torch.Tensor) -> None:
+    """Scatter latent rows (c_kv ‖ k_pe) into the paged MLA cache;
+    negative slots (graph-bucket padding) are skipped."""
+    _require_ext()
+    torch.ops.kaito.mla_cache_write(cache, c_kv, k_pe, slots)
+
+
 def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                       cu_seqlens: torch.Tensor, scale: float,
                       max_seqlen: int | None = None, window: int = 0,

@@ -24,6 +24,8 @@ void paged_attention(at::Tensor out, at::Tensor query, at::Tensor k_cache,
                      at::Tensor seq_lens, double scale);
 void mla_decode(at::Tensor out, at::Tensor q, at::Tensor cache,
                 at::Tensor block_tables, at::Tensor seq_lens, double scale);
+void mla_cache_write(at::Tensor cache, at::Tensor c_kv, at::Tensor k_pe,
+                     at::Tensor slots);
 void paged_attention_sp(at::Tensor out, at::Tensor query, at::Tensor k_cache,
                         at::Tensor v_cache, at::Tensor block_tables,
                         at::Tensor seq_lens, double scale, int64_t window,
@@ -82,6 +84,7 @@ TORCH_LIBRARY(kaito, m) {
   m.def("paged_attention(Tensor(a!) out, Tensor query, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, float scale) -> ()");
   m.def("paged_attention_sp(Tensor(a!) out, Tensor query, Tensor k_cache, Tensor v_cache, Tensor block_tables, Tensor seq_lens, float scale, int window, Tensor sinks) -> ()");
   m.def("mla_decode(Tensor(a!) out, Tensor q, Tensor cache, Tensor block_tables, Tensor seq_lens, float scale) -> ()");
+  m.def("mla_cache_write(Tensor(a!) cache, Tensor c_kv, Tensor k_pe, Tensor slots) -> ()");
   m.def("prefill_attention(Tensor(a!) out, Tensor q, Tensor k, Tensor v, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens, float scale, int window, Tensor sinks) -> ()");
   m.def("context_attention(Tensor(a!) out, Tensor q, Tensor k_cache, Tensor v_cache, Tensor tile_seq, Tensor tile_qbase, Tensor cu_seqlens_q, Tensor kv_lens, Tensor block_tables, float scale, int window, Tensor sinks) -> ()");
   m.def("mfma_tile_gemm(Tensor a, Tensor b) -> Tensor");
@@ -114,6 +117,7 @@ TORCH_LIBRARY_IMPL(kaito, CUDA, m) {
   m.impl("paged_attention", &kaito::paged_attention);
   m.impl("paged_attention_sp", &kaito::paged_attention_sp);
   m.impl("mla_decode", &kaito::mla_decode);
+  m.impl("mla_cache_write", &kaito::mla_cache_write);
   m.impl("prefill_attention", &kaito::prefill_attention);
   m.impl("context_attention", &kaito::context_attention);
   m.impl("mfma_tile_gemm", &kaito::mfma_tile_gemm);

@@ -223,4 +223,56 @@ void mla_decode(at::Tensor out, at::Tensor q, at::Tensor cache,
   }
 }
 
+// Fused MLA cache write: latent row (c_kv ‖ k_rope) scattered to the
+// paged cache in ONE kernel. Replaces a torch where + cat + index_copy_
+// chain that showed up at ~7% of a v2-lite serving run
+// (profiles/r02_mla_kernels.md). One wave per token: 64 lanes x 16 B
+// covers the 512-dim c_kv; lanes 0-7 append the 64 rope dims.
+template <int R, int P>
+__global__ __launch_bounds__(64, 8)
+void mla_cache_write_kernel(
+    short* __restrict__ cache,          // [NB, BS, R+P] bf16
+    const short* __restrict__ c_kv,     // [T, R]
+    const short* __restrict__ k_pe,     // [T, P] (row stride pe_stride)
+    const int64_t* __restrict__ slots,  // [T]; <0 = skip (padding)
+    const int64_t pe_stride) {
+  constexpr int DT = R + P;
+  const int t = blockIdx.x;
+  const int lane = threadIdx.x;
+  const int64_t slot = slots[t];
+  if (slot < 0) return;
+  short* dst = cache + slot * DT;
+  static_assert(R == 64 * 8 && P == 8 * 8);
+  *reinterpret_cast<short8_t*>(dst + lane * 8) =
+      *reinterpret_cast<const short8_t*>(c_kv + (int64_t)t * R + lane * 8);
+  if (lane < P / 8)
+    *reinterpret_cast<short8_t*>(dst + R + lane * 8) =
+        *reinterpret_cast<const short8_t*>(k_pe + t * pe_stride + lane * 8);
+}
+
+void mla_cache_write(at::Tensor cache, at::Tensor c_kv, at::Tensor k_pe,
+                     at::Tensor slots) {
+  TORCH_CHECK(cache.is_cuda() && cache.dtype() == at::kBFloat16
+              && cache.is_contiguous());
+  TORCH_CHECK(c_kv.is_contiguous() && c_kv.dtype() == at::kBFloat16);
+  TORCH_CHECK(k_pe.stride(-1) == 1 && k_pe.dtype() == at::kBFloat16);
+  TORCH_CHECK(slots.dtype() == at::kLong);
+  const int T = c_kv.size(0);
+  const int R = c_kv.size(1);
+  const int P = k_pe.size(1);
+  TORCH_CHECK(cache.size(2) == R + P && k_pe.size(0) == T
+              && slots.size(0) == T);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (T == 0) return;
+  if (R == 512 && P == 64) {
+    hipLaunchKernelGGL((mla_cache_write_kernel<512, 64>), dim3(T), dim3(64),
+                       0, stream, (short*)cache.data_ptr(),
+                       (const short*)c_kv.data_ptr(),
+                       (const short*)k_pe.data_ptr(),
+                       slots.data_ptr<int64_t>(), k_pe.stride(0));
+  } else {
+    TORCH_CHECK(false, "unsupported MLA cache dims ", R, "+", P);
+  }
+}
+
 }  // namespace kaito

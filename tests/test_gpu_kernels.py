@@ -372,6 +372,25 @@ def test_mla_decode_kernel_matches_ref():
         _close(out, ref.to(DEV), atol=2e-2)
 
 
+def test_mla_cache_write_kernel():
+    """Fused latent-row scatter vs index_copy_, incl. -1 padding skip."""
+    torch.manual_seed(3)
+    NB, BS, R_, P_ = 8, 16, 512, 64
+    cache = torch.zeros(NB, BS, R_ + P_, dtype=torch.bfloat16, device=DEV)
+    ref = cache.clone()
+    T = 21
+    c_kv = _bf16(T, R_)
+    k_pe = _bf16(T, P_)
+    slots = torch.randperm(NB * BS, device=DEV)[:T].long()
+    slots[3] = -1
+    slots[17] = -1
+    ops.mla_cache_write(cache, c_kv, k_pe, slots)
+    keep = slots >= 0
+    ref.view(-1, R_ + P_).index_copy_(
+        0, slots[keep], torch.cat([c_kv, k_pe], -1)[keep])
+    assert torch.equal(cache, ref)
+
+
 @pytest.mark.parametrize("name", ["tiny-phi2-test", "tiny-gemma3-test",
                                   "tiny-gptoss-test", "tiny-deepseek-test"])
 def test_engine_gpu_model_variants(name):
