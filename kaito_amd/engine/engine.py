@@ -52,20 +52,13 @@ class LLMEngine:
     def __init__(self, cfg: EngineConfig, weights_path: Optional[str] = None):
         self.cfg = cfg
         if cfg.model.is_mla:
-            # MLA (deepseek): prompts are prefillled whole over the
-            # compressed latent cache — no context-attention suffix path,
-            # so no chunked prefill, prefix-cache partial hits, offload
-            # restore, or mixed (overlapped) steps (models/mla.py)
-            cfg.enable_prefix_caching = False
+            # MLA (deepseek): chunked prefill + prefix caching work over
+            # the latent cache (MLAAttention._context), but host offload
+            # stays off — the aliased (c, c) cache pair would double the
+            # offloaded bytes in KVOffloadManager (models/mla.py)
             cfg.kv_offload = False
-            cfg.enable_mixed_batch = False
         self._maybe_enable_one_shot(cfg)
         self.runner = ModelRunner(cfg).load_model(weights_path, cfg.seed)
-        if cfg.model.is_mla:
-            # whole-prompt prefill must always fit one step's budget (incl.
-            # preemption recompute of prompt+generated up to max_model_len)
-            cfg.max_num_batched_tokens = max(cfg.max_num_batched_tokens,
-                                             self.runner.max_model_len)
         self.runner.setup_tunable()
         self.runner.profile_and_allocate_kv()
         if cfg.enable_prefix_caching:

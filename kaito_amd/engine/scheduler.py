@@ -65,11 +65,10 @@ class Scheduler:
         self.running: List[Sequence] = []
         # called at admission: restore_cb(seq) -> covered prefix tokens
         self.restore_cb = restore_cb
-        # MLA models prefill prompts WHOLE (no context-attention suffix
-        # path over the latent cache): an admission that exceeds the
-        # step's remaining token budget waits for the next step instead
-        # of being chunked (models/mla.py)
-        self.whole_prompt_only = cfg.model.is_mla
+        # opt-in: admit only prompts that fit the step budget whole
+        # (kept as a scheduling mode; MLA no longer needs it — the
+        # latent-cache context path handles suffix chunks)
+        self.whole_prompt_only = False
 
     # ---- queue state (serves the rate limiter / metrics) ----
     @property

@@ -168,10 +168,11 @@ class MLAAttention(nn.Module):
 
         if meta.is_prefill:
             if meta.kv_lens is not None:
-                raise NotImplementedError(
-                    "MLA prompts are scheduled whole "
-                    "(scheduler.whole_prompt_only)")
-            out = self._prefill(q_nope, q_pe, c_kv, k_pe, meta)
+                # chunked-prefill continuation / prefix-cache restore:
+                # suffix queries attend to the paged latent cache
+                out = self._context(q_nope, q_pe, kv_cache[0], meta)
+            else:
+                out = self._prefill(q_nope, q_pe, c_kv, k_pe, meta)
         else:
             out = self._decode(q_nope, q_pe, kv_cache[0], meta)
         return self.o_proj(out.reshape(T, -1))
@@ -190,6 +191,39 @@ class MLAAttention(nn.Module):
         # latent -> per-head v: [Hl, T, r] x [Hl, r, v]
         o = torch.bmm(out_c.transpose(0, 1), self.w_vc)
         return o.transpose(0, 1).reshape(T, -1)         # one copy at most
+
+    # ---- absorbed suffix attention over the paged latent cache ----------
+    def _context(self, q_nope, q_pe, cache, meta) -> torch.Tensor:
+        """Chunked-prefill continuation (the GQA engine's
+        context_attention analog). Runs in the ABSORBED space — no K/V
+        decompression: gather each sequence's latent rows once, causal-
+        mask the suffix, accumulate in latent space. Eager torch ops
+        (prefill steps are eager; these chunks are the budget-split
+        tail, not the hot path)."""
+        BS = cache.size(1)
+        DT = cache.size(2)
+        q_c = torch.bmm(q_nope.transpose(0, 1), self.w_kc)  # [Hl, Tq, r]
+        qf = torch.cat([q_c, q_pe.transpose(0, 1)], -1).float()
+        outs = []
+        cs = meta.cu_seqlens.tolist()
+        for b in range(len(cs) - 1):
+            s0, s1 = cs[b], cs[b + 1]
+            n = s1 - s0
+            L = int(meta.kv_lens[b])
+            nb = (L + BS - 1) // BS
+            rows = cache[meta.block_tables[b, :nb].long()]
+            rows = rows.reshape(-1, DT)[:L].float()         # [L, DT]
+            att = torch.einsum("hqd,ld->hql", qf[:, s0:s1], rows)
+            att = att * self.scale
+            qpos = torch.arange(L - n, L, device=att.device)
+            mask = torch.arange(L, device=att.device).unsqueeze(0) \
+                > qpos.unsqueeze(1)                         # [n, L]
+            att = att.masked_fill(mask.unsqueeze(0), float("-inf"))
+            p = torch.softmax(att, dim=-1)
+            outs.append(torch.einsum("hql,lr->hqr", p, rows[:, :self.r]))
+        out_c = torch.cat(outs, 1).to(self.w_vc.dtype)      # [Hl, Tq, r]
+        o = torch.bmm(out_c, self.w_vc)                     # [Hl, Tq, v]
+        return o.transpose(0, 1)
 
     # ---- non-absorbed whole-prompt prefill ------------------------------
     def _prefill(self, q_nope, q_pe, c_kv, k_pe, meta) -> torch.Tensor:

@@ -82,39 +82,44 @@ def _cfg(**kw):
 
 
 def test_mla_engine_config_guards():
-    cfg = _cfg(enable_prefix_caching=True, kv_offload=True,
-               enable_mixed_batch=True)
+    cfg = _cfg(enable_prefix_caching=True, kv_offload=True)
     eng = LLMEngine(cfg)
-    # MLA disables the context-attention-dependent features
+    # host offload stays off (aliased cache pair would double the bytes);
+    # prefix caching + chunked prefill now work over the latent cache
     assert eng.kv_offload is None
-    assert not cfg.enable_prefix_caching and not cfg.enable_mixed_batch
-    assert eng.scheduler.whole_prompt_only
-    # whole-prompt budget covers max_model_len
-    assert cfg.max_num_batched_tokens >= eng.runner.max_model_len
-    # latent cache: aliased (c, c) pair, [NB, BS, r+rope]
+    assert cfg.enable_prefix_caching
+    # latent cache: aliased (c, c) pair, [NB+1, BS, r+rope]
     k, v = eng.runner.kv_caches[0]
     assert k.data_ptr() == v.data_ptr()
     assert k.shape[-1] == cfg.model.kv_cache_row
     assert k.dim() == 3
 
 
-def test_mla_whole_prompt_admission_defers_over_budget():
-    """A prompt that exceeds the step budget waits instead of chunking."""
-    cfg = _cfg()
+def test_mla_chunked_prefill_matches_whole():
+    """A prompt longer than the step budget is CHUNKED (latent-cache
+    context attention, MLAAttention._context) and must produce the same
+    greedy tokens as an engine with a budget that fits it whole."""
+    cfg = _cfg(max_num_batched_tokens=32, max_model_len=96)
     eng = LLMEngine(cfg)
-    cfg.max_num_batched_tokens = 32     # force a tiny budget post-init
-    sp = SamplingParams(max_tokens=2, ignore_eos=True)
-    eng.add_request(list(range(2, 50)), sp)     # 48 > 32: never admitted...
-    eng.add_request([5, 6, 7], sp)              # ...but this one runs
-    cfg.max_num_batched_tokens = 64
-    done = []
-    for _ in range(30):
-        done += eng.step()
-        if len(done) == 2:
-            break
-    assert len(done) == 2
-    for s in done:
-        assert len(s.output_token_ids) == 2
+    prompt = list(range(2, 80))                 # 78 tokens -> 3 chunks
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+    got = eng.generate([prompt], sp)[0].output_token_ids
+    cfg2 = _cfg(max_num_batched_tokens=512, max_model_len=96)
+    eng2 = LLMEngine(cfg2)
+    want = eng2.generate([prompt], sp)[0].output_token_ids
+    assert got == want
+
+
+def test_mla_prefix_cache_hit_matches_cold():
+    """Prefix-cache revival over the latent cache: a repeated prompt
+    skips prefill (covered blocks) and still decodes identically."""
+    cfg = _cfg(enable_prefix_caching=True)
+    eng = LLMEngine(cfg)
+    prompt = list(range(3, 3 + 48))             # 3 full blocks
+    sp = SamplingParams(max_tokens=5, ignore_eos=True)
+    first = eng.generate([prompt], sp)[0].output_token_ids
+    again = eng.generate([prompt], sp)[0].output_token_ids
+    assert first == again
 
 
 def test_mla_kv_bytes_per_token():
