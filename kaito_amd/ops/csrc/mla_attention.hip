@@ -38,6 +38,7 @@
 // the launch underfills 256 CUs; a flash-decoding token-split tier is
 // the designated fix (docs/ROADMAP.md).
 #include "common.h"
+#include <cstdlib>
 #include <torch/library.h>
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
@@ -46,8 +47,14 @@ namespace kaito {
 
 typedef short short4_t __attribute__((ext_vector_type(4)));
 
-template <int R, int P, int BS>
-__global__ __launch_bounds__(256, 4)
+// OCC: workgroups/CU. LDS (19.3 KB/wg) allows 8; VGPRs decide the
+// real bound — selectable via KAITO_MLA_OCC for A/B on hardware.
+// DB: software double-buffer THROUGH REGISTERS — the next block's
+// global loads (5 x b128 per thread) issue before the current block's
+// compute, hiding the HBM latency the single-buffer stage exposes at
+// every __syncthreads; LDS footprint unchanged (KAITO_MLA_DB).
+template <int R, int P, int BS, int OCC = 4, bool DB = false>
+__global__ __launch_bounds__(256, OCC)
 void mla_decode_kernel(
     short* __restrict__ out,            // [T, H, R] bf16
     const short* __restrict__ q,        // [T, H, R+P] bf16
@@ -59,7 +66,8 @@ void mla_decode_kernel(
   constexpr int SL = DT / 16;           // 36 dims per score slice
   constexpr int AD = R / 16;            // 32 dims per accum slice
   constexpr int HT = 16;                // heads per workgroup
-  static_assert(DT % 16 == 0 && R % 16 == 0 && SL % 4 == 0 && AD % 8 == 0);
+  static_assert(DT % 16 == 0 && R % 16 == 0 && SL % 12 == 0
+                && AD % 8 == 0);
 
   const int seq = blockIdx.x;
   const int h0 = blockIdx.y * HT;
@@ -107,33 +115,82 @@ void mla_decode_kernel(
 
   typedef __bf16 bf16x2_t __attribute__((ext_vector_type(2)));
 
+  constexpr int NCH = (BS * DT / 8 + 255) / 256;   // b128 chunks/thread
+  short8_t pf[DB ? NCH : 1];
+  if constexpr (DB) {   // prologue: block 0 loads in flight
+    const short8_t* src = reinterpret_cast<const short8_t*>(
+        cache + (int64_t)bt[0] * (BS * DT));
+#pragma unroll
+    for (int c = 0; c < NCH; c++) {
+      const int idx = c * 256 + tid;
+      if (idx < BS * DT / 8) pf[c] = src[idx];
+    }
+  }
+
   for (int b = 0; b < nblocks; b++) {
-    const int blk = bt[b];
     // ---- stage one KV block: 1152 x b128, 256 threads ----
-    {
-      const short8_t* src = reinterpret_cast<const short8_t*>(
-          cache + (int64_t)blk * (BS * DT));
+    if constexpr (DB) {
       short8_t* dst = reinterpret_cast<short8_t*>(s_kv);
 #pragma unroll
-      for (int c = 0; c < (BS * DT / 8 + 255) / 256; c++) {
+      for (int c = 0; c < NCH; c++) {
+        const int idx = c * 256 + tid;
+        if (idx < BS * DT / 8) dst[idx] = pf[c];
+      }
+    } else {
+      const short8_t* src = reinterpret_cast<const short8_t*>(
+          cache + (int64_t)bt[b] * (BS * DT));
+      short8_t* dst = reinterpret_cast<short8_t*>(s_kv);
+#pragma unroll
+      for (int c = 0; c < NCH; c++) {
         const int idx = c * 256 + tid;
         if (idx < BS * DT / 8) dst[idx] = src[idx];
       }
     }
     __syncthreads();
+    if constexpr (DB) {   // issue NEXT block's loads before computing
+      if (b + 1 < nblocks) {
+        const short8_t* src = reinterpret_cast<const short8_t*>(
+            cache + (int64_t)bt[b + 1] * (BS * DT));
+#pragma unroll
+        for (int c = 0; c < NCH; c++) {
+          const int idx = c * 256 + tid;
+          if (idx < BS * DT / 8) pf[c] = src[idx];
+        }
+      }
+    }
 
     // ---- scores for the 16 tokens (intra-wave handoff to exp) ----
+    // 3 independent dot accumulators per token + 2-token unroll: the
+    // naive single-accumulator form is an 18-deep serial v_dot2 chain
+    // (~170 dependent cycles/token) that stalls the SIMD
+#pragma unroll 2
     for (int t = 0; t < BS; t++) {
       const short* kp = s_kv + t * DT + slice * SL;
-      float part = 0.f;
+      float p0 = 0.f, p1 = 0.f, p2 = 0.f;
 #pragma unroll
-      for (int j = 0; j < SL / 4; j++) {
-        const short4_t kv4 = *reinterpret_cast<const short4_t*>(kp + j * 4);
-        const bf16x2_t* k2 = reinterpret_cast<const bf16x2_t*>(&kv4);
-        const bf16x2_t* q2 = reinterpret_cast<const bf16x2_t*>(&qv[j]);
-        part = __builtin_amdgcn_fdot2_f32_bf16(q2[0], k2[0], part, false);
-        part = __builtin_amdgcn_fdot2_f32_bf16(q2[1], k2[1], part, false);
+      for (int j = 0; j < SL / 12; j++) {
+        const short4_t a = *reinterpret_cast<const short4_t*>(kp + j * 12);
+        const short4_t bq = *reinterpret_cast<const short4_t*>(
+            kp + j * 12 + 4);
+        const short4_t c = *reinterpret_cast<const short4_t*>(
+            kp + j * 12 + 8);
+        const bf16x2_t* ka = reinterpret_cast<const bf16x2_t*>(&a);
+        const bf16x2_t* kb = reinterpret_cast<const bf16x2_t*>(&bq);
+        const bf16x2_t* kc = reinterpret_cast<const bf16x2_t*>(&c);
+        const bf16x2_t* qa = reinterpret_cast<const bf16x2_t*>(
+            &qv[j * 3]);
+        const bf16x2_t* qb = reinterpret_cast<const bf16x2_t*>(
+            &qv[j * 3 + 1]);
+        const bf16x2_t* qc = reinterpret_cast<const bf16x2_t*>(
+            &qv[j * 3 + 2]);
+        p0 = __builtin_amdgcn_fdot2_f32_bf16(qa[0], ka[0], p0, false);
+        p1 = __builtin_amdgcn_fdot2_f32_bf16(qb[0], kb[0], p1, false);
+        p2 = __builtin_amdgcn_fdot2_f32_bf16(qc[0], kc[0], p2, false);
+        p0 = __builtin_amdgcn_fdot2_f32_bf16(qa[1], ka[1], p0, false);
+        p1 = __builtin_amdgcn_fdot2_f32_bf16(qb[1], kb[1], p1, false);
+        p2 = __builtin_amdgcn_fdot2_f32_bf16(qc[1], kc[1], p2, false);
       }
+      float part = (p0 + p1) + p2;
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1)
         part += __shfl_xor(part, off, 64);
@@ -165,18 +222,25 @@ void mla_decode_kernel(
     }
 
     // ---- accumulate c_kv into the latent-space output ----
+    // packed v_pk_fma_f32 (2 fma/instr) like paged_attention_sp phase C
     {
+      float2_t* a2 = reinterpret_cast<float2_t*>(acc);
+      const float2_t corr2 = {corr, corr};
 #pragma unroll
-      for (int j = 0; j < AD; j++) acc[j] *= corr;
+      for (int j = 0; j < AD / 2; j++) a2[j] *= corr2;
       for (int t = 0; t < BS; t++) {
         const float p = s_p[t][ac_h];
+        const float2_t p2 = {p, p};
         const short* vp = s_kv + t * DT + ac_s * AD;
 #pragma unroll
         for (int c = 0; c < AD / 8; c++) {
           const short8_t v8 = *reinterpret_cast<const short8_t*>(vp + c * 8);
+          float vreg[8];
 #pragma unroll
-          for (int j = 0; j < 8; j++)
-            acc[c * 8 + j] = fmaf(p, bf16_to_f32(v8[j]), acc[c * 8 + j]);
+          for (int j = 0; j < 8; j++) vreg[j] = bf16_to_f32(v8[j]);
+          const float2_t* v2 = reinterpret_cast<const float2_t*>(vreg);
+#pragma unroll
+          for (int k = 0; k < 4; k++) a2[c * 4 + k] += p2 * v2[k];
         }
       }
     }
@@ -210,13 +274,39 @@ void mla_decode(at::Tensor out, at::Tensor q, at::Tensor cache,
   auto stream = at::hip::getCurrentHIPStream();
   if (T == 0) return;
   const dim3 grid(T, (H + 15) / 16);
+  static const int occ_env = []() {
+    const char* e = getenv("KAITO_MLA_OCC");
+    return e ? atoi(e) : 0;
+  }();
+  // register double-buffer measured faster at every shape (247.9 vs
+  // 258.8 us bs=768, 183.5 vs 209.7 bs=256) — default ON
+  static const bool db_env = []() {
+    const char* e = getenv("KAITO_MLA_DB");
+    return e == nullptr || atoi(e) != 0;
+  }();
   if (R == 512 && DT == 576) {
-    hipLaunchKernelGGL((mla_decode_kernel<512, 64, 16>), grid, dim3(256), 0,
-                       stream, (short*)out.data_ptr(),
-                       (const short*)q.data_ptr(),
-                       (const short*)cache.data_ptr(),
-                       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                       (float)scale, H, max_blocks);
+#define MLA_LAUNCH(OCC_, DB_)                                                 \
+    hipLaunchKernelGGL((mla_decode_kernel<512, 64, 16, OCC_, DB_>), grid,      \
+                       dim3(256), 0, stream, (short*)out.data_ptr(),          \
+                       (const short*)q.data_ptr(),                            \
+                       (const short*)cache.data_ptr(),                        \
+                       block_tables.data_ptr<int>(),                          \
+                       seq_lens.data_ptr<int>(), (float)scale, H, max_blocks)
+    if (db_env) {
+      switch (occ_env) {
+        case 5: MLA_LAUNCH(5, true); break;
+        case 6: MLA_LAUNCH(6, true); break;
+        default: MLA_LAUNCH(4, true); break;
+      }
+    } else {
+      switch (occ_env) {
+        case 5: MLA_LAUNCH(5, false); break;
+        case 6: MLA_LAUNCH(6, false); break;
+        case 8: MLA_LAUNCH(8, false); break;
+        default: MLA_LAUNCH(4, false); break;
+      }
+    }
+#undef MLA_LAUNCH
   } else {
     TORCH_CHECK(false, "unsupported MLA dims r=", R, " r+rope=", DT,
                 " (deepseek family is 512/576)");
