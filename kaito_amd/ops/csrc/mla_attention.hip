@@ -53,7 +53,19 @@ typedef short short4_t __attribute__((ext_vector_type(4)));
 // global loads (5 x b128 per thread) issue before the current block's
 // compute, hiding the HBM latency the single-buffer stage exposes at
 // every __syncthreads; LDS footprint unchanged (KAITO_MLA_DB).
-template <int R, int P, int BS, int OCC = 4, bool DB = false>
+// PH: diagnostic phase mask (1 = stage only, 2 = +scores, 3 = full) —
+// timing ablation via KAITO_MLA_PH; output is garbage for PH<3.
+// MF: MFMA score phase — C[16h x 16t] via mfma_f32_16x16x32_bf16 with
+// the 18 K-steps split 4/5/4/5 across the 4 waves (partials reduced
+// through LDS), Q K-range PRELOADED per wave (Q is block-invariant),
+// s_kv rows padded +16 B so B-fragment b128 reads hit the 2-cycle
+// LDS minimum instead of a 16-way bank conflict (KAITO_MLA_MF).
+// MV: MFMA V-accumulate — out[16h x 512d] += P[16h x 16t(+16 zero-pad
+// k)] @ C_kv[16t x 512d] via 8 d-tiles per wave; V staged TRANSPOSED
+// (s_vt[dim][token]) during the cooperative copy so B-fragments are
+// contiguous b128 reads (same trick as prefill_attention); requires MF.
+template <int R, int P, int BS, int OCC = 4, bool DB = false, int PH = 3,
+          bool MF = false, bool MV = false>
 __global__ __launch_bounds__(256, OCC)
 void mla_decode_kernel(
     short* __restrict__ out,            // [T, H, R] bf16
@@ -85,21 +97,45 @@ void mla_decode_kernel(
   const int ac_h = tid >> 4;                      // == sc_head
   const int ac_s = tid & 15;
 
-  __shared__ short s_kv[BS * DT];                 // 18 KB staged block
+  constexpr int RS = MF ? DT + 8 : DT;   // padded row (bank spread)
+  constexpr int TP = 20;                 // s_vt token stride (16+4 pad)
+  __shared__ short s_kv[BS * RS];                 // staged block
   __shared__ float s_p[BS][HT];                   // scores→probs
+  __shared__ float s_part[MF ? 4 : 1][HT][HT];    // per-wave C partials
+  __shared__ short s_vt[MV ? R * TP : 1];         // V transposed (20.5KB)
+  __shared__ short s_pb[MV ? HT : 1][TP];         // P row-major bf16
+  __shared__ float s_corr[HT], s_l[HT];
 
   // per-head online-softmax state, held REDUNDANTLY in registers by all
   // 16 lanes of the head's thread group (every lane derives identical
   // values from the same shfl reductions) — no cross-lane LDS handoff
   float M = -1e30f, l = 0.f;
 
-  // q slice (36 bf16 = 9 x b64) for the score phase; zero for padded
-  // heads so their scores stay finite (their output is never written)
-  short4_t qv[SL / 4];
-  const bool head_ok = h0 + sc_head < H;
-  {
-    // clamp the ROW (not just the value) so a potentially if-converted
-    // load never dereferences past the q tensor for padded head tiles
+  constexpr int KSTEPS = DT / 32;                 // MFMA K-chunks (18)
+  constexpr int KW = (KSTEPS + 3) / 4 + 1;        // per-wave upper bound
+  const int kb0 = (KSTEPS * wave) / 4;            // ragged 4/5/4/5 split
+  const int kb1 = (KSTEPS * (wave + 1)) / 4;
+  const int hi = lane >> 4;
+
+  // q for the score phase; zero for padded heads so their scores stay
+  // finite (their output is never written). Rows are CLAMPED (not just
+  // the value) so an if-converted load can't read past the q tensor.
+  short4_t qv[MF ? 1 : SL / 4];
+  bf16x8 af[MF ? KW : 1];
+  if constexpr (MF) {
+    // A-fragment rows: head = lane&15; k = kc*32 + (lane>>4)*8 + [0..7]
+    const bool ok = h0 + (lane & 15) < H;
+    const int hq = min(h0 + (lane & 15), H - 1);
+    const short* qp = q + ((int64_t)seq * H + hq) * DT;
+#pragma unroll
+    for (int j = 0; j < KW; j++) {
+      if (kb0 + j >= kb1) break;
+      af[j] = *reinterpret_cast<const bf16x8*>(
+          qp + (kb0 + j) * 32 + hi * 8);
+      if (!ok) af[j] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  } else {
+    const bool head_ok = h0 + sc_head < H;
     const int hq = min(h0 + sc_head, H - 1);
     const short* qp = q + ((int64_t)seq * H + hq) * DT + slice * SL;
 #pragma unroll
@@ -109,9 +145,15 @@ void mla_decode_kernel(
     }
   }
 
-  float acc[AD];
+  float acc[MV ? 1 : AD];
+  f32x4 accv[MV ? 8 : 1];
+  if constexpr (MV) {
 #pragma unroll
-  for (int j = 0; j < AD; j++) acc[j] = 0.f;
+    for (int j = 0; j < 8; j++) accv[j] = f32x4{0.f, 0.f, 0.f, 0.f};
+  } else {
+#pragma unroll
+    for (int j = 0; j < AD; j++) acc[j] = 0.f;
+  }
 
   typedef __bf16 bf16x2_t __attribute__((ext_vector_type(2)));
 
@@ -130,20 +172,33 @@ void mla_decode_kernel(
   for (int b = 0; b < nblocks; b++) {
     // ---- stage one KV block: 1152 x b128, 256 threads ----
     if constexpr (DB) {
-      short8_t* dst = reinterpret_cast<short8_t*>(s_kv);
 #pragma unroll
       for (int c = 0; c < NCH; c++) {
         const int idx = c * 256 + tid;
-        if (idx < BS * DT / 8) dst[idx] = pf[c];
+        if (idx < BS * DT / 8) {
+          *reinterpret_cast<short8_t*>(
+              s_kv + (idx / (DT / 8)) * RS + (idx % (DT / 8)) * 8) = pf[c];
+          if constexpr (MV) {
+            const int tok = idx / (DT / 8);
+            const int d0 = (idx % (DT / 8)) * 8;
+            if (d0 < R) {
+#pragma unroll
+              for (int j = 0; j < 8; j++)
+                s_vt[(d0 + j) * TP + tok] = pf[c][j];
+            }
+          }
+        }
       }
     } else {
       const short8_t* src = reinterpret_cast<const short8_t*>(
           cache + (int64_t)bt[b] * (BS * DT));
-      short8_t* dst = reinterpret_cast<short8_t*>(s_kv);
 #pragma unroll
       for (int c = 0; c < NCH; c++) {
         const int idx = c * 256 + tid;
-        if (idx < BS * DT / 8) dst[idx] = src[idx];
+        if (idx < BS * DT / 8)
+          *reinterpret_cast<short8_t*>(
+              s_kv + (idx / (DT / 8)) * RS + (idx % (DT / 8)) * 8) =
+              src[idx];
       }
     }
     __syncthreads();
@@ -163,9 +218,25 @@ void mla_decode_kernel(
     // 3 independent dot accumulators per token + 2-token unroll: the
     // naive single-accumulator form is an 18-deep serial v_dot2 chain
     // (~170 dependent cycles/token) that stalls the SIMD
+    if constexpr (MF) {
+      f32x4 cp = {0.f, 0.f, 0.f, 0.f};
+      const int tcol = lane & 15;
+#pragma unroll
+      for (int j = 0; j < KW; j++) {
+        if (kb0 + j >= kb1) break;
+        const bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+            s_kv + tcol * RS + (kb0 + j) * 32 + hi * 8);
+        cp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[j], bfr, cp,
+                                                     0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; r++)
+        s_part[wave][hi * 4 + r][tcol] = cp[r];
+      __syncthreads();                 // cross-wave partial handoff
+    } else if constexpr (PH >= 2) {
 #pragma unroll 2
     for (int t = 0; t < BS; t++) {
-      const short* kp = s_kv + t * DT + slice * SL;
+      const short* kp = s_kv + t * RS + slice * SL;
       float p0 = 0.f, p1 = 0.f, p2 = 0.f;
 #pragma unroll
       for (int j = 0; j < SL / 12; j++) {
@@ -199,12 +270,20 @@ void mla_decode_kernel(
         s_p[t][sc_head] = valid ? part * scale : -1e30f;
       }
     }
+    }
 
     // ---- online softmax update (thread (h, t); state in registers) ----
-    float corr;
-    {
+    float corr = 1.f;
+    if constexpr (PH >= 3) {
       const int t = ac_s;
-      float s = s_p[t][ac_h];
+      float s;
+      if constexpr (MF) {
+        s = ((s_part[0][ac_h][t] + s_part[1][ac_h][t])
+             + (s_part[2][ac_h][t] + s_part[3][ac_h][t])) * scale;
+        if (b * BS + t >= seq_len) s = -1e30f;
+      } else {
+        s = s_p[t][ac_h];
+      }
       float bm = s;
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1)
@@ -219,11 +298,36 @@ void mla_decode_kernel(
       corr = Mnew > M ? __expf(M - Mnew) : 1.f;
       l = l * corr + bsum;
       M = Mnew;
+      if constexpr (MV) {
+        s_pb[ac_h][t] = f32_to_bf16(p);
+        if (t == 0) { s_corr[ac_h] = corr; s_l[ac_h] = l; }
+      }
     }
+    if constexpr (MV) __syncthreads();   // s_pb/s_corr to all waves
 
     // ---- accumulate c_kv into the latent-space output ----
-    // packed v_pk_fma_f32 (2 fma/instr) like paged_attention_sp phase C
-    {
+    if constexpr (MV) {
+      // rescale C fragments: row hi*4+r is head hi*4+r
+      float cr[4];
+#pragma unroll
+      for (int r = 0; r < 4; r++) cr[r] = s_corr[hi * 4 + r];
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+#pragma unroll
+        for (int r = 0; r < 4; r++) accv[j][r] *= cr[r];
+      const int tok0 = (hi & 1) * 8;    // hi>=2: A is zero, B ignored
+      const bf16x8 pa = hi < 2
+          ? *reinterpret_cast<const bf16x8*>(&s_pb[lane & 15][hi * 8])
+          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+      for (int j = 0; j < 8; j++) {     // 8 dim-tiles per wave
+        const int dim = (wave * 8 + j) * 16 + (lane & 15);
+        const bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+            &s_vt[dim * TP + tok0]);
+        accv[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, accv[j],
+                                                          0, 0, 0);
+      }
+    } else if constexpr (PH >= 3) {
       float2_t* a2 = reinterpret_cast<float2_t*>(acc);
       const float2_t corr2 = {corr, corr};
 #pragma unroll
@@ -231,7 +335,7 @@ void mla_decode_kernel(
       for (int t = 0; t < BS; t++) {
         const float p = s_p[t][ac_h];
         const float2_t p2 = {p, p};
-        const short* vp = s_kv + t * DT + ac_s * AD;
+        const short* vp = s_kv + t * RS + ac_s * AD;
 #pragma unroll
         for (int c = 0; c < AD / 8; c++) {
           const short8_t v8 = *reinterpret_cast<const short8_t*>(vp + c * 8);
@@ -247,7 +351,30 @@ void mla_decode_kernel(
     __syncthreads();   // s_kv free for the next block's staging
   }
 
+  if constexpr (PH < 3) {
+    // ablation variants: fake consumer keeps the staging/score stores
+    // alive (never taken at runtime — seq_len >= 1)
+    if (seq_len < 0)
+      out[tid] = s_kv[tid] + (short)s_p[tid & 15][tid >> 4];
+    return;
+  }
+
   // ---- epilogue: normalize + write [T, H, R] ----
+  if constexpr (MV) {
+    // C-fragment layout: row hi*4+r = head, col = dim-within-tile
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      const int dim = (wave * 8 + j) * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const int h = hi * 4 + r;
+        if (h0 + h < H)
+          out[((int64_t)seq * H + (h0 + h)) * R + dim] =
+              f32_to_bf16(accv[j][r] / fmaxf(s_l[h], 1e-20f));
+      }
+    }
+    return;
+  }
   if (h0 + ac_h < H) {
     const float inv = 1.f / fmaxf(l, 1e-20f);
     short* op = out + ((int64_t)seq * H + (h0 + ac_h)) * R + ac_s * AD;
@@ -292,7 +419,51 @@ void mla_decode(at::Tensor out, at::Tensor q, at::Tensor cache,
                        (const short*)cache.data_ptr(),                        \
                        block_tables.data_ptr<int>(),                          \
                        seq_lens.data_ptr<int>(), (float)scale, H, max_blocks)
-    if (db_env) {
+    static const int ph_env = []() {
+      const char* e = getenv("KAITO_MLA_PH");
+      return e ? atoi(e) : 3;
+    }();
+    // MFMA scores measured 159 vs 245 us (bs=768), 86 vs 176 (bs=256),
+    // 184 vs 295 (bs=1024) — default ON (KAITO_MLA_MF=0 reverts)
+    static const bool mf_env = []() {
+      const char* e = getenv("KAITO_MLA_MF");
+      return e == nullptr || atoi(e) != 0;
+    }();
+    static const bool mv_env = []() {
+      const char* e = getenv("KAITO_MLA_MV");
+      return e && atoi(e) != 0;
+    }();
+    if (ph_env == 1) {
+      hipLaunchKernelGGL((mla_decode_kernel<512, 64, 16, 4, true, 1>), grid,
+                         dim3(256), 0, stream, (short*)out.data_ptr(),
+                         (const short*)q.data_ptr(),
+                         (const short*)cache.data_ptr(),
+                         block_tables.data_ptr<int>(),
+                         seq_lens.data_ptr<int>(), (float)scale, H,
+                         max_blocks);
+    } else if (ph_env == 2) {
+      hipLaunchKernelGGL((mla_decode_kernel<512, 64, 16, 4, true, 2>), grid,
+                         dim3(256), 0, stream, (short*)out.data_ptr(),
+                         (const short*)q.data_ptr(),
+                         (const short*)cache.data_ptr(),
+                         block_tables.data_ptr<int>(),
+                         seq_lens.data_ptr<int>(), (float)scale, H,
+                         max_blocks);
+    } else if (mf_env && mv_env) {
+      hipLaunchKernelGGL(
+          (mla_decode_kernel<512, 64, 16, 4, true, 3, true, true>), grid,
+          dim3(256), 0, stream, (short*)out.data_ptr(),
+          (const short*)q.data_ptr(), (const short*)cache.data_ptr(),
+          block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
+          (float)scale, H, max_blocks);
+    } else if (mf_env) {
+      hipLaunchKernelGGL(
+          (mla_decode_kernel<512, 64, 16, 4, true, 3, true>), grid,
+          dim3(256), 0, stream, (short*)out.data_ptr(),
+          (const short*)q.data_ptr(), (const short*)cache.data_ptr(),
+          block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
+          (float)scale, H, max_blocks);
+    } else if (db_env) {
       switch (occ_env) {
         case 5: MLA_LAUNCH(5, true); break;
         case 6: MLA_LAUNCH(6, true); break;

@@ -18,9 +18,12 @@ def run_one():
     torch.manual_seed(0)
     dev = "cuda"
     tag = os.environ.get("KAITO_MLA_OCC", "dflt") + \
-        ("+db" if os.environ.get("KAITO_MLA_DB") else "")
+        ("+db" if os.environ.get("KAITO_MLA_DB") else "") + \
+        ("+ph" + os.environ.get("KAITO_MLA_PH")
+         if os.environ.get("KAITO_MLA_PH") else "") + \
+        ("+mf" if os.environ.get("KAITO_MLA_MF") else "")
     DT = R_ + P_
-    first = True
+    first = os.environ.get("KAITO_MLA_PH", "3") == "3"
     for bs, L in SHAPES:
         bps = (L + BS - 1) // BS
         nblocks = bs * bps + 1
@@ -56,11 +59,7 @@ def main():
     if "--sweep" not in sys.argv:
         run_one()
         return
-    variants = [{"KAITO_MLA_OCC": "4"}, {"KAITO_MLA_OCC": "5"},
-                {"KAITO_MLA_OCC": "6"}, {"KAITO_MLA_OCC": "8"},
-                {"KAITO_MLA_OCC": "4", "KAITO_MLA_DB": "1"},
-                {"KAITO_MLA_OCC": "5", "KAITO_MLA_DB": "1"},
-                {"KAITO_MLA_OCC": "6", "KAITO_MLA_DB": "1"}]
+    variants = [{}, {"KAITO_MLA_MF": "1"}]
     for v in variants:
         env = {k: val for k, val in os.environ.items()
                if not k.startswith("KAITO_MLA")}
