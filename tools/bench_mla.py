@@ -21,7 +21,8 @@ def run_one():
         ("+db" if os.environ.get("KAITO_MLA_DB") else "") + \
         ("+ph" + os.environ.get("KAITO_MLA_PH")
          if os.environ.get("KAITO_MLA_PH") else "") + \
-        ("+mf" if os.environ.get("KAITO_MLA_MF") else "")
+        ("+mf" if os.environ.get("KAITO_MLA_MF") else "") + \
+        ("+mv" if os.environ.get("KAITO_MLA_MV") else "")
     DT = R_ + P_
     first = os.environ.get("KAITO_MLA_PH", "3") == "3"
     for bs, L in SHAPES:
@@ -59,7 +60,7 @@ def main():
     if "--sweep" not in sys.argv:
         run_one()
         return
-    variants = [{}, {"KAITO_MLA_MF": "1"}]
+    variants = [{}, {"KAITO_MLA_MV": "1"}]
     for v in variants:
         env = {k: val for k, val in os.environ.items()
                if not k.startswith("KAITO_MLA")}
