@@ -60,7 +60,9 @@ def main():
     if "--sweep" not in sys.argv:
         run_one()
         return
-    variants = [{}, {"KAITO_MLA_MV": "1"}]
+    # default (MF+MV) vs regression references
+    variants = [{}, {"KAITO_MLA_MV": "0"},
+                {"KAITO_MLA_MF": "0", "KAITO_MLA_MV": "0"}]
     for v in variants:
         env = {k: val for k, val in os.environ.items()
                if not k.startswith("KAITO_MLA")}
