@@ -81,10 +81,19 @@ def main():
         import dataclasses
         mc = dataclasses.replace(mc, quant_method=args.quantization)
     max_len = args.in_tokens + args.out_tokens + 16
+    extra = {}
+    if args.max_num_seqs > 2048:
+        # extend the decode-graph buckets past the default ladder so
+        # saturated steps still replay a captured graph
+        from kaito_amd.engine.config import EngineConfig as _EC
+        base = _EC.__dataclass_fields__["graph_batch_sizes"].default
+        extra["graph_batch_sizes"] = tuple(
+            list(base) + [args.max_num_seqs])
     cfg = EngineConfig(
         model=mc,
         device="cuda" if torch.cuda.is_available() else "cpu",
         max_num_seqs=args.max_num_seqs,
+        **extra,
         max_model_len=max_len,
         tensor_parallel_size=args.tp,
         enforce_eager=args.eager,
