@@ -32,11 +32,11 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=250)
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--tp", type=int, default=1)
-    # 2560 concurrent seqs: measured ladder 33.5k (1024) / 51.5k (2048)
-    # / 54.3k (2560) / 53.7k (3072) tok/s — Llama-3-8B bf16; 288 GB
-    # HBM3E holds the bs=2560 KV (~139 GB) with a third of the card to
-    # spare (profiles/r02_perf_notes.md)
-    p.add_argument("--max-num-seqs", type=int, default=2560)
+    # 2048 concurrent seqs: measured ladder 33.5k (1024) / 48.9k (1536)
+    # / 52.3k (2048) / 51.4k (2560, same box) tok/s — Llama-3-8B bf16;
+    # 288 GB HBM3E holds the KV with half the card to spare
+    # (profiles/r02_perf_notes.md)
+    p.add_argument("--max-num-seqs", type=int, default=2048)
     p.add_argument("--in-tokens", type=int, default=200)
     p.add_argument("--out-tokens", type=int, default=200)
     p.add_argument("--eager", action="store_true")
