@@ -7,10 +7,15 @@ Sources:
   file://path or plain path  → used in place
   http(s)://...              → chunked streaming download with progress
                                callbacks (drives kaito_model_download_*)
-  az://account/container/blob → resolved to the blob HTTPS endpoint; a SAS
-                               token from AZURE_STORAGE_SAS_TOKEN (the
-                               fetch-sas init-container contract) is
-                               appended
+  az://account/container/blob → resolved to the blob HTTPS endpoint;
+                               auth is SAS (AZURE_STORAGE_SAS_TOKEN, the
+                               fetch-sas init-container contract) or AKS
+                               WORKLOAD IDENTITY: the projected federated
+                               token (AZURE_FEDERATED_TOKEN_FILE +
+                               AZURE_CLIENT_ID + AZURE_TENANT_ID) is
+                               exchanged at AAD for a storage bearer
+                               token (client_assertion grant) and sent as
+                               Authorization: Bearer
 Downloads go to a local cache dir (the NVMe PVC mount in-cluster).
 """
 from __future__ import annotations
@@ -36,12 +41,55 @@ def resolve_azure_url(url: str) -> str:
     return base
 
 
+def azure_workload_identity_token(
+        scope: str = "https://storage.azure.com/.default") -> Optional[str]:
+    """AKS workload-identity federation: exchange the projected service-
+    account token for an AAD access token (the reference streams az://
+    weights under workload identity when no SAS is provisioned). Returns
+    None when the identity env is absent. AZURE_AUTHORITY_HOST overrides
+    the login endpoint (tests point it at a local mock)."""
+    token_file = os.environ.get("AZURE_FEDERATED_TOKEN_FILE")
+    client_id = os.environ.get("AZURE_CLIENT_ID")
+    tenant = os.environ.get("AZURE_TENANT_ID")
+    if not (token_file and client_id and tenant):
+        return None
+    import httpx
+    assertion = Path(token_file).read_text().strip()
+    authority = os.environ.get("AZURE_AUTHORITY_HOST",
+                               "https://login.microsoftonline.com")
+    r = httpx.post(
+        f"{authority.rstrip('/')}/{tenant}/oauth2/v2.0/token",
+        data={
+            "grant_type": "client_credentials",
+            "client_id": client_id,
+            "scope": scope,
+            "client_assertion_type":
+                "urn:ietf:params:oauth:client-assertion-type:jwt-bearer",
+            "client_assertion": assertion,
+        }, timeout=30)
+    r.raise_for_status()
+    return r.json()["access_token"]
+
+
+def azure_auth_headers() -> dict:
+    """Bearer headers for blob GETs when workload identity is active and
+    no SAS token is configured (SAS rides the URL instead)."""
+    if os.environ.get("AZURE_STORAGE_SAS_TOKEN"):
+        return {}
+    tok = azure_workload_identity_token()
+    if tok is None:
+        return {}
+    return {"Authorization": f"Bearer {tok}", "x-ms-version": "2021-08-06"}
+
+
 def _download_http(url: str, dest: Path, progress: Optional[ProgressCb],
-                   chunk_bytes: int = 8 << 20) -> Path:
+                   chunk_bytes: int = 8 << 20,
+                   headers: Optional[dict] = None) -> Path:
     import httpx
     dest.parent.mkdir(parents=True, exist_ok=True)
     tmp = dest.with_suffix(dest.suffix + ".part")
-    with httpx.stream("GET", url, follow_redirects=True, timeout=600) as r:
+    with httpx.stream("GET", url, follow_redirects=True, timeout=600,
+                      headers=headers or {}) as r:
         r.raise_for_status()
         total = int(r.headers.get("content-length", 0)) or None
         got = 0
@@ -70,9 +118,11 @@ def fetch_weights(source: str, cache_dir: str = "/workspace/weights",
         if progress:
             progress(1.0)
         return path
+    headers = None
     if p.scheme == "az":
         source = resolve_azure_url(source)
         p = urlparse(source)
+        headers = azure_auth_headers() or None
     if p.scheme in ("http", "https"):
         base = source.rstrip("/")
         names = files or ["model.safetensors", "config.json",
@@ -84,7 +134,8 @@ def fetch_weights(source: str, cache_dir: str = "/workspace/weights",
                 if progress:
                     progress((i + frac) / n)
             try:
-                _download_http(f"{base}/{name}", out / name, sub)
+                _download_http(f"{base}/{name}", out / name, sub,
+                               headers=headers)
             except Exception:  # noqa: BLE001 — optional aux files
                 if name.endswith(".safetensors"):
                     raise
