@@ -129,15 +129,33 @@ def fetch_weights(source: str, cache_dir: str = "/workspace/weights",
                           "tokenizer.json", "tokenizer_config.json"]
         out = Path(cache_dir)
         n = len(names)
-        for i, name in enumerate(names):
+        # concurrent streaming (the RunAI-streamer analog: KAITO_STREAM_
+        # CONCURRENCY parallel connections; sharded checkpoints with many
+        # model-xxxxx-of-yyyyy.safetensors files saturate NVMe this way)
+        workers = max(1, int(os.environ.get("KAITO_STREAM_CONCURRENCY",
+                                            "4")))
+        fracs = [0.0] * n
+        lock = __import__("threading").Lock()
+
+        def fetch_one(i: int, name: str):
             def sub(frac, i=i):
                 if progress:
-                    progress((i + frac) / n)
+                    with lock:
+                        fracs[i] = frac
+                        progress(sum(fracs) / n)
             try:
                 _download_http(f"{base}/{name}", out / name, sub,
                                headers=headers)
             except Exception:  # noqa: BLE001 — optional aux files
                 if name.endswith(".safetensors"):
                     raise
+
+        if workers == 1 or n == 1:
+            for i, name in enumerate(names):
+                fetch_one(i, name)
+        else:
+            from concurrent.futures import ThreadPoolExecutor
+            with ThreadPoolExecutor(max_workers=workers) as ex:
+                list(ex.map(lambda t: fetch_one(*t), enumerate(names)))
         return str(out)
     raise ValueError(f"unsupported weight source scheme {p.scheme!r}")

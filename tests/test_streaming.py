@@ -65,3 +65,37 @@ def test_workload_identity_absent_returns_none(monkeypatch):
         monkeypatch.delenv(k, raising=False)
     assert streaming.azure_workload_identity_token() is None
     assert streaming.azure_auth_headers() == {}
+
+
+def test_concurrent_fetch_sharded_checkpoint(tmp_path, monkeypatch):
+    """Sharded safetensors fetch runs KAITO_STREAM_CONCURRENCY parallel
+    connections and aggregates progress monotonically to 1.0."""
+    import functools
+    import http.server
+    import threading
+
+    from kaito_amd.models.streaming import fetch_weights
+
+    src = tmp_path / "src"
+    src.mkdir()
+    names = [f"model-{i:05d}-of-00004.safetensors" for i in range(1, 5)]
+    for nm in names:
+        (src / nm).write_bytes(bytes([len(nm) % 251]) * 65536)
+
+    handler = functools.partial(
+        http.server.SimpleHTTPRequestHandler, directory=str(src))
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), handler)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    try:
+        monkeypatch.setenv("KAITO_STREAM_CONCURRENCY", "4")
+        seen = []
+        dest = tmp_path / "cache"
+        got = fetch_weights(
+            f"http://127.0.0.1:{srv.server_address[1]}",
+            cache_dir=str(dest), files=names, progress=seen.append)
+        for nm in names:
+            assert (dest / nm).read_bytes() == (src / nm).read_bytes()
+        assert got == str(dest)
+        assert seen and abs(seen[-1] - 1.0) < 1e-6
+    finally:
+        srv.shutdown()
