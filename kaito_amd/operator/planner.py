@@ -50,12 +50,16 @@ def configure_parallelism(model: ModelConfig, gpu: GPUConfig,
             return ParallelPlan(data_parallel=gpus, tensor_parallel=1,
                                 num_nodes=1, gpus_per_node=gpus,
                                 kv_offload=False)
-        # tier 2: shard across the xGMI mesh
+        # tier 2: shard across the xGMI mesh. MLA models never offload:
+        # the engine keeps the compressed latent cache on-device (its
+        # aliased (c, c) pair would double offloaded bytes — engine.py)
         return ParallelPlan(tensor_parallel=gpus, num_nodes=1,
-                            gpus_per_node=gpus)
+                            gpus_per_node=gpus,
+                            kv_offload=not model.is_mla)
     # tier 3: pipeline across nodes, TP within each node
     return ParallelPlan(tensor_parallel=gpus, pipeline_parallel=num_nodes,
-                        num_nodes=num_nodes, gpus_per_node=gpus)
+                        num_nodes=num_nodes, gpus_per_node=gpus,
+                        kv_offload=not model.is_mla)
 
 
 def build_inference_command(model: ModelConfig, gpu: GPUConfig,
