@@ -372,6 +372,22 @@ def test_mla_decode_kernel_matches_ref():
         _close(out, ref.to(DEV), atol=2e-2)
 
 
+def test_engine_gpu_mla_chunked_prefill_matches_whole():
+    """MLA chunked prefill on GPU (absorbed latent-cache context
+    attention) must equal the whole-prompt prefill result."""
+    from kaito_amd.engine import SamplingParams
+    from kaito_amd.models import get_model_config
+    mc = get_model_config("tiny-deepseek-test")
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+    prompt = list(range(2, 80))                  # 78 tokens
+    eng = _gpu_engine(model=mc, enforce_eager=True, max_model_len=96,
+                      max_num_batched_tokens=32)  # -> 3 chunks
+    got = eng.generate([prompt], sp)[0].output_token_ids
+    eng2 = _gpu_engine(model=mc, enforce_eager=True, max_model_len=96)
+    want = eng2.generate([prompt], sp)[0].output_token_ids
+    assert got == want
+
+
 def test_mla_cache_write_kernel():
     """Fused latent-row scatter vs index_copy_, incl. -1 padding skip."""
     torch.manual_seed(3)
