@@ -800,3 +800,52 @@ def test_ipc_handle_roundtrip_bytes():
     t = torch.empty(128, device=DEV)
     h = torch.ops.kaito.ipc_handle(t)
     assert h.numel() == 64 and h.dtype == torch.uint8
+
+
+@pytest.mark.timeout(300)
+def test_engine_gpu_threaded_submitters_with_graphs():
+    """Concurrency stress ON the GPU engine with hipGraph decode replay:
+    6 threads x 3 greedy requests through one AsyncLLMEngine; repeated
+    prompts must be deterministic and nothing may wedge (the Go -race
+    analog exercised against the real device path)."""
+    import asyncio
+    import threading
+    from kaito_amd.server.async_engine import AsyncLLMEngine
+    from kaito_amd.engine import SamplingParams
+
+    eng = _gpu_engine()            # graphs enabled (no enforce_eager)
+    eng.capture_graphs()
+    aeng = AsyncLLMEngine(eng).start()
+    prompts = [[3 + i, 7, 11, 15] for i in range(3)]
+    results = {}
+    lock = threading.Lock()
+    errors = []
+
+    def worker(widx):
+        async def run():
+            for r in range(3):
+                p = prompts[(widx + r) % len(prompts)]
+                toks = []
+                async for item in aeng.generate(
+                        list(p), SamplingParams(max_tokens=6,
+                                                ignore_eos=True)):
+                    if not item.finished:
+                        toks.append(item.token_id)
+                with lock:
+                    results.setdefault(tuple(p), []).append(tuple(toks))
+        try:
+            asyncio.run(run())
+        except Exception as e:  # noqa: BLE001
+            errors.append(e)
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=240)
+        assert not t.is_alive(), "submitter wedged"
+    aeng.shutdown()
+    assert not errors, errors
+    assert sum(len(v) for v in results.values()) == 18
+    for p, outs in results.items():
+        assert len(set(outs)) == 1, (p, set(outs))
