@@ -38,6 +38,7 @@
 // the launch underfills 256 CUs; a flash-decoding token-split tier is
 // the designated fix (docs/ROADMAP.md).
 #include "common.h"
+#include <algorithm>
 #include <cstdlib>
 #include <torch/library.h>
 #include <ATen/ATen.h>
@@ -64,8 +65,12 @@ typedef short short4_t __attribute__((ext_vector_type(4)));
 // k)] @ C_kv[16t x 512d] via 8 d-tiles per wave; V staged TRANSPOSED
 // (s_vt[dim][token]) during the cooperative copy so B-fragments are
 // contiguous b128 reads (same trick as prefill_attention); requires MF.
+// SP: flash-decoding token-split — blockIdx.z picks a contiguous
+// partition of the KV blocks; each split writes UN-normalized partials
+// (acc, M, l) to scratch and mla_merge_kernel combines them. Fills the
+// 256-CU chip when T x head-tiles alone can't (docs/ROADMAP.md).
 template <int R, int P, int BS, int OCC = 4, bool DB = false, int PH = 3,
-          bool MF = false, bool MV = false>
+          bool MF = false, bool MV = false, bool SP = false>
 __global__ __launch_bounds__(256, OCC)
 void mla_decode_kernel(
     short* __restrict__ out,            // [T, H, R] bf16
@@ -73,7 +78,10 @@ void mla_decode_kernel(
     const short* __restrict__ cache,    // [NB, BS, R+P] bf16
     const int* __restrict__ block_tables,  // [T, max_blocks]
     const int* __restrict__ seq_lens,   // [T]
-    const float scale, const int H, const int max_blocks) {
+    const float scale, const int H, const int max_blocks,
+    float* __restrict__ part_acc = nullptr,   // [T, S, 16, R]
+    float* __restrict__ part_ml = nullptr,    // [T, S, 16, 2]
+    const int S = 1) {
   constexpr int DT = R + P;             // 576
   constexpr int SL = DT / 16;           // 36 dims per score slice
   constexpr int AD = R / 16;            // 32 dims per accum slice
@@ -83,11 +91,14 @@ void mla_decode_kernel(
 
   const int seq = blockIdx.x;
   const int h0 = blockIdx.y * HT;
+  const int split = SP ? blockIdx.z : 0;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int seq_len = seq_lens[seq];
-  const int nblocks = (seq_len + BS - 1) / BS;
+  const int nb_all = (seq_len + BS - 1) / BS;
+  const int b_lo = SP ? (nb_all * split) / S : 0;
+  const int nblocks = SP ? (nb_all * (split + 1)) / S : nb_all;
   const int* bt = block_tables + (int64_t)seq * max_blocks;
 
   // score-phase role: 4 heads per wave, 16 dim-slices per head
@@ -105,6 +116,7 @@ void mla_decode_kernel(
   __shared__ short s_vt[MV ? R * TP : 1];         // V transposed (20.5KB)
   __shared__ short s_pb[MV ? HT : 1][TP];         // P row-major bf16
   __shared__ float s_corr[HT], s_l[HT];
+  __shared__ float s_M2[SP ? HT : 1];             // running max (splits)
 
   // per-head online-softmax state, held REDUNDANTLY in registers by all
   // 16 lanes of the head's thread group (every lane derives identical
@@ -159,9 +171,9 @@ void mla_decode_kernel(
 
   constexpr int NCH = (BS * DT / 8 + 255) / 256;   // b128 chunks/thread
   short8_t pf[DB ? NCH : 1];
-  if constexpr (DB) {   // prologue: block 0 loads in flight
+  if constexpr (DB) {   // prologue: first block's loads in flight
     const short8_t* src = reinterpret_cast<const short8_t*>(
-        cache + (int64_t)bt[0] * (BS * DT));
+        cache + (int64_t)bt[b_lo] * (BS * DT));
 #pragma unroll
     for (int c = 0; c < NCH; c++) {
       const int idx = c * 256 + tid;
@@ -169,7 +181,7 @@ void mla_decode_kernel(
     }
   }
 
-  for (int b = 0; b < nblocks; b++) {
+  for (int b = b_lo; b < nblocks; b++) {
     // ---- stage one KV block: 1152 x b128, 256 threads ----
     if constexpr (DB) {
 #pragma unroll
@@ -300,7 +312,11 @@ void mla_decode_kernel(
       M = Mnew;
       if constexpr (MV) {
         s_pb[ac_h][t] = f32_to_bf16(p);
-        if (t == 0) { s_corr[ac_h] = corr; s_l[ac_h] = l; }
+        if (t == 0) {
+          s_corr[ac_h] = corr;
+          s_l[ac_h] = l;
+          if constexpr (SP) s_M2[ac_h] = M;
+        }
       }
     }
     if constexpr (MV) __syncthreads();   // s_pb/s_corr to all waves
@@ -359,6 +375,35 @@ void mla_decode_kernel(
     return;
   }
 
+  if constexpr (SP) {
+    // split partials: UN-normalized latent acc + (M, l) per head; the
+    // merge kernel rescales across splits. Empty split (b_lo >=
+    // nblocks, short sequences) contributes exp(-1e30)=0.
+    const bool empty = b_lo >= nblocks;
+    const int HTOT = gridDim.y * HT;     // scratch covers ALL head tiles
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      const int dim = (wave * 8 + j) * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const int h = hi * 4 + r;
+        part_acc[(((int64_t)seq * S + split) * HTOT + h0 + h) * R + dim] =
+            accv[j][r];
+      }
+    }
+    if (wave == 0 && (lane & 15) == 0) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const int h = hi * 4 + r;
+        float* ml = part_ml
+            + (((int64_t)seq * S + split) * HTOT + h0 + h) * 2;
+        ml[0] = empty ? -1e30f : s_M2[h];
+        ml[1] = empty ? 0.f : s_l[h];
+      }
+    }
+    return;
+  }
+
   // ---- epilogue: normalize + write [T, H, R] ----
   if constexpr (MV) {
     // C-fragment layout: row hi*4+r = head, col = dim-within-tile
@@ -381,6 +426,44 @@ void mla_decode_kernel(
 #pragma unroll
     for (int j = 0; j < AD; j++) op[j] = f32_to_bf16(acc[j] * inv);
   }
+}
+
+// Combine flash-decoding split partials: out[h] = sum_s w_s*acc_s /
+// sum_s w_s*l_s with w_s = exp(M_s - max M). Thread (head = tid>>4,
+// dim-slice = tid&15) mirrors the main kernel's accum layout.
+template <int R>
+__global__ __launch_bounds__(256, 8)
+void mla_merge_kernel(short* __restrict__ out,
+                      const float* __restrict__ part_acc,
+                      const float* __restrict__ part_ml,
+                      const int H, const int S) {
+  constexpr int HT = 16;
+  constexpr int AD = R / 16;
+  const int seq = blockIdx.x;
+  const int h = (int)(threadIdx.x >> 4) + blockIdx.y * HT;
+  const int ds = threadIdx.x & 15;
+  if (h >= H) return;
+  const int HTOT = gridDim.y * HT;
+  const int64_t base = (int64_t)seq * S;
+  float gm = -1e30f;
+  for (int sp = 0; sp < S; sp++)
+    gm = fmaxf(gm, part_ml[((base + sp) * HTOT + h) * 2]);
+  float den = 0.f;
+  float num[AD];
+#pragma unroll
+  for (int j = 0; j < AD; j++) num[j] = 0.f;
+  for (int sp = 0; sp < S; sp++) {
+    const float* ml = part_ml + ((base + sp) * HTOT + h) * 2;
+    const float w = __expf(ml[0] - gm);
+    den += w * ml[1];
+    const float* pa = part_acc + ((base + sp) * HTOT + h) * R + ds * AD;
+#pragma unroll
+    for (int j = 0; j < AD; j++) num[j] += w * pa[j];
+  }
+  const float inv = 1.f / fmaxf(den, 1e-20f);
+  short* op = out + ((int64_t)seq * H + h) * R + ds * AD;
+#pragma unroll
+  for (int j = 0; j < AD; j++) op[j] = f32_to_bf16(num[j] * inv);
 }
 
 void mla_decode(at::Tensor out, at::Tensor q, at::Tensor cache,
@@ -452,12 +535,41 @@ void mla_decode(at::Tensor out, at::Tensor q, at::Tensor cache,
                          seq_lens.data_ptr<int>(), (float)scale, H,
                          max_blocks);
     } else if (mf_env && mv_env) {
-      hipLaunchKernelGGL(
-          (mla_decode_kernel<512, 64, 16, 4, true, 3, true, true>), grid,
-          dim3(256), 0, stream, (short*)out.data_ptr(),
-          (const short*)q.data_ptr(), (const short*)cache.data_ptr(),
-          block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
-          (float)scale, H, max_blocks);
+      // flash-decoding token-split when the plain grid underfills the
+      // 256-CU chip (KAITO_MLA_SPLIT overrides; 0 disables)
+      const int tiles = (H + 15) / 16;
+      int S = 1;
+      const char* se = getenv("KAITO_MLA_SPLIT");
+      if (se != nullptr) {
+        S = atoi(se);
+      } else if (T * tiles < 320) {
+        S = std::min<int>(8, std::max<int>(2, 512 / std::max(T * tiles, 1)));
+      }
+      if (S > 1) {
+        auto opts = at::TensorOptions()
+            .dtype(at::kFloat).device(q.device());
+        at::Tensor pacc = at::empty(
+            {(int64_t)T * S * tiles * 16 * 512}, opts);
+        at::Tensor pml = at::empty({(int64_t)T * S * tiles * 16 * 2}, opts);
+        hipLaunchKernelGGL(
+            (mla_decode_kernel<512, 64, 16, 4, true, 3, true, true, true>),
+            dim3(T, tiles, S), dim3(256), 0, stream,
+            (short*)out.data_ptr(), (const short*)q.data_ptr(),
+            (const short*)cache.data_ptr(), block_tables.data_ptr<int>(),
+            seq_lens.data_ptr<int>(), (float)scale, H, max_blocks,
+            pacc.data_ptr<float>(), pml.data_ptr<float>(), S);
+        hipLaunchKernelGGL((mla_merge_kernel<512>), dim3(T, tiles),
+                           dim3(256), 0, stream, (short*)out.data_ptr(),
+                           pacc.data_ptr<float>(), pml.data_ptr<float>(),
+                           H, S);
+      } else {
+        hipLaunchKernelGGL(
+            (mla_decode_kernel<512, 64, 16, 4, true, 3, true, true>), grid,
+            dim3(256), 0, stream, (short*)out.data_ptr(),
+            (const short*)q.data_ptr(), (const short*)cache.data_ptr(),
+            block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
+            (float)scale, H, max_blocks);
+      }
     } else if (mf_env) {
       hipLaunchKernelGGL(
           (mla_decode_kernel<512, 64, 16, 4, true, 3, true>), grid,
